@@ -1,0 +1,194 @@
+"""Audit trail (JSONL + Merkle), engine pipeline, output validation."""
+
+import hashlib
+import json
+import os
+
+import pytest
+
+from vainplex_openclaw_amd.governance.audit import AuditTrail, derive_controls, merkle_root
+from vainplex_openclaw_amd.governance.engine import GovernanceEngine
+from vainplex_openclaw_amd.governance.output_validator import OutputValidator, more_restrictive
+from vainplex_openclaw_amd.governance.claims import detect_claims
+from vainplex_openclaw_amd.governance.facts import FactRegistry, check_claims
+
+
+def test_merkle_root_shapes():
+    # single leaf: root == sha256(leaf-hash || leaf-hash)? No: single level -> leaf hash itself
+    leaf = b"hello"
+    assert merkle_root([leaf]) == hashlib.sha256(leaf).hexdigest()
+    # two leaves
+    h0 = hashlib.sha256(b"a").digest()
+    h1 = hashlib.sha256(b"b").digest()
+    assert merkle_root([b"a", b"b"]) == hashlib.sha256(h0 + h1).hexdigest()
+    # odd count duplicates last
+    h2 = hashlib.sha256(b"c").digest()
+    inner0 = hashlib.sha256(h0 + h1).digest()
+    inner1 = hashlib.sha256(h2 + h2).digest()
+    assert merkle_root([b"a", b"b", b"c"]) == hashlib.sha256(inner0 + inner1).hexdigest()
+
+
+def test_derive_controls_deny_baseline():
+    mp = [{"controls": ["A.8.6", "A.7.1"]}]
+    assert derive_controls(mp, "deny") == ["A.5.24", "A.5.28", "A.7.1", "A.8.6"]
+    assert derive_controls(mp, "allow") == ["A.7.1", "A.8.6"]
+
+
+def test_audit_record_format_and_flush(workspace):
+    trail = AuditTrail({}, workspace)
+    trail.load()
+    rec = trail.record(
+        "deny",
+        "test reason",
+        {"hook": "before_tool_call", "agentId": "a1", "toolParams": {"token": "api_key=supersecret123"}},
+        {"score": 35, "tier": "untrusted"},
+        {"level": "high", "score": 60},
+        [{"policyId": "p", "ruleId": "r", "effect": {"action": "deny"}, "controls": ["A.8.6"]}],
+        123,
+    )
+    for key in ("id", "timestamp", "timestampIso", "verdict", "reason", "context",
+                "trust", "risk", "matchedPolicies", "evaluationUs", "controls"):
+        assert key in rec
+    # redaction applied to context
+    assert "supersecret123" not in json.dumps(rec["context"])
+    trail.flush()
+    files = os.listdir(os.path.join(workspace, "governance", "audit"))
+    day_files = [f for f in files if f.endswith(".jsonl") and ".merkle." not in f]
+    assert len(day_files) == 1
+    day = day_files[0].replace(".jsonl", "")
+    assert trail.verify_merkle(day)
+
+
+def test_audit_query_filters(workspace):
+    trail = AuditTrail({}, workspace)
+    trail.load()
+    for i, verdict in enumerate(["allow", "deny", "allow"]):
+        trail.record(verdict, "r", {"agentId": f"a{i % 2}"}, {"score": 0, "tier": "untrusted"},
+                     {"level": "low", "score": 0}, [], 1)
+    trail.flush()
+    assert len(trail.query({"verdict": "deny"})) == 1
+    assert len(trail.query({"agentId": "a0"})) == 2
+    assert len(trail.query({"limit": 1})) == 1
+
+
+def test_audit_merkle_chain_continuity(workspace):
+    trail = AuditTrail({}, workspace)
+    trail.load()
+    trail.record("allow", "r1", {}, {"score": 0, "tier": "untrusted"}, {"level": "low", "score": 0}, [], 1)
+    trail.flush()
+    trail.record("allow", "r2", {}, {"score": 0, "tier": "untrusted"}, {"level": "low", "score": 0}, [], 1)
+    trail.flush()
+    audit_dir = os.path.join(workspace, "governance", "audit")
+    mfile = [f for f in os.listdir(audit_dir) if f.endswith(".merkle.jsonl")][0]
+    with open(os.path.join(audit_dir, mfile)) as fh:
+        entries = [json.loads(ln) for ln in fh if ln.strip()]
+    assert len(entries) == 2
+    assert entries[1]["prevRoot"] == entries[0]["chained"]
+
+
+def test_engine_deny_records_violation_and_audit(workspace):
+    engine = GovernanceEngine(
+        {"builtinPolicies": {"credentialGuard": True}, "trust": {"enabled": True, "defaultScore": 40}},
+        workspace,
+    )
+    engine.start()
+    try:
+        ctx = engine.build_context(
+            "before_tool_call", "a1", tool_name="read", tool_params={"file_path": "/x/.env"}
+        )
+        verdict = engine.evaluate(ctx)
+        assert verdict["action"] == "deny"
+        assert engine.trust_manager.score("a1") == 38  # violation -2
+        assert engine.stats["denies"] == 1
+        assert verdict["evaluationUs"] > 0
+        engine.audit_trail.flush()
+        recs = engine.audit_trail.query({"verdict": "deny"})
+        assert len(recs) == 1
+        assert "A.5.24" in recs[0]["controls"]
+    finally:
+        engine.stop()
+
+
+def test_engine_night_mode_no_trust_penalty(workspace):
+    t = [0.0]
+
+    # fixed clock at 23:30 local time
+    import datetime as dt
+    base = dt.datetime(2026, 1, 1, 23, 30).timestamp()
+    engine = GovernanceEngine(
+        {"builtinPolicies": {"nightMode": True}},
+        workspace,
+        clock=lambda: base,
+    )
+    engine.start()
+    try:
+        ctx = engine.build_context("before_tool_call", "a1", tool_name="exec", tool_params={"command": "ls"})
+        verdict = engine.evaluate(ctx)
+        assert verdict["action"] == "deny"
+        assert engine.trust_manager.score("a1") == 40  # night-mode deny: no violation
+    finally:
+        engine.stop()
+
+
+def test_engine_fail_modes(workspace):
+    engine = GovernanceEngine({"failMode": "closed"}, workspace)
+    engine.start()
+    try:
+        # force a pipeline crash
+        engine.risk_assessor = None
+        verdict = engine.evaluate(engine.build_context("before_tool_call", "a1", tool_name="exec"))
+        assert verdict["action"] == "deny"
+        assert "fail-closed" in verdict["reason"]
+    finally:
+        engine.stop()
+    engine2 = GovernanceEngine({"failMode": "open"}, workspace)
+    engine2.start()
+    try:
+        engine2.risk_assessor = None
+        verdict = engine2.evaluate(engine2.build_context("before_tool_call", "a1", tool_name="exec"))
+        assert verdict["action"] == "allow"
+    finally:
+        engine2.stop()
+
+
+def test_claim_detection_families():
+    text = (
+        "The nginx-service is running. The server named web-01 is fine. "
+        "backup.db does not exist. The queue has 255,908 items. disk is at 80%. "
+        "I am DeployBot, and I have admin capabilities.\n"
+    )
+    claims = detect_claims(text)
+    types = {c["type"] for c in claims}
+    assert "system_state" in types
+    assert "entity_name" in types
+    assert "existence" in types
+    assert "operational_status" in types
+    assert "self_referential" in types
+    # common word filter
+    assert not any(c["subject"] == "it" for c in detect_claims("it is running"))
+
+
+def test_fact_check_verdicts():
+    reg = FactRegistry([{"facts": [
+        {"subject": "nginx", "predicate": "state", "value": "running"},
+        {"subject": "queue", "predicate": "count", "value": "255908"},
+    ]}])
+    claims = detect_claims("nginx is stopped. queue count is 255908.")
+    results = check_claims(claims, reg)
+    by_subject = {r["claim"]["subject"]: r["status"] for r in results}
+    assert by_subject["nginx"] == "contradicted"
+    assert by_subject["queue"] == "verified"  # fuzzy numeric
+    unv = check_claims(detect_claims("mystery-svc is running."), reg)
+    assert unv[0]["status"] == "unverified"
+
+
+def test_output_validator_trust_proportional():
+    cfg = {"factRegistries": [{"facts": [{"subject": "nginx", "predicate": "state", "value": "running"}]}]}
+    ov = OutputValidator(cfg)
+    text = "nginx is stopped."
+    assert ov.validate(text, trust_score=30)["verdict"] == "block"
+    assert ov.validate(text, trust_score=50)["verdict"] == "flag"
+    assert ov.validate(text, trust_score=70)["verdict"] == "pass"
+    assert ov.validate("all good here", trust_score=10)["verdict"] == "pass"
+    assert more_restrictive("flag", "block") == "block"
+    assert more_restrictive("pass", "flag") == "flag"

@@ -1,0 +1,108 @@
+"""Core runtime: hook bus, plugin API, config layering, gateway."""
+
+import json
+import os
+
+from vainplex_openclaw_amd.core.api import HookBus, PluginApi, NullLogger
+from vainplex_openclaw_amd.core.config import (
+    load_plugin_config,
+    parse_jsonc,
+    plugin_enabled,
+    resolve_defaults,
+)
+from vainplex_openclaw_amd.core.gateway import Gateway
+
+
+def test_hook_priority_order():
+    bus = HookBus()
+    order = []
+    bus.on("before_tool_call", lambda ev: order.append("low"), priority=5)
+    bus.on("before_tool_call", lambda ev: order.append("high"), priority=1000)
+    bus.on("before_tool_call", lambda ev: order.append("mid"), priority=900)
+    bus.emit("before_tool_call", {})
+    assert order == ["high", "mid", "low"]
+
+
+def test_hook_result_merge_and_short_circuit():
+    bus = HookBus()
+    bus.on("before_tool_call", lambda ev: {"block": True, "blockReason": "denied"}, priority=1000)
+    called = []
+    bus.on("before_tool_call", lambda ev: called.append(1), priority=5)
+    ev = bus.emit("before_tool_call", {"toolName": "exec"})
+    assert ev["block"] is True
+    assert ev["blockReason"] == "denied"
+    assert called == []  # short-circuited after block
+
+
+def test_hook_fail_open_counts_errors():
+    bus = HookBus()
+
+    def boom(ev):
+        raise RuntimeError("x")
+
+    bus.on("message_received", boom)
+    out = bus.emit("message_received", {"content": "hi"})
+    assert out["content"] == "hi"
+    assert bus.diagnostics["message_received"].errors == 1
+
+
+def test_jsonc_parse():
+    text = '{\n  // comment\n  "a": 1, /* inline */\n  "b": [1, 2,],\n}'
+    assert parse_jsonc(text) == {"a": 1, "b": [1, 2]}
+
+
+def test_plugin_config_external_file_first(tmp_path):
+    home = str(tmp_path)
+    d = os.path.join(home, "plugins", "my-plugin")
+    os.makedirs(d)
+    with open(os.path.join(d, "config.json"), "w") as fh:
+        json.dump({"x": 1}, fh)
+    assert load_plugin_config("my-plugin", fallback={"x": 2}, home=home) == {"x": 1}
+    assert load_plugin_config("other", fallback={"x": 2}, home=home) == {"x": 2}
+
+
+def test_resolve_defaults_recursive():
+    out = resolve_defaults({"a": {"b": 1}}, {"a": {"b": 0, "c": 2}, "d": 3})
+    assert out == {"a": {"b": 1, "c": 2}, "d": 3}
+
+
+def test_plugin_enabled_allow_list():
+    cfg = {"plugins": {"entries": {"p1": {"enabled": True}, "p2": {"enabled": True}}, "allow": ["p1"]}}
+    assert plugin_enabled(cfg, "p1")
+    assert not plugin_enabled(cfg, "p2")
+    assert not plugin_enabled(cfg, "p3")
+
+
+class _DemoPlugin:
+    id = "demo"
+    name = "Demo"
+    version = "1.0.0"
+
+    def __init__(self):
+        self.started = False
+
+    def register(self, api: PluginApi):
+        api.register_service({"start": self._start, "stop": self._stop})
+        api.on("message_received", lambda ev: {"seen": True})
+        api.register_command("demo", lambda: "ok")
+        api.register_gateway_method("demo.status", lambda: {"ok": True})
+
+    def _start(self):
+        self.started = True
+
+    def _stop(self):
+        self.started = False
+
+
+def test_gateway_lifecycle():
+    gw = Gateway(config={}, logger=NullLogger())
+    plugin = _DemoPlugin()
+    gw.load(plugin, plugin_config={})
+    gw.start()
+    assert plugin.started
+    ev = gw.emit("message_received", {"content": "x"})
+    assert ev["seen"] is True
+    assert gw.command("demo") == "ok"
+    assert gw.gateway_method("demo.status") == {"ok": True}
+    gw.stop()
+    assert not plugin.started
