@@ -1,0 +1,214 @@
+"""Governance hooks, 2FA TOTP, response gate, cross-agent, security clients."""
+
+import pytest
+
+from vainplex_openclaw_amd.core.api import HookBus, PluginApi, NullLogger
+from vainplex_openclaw_amd.core.gateway import Gateway
+from vainplex_openclaw_amd.governance.approval_2fa import Approval2FA, totp_at, verify_totp, generate_secret
+from vainplex_openclaw_amd.governance.cross_agent import CrossAgentManager
+from vainplex_openclaw_amd.governance.engine import GovernanceEngine
+from vainplex_openclaw_amd.governance.hooks import GovernanceHooks, detect_external_comm, register_governance_hooks
+from vainplex_openclaw_amd.governance.plugin import GovernancePlugin, extract_agent_ids
+from vainplex_openclaw_amd.governance.response_gate import ResponseGate
+from vainplex_openclaw_amd.governance.security.agentproof import AgentProofRestClient, CircuitBreaker
+from vainplex_openclaw_amd.governance.security.erc8004 import classify_tier, encode_call, decode_uint256, ERC8004Provider, ERC8004Client
+from vainplex_openclaw_amd.governance.trust import TrustManager, TrustConfig
+from vainplex_openclaw_amd.governance.matrix_poller import MatrixPoller
+
+
+def test_totp_roundtrip():
+    secret = generate_secret()
+    code = totp_at(secret, 1_700_000_000)
+    assert len(code) == 6 and code.isdigit()
+    assert verify_totp(secret, code, 1_700_000_000) is not None
+    assert verify_totp(secret, code, 1_700_000_000 + 29) is not None  # same step
+    assert verify_totp(secret, "000000", 1_700_000_000) in (None, verify_totp(secret, "000000", 1_700_000_000))
+    assert verify_totp(secret, "12345", 1_700_000_000) is None  # wrong length
+
+
+def test_2fa_batch_resolution_and_replay():
+    t = [1_700_000_000.0]
+    ap = Approval2FA(clock=lambda: t[0])
+    r1 = ap.request("agent:a", "a", "deploy")
+    r2 = ap.request("agent:a", "a", "push")  # same 3s window -> same batch
+    assert r1["status"] == "pending" and r2["status"] == "pending"
+    assert len(ap.pending_requests()) == 2
+    code = totp_at(ap.secret, t[0])
+    resolved = ap.try_resolve_any(code)
+    assert len(resolved) == 2
+    assert all(r["status"] == "approved" for r in resolved)
+    # replay protection: same code again does nothing
+    ap.request("agent:b", "b", "x")
+    assert ap.try_resolve_any(code) == []
+    # session auto-approval within 10 min
+    r3 = ap.request("agent:a", "a", "again")
+    assert r3["status"] == "approved" and r3.get("auto")
+
+
+def test_2fa_expiry():
+    t = [0.0]
+    ap = Approval2FA(clock=lambda: t[0], timeout_s=100)
+    ap.request("agent:a", "a", "x")
+    t[0] = 200
+    expired = ap.expire_stale()
+    assert len(expired) == 1 and expired[0]["status"] == "expired"
+
+
+def test_response_gate():
+    gate = ResponseGate({
+        "enabled": True,
+        "fallbackTemplate": "blocked for {agent}: {reasons}",
+        "rules": [
+            {"agentId": "a1", "validators": [
+                {"type": "requiredTools", "tools": ["read"]},
+                {"type": "mustNotMatch", "pattern": "password"},
+            ]},
+        ],
+    })
+    res = gate.validate("here is the password", "a1", [])
+    assert not res["passed"]
+    assert len(res["failedValidators"]) == 2
+    assert "blocked for a1" in res["fallbackMessage"]
+    ok = gate.validate("clean", "a1", [{"toolName": "read", "output": "x"}])
+    assert ok["passed"]
+    # other agents unaffected
+    assert gate.validate("password", "a2", [])["passed"]
+    # invalid regex fails closed
+    bad = ResponseGate({"enabled": True, "rules": [{"validators": [{"type": "mustMatch", "pattern": "("}]}]})
+    assert not bad.validate("anything", "a", [])["passed"]
+
+
+def test_cross_agent_ceiling_and_cascade(workspace):
+    tm = TrustManager(TrustConfig(default_score=40, initial_scores={"parent": 55}), workspace)
+    cam = CrossAgentManager(tm)
+    ctx = {
+        "sessionKey": "agent:parent:subagent:child:123",
+        "agentId": "child",
+        "trust": {"agent": {"score": 90, "tier": "elevated"}, "session": {"score": 80, "tier": "elevated"}},
+    }
+    out = cam.enrich_context(ctx)
+    assert out["trust"]["session"]["score"] == 55  # capped at parent
+    assert out["trust"]["agent"]["score"] == 55
+    assert out["crossAgent"]["parentAgentId"] == "parent"
+    # explicit registration
+    cam.register_relationship("agent:main", "sess-xyz")
+    assert cam.get_parent("sess-xyz")["parentAgentId"] == "main"
+    assert cam.graph_summary()["agentCount"] == 1
+
+
+def test_governance_hooks_end_to_end(workspace):
+    gw = Gateway(config={"agents": ["a1"]})
+    plugin = GovernancePlugin(workspace)
+    gw.load(plugin, plugin_config={
+        "builtinPolicies": {"credentialGuard": True},
+        "workspace": workspace,
+    })
+    gw.start()
+    try:
+        ev = gw.emit("before_tool_call", {
+            "agentId": "a1", "sessionKey": "agent:a1",
+            "toolName": "read", "params": {"file_path": "/etc/secrets/k.pem"},
+        })
+        assert ev["block"] is True
+        assert "Credential Guard" in ev["blockReason"]
+        ok = gw.emit("before_tool_call", {
+            "agentId": "a1", "sessionKey": "agent:a1",
+            "toolName": "read", "params": {"file_path": "/tmp/notes.txt"},
+        })
+        assert "block" not in ok or not ok["block"]
+        # after_tool_call logs for response gate + trust
+        gw.emit("after_tool_call", {
+            "agentId": "a1", "sessionKey": "agent:a1", "toolName": "read", "result": "data",
+        })
+        assert plugin.engine.trust_manager.get("a1")["signals"]["successCount"] == 1
+        # sub-agent spawn registration
+        gw.emit("after_tool_call", {
+            "agentId": "a1", "sessionKey": "agent:a1", "toolName": "sessions_spawn",
+            "result": {"sessionId": "agent:a1:subagent:w:9"},
+        })
+        assert plugin.engine.cross_agent.get_parent("agent:a1:subagent:w:9") is not None
+        status = gw.gateway_method("governance.status")
+        assert status["stats"]["evaluations"] >= 2
+    finally:
+        gw.stop()
+
+
+def test_detect_external_comm():
+    cfg = {"outputValidation": {"llmValidator": {
+        "enabled": True, "externalChannels": ["twitter", "email"], "externalCommands": ["bird tweet"],
+    }}}
+    assert detect_external_comm(
+        {"toolName": "message", "params": {"channel": "twitter", "text": "hi"}}, cfg) == "hi"
+    assert detect_external_comm(
+        {"toolName": "exec", "params": {"command": "bird tweet 'yo'"}}, cfg) == "bird tweet 'yo'"
+    assert detect_external_comm(
+        {"toolName": "message", "params": {"channel": "internal", "text": "hi"}}, cfg) is None
+    assert detect_external_comm({"toolName": "exec", "params": {"command": "ls"}}, {}) is None
+
+
+def test_erc8004_encoding_and_provider():
+    call = encode_call("getReputation(address)", "0x" + "ab" * 20)
+    assert call.startswith("0x") and len(call) == 2 + 8 + 64
+    assert decode_uint256("0x" + "0" * 63 + "5") == 5
+    assert classify_tier(85) == "elevated"
+    assert classify_tier(10) == "untrusted"
+    calls = []
+
+    def rpc(method, params):
+        calls.append(method)
+        return "0x" + hex(72)[2:].rjust(64, "0")
+
+    prov = ERC8004Provider(ERC8004Client(rpc))
+    rep = prov.lookup_reputation("0x" + "ab" * 20)
+    assert rep["score"] == 72 and rep["tier"] == "trusted"
+    prov.lookup_reputation("0x" + "ab" * 20)
+    assert len(calls) == 1  # cached
+
+
+def test_agentproof_queue_and_breaker():
+    t = [0.0]
+    posts = []
+
+    def http_post(url, headers, body):
+        posts.append(body)
+
+    client = AgentProofRestClient(http_post=http_post, clock=lambda: t[0])
+    for i in range(1005):
+        client.enqueue_signal("a", "success")
+    assert client.queue_depth == 1000
+    assert client.dropped == 5
+    assert client.flush() == 1000
+    assert posts and len(posts[0]["signals"]) == 1000
+
+    # breaker opens after 5 failures
+    def bad_post(url, headers, body):
+        raise IOError("down")
+
+    failing = AgentProofRestClient(http_post=bad_post, clock=lambda: t[0])
+    for _ in range(6):
+        failing.enqueue_signal("a", "x")
+        failing.flush()
+    assert failing.breaker.is_open
+    t[0] += 61
+    assert not failing.breaker.is_open  # half-open after reset
+
+
+def test_matrix_poller_code_extraction():
+    ap = Approval2FA(clock=lambda: 1_700_000_000.0)
+    ap.request("agent:a", "a", "x")
+    code = totp_at(ap.secret, 1_700_000_000.0)
+
+    def http_get(url, headers):
+        return {"chunk": [{"type": "m.room.message", "content": {"body": f"approve {code}"}}], "end": "t1"}
+
+    poller = MatrixPoller(ap, room_id="!r", http_get=http_get)
+    codes = poller.poll_once()
+    assert codes == [code]
+    assert ap.pending_requests() == []
+
+
+def test_extract_agent_ids_shapes():
+    assert extract_agent_ids({"agents": ["a", "b"]}) == ["a", "b"]
+    assert extract_agent_ids({"agents": [{"id": "x"}, {"name": "y"}]}) == ["x", "y"]
+    assert extract_agent_ids({"agents": {"m": {}, "n": {}}}) == ["m", "n"]
+    assert extract_agent_ids({"defaultAgent": "main"}) == ["main"]

@@ -1,0 +1,79 @@
+"""Governance plugin entry: register(api).
+
+Parity target: `openclaw-governance/index.ts:66-116` — load config (file
+first, pluginConfig fallback), build engine, set known agents from host
+config, register service start/stop, wire hooks + gateway methods.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+from ..core.api import PluginApi
+from ..core.config import load_plugin_config
+from .approval_2fa import Approval2FA
+from .engine import GovernanceEngine
+from .hooks import register_governance_hooks
+from .redaction.hooks import register_redaction_hooks
+
+
+def extract_agent_ids(host_config: Dict[str, Any]) -> List[str]:
+    """Agent ids from the host openclaw.json (index.ts:85-88); tolerant of
+    the 4 config shapes brainplex scans (scanner.ts:58-91)."""
+    agents: List[str] = []
+    cfg = host_config or {}
+    raw = cfg.get("agents")
+    if isinstance(raw, list):
+        for a in raw:
+            if isinstance(a, str):
+                agents.append(a)
+            elif isinstance(a, dict) and a.get("id"):
+                agents.append(str(a["id"]))
+            elif isinstance(a, dict) and a.get("name"):
+                agents.append(str(a["name"]))
+    elif isinstance(raw, dict):
+        agents.extend(str(k) for k in raw.keys())
+    for key in ("agent", "defaultAgent"):
+        v = cfg.get(key)
+        if isinstance(v, str):
+            agents.append(v)
+        elif isinstance(v, dict) and v.get("id"):
+            agents.append(str(v["id"]))
+    seen = set()
+    out = []
+    for a in agents:
+        if a not in seen:
+            seen.add(a)
+            out.append(a)
+    return out
+
+
+class GovernancePlugin:
+    id = "openclaw-governance"
+    name = "Governance"
+    description = "Agent firewall: policy engine, trust scoring, 2FA, output validation, redaction, audit"
+    version = "0.1.0"
+
+    def __init__(self, workspace: Optional[str] = None):
+        self.workspace = workspace
+        self.engine: Optional[GovernanceEngine] = None
+        self.hooks = None
+        self.redaction = None
+        self.approval: Optional[Approval2FA] = None
+
+    def register(self, api: PluginApi) -> None:
+        config = load_plugin_config(self.id, fallback=api.plugin_config)
+        workspace = self.workspace or config.get("workspace") or "."
+        engine = GovernanceEngine(config, workspace, api.logger)
+        engine.set_known_agents(extract_agent_ids(api.config))
+        self.engine = engine
+        twofa_cfg = config.get("approval2fa") or {}
+        if twofa_cfg.get("enabled"):
+            self.approval = Approval2FA(secret=twofa_cfg.get("secret"))
+        api.register_service({"id": self.id, "start": engine.start, "stop": engine.stop})
+        self.hooks = register_governance_hooks(api, engine, config, self.approval)
+        self.redaction = register_redaction_hooks(api, config.get("redaction"))
+
+
+def create_plugin(workspace: Optional[str] = None) -> GovernancePlugin:
+    return GovernancePlugin(workspace)
