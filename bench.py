@@ -1,0 +1,139 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: messages/sec through the full
+Governance -> Membrane -> Cortex/KE firewall pipeline (BASELINE.json
+headline metric).
+
+Single GPU:   python bench.py --steps 10 --warmup 3
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Each rank owns one GPU (RCCL over xGMI), a shard of the 50M x 1024-d
+Membrane index (shard = total/world: per-GPU recall work is constant as
+ranks grow -> weak scaling), and processes its own batch of 4096 synthetic
+messages per step through every stage: DFA pattern scans, encoder,
+classifier head, full-index cosine top-k recall (queries all-gathered),
+verdict + trust update, audit Merkle root (roots combined across ranks).
+
+Rank 0 prints ONE JSON line with the whole-job aggregate messages/sec.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from vainplex_openclaw_amd.pipeline.engine import FirewallPipeline, PipelineConfig
+from vainplex_openclaw_amd.pipeline.synth import synthetic_batch
+
+BASELINE_MSG_S = 9000.0  # reference headline: NATS concurrent publish ~9k msg/s
+                         # (openclaw-nats-eventstore/README.md:261, other hardware)
+
+TOTAL_INDEX = 50_000_000  # BASELINE config: cosine-kNN over 50M x 1024-d
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--batch", type=int, default=4096)
+    ap.add_argument("--dim", type=int, default=1024)
+    ap.add_argument("--index", type=int, default=TOTAL_INDEX, help="TOTAL index rows across ranks")
+    ap.add_argument("--topk", type=int, default=16)
+    ap.add_argument("--pool", type=int, default=4, help="pre-generated batch pool size")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world > 1
+    if distributed:
+        torch.distributed.init_process_group("nccl", rank=rank, world_size=world)
+    device = f"cuda:{local_rank}"
+    torch.cuda.set_device(device)
+
+    shard = max(1, args.index // world)
+    # round shard to a multiple of the 128-row recall tile
+    shard = (shard // 128) * 128
+    cfg = PipelineConfig(
+        batch=args.batch, dim=args.dim, index_size=shard, topk=args.topk,
+    )
+    t0 = time.time()
+    pipe = FirewallPipeline(cfg, device=device, world_size=world, rank=rank)
+    torch.cuda.synchronize()
+    setup_s = time.time() - t0
+
+    # pre-generated, pre-staged synthetic batch pool (rotated; every step
+    # still runs the full pipeline on real message bytes)
+    pool = []
+    for i in range(args.pool):
+        batch = synthetic_batch(args.batch, seed=1000 * rank + i, n_agents=cfg.n_agents)
+        pool.append((batch, pipe.stage(batch)))
+
+    def barrier():
+        if distributed:
+            torch.distributed.barrier()
+
+    # warmup
+    for i in range(args.warmup):
+        b, s = pool[i % len(pool)]
+        pipe.step(b, staged=s)
+    barrier()
+    torch.cuda.synchronize()
+
+    # timed region: exactly --steps full pipeline steps
+    t_start = time.perf_counter()
+    for i in range(args.steps):
+        b, s = pool[i % len(pool)]
+        pipe.step(b, staged=s)
+    barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t_start
+
+    # MAX elapsed over ranks
+    if distributed:
+        t = torch.tensor([elapsed], device=device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = t.item()
+
+    total_msgs = world * args.batch * args.steps
+    msg_per_s = total_msgs / elapsed
+    ms_per_step = elapsed / args.steps * 1000
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "messages/sec through Governance->Membrane->Cortex/KE pipeline",
+            "value": round(msg_per_s, 2),
+            "unit": "msg/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(msg_per_s / BASELINE_MSG_S, 3),
+            "dtype": "bf16",
+            "data": "synthetic",
+            "setup_s": round(setup_s, 1),
+            "config": {
+                "model": "firewall-pipeline (dfa-scan + encoder + classifier head + cosine-kNN recall + merkle audit)",
+                "global_batch": world * args.batch,
+                "seq_len": 500,
+                "parallelism": f"dp{world}",
+                "index_rows_total": shard * world,
+                "index_dim": args.dim,
+                "topk": args.topk,
+            },
+        }))
+    if distributed:
+        torch.distributed.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

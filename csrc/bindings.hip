@@ -1,0 +1,249 @@
+// Torch extension bindings for the MI355X kernels.
+//
+// All entry points take torch tensors on the current CUDA(HIP) device and
+// launch on the current stream. u64 masks travel as int64 tensors
+// (bitwise-identical).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <vector>
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+extern "C" {
+__global__ void sha256_leaves_kernel(const uint8_t*, const int32_t*, uint8_t*, int);
+__global__ void merkle_level_kernel(const uint8_t*, uint8_t*, int);
+__global__ void dfa_scan_kernel(const uint8_t*, const int32_t*, const uint16_t*,
+                                const unsigned long long*, const unsigned long long*,
+                                const uint8_t*, const int32_t*, int,
+                                unsigned long long*, int);
+__global__ void encode_messages_kernel(const uint8_t*, const int32_t*, const __bf16*,
+                                       int, int, __bf16*, int, int);
+__global__ void gemm_nt_bf16_kernel(const __bf16*, const __bf16*, float*, __bf16*,
+                                    const float*, int, int, int, int, int);
+__global__ void topk_recall_kernel(const __bf16*, const __bf16*, int, int, int, int,
+                                   int, float*, int32_t*);
+__global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
+                                  float*, int32_t*);
+__global__ void firewall_verdict_kernel(const unsigned long long*, const unsigned long long*,
+                                        const float*, int, const int32_t*, const float*,
+                                        const float*, const int32_t*, int,
+                                        unsigned long long, float, int8_t*, float*,
+                                        float*, float*, int);
+__global__ void trust_recompute_kernel(float*, float*, const float*, const float*,
+                                       const float*, float*, const float*, float*, int);
+struct AuditRecord64;
+__global__ void audit_pack_kernel(const int8_t*, const float*, const unsigned long long*,
+                                  const unsigned long long*, const int32_t*, const float*,
+                                  const float*, long long, long long, uint32_t,
+                                  AuditRecord64*, int);
+}
+
+static hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+torch::Tensor sha256_leaves(torch::Tensor bytes, torch::Tensor offsets) {
+  CHECK_GPU(bytes); CHECK_CONTIG(bytes); CHECK_GPU(offsets); CHECK_CONTIG(offsets);
+  TORCH_CHECK(bytes.dtype() == torch::kUInt8 && offsets.dtype() == torch::kInt32);
+  int n = offsets.numel() - 1;
+  auto out = torch::empty({n, 32}, bytes.options());
+  int threads = 128;
+  int blocks = (n + threads - 1) / threads;
+  hipLaunchKernelGGL(sha256_leaves_kernel, dim3(blocks), dim3(threads), 0, cur_stream(),
+                     bytes.data_ptr<uint8_t>(), offsets.data_ptr<int32_t>(),
+                     out.data_ptr<uint8_t>(), n);
+  return out;
+}
+
+torch::Tensor merkle_root_gpu(torch::Tensor digests) {
+  CHECK_GPU(digests); CHECK_CONTIG(digests);
+  TORCH_CHECK(digests.dtype() == torch::kUInt8 && digests.size(1) == 32);
+  int n = digests.size(0);
+  TORCH_CHECK(n >= 1);
+  auto cur = digests;
+  while (n > 1) {
+    int n_out = (n + 1) / 2;
+    auto nxt = torch::empty({n_out, 32}, digests.options());
+    int threads = 128;
+    int blocks = (n_out + threads - 1) / threads;
+    hipLaunchKernelGGL(merkle_level_kernel, dim3(blocks), dim3(threads), 0, cur_stream(),
+                       cur.data_ptr<uint8_t>(), nxt.data_ptr<uint8_t>(), n);
+    cur = nxt;
+    n = n_out;
+  }
+  return cur.reshape({32});
+}
+
+torch::Tensor dfa_scan(torch::Tensor bytes, torch::Tensor offsets, torch::Tensor next_tab,
+                       torch::Tensor accept, torch::Tensor eof_mask,
+                       torch::Tensor class_maps, torch::Tensor meta) {
+  CHECK_GPU(bytes); CHECK_GPU(offsets); CHECK_GPU(next_tab); CHECK_GPU(accept);
+  CHECK_GPU(eof_mask); CHECK_GPU(class_maps); CHECK_GPU(meta);
+  TORCH_CHECK(next_tab.dtype() == torch::kInt16 || next_tab.dtype() == torch::kUInt8 ||
+              next_tab.scalar_type() == at::kShort, "next_tab must be int16 view of u16");
+  int n = offsets.numel() - 1;
+  int n_dfas = meta.size(0);
+  auto hits = torch::zeros({n}, torch::dtype(torch::kInt64).device(bytes.device()));
+  int threads = 256;
+  int blocks = (n + threads - 1) / threads;
+  size_t lds = (size_t)n_dfas * 256;
+  hipLaunchKernelGGL(dfa_scan_kernel, dim3(blocks), dim3(threads), lds, cur_stream(),
+                     bytes.data_ptr<uint8_t>(), offsets.data_ptr<int32_t>(),
+                     reinterpret_cast<const uint16_t*>(next_tab.data_ptr()),
+                     reinterpret_cast<const unsigned long long*>(accept.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(eof_mask.data_ptr<int64_t>()),
+                     class_maps.data_ptr<uint8_t>(), meta.data_ptr<int32_t>(), n_dfas,
+                     reinterpret_cast<unsigned long long*>(hits.data_ptr<int64_t>()), n);
+  return hits;
+}
+
+torch::Tensor encode_messages(torch::Tensor bytes, torch::Tensor offsets,
+                              torch::Tensor embed, bool normalize) {
+  CHECK_GPU(bytes); CHECK_GPU(offsets); CHECK_GPU(embed); CHECK_CONTIG(embed);
+  TORCH_CHECK(embed.dtype() == torch::kBFloat16);
+  int vocab = embed.size(0);
+  int dim = embed.size(1);
+  TORCH_CHECK((vocab & (vocab - 1)) == 0, "vocab must be a power of two");
+  TORCH_CHECK(dim % 1024 == 0, "dim must be a multiple of 1024");
+  int n = offsets.numel() - 1;
+  auto out = torch::empty({n, dim}, embed.options());
+  hipLaunchKernelGGL(encode_messages_kernel, dim3(n), dim3(256), 0, cur_stream(),
+                     bytes.data_ptr<uint8_t>(), offsets.data_ptr<int32_t>(),
+                     reinterpret_cast<const __bf16*>(embed.data_ptr()), vocab - 1, dim,
+                     reinterpret_cast<__bf16*>(out.data_ptr()), n, normalize ? 1 : 0);
+  return out;
+}
+
+torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B,
+                      c10::optional<torch::Tensor> bias, int64_t act, bool out_bf16) {
+  CHECK_GPU(A); CHECK_CONTIG(A); CHECK_GPU(B); CHECK_CONTIG(B);
+  TORCH_CHECK(A.dtype() == torch::kBFloat16 && B.dtype() == torch::kBFloat16);
+  int M = A.size(0), K = A.size(1), N = B.size(0);
+  TORCH_CHECK(B.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
+  auto opts = torch::dtype(out_bf16 ? torch::kBFloat16 : torch::kFloat32).device(A.device());
+  auto C = torch::empty({M, N}, opts);
+  const float* bias_ptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_GPU(bias.value());
+    TORCH_CHECK(bias->dtype() == torch::kFloat32 && bias->numel() == N);
+    bias_ptr = bias->data_ptr<float>();
+  }
+  dim3 grid((M + 127) / 128, (N + 127) / 128);
+  hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const __bf16*>(A.data_ptr()),
+                     reinterpret_cast<const __bf16*>(B.data_ptr()),
+                     out_bf16 ? nullptr : C.data_ptr<float>(),
+                     out_bf16 ? reinterpret_cast<__bf16*>(C.data_ptr()) : nullptr,
+                     bias_ptr, M, N, K, (int)act, out_bf16 ? 1 : 0);
+  return C;
+}
+
+std::vector<torch::Tensor> topk_recall(torch::Tensor Q, torch::Tensor X, int64_t k,
+                                       int64_t n_swaths) {
+  CHECK_GPU(Q); CHECK_CONTIG(Q); CHECK_GPU(X); CHECK_CONTIG(X);
+  TORCH_CHECK(Q.dtype() == torch::kBFloat16 && X.dtype() == torch::kBFloat16);
+  int nq = Q.size(0), D = Q.size(1);
+  long long nx = X.size(0);
+  TORCH_CHECK(X.size(1) == D && D % 32 == 0);
+  TORCH_CHECK(k >= 1 && k <= 32, "k in [1,32]");
+  int n_qblocks = (nq + 127) / 128;
+  auto f32opts = torch::dtype(torch::kFloat32).device(Q.device());
+  auto i32opts = torch::dtype(torch::kInt32).device(Q.device());
+  auto cand_s = torch::empty({(long long)n_qblocks * n_swaths * 128 * k}, f32opts);
+  auto cand_i = torch::empty({(long long)n_qblocks * n_swaths * 128 * k}, i32opts);
+  dim3 grid(n_qblocks, n_swaths);
+  hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const __bf16*>(Q.data_ptr()),
+                     reinterpret_cast<const __bf16*>(X.data_ptr()), nq, (int)nx, D,
+                     (int)k, (int)n_swaths, cand_s.data_ptr<float>(),
+                     cand_i.data_ptr<int32_t>());
+  auto out_s = torch::empty({nq, k}, f32opts);
+  auto out_i = torch::empty({nq, k}, i32opts);
+  int waves_per_block = 4;
+  int blocks = (nq + waves_per_block - 1) / waves_per_block;
+  hipLaunchKernelGGL(topk_merge_kernel, dim3(blocks), dim3(waves_per_block * 64), 0,
+                     cur_stream(), cand_s.data_ptr<float>(), cand_i.data_ptr<int32_t>(),
+                     nq, (int)k, (int)n_swaths, out_s.data_ptr<float>(),
+                     out_i.data_ptr<int32_t>());
+  return {out_s, out_i};
+}
+
+std::vector<torch::Tensor> firewall_verdict(
+    torch::Tensor inj_hits, torch::Tensor red_hits, torch::Tensor logits,
+    torch::Tensor agent_idx, torch::Tensor agent_trust, torch::Tensor tool_risk,
+    torch::Tensor freq_count, int64_t hour, int64_t cred_bits, double inj_threshold,
+    int64_t n_agents) {
+  CHECK_GPU(inj_hits); CHECK_GPU(red_hits); CHECK_GPU(logits); CHECK_GPU(agent_idx);
+  CHECK_GPU(agent_trust); CHECK_GPU(tool_risk); CHECK_GPU(freq_count);
+  int B = inj_hits.numel();
+  int n_cls = logits.size(1);
+  auto verdict = torch::empty({B}, torch::dtype(torch::kInt8).device(inj_hits.device()));
+  auto risk = torch::empty({B}, torch::dtype(torch::kFloat32).device(inj_hits.device()));
+  auto sdelta = torch::zeros({n_agents}, torch::dtype(torch::kFloat32).device(inj_hits.device()));
+  auto vdelta = torch::zeros({n_agents}, torch::dtype(torch::kFloat32).device(inj_hits.device()));
+  int threads = 256;
+  int blocks = (B + threads - 1) / threads;
+  hipLaunchKernelGGL(firewall_verdict_kernel, dim3(blocks), dim3(threads), 0, cur_stream(),
+                     reinterpret_cast<const unsigned long long*>(inj_hits.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(red_hits.data_ptr<int64_t>()),
+                     logits.data_ptr<float>(), n_cls, agent_idx.data_ptr<int32_t>(),
+                     agent_trust.data_ptr<float>(), tool_risk.data_ptr<float>(),
+                     freq_count.data_ptr<int32_t>(), (int)hour,
+                     (unsigned long long)cred_bits, (float)inj_threshold,
+                     verdict.data_ptr<int8_t>(), risk.data_ptr<float>(),
+                     sdelta.data_ptr<float>(), vdelta.data_ptr<float>(), B);
+  return {verdict, risk, sdelta, vdelta};
+}
+
+void trust_recompute(torch::Tensor success_count, torch::Tensor violation_count,
+                     torch::Tensor success_delta, torch::Tensor violation_delta,
+                     torch::Tensor age_days, torch::Tensor clean_streak,
+                     torch::Tensor manual_adj, torch::Tensor score) {
+  int A = score.numel();
+  int threads = 256;
+  int blocks = (A + threads - 1) / threads;
+  hipLaunchKernelGGL(trust_recompute_kernel, dim3(blocks), dim3(threads), 0, cur_stream(),
+                     success_count.data_ptr<float>(), violation_count.data_ptr<float>(),
+                     success_delta.data_ptr<float>(), violation_delta.data_ptr<float>(),
+                     age_days.data_ptr<float>(), clean_streak.data_ptr<float>(),
+                     manual_adj.data_ptr<float>(), score.data_ptr<float>(), A);
+}
+
+torch::Tensor audit_pack(torch::Tensor verdict, torch::Tensor risk, torch::Tensor inj_hits,
+                         torch::Tensor red_hits, torch::Tensor agent_idx,
+                         torch::Tensor agent_trust, torch::Tensor inj_score,
+                         int64_t ts_ms, int64_t msg_id0, int64_t batch_seq) {
+  int B = verdict.numel();
+  auto out = torch::empty({B, 64}, torch::dtype(torch::kUInt8).device(verdict.device()));
+  int threads = 256;
+  int blocks = (B + threads - 1) / threads;
+  hipLaunchKernelGGL(audit_pack_kernel, dim3(blocks), dim3(threads), 0, cur_stream(),
+                     verdict.data_ptr<int8_t>(), risk.data_ptr<float>(),
+                     reinterpret_cast<const unsigned long long*>(inj_hits.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(red_hits.data_ptr<int64_t>()),
+                     agent_idx.data_ptr<int32_t>(), agent_trust.data_ptr<float>(),
+                     inj_score.data_ptr<float>(), (long long)ts_ms, (long long)msg_id0,
+                     (uint32_t)batch_seq,
+                     reinterpret_cast<AuditRecord64*>(out.data_ptr<uint8_t>()), B);
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sha256_leaves", &sha256_leaves, "Batched SHA-256 leaf digests");
+  m.def("merkle_root", &merkle_root_gpu, "Merkle root over leaf digests");
+  m.def("dfa_scan", &dfa_scan, "Multi-pattern DFA scan");
+  m.def("encode_messages", &encode_messages, "4-gram hash embedding encoder");
+  m.def("gemm_nt", &gemm_nt, "bf16 NT GEMM with fused epilogue",
+        py::arg("A"), py::arg("B"), py::arg("bias") = c10::nullopt,
+        py::arg("act") = 0, py::arg("out_bf16") = false);
+  m.def("topk_recall", &topk_recall, "Fused cosine top-k recall");
+  m.def("firewall_verdict", &firewall_verdict, "Fused verdict/risk/trust-delta");
+  m.def("trust_recompute", &trust_recompute, "Agent trust score recompute");
+  m.def("audit_pack", &audit_pack, "Pack 64-byte audit records");
+}

@@ -180,21 +180,23 @@ def test_multidfa_pack_roundtrip():
     mdfa = ps.get_family("redaction")
     packed = mdfa.pack()
     total = mdfa.n_states
-    assert packed["next32"].shape == (total * 256,)
     assert packed["accept"].shape == (total,)
-    assert len(packed["starts"]) == len(mdfa.dfas)
+    assert packed["meta"].shape == (len(mdfa.dfas), 4)
+    assert packed["class_maps"].shape == (len(mdfa.dfas), 256)
 
-    # packed-table reference scan must agree with per-DFA scan
+    # packed-table reference scan (mirrors csrc/pattern_scan.hip) must
+    # agree with per-DFA scan
     def packed_scan(data: bytes) -> int:
-        nxt = packed["next32"].reshape(total, 256)
         mask = 0
-        for s0 in packed["starts"]:
-            state = int(s0)
-            mask |= int(packed["accept"][state])
+        for d in range(len(mdfa.dfas)):
+            next_base, state_base, ncls, cmap_row = packed["meta"][d]
+            cmap = packed["class_maps"][cmap_row]
+            state = 0
+            mask |= int(packed["accept"][state_base])
             for b in data:
-                state = int(nxt[state, b])
-                mask |= int(packed["accept"][state])
-            mask |= int(packed["eof"][state])
+                state = int(packed["next"][next_base + state * ncls + cmap[b]])
+                mask |= int(packed["accept"][state_base + state])
+            mask |= int(packed["eof"][state_base + state])
         return mask
 
     for text in SECRET_SAMPLES + CLEAN_SAMPLES:

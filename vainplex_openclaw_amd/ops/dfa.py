@@ -745,33 +745,43 @@ class MultiDFA:
         return mask
 
     def pack(self):
-        """Concatenate tables into flat arrays for the GPU kernel:
-        returns dict with
-          next32:  uint32 [sum_states * 256]  (per-BYTE, class fold applied,
-                   state ids offset by sub-DFA base)
-          accept:  uint64 [sum_states]
-          eof:     uint64 [sum_states]
-          starts:  int32  [n_dfas]  start state id of each sub-DFA
-        Folding byte->class into a per-byte table costs memory
-        (states*256*4B) but removes one dependent load per step in the
-        kernel's inner loop."""
+        """Concatenate tables into flat arrays for the GPU kernel
+        (csrc/pattern_scan.hip):
+          next:       uint16 flat — each sub-DFA's [n_states, n_classes]
+                      class-transition table (LOCAL state ids)
+          accept,eof: uint64 [sum_states] (indexed by state_base + state)
+          class_maps: uint8 [n_dfas, 256] byte -> class
+          meta:       int32 [n_dfas, 4] = (next_base_in_u16s, state_base,
+                      n_classes, class_map_row)
+        The kernel keeps class_maps in LDS and walks `next` out of L2
+        (the tables are a few hundred KB — L2-resident)."""
         total = self.n_states
-        next32 = np.zeros((total, 256), dtype=np.uint32)
         accept = np.zeros(total, dtype=np.uint64)
         eof = np.zeros(total, dtype=np.uint64)
-        starts = np.zeros(len(self.dfas), dtype=np.int32)
-        base = 0
+        meta = np.zeros((len(self.dfas), 4), dtype=np.int32)
+        class_maps = np.zeros((len(self.dfas), 256), dtype=np.uint8)
+        next_parts = []
+        state_base = 0
+        next_base = 0
         for i, d in enumerate(self.dfas):
-            starts[i] = base
-            # expand class table to byte table with offset state ids
-            per_byte = d.next_state[:, d.byte_class].astype(np.uint32) + np.uint32(base)
-            next32[base : base + d.n_states] = per_byte
-            accept[base : base + d.n_states] = d.accept_mask
+            if d.n_states > 65535:
+                raise RegexError("sub-DFA exceeds uint16 state ids")
+            meta[i] = (next_base, state_base, d.n_classes, i)
+            class_maps[i] = d.byte_class
+            next_parts.append(d.next_state.astype(np.uint16).reshape(-1))
+            accept[state_base : state_base + d.n_states] = d.accept_mask
             e = getattr(d, "eof_mask", None)
             if e is not None:
-                eof[base : base + d.n_states] = e
-            base += d.n_states
-        return {"next32": next32.reshape(-1), "accept": accept, "eof": eof, "starts": starts}
+                eof[state_base : state_base + d.n_states] = e
+            state_base += d.n_states
+            next_base += d.n_states * d.n_classes
+        return {
+            "next": np.concatenate(next_parts),
+            "accept": accept,
+            "eof": eof,
+            "class_maps": class_maps,
+            "meta": meta,
+        }
 
 
 def compile_multi(

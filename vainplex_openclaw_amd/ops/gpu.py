@@ -1,0 +1,217 @@
+"""GPU op wrappers: the bridge between the engines and csrc/ kernels.
+
+Every op here REQUIRES the in-tree `_hip_ops` extension when running on a
+GPU — there is no silent eager fallback (a GPU box without the extension
+raises immediately). CPU reference implementations used by the numerics
+tests live in `reference_*` functions.
+"""
+
+from __future__ import annotations
+
+import hashlib
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from .dfa import MultiDFA
+from . import pattern_sets
+
+_EXT = None
+
+
+def ext():
+    """The compiled extension; raises loudly when missing."""
+    global _EXT
+    if _EXT is None:
+        try:
+            from .. import _hip_ops  # type: ignore
+
+            _EXT = _hip_ops
+        except ImportError as exc:  # pragma: no cover
+            raise RuntimeError(
+                "vainplex_openclaw_amd._hip_ops is not built. Run "
+                "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950). "
+                "GPU ops never fall back to eager."
+            ) from exc
+    return _EXT
+
+
+def have_ext() -> bool:
+    try:
+        ext()
+        return True
+    except RuntimeError:
+        return False
+
+
+# -- message packing --------------------------------------------------------
+
+def pack_messages(messages: Sequence[bytes], device="cuda") -> Tuple[torch.Tensor, torch.Tensor]:
+    """Concatenate messages into (bytes u8 [total], offsets i32 [n+1])."""
+    offsets = np.zeros(len(messages) + 1, dtype=np.int32)
+    for i, m in enumerate(messages):
+        offsets[i + 1] = offsets[i] + len(m)
+    blob = b"".join(messages)
+    b = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+    o = torch.from_numpy(offsets)
+    if device != "cpu":
+        b = b.to(device, non_blocking=True)
+        o = o.to(device, non_blocking=True)
+    return b, o
+
+
+# -- DFA scan ---------------------------------------------------------------
+
+class DeviceDFA:
+    """Packed MultiDFA tables resident on one device."""
+
+    def __init__(self, mdfa: MultiDFA, device) -> None:
+        packed = mdfa.pack()
+        self.next = torch.from_numpy(packed["next"].view(np.int16)).to(device)
+        self.accept = torch.from_numpy(packed["accept"].view(np.int64)).to(device)
+        self.eof = torch.from_numpy(packed["eof"].view(np.int64)).to(device)
+        self.class_maps = torch.from_numpy(packed["class_maps"].reshape(-1)).to(device)
+        self.meta = torch.from_numpy(packed["meta"].reshape(-1)).to(device)
+        self.n_dfas = packed["meta"].shape[0]
+        self.mdfa = mdfa
+
+
+_device_dfas: Dict[Tuple[str, int], DeviceDFA] = {}
+
+
+def device_family(name: str, device) -> DeviceDFA:
+    dev = torch.device(device)
+    key = (name, dev.index if dev.index is not None else -1)
+    if key not in _device_dfas:
+        _device_dfas[key] = DeviceDFA(pattern_sets.get_family(name), dev)
+    return _device_dfas[key]
+
+
+def dfa_scan(bytes_t: torch.Tensor, offsets: torch.Tensor, family: str) -> torch.Tensor:
+    """Per-message u64 hit masks (as int64 tensor) for one pattern family."""
+    d = device_family(family, bytes_t.device)
+    meta2d = d.meta.view(d.n_dfas, 4)
+    return ext().dfa_scan(bytes_t, offsets, d.next, d.accept, d.eof, d.class_maps, meta2d)
+
+
+def reference_dfa_scan(messages: Sequence[bytes], family: str) -> np.ndarray:
+    mdfa = pattern_sets.get_family(family)
+    return np.array([mdfa.scan(m) for m in messages], dtype=np.uint64).view(np.int64)
+
+
+# -- SHA-256 / Merkle -------------------------------------------------------
+
+def sha256_leaves(bytes_t: torch.Tensor, offsets: torch.Tensor) -> torch.Tensor:
+    return ext().sha256_leaves(bytes_t, offsets)
+
+
+def merkle_root(digests: torch.Tensor) -> torch.Tensor:
+    return ext().merkle_root(digests)
+
+
+def reference_sha256_leaves(messages: Sequence[bytes]) -> np.ndarray:
+    return np.stack([np.frombuffer(hashlib.sha256(m).digest(), dtype=np.uint8) for m in messages])
+
+
+def reference_merkle_root(messages: Sequence[bytes]) -> bytes:
+    from ..governance.audit import merkle_root as cpu_root
+
+    return bytes.fromhex(cpu_root(list(messages)))
+
+
+# -- encoder ----------------------------------------------------------------
+
+def encode_messages(
+    bytes_t: torch.Tensor, offsets: torch.Tensor, embed: torch.Tensor, normalize: bool = True
+) -> torch.Tensor:
+    return ext().encode_messages(bytes_t, offsets, embed, normalize)
+
+
+def reference_encode(messages: Sequence[bytes], embed: np.ndarray, normalize: bool = True) -> np.ndarray:
+    """CPU reference of csrc/encoder.hip (fnv1a 4-gram, stride-subsampled
+    mean, L2 norm)."""
+    vocab, dim = embed.shape
+    out = np.zeros((len(messages), dim), dtype=np.float32)
+    max_tokens = 512
+    for i, m in enumerate(messages):
+        n_pos = max(len(m) - 3, 0)
+        stride = max((n_pos + max_tokens - 1) // max_tokens, 1)
+        toks = []
+        for t in range((n_pos + stride - 1) // stride if n_pos else 0):
+            p = t * stride
+            h = np.uint32(2166136261)
+            for b in m[p : p + 4]:
+                h = np.uint32((np.uint32(h ^ np.uint32(b)) * np.uint32(16777619)) & 0xFFFFFFFF)
+            toks.append(int(h) & (vocab - 1))
+        if toks:
+            f = embed[toks].astype(np.float32).mean(axis=0)
+        else:
+            f = np.zeros(dim, dtype=np.float32)
+        if normalize:
+            n = np.sqrt(max((f * f).sum(), 1e-12))
+            f = f / n
+        out[i] = f
+    return out
+
+
+# -- GEMM / top-k -----------------------------------------------------------
+
+def gemm_nt(
+    A: torch.Tensor,
+    B: torch.Tensor,
+    bias: Optional[torch.Tensor] = None,
+    act: int = 0,
+    out_bf16: bool = False,
+) -> torch.Tensor:
+    return ext().gemm_nt(A, B, bias, act, out_bf16)
+
+
+def topk_recall(Q: torch.Tensor, X: torch.Tensor, k: int, n_swaths: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:
+    if n_swaths <= 0:
+        # enough blocks to fill the chip: 256 CUs / qblocks, >=1
+        qblocks = (Q.shape[0] + 127) // 128
+        n_swaths = max(1, min(512 // max(qblocks, 1), 64))
+        n_swaths = min(n_swaths, max(1, X.shape[0] // 128))
+    s, i = ext().topk_recall(Q, X, k, n_swaths)
+    return s, i
+
+
+# -- firewall tail ----------------------------------------------------------
+
+def firewall_verdict(
+    inj_hits: torch.Tensor,
+    red_hits: torch.Tensor,
+    logits: torch.Tensor,
+    agent_idx: torch.Tensor,
+    agent_trust: torch.Tensor,
+    tool_risk: torch.Tensor,
+    freq_count: torch.Tensor,
+    hour: int,
+    n_agents: int,
+    cred_bits: int = pattern_sets.REDACTION_CREDENTIAL_BITS,
+    inj_threshold: float = 0.9,
+):
+    v, r, sd, vd = ext().firewall_verdict(
+        inj_hits, red_hits, logits, agent_idx, agent_trust, tool_risk, freq_count,
+        hour, cred_bits, inj_threshold, n_agents,
+    )
+    return v, r, sd, vd
+
+
+def trust_recompute(state: Dict[str, torch.Tensor], sdelta: torch.Tensor, vdelta: torch.Tensor) -> None:
+    ext().trust_recompute(
+        state["success"], state["violation"], sdelta, vdelta,
+        state["age_days"], state["clean_streak"], state["manual_adj"], state["score"],
+    )
+
+
+def audit_pack(
+    verdict: torch.Tensor, risk: torch.Tensor, inj_hits: torch.Tensor, red_hits: torch.Tensor,
+    agent_idx: torch.Tensor, agent_trust: torch.Tensor, inj_score: torch.Tensor,
+    ts_ms: int, msg_id0: int, batch_seq: int,
+) -> torch.Tensor:
+    return ext().audit_pack(
+        verdict, risk, inj_hits, red_hits, agent_idx, agent_trust, inj_score,
+        ts_ms, msg_id0, batch_seq,
+    )

@@ -1,0 +1,268 @@
+"""The batched GPU firewall pipeline: Governance -> Membrane -> Cortex/KE.
+
+This is the MI355X-native high-throughput mode of the suite (the Python
+plugin engines in governance/, cortex/, knowledge/ are the per-message
+interactive mode with identical semantics). Per step, for a batch of
+messages:
+
+  1. pack + H2D copy of message bytes
+  2. DFA scans (csrc/pattern_scan.hip): redaction, injection, claims,
+     entity families -> u64 hit masks per message
+  3. encoder (csrc/encoder.hip): 4-gram hash features [B, D] bf16
+  4. classifier head (csrc/gemm_nt.hip, fused sigmoid): injection /
+     threat logits
+  5. Membrane recall (csrc/topk_recall.hip): cosine top-k against the
+     HBM-resident embedding index shard
+  6. firewall verdict + trust updates (csrc/firewall.hip) implementing
+     the reference's risk weights / verdict precedence / trust formula
+  7. audit: 64-B records packed on GPU, SHA-256 leaves + Merkle root
+     (csrc/sha256_merkle.hip)
+
+Multi-GPU (one process per GPU, torch.distributed over RCCL/xGMI):
+the index is sharded across ranks; recall queries are all-gathered so
+every message searches the FULL index, per-rank top-k candidates are
+all-gathered and merged back, and the per-batch Merkle roots are combined
+into a global root. Per-GPU work is constant as ranks grow (weak scaling):
+each rank scores (world * batch) queries against (total / world) index
+rows.
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from dataclasses import dataclass
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..ops import gpu as g
+from ..ops import pattern_sets
+from .synth import SynthBatch
+
+
+@dataclass
+class PipelineConfig:
+    batch: int = 4096
+    dim: int = 1024
+    vocab: int = 65536
+    n_classes: int = 8
+    n_agents: int = 64
+    index_size: int = 6_250_000  # per-GPU shard (8 GPUs x 6.25M = 50M)
+    topk: int = 16
+    inj_threshold: float = 0.9
+    seed: int = 1234
+    families: tuple = ("redaction", "injection", "claims", "entity")
+
+
+class FirewallPipeline:
+    def __init__(self, cfg: PipelineConfig, device: str = "cuda:0", world_size: int = 1, rank: int = 0):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        self.world_size = world_size
+        self.rank = rank
+        self.batch_seq = 0
+        torch.manual_seed(cfg.seed + rank)
+
+        with torch.cuda.device(self.device):
+            # model state (random init; no network for checkpoints)
+            self.embed = (torch.randn(cfg.vocab, cfg.dim, dtype=torch.float32, device=self.device) * 0.05).bfloat16()
+            self.head = (torch.randn(cfg.n_classes, cfg.dim, dtype=torch.float32, device=self.device) * 0.05).bfloat16()
+            self.head_bias = torch.zeros(cfg.n_classes, device=self.device)
+
+            # Membrane index shard: L2-normalized rows, built in chunks to
+            # bound peak memory during init
+            chunks = []
+            chunk = 1_000_000
+            gen = torch.Generator(device="cuda")
+            gen.manual_seed(cfg.seed * 7919 + rank)
+            for i in range(0, cfg.index_size, chunk):
+                n = min(chunk, cfg.index_size - i)
+                x = torch.randn(n, cfg.dim, generator=gen, dtype=torch.float32, device=self.device)
+                x = torch.nn.functional.normalize(x, dim=1)
+                chunks.append(x.bfloat16())
+            self.index = torch.cat(chunks, dim=0) if len(chunks) > 1 else chunks[0]
+            del chunks
+
+            # salience state: recall strength + decay (Membrane semantics)
+            self.salience = torch.ones(cfg.index_size, device=self.device)
+
+            # trust state vectors (mirrors governance TrustManager fields)
+            A = cfg.n_agents
+            self.trust_state = {
+                "success": torch.zeros(A, device=self.device),
+                "violation": torch.zeros(A, device=self.device),
+                "age_days": torch.zeros(A, device=self.device),
+                "clean_streak": torch.zeros(A, device=self.device),
+                "manual_adj": torch.full((A,), 40.0, device=self.device),
+                "score": torch.full((A,), 40.0, device=self.device),
+            }
+            # warm the DFA tables onto the device
+            for fam in cfg.families:
+                g.device_family(fam, self.device)
+
+        self.audit_sink: Optional[Any] = None  # callable(records_cpu, root_hex)
+
+    # -- input staging -----------------------------------------------------
+    def stage(self, batch: SynthBatch) -> Dict[str, torch.Tensor]:
+        b, o = g.pack_messages(batch.messages, device=self.device)
+        return {
+            "bytes": b,
+            "offsets": o,
+            "agent_idx": torch.from_numpy(batch.agent_idx).to(self.device),
+            "tool_risk": torch.from_numpy(batch.tool_risk).to(self.device),
+        }
+
+    # -- the step ----------------------------------------------------------
+    def step(self, batch, staged: Optional[Dict[str, torch.Tensor]] = None) -> Dict[str, Any]:
+        cfg = self.cfg
+        s = staged if staged is not None else self.stage(batch)
+        bytes_t, offsets = s["bytes"], s["offsets"]
+        agent_idx, tool_risk = s["agent_idx"], s["tool_risk"]
+        B = offsets.numel() - 1
+
+        # 2. pattern scans
+        hits = {fam: g.dfa_scan(bytes_t, offsets, fam) for fam in cfg.families}
+
+        # 3. encoder
+        feats = g.encode_messages(bytes_t, offsets, self.embed, normalize=True)
+
+        # 4. classifier head (fused sigmoid)
+        logits = g.gemm_nt(feats, self.head, bias=self.head_bias, act=1)
+
+        # 5. Membrane recall (full index across ranks)
+        if self.world_size > 1 and torch.distributed.is_initialized():
+            q_all = torch.empty(
+                self.world_size * B, cfg.dim, dtype=torch.bfloat16, device=self.device
+            )
+            torch.distributed.all_gather_into_tensor(q_all, feats)
+            scores, ids = g.topk_recall(q_all, self.index, cfg.topk)
+            ids = ids + self.rank * cfg.index_size  # globalize shard ids
+            # gather every rank's candidates, keep my rows, merge
+            cand_s = torch.empty(self.world_size, q_all.shape[0], cfg.topk, device=self.device)
+            cand_i = torch.empty(
+                self.world_size, q_all.shape[0], cfg.topk, dtype=torch.int32, device=self.device
+            )
+            torch.distributed.all_gather_into_tensor(cand_s, scores.unsqueeze(0).contiguous())
+            torch.distributed.all_gather_into_tensor(cand_i, ids.unsqueeze(0).contiguous())
+            my0 = self.rank * B
+            mine_s = cand_s[:, my0 : my0 + B].permute(1, 0, 2).reshape(B, -1)
+            mine_i = cand_i[:, my0 : my0 + B].permute(1, 0, 2).reshape(B, -1)
+            top = torch.topk(mine_s, cfg.topk, dim=1)
+            recall_scores = top.values
+            recall_ids = torch.gather(mine_i, 1, top.indices)
+        else:
+            recall_scores, recall_ids = g.topk_recall(feats, self.index, cfg.topk)
+
+        # salience reinforcement + decay (Membrane recall semantics)
+        flat_local = recall_ids.reshape(-1)
+        if self.world_size > 1:
+            base = self.rank * cfg.index_size
+            local_mask = (flat_local >= base) & (flat_local < base + cfg.index_size)
+            flat_local = flat_local[local_mask] - base
+        self.salience.mul_(0.9999)
+        self.salience.index_add_(
+            0, flat_local.long(), torch.full((flat_local.numel(),), 0.01, device=self.device)
+        )
+
+        # 6. verdict + trust
+        hour = time.localtime().tm_hour
+        freq = torch.bincount(agent_idx.long(), minlength=cfg.n_agents).to(torch.int32)
+        freq_count = freq[agent_idx.long()]
+        verdict, risk, sdelta, vdelta = g.firewall_verdict(
+            hits["injection"], hits["redaction"], logits, agent_idx.to(torch.int32),
+            self.trust_state["score"], tool_risk, freq_count, hour, cfg.n_agents,
+            inj_threshold=cfg.inj_threshold,
+        )
+        g.trust_recompute(self.trust_state, sdelta, vdelta)
+
+        # 7. audit Merkle
+        inj_score = logits.max(dim=1).values
+        records = g.audit_pack(
+            verdict, risk, hits["injection"], hits["redaction"], agent_idx.to(torch.int32),
+            self.trust_state["score"], inj_score,
+            ts_ms=int(time.time() * 1000), msg_id0=self.batch_seq * B, batch_seq=self.batch_seq,
+        )
+        flat = records.reshape(-1)
+        offs64 = torch.arange(0, (B + 1) * 64, 64, dtype=torch.int32, device=self.device)
+        leaves = g.sha256_leaves(flat, offs64)
+        root = g.merkle_root(leaves)
+
+        if self.world_size > 1 and torch.distributed.is_initialized():
+            roots = torch.empty(self.world_size, 32, dtype=torch.uint8, device=self.device)
+            torch.distributed.all_gather_into_tensor(roots, root.unsqueeze(0))
+            root = g.merkle_root(roots)
+
+        self.batch_seq += 1
+        if self.audit_sink is not None:
+            self.audit_sink(records, root)
+
+        return {
+            "verdict": verdict,
+            "risk": risk,
+            "logits": logits,
+            "hits": hits,
+            "features": feats,
+            "recall_scores": recall_scores,
+            "recall_ids": recall_ids,
+            "merkle_root": root,
+            "trust_scores": self.trust_state["score"],
+        }
+
+
+class AsyncAuditWriter:
+    """Background writer: binary audit records + JSONL batch manifest with
+    the Merkle chain (byte format shared with governance.audit)."""
+
+    def __init__(self, audit_dir: str):
+        import os
+
+        self.audit_dir = audit_dir
+        os.makedirs(audit_dir, exist_ok=True)
+        self._q: List = []
+        self._cv = threading.Condition()
+        self._stop = False
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+        self.batches_written = 0
+
+    def __call__(self, records: torch.Tensor, root: torch.Tensor) -> None:
+        rec_cpu = records.to("cpu", non_blocking=True)
+        root_cpu = root.to("cpu", non_blocking=True)
+        with self._cv:
+            self._q.append((rec_cpu, root_cpu, time.time()))
+            self._cv.notify()
+
+    def _run(self) -> None:
+        import json
+        import os
+
+        bin_path = os.path.join(self.audit_dir, "audit-records.bin")
+        manifest = os.path.join(self.audit_dir, "audit-manifest.jsonl")
+        while True:
+            with self._cv:
+                while not self._q and not self._stop:
+                    self._cv.wait(0.2)
+                if self._stop and not self._q:
+                    return
+                items = self._q
+                self._q = []
+            with open(bin_path, "ab") as bf, open(manifest, "a") as mf:
+                for rec, root, ts in items:
+                    torch.cuda.synchronize()
+                    raw = rec.numpy().tobytes()
+                    bf.write(raw)
+                    mf.write(
+                        json.dumps(
+                            {"ts": int(ts * 1000), "count": rec.shape[0], "root": bytes(root.numpy()).hex()}
+                        )
+                        + "\n"
+                    )
+                    self.batches_written += 1
+
+    def close(self) -> None:
+        with self._cv:
+            self._stop = True
+            self._cv.notify()
+        self._thread.join(timeout=5.0)
