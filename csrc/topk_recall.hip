@@ -1,172 +1,205 @@
 // Fused cosine-scores + streaming top-k: the Membrane salience recall
-// kernel.
+// kernel (v2).
 //
 // Replaces the reference Membrane plugin's salience retrieval (external
 // repo; config surface in brainplex configurator.ts:137-148) with an
-// LDS-tiled MFMA scan over the HBM-resident embedding shard: for each
-// 128-query block the kernel walks its swath of the [N, D] bf16 index,
-// computes a 128x128 score tile per step (queries and index rows are
-// L2-normalized, so dot = cosine), and maintains per-query top-k lists in
-// LDS with a running-threshold filter — scores are NEVER materialized to
-// HBM (a 4096 x 50M f32 score matrix would be 800 GB of traffic per
-// step). A second small kernel merges the per-swath candidate lists.
+// LDS-tiled MFMA scan over the HBM-resident embedding shard. Scores are
+// never materialized to HBM (4096 x 50M f32 would be 800 GB of traffic
+// per step): each 128-query block walks its swath of the [N, D] bf16
+// index, computes 128x256 score tiles with v_mfma_f32_16x16x32_bf16 and
+// filters them against per-row running top-k thresholds straight from the
+// accumulator registers.
 //
-// Grid: (Q/128) x n_swaths. Each block scans N/n_swaths index rows.
-// LDS: staging tiles + score tile + top-k lists ~= 100 KB -> 1 block/CU;
-// the 16 independent MFMA accumulators keep the matrix pipe busy at one
-// wave/SIMD.
+// Structure (cdna_hip_programming.md §5 "step 3"):
+// - 8 waves (512 thr), block tile BM=128 x BN=256, BK=32, wave grid 2x4
+//   (each wave a 64x64 subtile = 4x4 fragments, 16 independent
+//   accumulators keep the matrix pipe busy at 2 waves/SIMD).
+// - global_load_lds width 16 staging, LDS image lane-linear with the
+//   XOR swizzle applied on the SOURCE k-group and again on the ds_read
+//   offset (guide rule 21): slot ^= (row>>2)&3 makes the 16-lane
+//   ds_read_b128 groups conflict-free without row padding.
+// - double-buffered LDS, one barrier pair per K-step (the deep-pipelined
+//   8-phase schedule is the next rung once this baseline is profiled).
+// - top-k: per-row threshold in LDS; lanes test their 16 accumulator
+//   values (C/D map row=(lane>>4)*4+r, col=lane&15), survivors go through
+//   a bounded LDS candidate queue drained by row-owning lanes.
 #include "common.hpp"
 
 #define BM 128
-#define BN 128
+#define BN 256
 #define BK 32
-#define PAD 8
-#define LDS_STRIDE (BK + PAD)
-#define TK_THREADS 256
+#define TK_THREADS 512
 #define TOPK_MAX 32
+#define QCAP 2048
 
-DEVINL void stage_tile_tk(const bf16* __restrict__ src, int ld, int row0,
-                          int rows, int k0, bf16* lds) {
-  int tid = threadIdx.x;
-  int r = tid >> 1;
-  int half = (tid & 1) * 16;
-  bf16x8 v0 = {}, v1 = {};
-  int gr = row0 + r;
-  if (gr < rows) {
-    const bf16* p = src + (size_t)gr * ld + k0 + half;
-    v0 = *(const bf16x8*)(p);
-    v1 = *(const bf16x8*)(p + 8);
-  }
-  *(bf16x8*)(lds + r * LDS_STRIDE + half) = v0;
-  *(bf16x8*)(lds + r * LDS_STRIDE + half + 8) = v1;
+// LDS tile addressing: row-major [rows][BK] bf16, 64 B per row = 4 slots
+// of 16 B. Swizzle: slot' = slot ^ ((row>>2)&3).
+DEVINL uint32_t lds_off_bytes(uint32_t row, uint32_t slot) {
+  return (row * 4u + (slot ^ ((row >> 2u) & 3u))) * 16u;
 }
 
-// candidates layout: [n_qblocks][n_swaths][BM][k] for scores f32 and ids i32
+// Stage a [rows x BK] tile via global_load_lds: thread t writes LDS bytes
+// [t*16, t*16+16) = row t/4, slot t%4; the matching SOURCE k-group is the
+// swizzled slot (involution).
+DEVINL void stage_glds(const bf16* __restrict__ src, long long ld,
+                       long long row0, long long row_max, int k0,
+                       bf16* lds_base, int tile_rows) {
+  int t = threadIdx.x;
+  int n_lanes = tile_rows * 4;  // 16B pieces in the tile
+  for (int piece = t; piece < n_lanes; piece += TK_THREADS) {
+    uint32_t r = piece >> 2;
+    uint32_t slot = piece & 3;
+    uint32_t src_slot = slot ^ ((r >> 2u) & 3u);
+    long long gr = row0 + r;
+    if (gr >= row_max) gr = row_max - 1;  // clamp: garbage filtered later
+    const bf16* p = src + gr * ld + k0 + src_slot * 8;
+    auto gsrc = (const __attribute__((address_space(1))) char*)p;
+    auto ldst = (__attribute__((address_space(3))) char*)lds_base + piece * 16;
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)gsrc,
+                                     (__attribute__((address_space(3))) void*)ldst, 16, 0, 0);
+  }
+}
+
 extern "C" __global__ void __launch_bounds__(TK_THREADS)
 topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
                    int nq, int nx, int D, int k, int n_swaths,
                    float* __restrict__ cand_scores,
                    int32_t* __restrict__ cand_ids) {
-  __shared__ bf16 Qs[2][BM * LDS_STRIDE];
-  __shared__ bf16 Xs[2][BN * LDS_STRIDE];
-  __shared__ float scores[BM][BN + 1];        // +1: avoid column-bank alignment
+  __shared__ bf16 lds_all[2 * (BM + BN) * BK];
+#define QS(buf) (lds_all + (buf) * BM * BK)
+#define XS(buf) (lds_all + 2 * BM * BK + (buf) * BN * BK)
   __shared__ float topk_s[BM][TOPK_MAX];
   __shared__ int32_t topk_i[BM][TOPK_MAX];
   __shared__ float row_min[BM];
   __shared__ int row_min_slot[BM];
+  __shared__ float q_score[QCAP];
+  __shared__ uint32_t q_meta[QCAP];  // (row<<16) | col_in_tile
+  __shared__ int q_count;
+  __shared__ int q_overflow;
 
   int qb = blockIdx.x;
   int swath = blockIdx.y;
-  int row0 = qb * BM;
+  long long row0 = (long long)qb * BM;
 
-  // swath range over the index
   long long per = ((long long)nx + n_swaths - 1) / n_swaths;
+  per = ((per + BN - 1) / BN) * BN;
   long long x_begin = (long long)swath * per;
   long long x_end = min((long long)nx, x_begin + per);
 
-  // init top-k lists
-  for (int i = threadIdx.x; i < BM * k; i += blockDim.x) {
-    topk_s[i / k][i % k] = -1e30f;
-    topk_i[i / k][i % k] = -1;
+  for (int i = threadIdx.x; i < BM * TOPK_MAX; i += blockDim.x) {
+    topk_s[i / TOPK_MAX][i % TOPK_MAX] = -1e30f;
+    topk_i[i / TOPK_MAX][i % TOPK_MAX] = -1;
   }
   for (int i = threadIdx.x; i < BM; i += blockDim.x) {
     row_min[i] = -1e30f;
     row_min_slot[i] = 0;
   }
+  if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
   __syncthreads();
 
-  int wid = wave_id();
-  int wm = wid >> 1, wn = wid & 1;
+  int wid = wave_id();          // 0..7
+  int wm = wid >> 2, wn = wid & 3;  // 2 x 4 wave grid
   int lane = lane_id();
   int lrow = lane & 15;
-  int kgrp = (lane >> 4) * 8;
+  int kslot = lane >> 4;        // 0..3 -> k-group (8 bf16 = 16 B)
   int nk = D / BK;
 
   for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
     f32x4 acc[4][4] = {};
-    stage_tile_tk(Q, D, row0, nq, 0, Qs[0]);
-    stage_tile_tk(X, D, (int)x0, (int)x_end, 0, Xs[0]);
+    stage_glds(Q, D, row0, nq, 0, QS(0), BM);
+    stage_glds(X, D, x0, (long long)nx, 0, XS(0), BN);
+    asm volatile("s_waitcnt vmcnt(0)");
     __syncthreads();
     for (int kt = 0; kt < nk; ++kt) {
       int cur = kt & 1, nxt = cur ^ 1;
       if (kt + 1 < nk) {
-        stage_tile_tk(Q, D, row0, nq, (kt + 1) * BK, Qs[nxt]);
-        stage_tile_tk(X, D, (int)x0, (int)x_end, (kt + 1) * BK, Xs[nxt]);
+        stage_glds(Q, D, row0, nq, (kt + 1) * BK, QS(nxt), BM);
+        stage_glds(X, D, x0, (long long)nx, (kt + 1) * BK, XS(nxt), BN);
       }
       bf16x8 qf[4], xf[4];
 #pragma unroll
-      for (int m = 0; m < 4; ++m)
-        qf[m] = *(const bf16x8*)(Qs[cur] + (wm * 64 + m * 16 + lrow) * LDS_STRIDE + kgrp);
+      for (int m = 0; m < 4; ++m) {
+        uint32_t r = wm * 64 + m * 16 + lrow;
+        qf[m] = *(const bf16x8*)((const char*)QS(cur) + lds_off_bytes(r, kslot));
+      }
 #pragma unroll
-      for (int n = 0; n < 4; ++n)
-        xf[n] = *(const bf16x8*)(Xs[cur] + (wn * 64 + n * 16 + lrow) * LDS_STRIDE + kgrp);
+      for (int n = 0; n < 4; ++n) {
+        uint32_t r = wn * 64 + n * 16 + lrow;
+        xf[n] = *(const bf16x8*)((const char*)XS(cur) + lds_off_bytes(r, kslot));
+      }
 #pragma unroll
       for (int m = 0; m < 4; ++m)
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[m], xf[n], acc[m][n], 0, 0, 0);
+      asm volatile("s_waitcnt vmcnt(0)");
       __syncthreads();
     }
 
-    // scores -> LDS (C/D map: col=lane&15, row=(lane>>4)*4+reg)
-#pragma unroll
-    for (int m = 0; m < 4; ++m)
-#pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        int col = wn * 64 + n * 16 + lrow;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int row = wm * 64 + m * 16 + (lane >> 4) * 4 + r;
-          scores[row][col] = acc[m][n][r];
+    // ---- streaming top-k from the accumulators -------------------------
+    // lane holds acc[m][n][r] at row = wm*64+m*16+(lane>>4)*4+r,
+    //                         col = wn*64+n*16+(lane&15)
+    // Rounds: push survivors into the queue; drain by row-owner lanes;
+    // repeat if the queue overflowed (only plausible on the first tile).
+    // pending bit (m*16 + n*4 + r): value not yet pushed/retired
+    unsigned long long pending = ~0ull;
+    for (int round = 0; ; ++round) {
+      unsigned long long still = 0ull;
+      unsigned long long todo = pending;
+      while (todo) {
+        int vi = __ffsll((long long)todo) - 1;
+        todo &= todo - 1;
+        int m = vi >> 4, n = (vi >> 2) & 3, r = vi & 3;
+        int row = wm * 64 + m * 16 + (lane >> 4) * 4 + r;
+        if ((row0 + row) >= nq) continue;
+        long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+        if (col >= x_end) continue;
+        float v = acc[m][n][r];
+        if (!(v > row_min[row])) continue;
+        int idx = atomicAdd(&q_count, 1);
+        if (idx < QCAP) {
+          q_score[idx] = v;
+          q_meta[idx] = (uint32_t(row) << 16) | uint32_t(col - x0);
+        } else {
+          still |= 1ull << vi;  // retry just this value next round
+          atomicExch(&q_overflow, 1);
         }
       }
-    __syncthreads();
-
-    // streaming top-k update: each wave owns 32 rows; per row the wave's
-    // 64 lanes scan 128 scores (2 each) against the row threshold, then
-    // lane 0 serially inserts the (rare) survivors.
-    for (int rr = 0; rr < 32; ++rr) {
-      int row = wid * 32 + rr;
-      float th = row_min[row];
-      float s0 = scores[row][lane];
-      float s1 = scores[row][lane + 64];
-      bool valid0 = (x0 + lane) < x_end;
-      bool valid1 = (x0 + lane + 64) < x_end;
-      bool c0 = valid0 && s0 > th;
-      bool c1 = valid1 && s1 > th;
-      unsigned long long b0 = __ballot(c0);
-      unsigned long long b1 = __ballot(c1);
-      if (b0 == 0 && b1 == 0) continue;
-      if (lane == 0) {
-        // serial insert by lane 0 via LDS scan (rare path)
-        for (int part = 0; part < 2; ++part) {
-          unsigned long long bits = part == 0 ? b0 : b1;
-          while (bits) {
-            int src_lane = __ffsll((long long)bits) - 1;
-            bits &= bits - 1;
-            int col = src_lane + part * 64;
-            float sv = scores[row][col];
-            float mn = row_min[row];
-            if (sv > mn) {
-              int slot = row_min_slot[row];
-              topk_s[row][slot] = sv;
-              topk_i[row][slot] = (int32_t)(x0 + col);
-              // recompute min
-              float new_mn = topk_s[row][0];
-              int new_slot = 0;
-              for (int j = 1; j < k; ++j)
-                if (topk_s[row][j] < new_mn) { new_mn = topk_s[row][j]; new_slot = j; }
-              row_min[row] = new_mn;
-              row_min_slot[row] = new_slot;
-            }
+      __syncthreads();
+      // drain: lanes 0..15 of each wave own rows wid*16 + (lane)
+      int total = min(q_count, QCAP);
+      if (lane < 16) {
+        int my_row = wid * 16 + lane;
+        for (int i = 0; i < total; ++i) {
+          uint32_t meta = q_meta[i];
+          int row = int(meta >> 16);
+          if (row != my_row) continue;
+          float v = q_score[i];
+          if (v > row_min[row]) {
+            int slot = row_min_slot[row];
+            topk_s[row][slot] = v;
+            topk_i[row][slot] = int32_t(x0 + (meta & 0xFFFFu));
+            float mn = topk_s[row][0];
+            int ms = 0;
+            for (int j = 1; j < k; ++j)
+              if (topk_s[row][j] < mn) { mn = topk_s[row][j]; ms = j; }
+            row_min[row] = mn;
+            row_min_slot[row] = ms;
           }
         }
       }
+      __syncthreads();
+      int of = q_overflow;
+      __syncthreads();  // all reads of q_overflow done before the reset
+      if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
+      pending = still;
+      if (!of) break;   // of is uniform (LDS): no divergence
+      __syncthreads();  // reset visible before next round's pushes
     }
     __syncthreads();
   }
 
-  // write candidates: [qb][swath][row][k]
+  // candidates out: [qb][swath][row][k]
   size_t base = (((size_t)qb * n_swaths) + swath) * BM * k;
   for (int i = threadIdx.x; i < BM * k; i += blockDim.x) {
     cand_scores[base + i] = topk_s[i / k][i % k];
@@ -183,10 +216,7 @@ extern "C" __global__ void topk_merge_kernel(
   if (q >= nq) return;
   int lane = lane_id();
   int qb = q / BM, row = q % BM;
-  int total = n_swaths * k;  // candidates for this query
-  // lane-local partial top-k via serial selection in registers:
-  // simple approach: k rounds of argmax over remaining (total <= 32*16=512)
-  // Each lane holds ceil(total/64) candidates.
+  int total = n_swaths * k;
   float my_s[16];
   int32_t my_i[16];
   int per_lane = (total + WAVE - 1) / WAVE;
@@ -203,12 +233,10 @@ extern "C" __global__ void topk_merge_kernel(
     }
   }
   for (int sel = 0; sel < k; ++sel) {
-    // local max
     float best = -1e30f;
     int bj = -1;
     for (int j = 0; j < per_lane; ++j)
       if (my_s[j] > best) { best = my_s[j]; bj = j; }
-    // wave max reduce
     float wbest = best;
     int wlane = lane;
     for (int off = 32; off; off >>= 1) {
@@ -218,14 +246,10 @@ extern "C" __global__ void topk_merge_kernel(
     }
     wbest = __shfl(wbest, 0);
     wlane = __shfl(wlane, 0);
-    if (lane == wlane && bj >= 0) {
-      if (lane == 0 || true) {
-        // winner lane writes and retires its candidate
-      }
+    if (lane == wlane) {
       out_scores[(size_t)q * k + sel] = wbest;
-      out_ids[(size_t)q * k + sel] = my_i[bj];
-      my_s[bj] = -1e30f;
+      out_ids[(size_t)q * k + sel] = (bj >= 0) ? my_i[bj] : -1;
+      if (bj >= 0) my_s[bj] = -1e30f;
     }
-    __syncthreads();
   }
 }
