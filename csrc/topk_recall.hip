@@ -38,25 +38,40 @@ DEVINL uint32_t lds_off_bytes(uint32_t row, uint32_t slot) {
   return (row * 4u + (slot ^ ((row >> 2u) & 3u))) * 16u;
 }
 
-// Stage a [rows x BK] tile via global_load_lds: thread t writes LDS bytes
-// [t*16, t*16+16) = row t/4, slot t%4; the matching SOURCE k-group is the
-// swizzled slot (involution).
+// Stage a [rows x BK] tile into LDS. Piece p (16 B) = row p/4, slot p%4;
+// the SOURCE k-group is the swizzled slot (involution with the read side).
+//
+// USE_GLDS=1: global_load_lds — the LDS operand must be the WAVE-UNIFORM
+// base of the wave's 64 consecutive pieces (the hardware adds lane*16;
+// a per-lane destination silently scatters / faults). The per-lane part
+// lives only in the global SOURCE address.
+#ifndef USE_GLDS
+#define USE_GLDS 1
+#endif
+
 DEVINL void stage_glds(const bf16* __restrict__ src, long long ld,
                        long long row0, long long row_max, int k0,
                        bf16* lds_base, int tile_rows) {
-  int t = threadIdx.x;
-  int n_lanes = tile_rows * 4;  // 16B pieces in the tile
-  for (int piece = t; piece < n_lanes; piece += TK_THREADS) {
+  int n_pieces = tile_rows * 4;  // 16B pieces in the tile
+  int w = wave_id();
+  int lane = lane_id();
+  for (int piece0 = w * WAVE; piece0 < n_pieces; piece0 += (TK_THREADS / WAVE) * WAVE) {
+    int piece = piece0 + lane;
     uint32_t r = piece >> 2;
     uint32_t slot = piece & 3;
     uint32_t src_slot = slot ^ ((r >> 2u) & 3u);
     long long gr = row0 + r;
     if (gr >= row_max) gr = row_max - 1;  // clamp: garbage filtered later
     const bf16* p = src + gr * ld + k0 + src_slot * 8;
-    auto gsrc = (const __attribute__((address_space(1))) char*)p;
-    auto ldst = (__attribute__((address_space(3))) char*)lds_base + piece * 16;
-    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)gsrc,
-                                     (__attribute__((address_space(3))) void*)ldst, 16, 0, 0);
+#if USE_GLDS
+    int piece0_u = __builtin_amdgcn_readfirstlane(piece0);
+    auto ldst = (__attribute__((address_space(3))) char*)lds_base + piece0_u * 16;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)p,
+        (__attribute__((address_space(3))) void*)ldst, 16, 0, 0);
+#else
+    *(bf16x8*)((char*)lds_base + piece * 16) = *(const bf16x8*)p;
+#endif
   }
 }
 
@@ -108,7 +123,7 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
     f32x4 acc[4][4] = {};
     stage_glds(Q, D, row0, nq, 0, QS(0), BM);
     stage_glds(X, D, x0, (long long)nx, 0, XS(0), BN);
-    asm volatile("s_waitcnt vmcnt(0)");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     for (int kt = 0; kt < nk; ++kt) {
       int cur = kt & 1, nxt = cur ^ 1;
@@ -132,7 +147,7 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[m], xf[n], acc[m][n], 0, 0, 0);
-      asm volatile("s_waitcnt vmcnt(0)");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
     }
 

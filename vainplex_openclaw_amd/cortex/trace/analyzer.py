@@ -1,0 +1,266 @@
+"""Trace analyzer: orchestrator, classifier, redactor, outputs, report,
+sources.
+
+Parity target: cortex `src/trace-analyzer/` —
+- analyzer.ts run(): connect source -> incremental fetch (lastProcessedTs
+  minus context window, `:171-194`) -> reconstruct -> detect -> classify
+  -> outputs -> report -> persist state (`:124-155`).
+- classifier.ts: optional triage LLM (keep? severity?) then analysis LLM,
+  chains redacted first (`:30-60`).
+- redactor.ts: strips credentials/PII from chains before any LLM call
+  (reuses the governance redaction registry).
+- output-generator.ts: group findings by normalized actionText ->
+  soul_rule / governance_policy / cortex_pattern with confidence
+  (`:13-27,36-50`).
+- report.ts: AnalysisReport + ProcessingState for incremental resume.
+- trace-source.ts / nats-trace-source.ts: source protocol; the NATS
+  source degrades to None gracefully when no client is available
+  (nats-trace-source.ts:1-12); a journal source reads the eventstore's
+  embedded journal.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+import uuid
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional
+
+from ...governance.redaction.engine import RedactionEngine
+from ...governance.redaction.registry import PatternRegistry
+from ...governance.redaction.vault import RedactionVault
+from ..storage import load_json, save_json
+from .chains import ConversationChain, reconstruct_chains
+from .events import NormalizedEvent, normalize_schema_a, normalize_schema_b
+from .signals import Finding, detect_all_signals
+
+
+# -- sources ----------------------------------------------------------------
+
+class MockTraceSource:
+    """Deterministic in-memory source (test helper parity:
+    openclaw-cortex/test/trace-analyzer/helpers.ts)."""
+
+    def __init__(self, events: Optional[List[NormalizedEvent]] = None):
+        self.events = list(events or [])
+
+    def fetch(self, since_ts: float = 0) -> List[NormalizedEvent]:
+        return [e for e in self.events if e.ts >= since_ts]
+
+
+class JournalTraceSource:
+    """Reads the embedded eventstore journal (Schema A envelopes)."""
+
+    def __init__(self, journal) -> None:
+        self.journal = journal  # eventstore.journal.EventJournal
+
+    def fetch(self, since_ts: float = 0) -> List[NormalizedEvent]:
+        out = []
+        for seq, env in self.journal.replay(since_ts=since_ts):
+            ne = normalize_schema_a(env, seq)
+            if ne:
+                out.append(ne)
+        return out
+
+
+def create_nats_source(url: Optional[str] = None):
+    """NATS JetStream source — returns None gracefully when no client is
+    available (nats-trace-source.ts:1-12). There is no network in this
+    environment; the journal source is the production path."""
+    return None
+
+
+# -- redactor ---------------------------------------------------------------
+
+class ChainRedactor:
+    def __init__(self) -> None:
+        self.engine = RedactionEngine(PatternRegistry(), RedactionVault())
+
+    def redact_chain(self, chain: ConversationChain) -> List[Dict[str, Any]]:
+        out = []
+        for ev in chain.events:
+            payload = dict(ev.payload)
+            for key in ("content", "toolError"):
+                if isinstance(payload.get(key), str):
+                    payload[key] = self.engine.scan_string(payload[key])["output"]
+            if isinstance(payload.get("toolParams"), dict):
+                payload["toolParams"] = self.engine.scan(payload["toolParams"])["output"]
+            out.append({"type": ev.type, "ts": ev.ts, "payload": payload})
+        return out
+
+
+# -- classifier -------------------------------------------------------------
+
+TRIAGE_PROMPT = (
+    "You triage agent-failure findings. Reply ONLY JSON: "
+    '{"keep": true|false, "severity": "low"|"medium"|"high"}\nFinding: '
+)
+ANALYSIS_PROMPT = (
+    "You analyze an agent failure. Reply ONLY JSON: "
+    '{"rootCause": "...", "actionText": "...", "confidence": 0.0}\n'
+)
+
+
+class FindingClassifier:
+    """Stage 2: optional triage + analysis LLM per finding (classifier.ts)."""
+
+    def __init__(
+        self,
+        call_llm: Optional[Callable[[str], str]] = None,
+        triage_enabled: bool = True,
+    ):
+        self.call_llm = call_llm
+        self.triage_enabled = triage_enabled
+        self.redactor = ChainRedactor()
+
+    @staticmethod
+    def _parse(raw: str) -> Optional[Dict[str, Any]]:
+        start, end = raw.find("{"), raw.rfind("}")
+        if start < 0 or end <= start:
+            return None
+        try:
+            out = json.loads(raw[start : end + 1])
+            return out if isinstance(out, dict) else None
+        except json.JSONDecodeError:
+            return None
+
+    def classify(self, findings: List[Finding], chains: Dict[str, ConversationChain]) -> List[Dict[str, Any]]:
+        out = []
+        for f in findings:
+            d = f.to_dict()
+            if self.call_llm is None:
+                d["actionText"] = default_action_text(f)
+                out.append(d)
+                continue
+            chain = chains.get(f.chain_id)
+            redacted = self.redactor.redact_chain(chain) if chain else []
+            context = json.dumps({"finding": d, "chain": redacted[-10:]}, default=str)[:4000]
+            try:
+                if self.triage_enabled:
+                    triage = self._parse(self.call_llm(TRIAGE_PROMPT + context)) or {}
+                    if triage.get("keep") is False:
+                        continue
+                    if triage.get("severity") in ("low", "medium", "high"):
+                        d["severity"] = triage["severity"]
+                analysis = self._parse(self.call_llm(ANALYSIS_PROMPT + context)) or {}
+                d["rootCause"] = analysis.get("rootCause", "")
+                d["actionText"] = analysis.get("actionText") or default_action_text(f)
+                if isinstance(analysis.get("confidence"), (int, float)):
+                    d["confidence"] = float(analysis["confidence"])
+            except Exception:
+                d["actionText"] = default_action_text(f)
+            out.append(d)
+        return out
+
+
+def default_action_text(f: Finding) -> str:
+    table = {
+        "doom_loop": f"Stop retrying {f.evidence.get('toolName', 'the tool')} after 2 consecutive identical failures; change approach instead",
+        "correction": "Re-read the user's original request before answering; confirm understanding on ambiguity",
+        "tool_fail": "Check tool preconditions before calling; validate inputs",
+        "dissatisfied": "Acknowledge the problem and summarize a concrete recovery plan",
+        "repeat_fail": f"Investigate the root cause of repeated {f.evidence.get('toolName', 'tool')} errors before retrying",
+        "hallucination": "Verify system state with a tool call before asserting it",
+        "unverified_claim": "Qualify claims or verify them with a tool call first",
+    }
+    return table.get(f.signal_type, "Review this failure pattern")
+
+
+# -- output generator -------------------------------------------------------
+
+def _norm_action(text: str) -> str:
+    return " ".join(text.lower().split())[:120]
+
+
+def generate_outputs(classified: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+    """Group by normalized actionText -> soul_rule / governance_policy /
+    cortex_pattern (output-generator.ts:13-50)."""
+    groups: Dict[str, List[Dict[str, Any]]] = {}
+    for c in classified:
+        groups.setdefault(_norm_action(c.get("actionText", "")), []).append(c)
+    outputs = []
+    for action, items in groups.items():
+        if not action:
+            continue
+        sig_types = {i["signalType"] for i in items}
+        confidence = min(0.95, max(i.get("confidence", 0.5) for i in items) + 0.1 * (len(items) - 1))
+        if sig_types & {"doom_loop", "repeat_fail", "tool_fail"}:
+            kind = "governance_policy"
+        elif sig_types & {"hallucination", "unverified_claim"}:
+            kind = "soul_rule"
+        else:
+            kind = "cortex_pattern"
+        outputs.append({
+            "id": f"out-{uuid.uuid4().hex[:10]}",
+            "kind": kind,
+            "actionText": items[0].get("actionText", ""),
+            "occurrences": len(items),
+            "agents": sorted({i["agent"] for i in items}),
+            "signalTypes": sorted(sig_types),
+            "confidence": round(confidence, 3),
+            "findingIds": [i["id"] for i in items],
+        })
+    outputs.sort(key=lambda o: -o["confidence"])
+    return outputs
+
+
+# -- report + orchestrator --------------------------------------------------
+
+@dataclass
+class AnalyzerConfig:
+    enabled: bool = True
+    detectors: Optional[List[str]] = None
+    incremental_context_window_ms: float = 30 * 60 * 1000
+    min_confidence: float = 0.0
+
+
+class TraceAnalyzer:
+    def __init__(
+        self,
+        workspace: str,
+        source,
+        config: Optional[AnalyzerConfig] = None,
+        call_llm: Optional[Callable[[str], str]] = None,
+        clock=time.time,
+    ):
+        self.workspace = workspace
+        self.source = source
+        self.config = config or AnalyzerConfig()
+        self.classifier = FindingClassifier(call_llm)
+        self.clock = clock
+        self.state_path = os.path.join(workspace, "memory", "reboot", "trace-analyzer-state.json")
+        self.report_path = os.path.join(workspace, "memory", "reboot", "trace-analysis-report.json")
+        self.state: Dict[str, Any] = load_json(self.state_path) or {}
+
+    def run(self) -> Dict[str, Any]:
+        """Full pipeline: fetch -> chains -> detect -> classify -> outputs
+        -> report -> persist state (analyzer.ts:124-155)."""
+        last_ts = float(self.state.get("lastProcessedTs", 0))
+        since = max(0.0, last_ts - self.config.incremental_context_window_ms)
+        events = self.source.fetch(since_ts=since)
+        chains = reconstruct_chains(events)
+        chain_map = {c.id: c for c in chains}
+        findings = detect_all_signals(chains, self.config.detectors)
+        classified = self.classifier.classify(findings, chain_map)
+        classified = [c for c in classified if c.get("confidence", 0) >= self.config.min_confidence]
+        outputs = generate_outputs(classified)
+        now_ms = self.clock() * 1000
+        report = {
+            "version": 1,
+            "generatedAt": int(now_ms),
+            "eventsAnalyzed": len(events),
+            "chains": len(chains),
+            "findings": classified,
+            "outputs": outputs,
+        }
+        self.state = {
+            "lastProcessedTs": max([e.ts for e in events], default=last_ts),
+            "lastProcessedSeq": max([e.seq for e in events], default=self.state.get("lastProcessedSeq", 0)),
+            "lastRunAt": int(now_ms),
+            "runsCompleted": int(self.state.get("runsCompleted", 0)) + 1,
+        }
+        save_json(self.report_path, report)
+        save_json(self.state_path, self.state)
+        return report
