@@ -1,0 +1,210 @@
+"""Trace analyzer: events, chains, 7 detectors, classifier, outputs, run()."""
+
+import pytest
+
+from vainplex_openclaw_amd.cortex.trace.analyzer import (
+    ChainRedactor,
+    FindingClassifier,
+    MockTraceSource,
+    TraceAnalyzer,
+    generate_outputs,
+)
+from vainplex_openclaw_amd.cortex.trace.chains import reconstruct_chains
+from vainplex_openclaw_amd.cortex.trace.events import (
+    NormalizedEvent,
+    normalize_schema_a,
+    normalize_schema_b,
+)
+from vainplex_openclaw_amd.cortex.trace.signals import (
+    detect_all_signals,
+    jaccard_similarity,
+    levenshtein,
+)
+
+_seq = [0]
+_ts = [1_000_000.0]
+
+
+def ev(etype, agent="main", session="s1", gap_ms=1000, **payload):
+    _seq[0] += 1
+    _ts[0] += gap_ms
+    return NormalizedEvent(
+        id=f"e{_seq[0]}", ts=_ts[0], agent=agent, session=session,
+        type=etype, payload=payload, seq=_seq[0],
+    )
+
+
+def tool_fail_pair(cmd="make build", err="exit 1", tool="exec"):
+    return [
+        ev("tool.call", toolName=tool, toolParams={"command": cmd}),
+        ev("tool.result", toolName=tool, toolError=err, toolIsError=True),
+    ]
+
+
+def test_normalize_schema_a_and_b():
+    env = {
+        "id": "evt-1", "type": "tool.call.before", "ts": 123,
+        "actor": {"id": "forge"}, "scope": {"sessionKey": "agent:forge:abc"},
+        "data": {"toolName": "exec", "params": {"command": "ls"}},
+    }
+    ne = normalize_schema_a(env)
+    assert ne.type == "tool.call" and ne.agent == "forge" and ne.session == "abc"
+    assert ne.payload["toolName"] == "exec"
+    nb = normalize_schema_b({"kind": "user_message", "ts": 5, "agentId": "m", "sessionId": "x", "body": {"text": "hi"}})
+    assert nb.type == "msg.in" and nb.payload["content"] == "hi"
+    assert normalize_schema_a({"type": "unknown.kind"}) is None
+
+
+def test_chain_reconstruction_splits():
+    events = []
+    events.append(ev("msg.in", content="start"))
+    events.append(ev("msg.out", content="ok"))
+    events.append(ev("msg.in", content="later", gap_ms=31 * 60 * 1000))  # 31-min gap
+    events.append(ev("session.start", gap_ms=100))  # lifecycle split
+    events.append(ev("msg.in", content="fresh"))
+    # different agent bucket
+    events.append(ev("msg.in", agent="other", content="x"))
+    chains = reconstruct_chains(events)
+    sessions = [(c.agent, len(c.events)) for c in chains]
+    assert len(chains) == 4
+    assert ("other", 1) in sessions
+    # dedupe by id
+    dup = events[0]
+    chains2 = reconstruct_chains(events + [dup])
+    assert sum(len(c.events) for c in chains2) == sum(len(c.events) for c in chains)
+
+
+def test_similarity_helpers():
+    assert jaccard_similarity({"a": 1, "timeout": 5}, {"a": 1, "timeout": 9}) == 1.0
+    assert jaccard_similarity({"a": 1}, {"a": 2}) == 0.0
+    assert levenshtein("kitten", "sitting") == 3
+    assert levenshtein("same", "same") == 0
+
+
+def test_doom_loop_detector():
+    events = [ev("msg.in", content="please build")]
+    for _ in range(4):
+        events.extend(tool_fail_pair("make build -j8"))
+    chains = reconstruct_chains(events)
+    findings = detect_all_signals(chains, ["doom_loop"])
+    assert len(findings) == 1
+    assert findings[0].signal_type == "doom_loop"
+    assert findings[0].evidence["count"] == 4
+    # dissimilar commands break the loop
+    events2 = [ev("msg.in", content="x")]
+    events2.extend(tool_fail_pair("make build"))
+    events2.extend(tool_fail_pair("rm -rf /tmp/cache && fetch artifacts --all"))
+    events2.extend(tool_fail_pair("curl http://somewhere/else/entirely"))
+    assert detect_all_signals(reconstruct_chains(events2), ["doom_loop"]) == []
+
+
+def test_correction_and_dissatisfied():
+    events = [
+        ev("msg.in", content="deploy please"),
+        ev("msg.out", content="deployed to prod-a"),
+        ev("msg.in", content="no, that's wrong — I wanted prod-b"),
+        ev("msg.in", content="this is not helpful at all, useless"),
+    ]
+    chains = reconstruct_chains(events)
+    types = {f.signal_type for f in detect_all_signals(chains, ["correction", "dissatisfied"])}
+    assert types == {"correction", "dissatisfied"}
+
+
+def test_tool_fail_and_repeat_fail():
+    events = [ev("msg.in", content="go")]
+    events.extend(tool_fail_pair("a b c", err="timeout"))
+    events.append(ev("tool.call", toolName="read", toolParams={"p": 1}))
+    events.append(ev("tool.result", toolName="read", toolResult="ok"))
+    events.extend(tool_fail_pair("x y z", err="timeout"))
+    events.extend(tool_fail_pair("q r s", err="timeout"))
+    chains = reconstruct_chains(events)
+    findings = detect_all_signals(chains, ["tool_fail", "repeat_fail"])
+    types = [f.signal_type for f in findings]
+    assert "tool_fail" in types
+    assert "repeat_fail" in types
+
+
+def test_hallucination_and_unverified():
+    events = [
+        ev("msg.out", content="nginx is running and healthy, all good"),
+        ev("tool.call", toolName="exec", toolParams={"command": "systemctl status nginx"}),
+        ev("tool.result", toolName="exec", toolError="nginx: service not found", toolIsError=True),
+        ev("msg.out", content="This will definitely work, 100% guaranteed."),
+    ]
+    chains = reconstruct_chains(events)
+    types = {f.signal_type for f in detect_all_signals(chains, ["hallucination", "unverified_claim"])}
+    assert "hallucination" in types
+    assert "unverified_claim" in types
+
+
+def test_redactor_strips_credentials():
+    events = [ev("msg.in", content="token ghp_" + "f" * 36 + " here")]
+    chains = reconstruct_chains(events)
+    red = ChainRedactor().redact_chain(chains[0])
+    assert "ghp_" + "f" * 36 not in str(red)
+
+
+def test_classifier_with_llm_triage():
+    events = [ev("msg.in", content="x")]
+    for _ in range(3):
+        events.extend(tool_fail_pair())
+    chains = reconstruct_chains(events)
+    findings = detect_all_signals(chains, ["doom_loop"])
+
+    def fake_llm(prompt):
+        if "triage" in prompt.lower() or '"keep"' in prompt:
+            return '{"keep": true, "severity": "high"}'
+        return '{"rootCause": "stale cache", "actionText": "clear the cache before retrying", "confidence": 0.9}'
+
+    out = FindingClassifier(fake_llm).classify(findings, {c.id: c for c in chains})
+    assert out[0]["actionText"] == "clear the cache before retrying"
+    assert out[0]["confidence"] == 0.9
+    # no-LLM path uses default action text
+    out2 = FindingClassifier(None).classify(findings, {})
+    assert out2[0]["actionText"]
+
+
+def test_output_grouping():
+    classified = [
+        {"id": "1", "signalType": "doom_loop", "agent": "a", "actionText": "Stop retrying X", "confidence": 0.8},
+        {"id": "2", "signalType": "doom_loop", "agent": "b", "actionText": "stop retrying  x", "confidence": 0.7},
+        {"id": "3", "signalType": "hallucination", "agent": "a", "actionText": "Verify before asserting", "confidence": 0.6},
+    ]
+    outputs = generate_outputs(classified)
+    assert len(outputs) == 2
+    gp = next(o for o in outputs if o["kind"] == "governance_policy")
+    assert gp["occurrences"] == 2 and set(gp["agents"]) == {"a", "b"}
+    assert any(o["kind"] == "soul_rule" for o in outputs)
+
+
+def test_analyzer_run_and_incremental_state(workspace):
+    events = [ev("msg.in", content="go")]
+    for _ in range(3):
+        events.extend(tool_fail_pair())
+    source = MockTraceSource(events)
+    analyzer = TraceAnalyzer(workspace, source)
+    report = analyzer.run()
+    assert report["eventsAnalyzed"] == len(events)
+    assert report["findings"]
+    assert report["outputs"]
+    assert analyzer.state["runsCompleted"] == 1
+    # second run resumes from state (context window re-reads the tail only)
+    analyzer2 = TraceAnalyzer(workspace, source)
+    report2 = analyzer2.run()
+    assert analyzer2.state["runsCompleted"] == 2
+    assert report2["eventsAnalyzed"] <= len(events)
+
+
+def test_trace_to_facts_bridge(workspace):
+    from vainplex_openclaw_amd.governance.facts import FactRegistry
+    from vainplex_openclaw_amd.governance.trace_to_facts import apply_report_to_registry
+
+    report = {"findings": [
+        {"id": "f1", "signalType": "hallucination", "confidence": 0.7,
+         "evidence": {"subject": "nginx", "predicate": "state", "value": "error"}},
+        {"id": "f2", "signalType": "correction", "confidence": 0.9, "evidence": {}},
+    ]}
+    reg = FactRegistry([])
+    n = apply_report_to_registry(report, reg)
+    assert n == 1
+    assert reg.lookup("nginx", "state")["value"] == "error"

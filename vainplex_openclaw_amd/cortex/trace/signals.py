@@ -308,7 +308,12 @@ def detect_unverified_claim(chain: ConversationChain) -> List[Finding]:
             continue
         content = str(ev.payload.get("content") or "")
         if _ABS_CLAIM_RX.search(content):
-            tool_before = any(e.type == "tool.result" for e in evs[max(0, i - 4):i])
+            # only a SUCCESSFUL tool result counts as verification
+            tool_before = any(
+                e.type == "tool.result"
+                and not (e.payload.get("toolError") or e.payload.get("toolIsError"))
+                for e in evs[max(0, i - 4):i]
+            )
             if not tool_before:
                 findings.append(_mk(chain, "unverified_claim", "low",
                                     "Absolute claim with no verifying tool call",
