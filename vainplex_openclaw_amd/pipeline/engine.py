@@ -71,19 +71,19 @@ class FirewallPipeline:
             self.head = (torch.randn(cfg.n_classes, cfg.dim, dtype=torch.float32, device=self.device) * 0.05).bfloat16()
             self.head_bias = torch.zeros(cfg.n_classes, device=self.device)
 
-            # Membrane index shard: L2-normalized rows, built in chunks to
-            # bound peak memory during init
-            chunks = []
+            # Membrane index shard: L2-normalized rows. Preallocate the bf16
+            # tensor and fill in chunks so peak extra memory is one fp32
+            # chunk (4 GB), not a second copy of the whole index.
             chunk = 1_000_000
             gen = torch.Generator(device="cuda")
             gen.manual_seed(cfg.seed * 7919 + rank)
+            self.index = torch.empty(cfg.index_size, cfg.dim, dtype=torch.bfloat16, device=self.device)
             for i in range(0, cfg.index_size, chunk):
                 n = min(chunk, cfg.index_size - i)
                 x = torch.randn(n, cfg.dim, generator=gen, dtype=torch.float32, device=self.device)
                 x = torch.nn.functional.normalize(x, dim=1)
-                chunks.append(x.bfloat16())
-            self.index = torch.cat(chunks, dim=0) if len(chunks) > 1 else chunks[0]
-            del chunks
+                self.index[i : i + n] = x.bfloat16()
+                del x
 
             # salience state: recall strength + decay (Membrane semantics)
             self.salience = torch.ones(cfg.index_size, device=self.device)
