@@ -158,7 +158,8 @@ std::vector<torch::Tensor> topk_recall(torch::Tensor Q, torch::Tensor X, int64_t
   auto cand_s = torch::empty({(long long)n_qblocks * n_swaths * 128 * k}, f32opts);
   auto cand_i = torch::empty({(long long)n_qblocks * n_swaths * 128 * k}, i32opts);
   dim3 grid(n_qblocks, n_swaths);
-  hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(256), 0, cur_stream(),
+  // 512 threads = the kernel's 8-wave 2x4 grid (TK_THREADS in topk_recall.hip)
+  hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(512), 0, cur_stream(),
                      reinterpret_cast<const __bf16*>(Q.data_ptr()),
                      reinterpret_cast<const __bf16*>(X.data_ptr()), nq, (int)nx, D,
                      (int)k, (int)n_swaths, cand_s.data_ptr<float>(),

@@ -140,10 +140,10 @@ def reference_encode(messages: Sequence[bytes], embed: np.ndarray, normalize: bo
         toks = []
         for t in range((n_pos + stride - 1) // stride if n_pos else 0):
             p = t * stride
-            h = np.uint32(2166136261)
+            h = 2166136261
             for b in m[p : p + 4]:
-                h = np.uint32((np.uint32(h ^ np.uint32(b)) * np.uint32(16777619)) & 0xFFFFFFFF)
-            toks.append(int(h) & (vocab - 1))
+                h = ((h ^ b) * 16777619) & 0xFFFFFFFF
+            toks.append(h & (vocab - 1))
         if toks:
             f = embed[toks].astype(np.float32).mean(axis=0)
         else:

@@ -161,6 +161,8 @@ class FirewallPipeline:
             base = self.rank * cfg.index_size
             local_mask = (flat_local >= base) & (flat_local < base + cfg.index_size)
             flat_local = flat_local[local_mask] - base
+        else:
+            flat_local = flat_local[flat_local >= 0]  # -1 = unfilled slot
         self.salience.mul_(0.9999)
         self.salience.index_add_(
             0, flat_local.long(), torch.full((flat_local.numel(),), 0.01, device=self.device)
