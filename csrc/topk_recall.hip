@@ -157,29 +157,35 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
     // Rounds: push survivors into the queue; drain by row-owner lanes;
     // repeat if the queue overflowed (only plausible on the first tile).
     // pending bit (m*16 + n*4 + r): value not yet pushed/retired
+    // STATIC m/n/r indexing only: a dynamic index into acc forces the
+    // compiler to spill all 64 accumulator VGPRs to scratch, and every
+    // MFMA then round-trips scratch memory (measured 46x slowdown).
     unsigned long long pending = ~0ull;
     for (int round = 0; ; ++round) {
       unsigned long long still = 0ull;
-      unsigned long long todo = pending;
-      while (todo) {
-        int vi = __ffsll((long long)todo) - 1;
-        todo &= todo - 1;
-        int m = vi >> 4, n = (vi >> 2) & 3, r = vi & 3;
-        int row = wm * 64 + m * 16 + (lane >> 4) * 4 + r;
-        if ((row0 + row) >= nq) continue;
-        long long col = x0 + wn * 64 + n * 16 + (lane & 15);
-        if (col >= x_end) continue;
-        float v = acc[m][n][r];
-        if (!(v > row_min[row])) continue;
-        int idx = atomicAdd(&q_count, 1);
-        if (idx < QCAP) {
-          q_score[idx] = v;
-          q_meta[idx] = (uint32_t(row) << 16) | uint32_t(col - x0);
-        } else {
-          still |= 1ull << vi;  // retry just this value next round
-          atomicExch(&q_overflow, 1);
-        }
-      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int vi = m * 16 + n * 4 + r;
+            if (!((pending >> vi) & 1ull)) continue;
+            int row = wm * 64 + m * 16 + (lane >> 4) * 4 + r;
+            if ((row0 + row) >= nq) continue;
+            long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+            if (col >= x_end) continue;
+            float v = acc[m][n][r];
+            if (!(v > row_min[row])) continue;
+            int idx = atomicAdd(&q_count, 1);
+            if (idx < QCAP) {
+              q_score[idx] = v;
+              q_meta[idx] = (uint32_t(row) << 16) | uint32_t(col - x0);
+            } else {
+              still |= 1ull << vi;  // retry just this value next round
+              atomicExch(&q_overflow, 1);
+            }
+          }
       __syncthreads();
       // drain: lanes 0..15 of each wave own rows wid*16 + (lane)
       int total = min(q_count, QCAP);

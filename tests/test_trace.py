@@ -43,7 +43,7 @@ def tool_fail_pair(cmd="make build", err="exit 1", tool="exec"):
 
 def test_normalize_schema_a_and_b():
     env = {
-        "id": "evt-1", "type": "tool.call.before", "ts": 123,
+        "id": "evt-1", "type": "tool.call.requested", "ts": 123,
         "actor": {"id": "forge"}, "scope": {"sessionKey": "agent:forge:abc"},
         "data": {"toolName": "exec", "params": {"command": "ls"}},
     }

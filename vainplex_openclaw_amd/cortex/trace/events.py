@@ -17,18 +17,31 @@ EVENT_TYPES = (
     "session.start", "session.end", "run.start", "run.end", "run.error",
 )
 
-# Schema A (eventstore canonical types) -> analyzer types
+# Schema A (eventstore canonical + legacy types, eventstore/events.py)
+# -> analyzer types
 SCHEMA_A_MAP = {
     "message.in.received": "msg.in",
     "message.out.sending": "msg.out",
     "message.out.sent": "msg.out",
-    "tool.call.before": "tool.call",
-    "tool.call.after": "tool.result",
+    "tool.call.requested": "tool.call",
+    "tool.call.executed": "tool.result",
+    "tool.call.failed": "tool.result",
     "session.started": "session.start",
     "session.ended": "session.end",
-    "agent.run.started": "run.start",
-    "agent.run.ended": "run.end",
-    "agent.run.error": "run.error",
+    "run.started": "run.start",
+    "run.ended": "run.end",
+    "run.failed": "run.error",
+    # legacy aliases (the envelope's `type` field)
+    "msg.in": "msg.in",
+    "msg.out": "msg.out",
+    "msg.sending": "msg.out",
+    "tool.call": "tool.call",
+    "tool.result": "tool.result",
+    "session.start": "session.start",
+    "session.end": "session.end",
+    "run.start": "run.start",
+    "run.end": "run.end",
+    "run.error": "run.error",
 }
 
 
@@ -53,10 +66,12 @@ def _norm_session(raw: Optional[str]) -> str:
 
 def normalize_schema_a(ev: Dict[str, Any], seq: int = 0) -> Optional[NormalizedEvent]:
     """ClawEvent envelope (eventstore/envelope.py) -> NormalizedEvent."""
-    etype = SCHEMA_A_MAP.get(str(ev.get("type", "")))
+    etype = SCHEMA_A_MAP.get(
+        str(ev.get("canonicalType") or ev.get("type", ""))
+    ) or SCHEMA_A_MAP.get(str(ev.get("type", "")))
     if etype is None:
         return None
-    data = ev.get("data") or {}
+    data = ev.get("data") or ev.get("payload") or {}
     actor = ev.get("actor") or {}
     scope = ev.get("scope") or {}
     payload: Dict[str, Any] = {}
@@ -78,7 +93,7 @@ def normalize_schema_a(ev: Dict[str, Any], seq: int = 0) -> Optional[NormalizedE
     return NormalizedEvent(
         id=str(ev.get("id", "")),
         ts=float(ts),
-        agent=str(actor.get("id") or ev.get("agent") or "unknown"),
+        agent=str(actor.get("id") or actor.get("agentId") or ev.get("agent") or "unknown"),
         session=_norm_session(scope.get("sessionKey") or ev.get("session")),
         type=etype,
         payload=payload,
