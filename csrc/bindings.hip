@@ -150,14 +150,23 @@ std::vector<torch::Tensor> topk_recall(torch::Tensor Q, torch::Tensor X, int64_t
   TORCH_CHECK(Q.dtype() == torch::kBFloat16 && X.dtype() == torch::kBFloat16);
   int nq = Q.size(0), D = Q.size(1);
   long long nx = X.size(0);
-  TORCH_CHECK(X.size(1) == D && D % 32 == 0);
+  TORCH_CHECK(X.size(1) == D && D % 64 == 0, "D must be a multiple of 64");
   TORCH_CHECK(k >= 1 && k <= 32, "k in [1,32]");
-  int n_qblocks = (nq + 127) / 128;
+  int n_qblocks = (nq + 255) / 256;  // BM=256 (topk_recall.hip v3)
+  // merge kernel holds n_swaths*k candidates in 16 regs x 64 lanes
+  TORCH_CHECK(n_swaths >= 1 && (long long)n_swaths * k <= 1024,
+              "n_swaths * k must be <= 1024");
   auto f32opts = torch::dtype(torch::kFloat32).device(Q.device());
   auto i32opts = torch::dtype(torch::kInt32).device(Q.device());
-  auto cand_s = torch::empty({(long long)n_qblocks * n_swaths * 128 * k}, f32opts);
-  auto cand_i = torch::empty({(long long)n_qblocks * n_swaths * 128 * k}, i32opts);
-  dim3 grid(n_qblocks, n_swaths);
+  // candidate state lives in global memory, host-initialized; the kernel
+  // only updates slots that beat the running per-row threshold
+  auto cand_s = torch::full({(long long)n_qblocks * n_swaths * 256 * k}, -1e30,
+                            f32opts);
+  auto cand_i = torch::full({(long long)n_qblocks * n_swaths * 256 * k}, -1,
+                            i32opts);
+  // flat grid: block f -> (qb = f / S, swath = f % S); S a multiple of 8
+  // keeps each co-sweeping same-swath group on one XCD
+  dim3 grid((unsigned)(n_qblocks * n_swaths));
   // 512 threads = the kernel's 8-wave 2x4 grid (TK_THREADS in topk_recall.hip)
   hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(512), 0, cur_stream(),
                      reinterpret_cast<const __bf16*>(Q.data_ptr()),

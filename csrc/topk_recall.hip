@@ -1,77 +1,74 @@
 // Fused cosine-scores + streaming top-k: the Membrane salience recall
-// kernel (v2).
+// kernel (v3).
 //
 // Replaces the reference Membrane plugin's salience retrieval (external
 // repo; config surface in brainplex configurator.ts:137-148) with an
 // LDS-tiled MFMA scan over the HBM-resident embedding shard. Scores are
 // never materialized to HBM (4096 x 50M f32 would be 800 GB of traffic
-// per step): each 128-query block walks its swath of the [N, D] bf16
-// index, computes 128x256 score tiles with v_mfma_f32_16x16x32_bf16 and
-// filters them against per-row running top-k thresholds straight from the
+// per step): each block walks its swath of the [N, D] bf16 index,
+// computes 256x256 score tiles with v_mfma_f32_16x16x32_bf16 and filters
+// them against per-row running top-k thresholds straight from the
 // accumulator registers.
 //
-// Structure (cdna_hip_programming.md §5 "step 3"):
-// - 8 waves (512 thr), block tile BM=128 x BN=256, BK=32, wave grid 2x4
-//   (each wave a 64x64 subtile = 4x4 fragments, 16 independent
-//   accumulators keep the matrix pipe busy at 2 waves/SIMD).
-// - global_load_lds width 16 staging, LDS image lane-linear with the
-//   XOR swizzle applied on the SOURCE k-group and again on the ds_read
-//   offset (guide rule 21): slot ^= (row>>2)&3 makes the 16-lane
-//   ds_read_b128 groups conflict-free without row padding.
-// - double-buffered LDS, one barrier pair per K-step (the deep-pipelined
-//   8-phase schedule is the next rung once this baseline is profiled).
-// - top-k: per-row threshold in LDS; lanes test their 16 accumulator
-//   values (C/D map row=(lane>>4)*4+r, col=lane&15), survivors go through
-//   a bounded LDS candidate queue drained by row-owning lanes.
+// v3 structure (cdna_hip_programming.md §5, 256² template geometry):
+// - 512 threads (8 waves, 2M x 4N wave grid), block tile BM=256 x BN=256,
+//   BK=64. Per wave: 128x64 output = 8x4 fragments of 16x16, 32
+//   independent f32x4 accumulators. 2x the queries per index pass vs the
+//   v2 128-tile: halves both HBM re-reads and LDS staging per FLOP.
+// - global_load_lds width-16 staging; LDS image [row][8 slots of 16 B]
+//   with slot ^= (row>>1)&7 — each 16-lane ds_read_b128 group covers 16
+//   distinct 16 B slots across its half of the 64-bank array:
+//   conflict-free (the v2 swizzle generalized to 128 B rows).
+// - double-buffered LDS, one vmcnt(0)+barrier pair per K-step.
+// - candidates + per-row top-k values live in GLOBAL memory
+//   ([qb][swath][BM][k], host-initialized to -1e30/-1) — LDS holds only
+//   tiles + thresholds + the push queue. Pushes are threshold-filtered
+//   from registers (STATIC m/n/r indexing only: a dynamic index into acc
+//   spills all accumulators to scratch — measured 46x), drained by one
+//   row-owner thread each.
+// - XCD-aware grid: flat blockIdx -> (qb = f / S, swath = f % S) with S a
+//   multiple of 8 puts every co-sweeping same-swath block group on one
+//   XCD (dispatch assigns XCDs round-robin), so the swath stream is
+//   shared through that XCD's L2 and the L3.
 #include "common.hpp"
 
-#define BM 128
+#define BM 256
 #define BN 256
-#define BK 32
+#define BK 64
 #define TK_THREADS 512
 #define TOPK_MAX 32
 #define QCAP 2048
+#define AS1 __attribute__((address_space(1)))
+#define AS3 __attribute__((address_space(3)))
 
-// LDS tile addressing: row-major [rows][BK] bf16, 64 B per row = 4 slots
-// of 16 B. Swizzle: slot' = slot ^ ((row>>2)&3).
+// LDS tile addressing: row-major [rows][BK] bf16, 128 B per row = 8 slots
+// of 16 B. Swizzle: slot' = slot ^ ((row>>1)&7).
 DEVINL uint32_t lds_off_bytes(uint32_t row, uint32_t slot) {
-  return (row * 4u + (slot ^ ((row >> 2u) & 3u))) * 16u;
+  return (row * 8u + (slot ^ ((row >> 1u) & 7u))) * 16u;
 }
 
-// Stage a [rows x BK] tile into LDS. Piece p (16 B) = row p/4, slot p%4;
-// the SOURCE k-group is the swizzled slot (involution with the read side).
-//
-// USE_GLDS=1: global_load_lds — the LDS operand must be the WAVE-UNIFORM
-// base of the wave's 64 consecutive pieces (the hardware adds lane*16;
-// a per-lane destination silently scatters / faults). The per-lane part
-// lives only in the global SOURCE address.
-#ifndef USE_GLDS
-#define USE_GLDS 1
-#endif
-
-DEVINL void stage_glds(const bf16* __restrict__ src, long long ld,
+// Stage a [rows x BK] tile into LDS via global_load_lds. Destination is
+// linear in piece index (the hardware adds lane*16 to the wave-uniform
+// base); the swizzle is applied on the SOURCE slot (involution with the
+// read side).
+DEVINL void stage_tile(const bf16* __restrict__ src, long long ld,
                        long long row0, long long row_max, int k0,
                        bf16* lds_base, int tile_rows) {
-  int n_pieces = tile_rows * 4;  // 16B pieces in the tile
+  int n_pieces = tile_rows * 8;  // 16 B pieces
   int w = wave_id();
   int lane = lane_id();
-  for (int piece0 = w * WAVE; piece0 < n_pieces; piece0 += (TK_THREADS / WAVE) * WAVE) {
+  for (int piece0 = w * WAVE; piece0 < n_pieces; piece0 += TK_THREADS) {
     int piece = piece0 + lane;
-    uint32_t r = piece >> 2;
-    uint32_t slot = piece & 3;
-    uint32_t src_slot = slot ^ ((r >> 2u) & 3u);
+    uint32_t r = piece >> 3;
+    uint32_t slot = piece & 7;
+    uint32_t src_slot = slot ^ ((r >> 1u) & 7u);
     long long gr = row0 + r;
     if (gr >= row_max) gr = row_max - 1;  // clamp: garbage filtered later
     const bf16* p = src + gr * ld + k0 + src_slot * 8;
-#if USE_GLDS
     int piece0_u = __builtin_amdgcn_readfirstlane(piece0);
-    auto ldst = (__attribute__((address_space(3))) char*)lds_base + piece0_u * 16;
+    auto ldst = (AS3 char*)lds_base + piece0_u * 16;
     __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)p,
-        (__attribute__((address_space(3))) void*)ldst, 16, 0, 0);
-#else
-    *(bf16x8*)((char*)lds_base + piece * 16) = *(const bf16x8*)p;
-#endif
+        (const AS1 void*)p, (AS3 void*)ldst, 16, 0, 0);
   }
 }
 
@@ -83,8 +80,6 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   __shared__ bf16 lds_all[2 * (BM + BN) * BK];
 #define QS(buf) (lds_all + (buf) * BM * BK)
 #define XS(buf) (lds_all + 2 * BM * BK + (buf) * BN * BK)
-  __shared__ float topk_s[BM][TOPK_MAX];
-  __shared__ int32_t topk_i[BM][TOPK_MAX];
   __shared__ float row_min[BM];
   __shared__ int row_min_slot[BM];
   __shared__ float q_score[QCAP];
@@ -92,19 +87,16 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   __shared__ int q_count;
   __shared__ int q_overflow;
 
-  int qb = blockIdx.x;
-  int swath = blockIdx.y;
+  int S = n_swaths;
+  int qb = blockIdx.x / S;      // same-swath groups land on one XCD
+  int swath = blockIdx.x % S;
   long long row0 = (long long)qb * BM;
 
-  long long per = ((long long)nx + n_swaths - 1) / n_swaths;
+  long long per = ((long long)nx + S - 1) / S;
   per = ((per + BN - 1) / BN) * BN;
   long long x_begin = (long long)swath * per;
   long long x_end = min((long long)nx, x_begin + per);
 
-  for (int i = threadIdx.x; i < BM * TOPK_MAX; i += blockDim.x) {
-    topk_s[i / TOPK_MAX][i % TOPK_MAX] = -1e30f;
-    topk_i[i / TOPK_MAX][i % TOPK_MAX] = -1;
-  }
   for (int i = threadIdx.x; i < BM; i += blockDim.x) {
     row_min[i] = -1e30f;
     row_min_slot[i] = 0;
@@ -112,66 +104,76 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
   __syncthreads();
 
-  int wid = wave_id();          // 0..7
+  int wid = wave_id();              // 0..7
   int wm = wid >> 2, wn = wid & 3;  // 2 x 4 wave grid
   int lane = lane_id();
   int lrow = lane & 15;
-  int kslot = lane >> 4;        // 0..3 -> k-group (8 bf16 = 16 B)
+  int kgrp = lane >> 4;             // 0..3 -> 16 B k-group within 32 elems
   int nk = D / BK;
+  // candidate slice this block owns: [qb][swath][BM][k]
+  size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
 
   for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
-    f32x4 acc[4][4] = {};
-    stage_glds(Q, D, row0, nq, 0, QS(0), BM);
-    stage_glds(X, D, x0, (long long)nx, 0, XS(0), BN);
+    f32x4 acc[8][4] = {};
+    stage_tile(Q, D, row0, nq, 0, QS(0), BM);
+    stage_tile(X, D, x0, (long long)nx, 0, XS(0), BN);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     for (int kt = 0; kt < nk; ++kt) {
       int cur = kt & 1, nxt = cur ^ 1;
       if (kt + 1 < nk) {
-        stage_glds(Q, D, row0, nq, (kt + 1) * BK, QS(nxt), BM);
-        stage_glds(X, D, x0, (long long)nx, (kt + 1) * BK, XS(nxt), BN);
-      }
-      bf16x8 qf[4], xf[4];
-#pragma unroll
-      for (int m = 0; m < 4; ++m) {
-        uint32_t r = wm * 64 + m * 16 + lrow;
-        qf[m] = *(const bf16x8*)((const char*)QS(cur) + lds_off_bytes(r, kslot));
+        stage_tile(Q, D, row0, nq, (kt + 1) * BK, QS(nxt), BM);
+        stage_tile(X, D, x0, (long long)nx, (kt + 1) * BK, XS(nxt), BN);
       }
 #pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        uint32_t r = wn * 64 + n * 16 + lrow;
-        xf[n] = *(const bf16x8*)((const char*)XS(cur) + lds_off_bytes(r, kslot));
+      for (int ks = 0; ks < 2; ++ks) {  // two 32-deep mfma steps per BK
+        int slot = ks * 4 + kgrp;
+        bf16x8 xf[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          uint32_t r = wn * 64 + n * 16 + lrow;
+          xf[n] = *(const bf16x8*)((const char*)XS(cur) + lds_off_bytes(r, slot));
+        }
+        // m-fragments in halves of 4: keeps peak live registers at
+        // acc(128) + xf(16) + qf(16) instead of + qf(32)
+#pragma unroll
+        for (int mh = 0; mh < 2; ++mh) {
+          bf16x8 qf[4];
+#pragma unroll
+          for (int mi = 0; mi < 4; ++mi) {
+            uint32_t r = wm * 128 + (mh * 4 + mi) * 16 + lrow;
+            qf[mi] = *(const bf16x8*)((const char*)QS(cur) + lds_off_bytes(r, slot));
+          }
+#pragma unroll
+          for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+            for (int n = 0; n < 4; ++n)
+              acc[mh * 4 + mi][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  qf[mi], xf[n], acc[mh * 4 + mi][n], 0, 0, 0);
+        }
       }
-#pragma unroll
-      for (int m = 0; m < 4; ++m)
-#pragma unroll
-        for (int n = 0; n < 4; ++n)
-          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[m], xf[n], acc[m][n], 0, 0, 0);
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
     }
 
     // ---- streaming top-k from the accumulators -------------------------
-    // lane holds acc[m][n][r] at row = wm*64+m*16+(lane>>4)*4+r,
-    //                         col = wn*64+n*16+(lane&15)
-    // Rounds: push survivors into the queue; drain by row-owner lanes;
-    // repeat if the queue overflowed (only plausible on the first tile).
-    // pending bit (m*16 + n*4 + r): value not yet pushed/retired
-    // STATIC m/n/r indexing only: a dynamic index into acc forces the
-    // compiler to spill all 64 accumulator VGPRs to scratch, and every
-    // MFMA then round-trips scratch memory (measured 46x slowdown).
-    unsigned long long pending = ~0ull;
+    // lane holds acc[m][n][r] at row = wm*128 + m*16 + (lane>>4)*4 + r,
+    //                         col = wn*64  + n*16 + (lane&15)
+    // Rounds: push survivors into the queue; drain by row-owner threads;
+    // repeat if the queue overflowed (only plausible on the first tiles).
+    unsigned long long pend0 = ~0ull, pend1 = ~0ull;  // 128 pending bits
     for (int round = 0; ; ++round) {
-      unsigned long long still = 0ull;
+      unsigned long long still0 = 0ull, still1 = 0ull;
 #pragma unroll
-      for (int m = 0; m < 4; ++m)
+      for (int m = 0; m < 8; ++m)
 #pragma unroll
         for (int n = 0; n < 4; ++n)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            int vi = m * 16 + n * 4 + r;
-            if (!((pending >> vi) & 1ull)) continue;
-            int row = wm * 64 + m * 16 + (lane >> 4) * 4 + r;
+            int vi = (m & 3) * 16 + n * 4 + r;
+            unsigned long long bit = 1ull << vi;
+            if (!(((m < 4) ? pend0 : pend1) & bit)) continue;
+            int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
             if ((row0 + row) >= nq) continue;
             long long col = x0 + wn * 64 + n * 16 + (lane & 15);
             if (col >= x_end) continue;
@@ -182,49 +184,52 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
               q_score[idx] = v;
               q_meta[idx] = (uint32_t(row) << 16) | uint32_t(col - x0);
             } else {
-              still |= 1ull << vi;  // retry just this value next round
+              if (m < 4) still0 |= bit; else still1 |= bit;
               atomicExch(&q_overflow, 1);
             }
           }
       __syncthreads();
-      // drain: lanes 0..15 of each wave own rows wid*16 + (lane)
+      // drain: thread t < BM owns row t; candidate state is in GLOBAL
+      // memory (block-private slice, L2-hot)
       int total = min(q_count, QCAP);
-      if (lane < 16) {
-        int my_row = wid * 16 + lane;
+      if (threadIdx.x < BM) {
+        int my_row = threadIdx.x;
+        float rmin = row_min[my_row];
+        int rslot = row_min_slot[my_row];
+        float* cs = cand_scores + cbase + (size_t)my_row * k;
+        int32_t* ci = cand_ids + cbase + (size_t)my_row * k;
+        bool touched = false;
         for (int i = 0; i < total; ++i) {
           uint32_t meta = q_meta[i];
-          int row = int(meta >> 16);
-          if (row != my_row) continue;
+          if (int(meta >> 16) != my_row) continue;
           float v = q_score[i];
-          if (v > row_min[row]) {
-            int slot = row_min_slot[row];
-            topk_s[row][slot] = v;
-            topk_i[row][slot] = int32_t(x0 + (meta & 0xFFFFu));
-            float mn = topk_s[row][0];
+          if (v > rmin) {
+            cs[rslot] = v;
+            ci[rslot] = int32_t(x0 + (meta & 0xFFFFu));
+            float mn = cs[0];
             int ms = 0;
             for (int j = 1; j < k; ++j)
-              if (topk_s[row][j] < mn) { mn = topk_s[row][j]; ms = j; }
-            row_min[row] = mn;
-            row_min_slot[row] = ms;
+              if (cs[j] < mn) { mn = cs[j]; ms = j; }
+            rmin = mn;
+            rslot = ms;
+            touched = true;
           }
+        }
+        if (touched) {
+          row_min[my_row] = rmin;
+          row_min_slot[my_row] = rslot;
         }
       }
       __syncthreads();
       int of = q_overflow;
       __syncthreads();  // all reads of q_overflow done before the reset
       if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
-      pending = still;
+      pend0 = still0;
+      pend1 = still1;
       if (!of) break;   // of is uniform (LDS): no divergence
       __syncthreads();  // reset visible before next round's pushes
     }
     __syncthreads();
-  }
-
-  // candidates out: [qb][swath][row][k]
-  size_t base = (((size_t)qb * n_swaths) + swath) * BM * k;
-  for (int i = threadIdx.x; i < BM * k; i += blockDim.x) {
-    cand_scores[base + i] = topk_s[i / k][i % k];
-    cand_ids[base + i] = topk_i[i / k][i % k];
   }
 }
 

@@ -168,11 +168,19 @@ def gemm_nt(
 
 
 def topk_recall(Q: torch.Tensor, X: torch.Tensor, k: int, n_swaths: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Cosine top-k of each Q row against X (csrc/topk_recall.hip v3).
+
+    Swath count: a multiple of 8 (XCD-aware grid mapping), sized so
+    qblocks*swaths fills the 256 CUs, capped by the merge kernel's
+    n_swaths*k <= 1024 register budget and the index size.
+    """
     if n_swaths <= 0:
-        # enough blocks to fill the chip: 256 CUs / qblocks, >=1
-        qblocks = (Q.shape[0] + 127) // 128
-        n_swaths = max(1, min(512 // max(qblocks, 1), 64))
-        n_swaths = min(n_swaths, max(1, X.shape[0] // 128))
+        qblocks = (Q.shape[0] + 255) // 256  # BM=256
+        want = max(1, 256 // max(qblocks, 1))
+        n_swaths = max(8, (want // 8) * 8)
+        n_swaths = min(n_swaths, (1024 // max(k, 1)) // 8 * 8)
+        n_swaths = min(n_swaths, max(1, X.shape[0] // 256))
+        n_swaths = max(1, n_swaths)
     s, i = ext().topk_recall(Q, X, k, n_swaths)
     return s, i
 
