@@ -77,9 +77,13 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
                    int nq, int nx, int D, int k, int n_swaths,
                    float* __restrict__ cand_scores,
                    int32_t* __restrict__ cand_ids) {
-  __shared__ bf16 lds_all[2 * (BM + BN) * BK];
-#define QS(buf) (lds_all + (buf) * BM * BK)
-#define XS(buf) (lds_all + 2 * BM * BK + (buf) * BN * BK)
+  // LDS holds ONLY the X double-buffer: Q fragments are read directly
+  // from global memory. The whole Q tensor (4096 x 1024 bf16 = 8 MB) is
+  // L3-resident and shared by every block, so direct a-frag loads (64 B
+  // coalesced segments) cost ~36 GB/s/CU from cache — while halving the
+  // LDS-DMA staging traffic, which is this kernel's measured bound.
+  __shared__ bf16 lds_all[2 * BN * BK];
+#define XS(buf) (lds_all + (buf) * BN * BK)
   __shared__ float row_min[BM];
   __shared__ int row_min_slot[BM];
   __shared__ float q_score[QCAP];
@@ -113,21 +117,29 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   // candidate slice this block owns: [qb][swath][BM][k]
   size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
 
+  // per-lane Q row pointers for the 8 m-fragments (row clamped once)
+  const bf16* qrow[8];
+#pragma unroll
+  for (int m = 0; m < 8; ++m) {
+    long long gr = row0 + wm * 128 + m * 16 + lrow;
+    if (gr >= nq) gr = nq - 1;  // garbage rows filtered in the push phase
+    qrow[m] = Q + gr * D;
+  }
+
   for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
     f32x4 acc[8][4] = {};
-    stage_tile(Q, D, row0, nq, 0, QS(0), BM);
     stage_tile(X, D, x0, (long long)nx, 0, XS(0), BN);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     for (int kt = 0; kt < nk; ++kt) {
       int cur = kt & 1, nxt = cur ^ 1;
       if (kt + 1 < nk) {
-        stage_tile(Q, D, row0, nq, (kt + 1) * BK, QS(nxt), BM);
         stage_tile(X, D, x0, (long long)nx, (kt + 1) * BK, XS(nxt), BN);
       }
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {  // two 32-deep mfma steps per BK
         int slot = ks * 4 + kgrp;
+        int qoff = kt * BK + ks * 32 + kgrp * 8;
         bf16x8 xf[4];
 #pragma unroll
         for (int n = 0; n < 4; ++n) {
@@ -140,10 +152,8 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
         for (int mh = 0; mh < 2; ++mh) {
           bf16x8 qf[4];
 #pragma unroll
-          for (int mi = 0; mi < 4; ++mi) {
-            uint32_t r = wm * 128 + (mh * 4 + mi) * 16 + lrow;
-            qf[mi] = *(const bf16x8*)((const char*)QS(cur) + lds_off_bytes(r, slot));
-          }
+          for (int mi = 0; mi < 4; ++mi)
+            qf[mi] = *(const bf16x8*)(qrow[mh * 4 + mi] + qoff);
 #pragma unroll
           for (int mi = 0; mi < 4; ++mi)
 #pragma unroll

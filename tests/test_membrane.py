@@ -195,8 +195,9 @@ def test_index_gpu_matches_cpu_reference():
 
     texts = [f"memory item number {i} about topic {i % 7}" for i in range(300)]
     ids = [f"r{i}" for i in range(300)]
-    cpu = SalienceIndex(dim=256, capacity=512)
-    gpu = SalienceIndex(dim=256, capacity=512, device="cuda:0")
+    # encoder kernel requires dim % 1024 == 0
+    cpu = SalienceIndex(dim=1024, capacity=512)
+    gpu = SalienceIndex(dim=1024, capacity=512, device="cuda:0")
     cpu.add("a", ids, texts)
     gpu.add("a", ids, texts)
     for q in ("memory about topic 3", "item number 250"):
