@@ -1,0 +1,262 @@
+"""Leuko tests: collectors, custom thresholds, aggregator (health/
+categories/delta/summary), anomaly detection, plugin surface."""
+
+import json
+import time
+
+import pytest
+
+from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+from vainplex_openclaw_amd.eventstore import EventJournal
+from vainplex_openclaw_amd.leuko import (
+    AnomalyDetector,
+    LeukoPlugin,
+    MetricHistory,
+    generate_sitrep,
+    resolve_config,
+    run_custom_collector,
+    safe_collect,
+    write_sitrep,
+)
+from vainplex_openclaw_amd.leuko.collectors import (
+    collect_errors,
+    collect_goals,
+    collect_journal,
+    collect_threads,
+    result,
+)
+
+
+class FakeClock:
+    def __init__(self, t=1_700_000_000.0):
+        self.t = t
+
+    def __call__(self):
+        return self.t
+
+
+def _iso(ts):
+    return time.strftime("%Y-%m-%dT%H:%M:%S", time.gmtime(ts)) + "Z"
+
+
+# -- safe_collect ------------------------------------------------------------
+
+def test_safe_collect_disabled_and_error():
+    assert safe_collect("x", lambda c: 1 / 0, {"enabled": False})["summary"] == "disabled"
+    r = safe_collect("x", lambda c: 1 / 0, {"enabled": True}, NullLogger())
+    assert r["status"] == "error" and "division" in r["error"]
+
+
+def test_safe_collect_times():
+    r = safe_collect("x", lambda c: result("ok", [], "fine"), {"enabled": True})
+    assert r["status"] == "ok" and r["duration_ms"] >= 0
+
+
+# -- collectors --------------------------------------------------------------
+
+def test_collect_goals_red_zone_and_stale(tmp_path):
+    now = time.time()
+    p = tmp_path / "goals.json"
+    p.write_text(json.dumps({"goals": [
+        {"id": "g1", "title": "Deploy", "zone": "red", "status": "approved"},
+        {"id": "g2", "title": "Old", "status": "proposed", "proposed_at": _iso(now - 100 * 3600)},
+        {"id": "g3", "title": "Fresh", "status": "proposed", "proposed_at": _iso(now)},
+    ]}))
+    r = collect_goals({"enabled": True, "goalsPath": str(p), "staleHours": 48})
+    ids = [i["id"] for i in r["items"]]
+    assert "goal-g1-red-approved" in ids and "goal-g2-stale" in ids
+    assert "goal-g3-stale" not in ids
+    assert r["status"] == "warn"
+
+
+def test_collect_threads_stale_and_high_priority(tmp_path):
+    now = time.time()
+    p = tmp_path / "threads.json"
+    p.write_text(json.dumps({"threads": [
+        {"id": "t1", "topic": "old stuff", "status": "open", "last_activity": _iso(now - 10 * 86400)},
+        {"id": "t2", "topic": "urgent", "status": "open", "priority": "high", "last_activity": _iso(now)},
+        {"id": "t3", "topic": "done", "status": "closed", "last_activity": _iso(now - 30 * 86400)},
+    ]}))
+    r = collect_threads({"enabled": True, "threadsPath": str(p), "staleDays": 7})
+    ids = [i["id"] for i in r["items"]]
+    assert "thread-t1-stale" in ids and "thread-t2-high" in ids
+    assert not any("t3" in i for i in ids)
+
+
+def test_collect_errors_recent_critical(tmp_path):
+    now = time.time()
+    p = tmp_path / "errors.json"
+    p.write_text(json.dumps([
+        {"id": "e1", "pattern": "OOM", "severity": "critical", "last_seen": _iso(now - 3600)},
+        {"id": "e2", "pattern": "old", "severity": "critical", "last_seen": _iso(now - 48 * 3600)},
+        {"id": "e3", "pattern": "meh", "severity": "low", "last_seen": _iso(now)},
+    ]))
+    r = collect_errors({"enabled": True, "patternsPath": str(p), "recentHours": 24})
+    assert [i["id"] for i in r["items"]] == ["error-e1"]
+    assert r["status"] == "critical"
+
+
+def test_collect_journal_counts_and_staleness():
+    j = EventJournal(durable=False)
+    j.publish("s", {"ts": (time.time() - 7200) * 1000})
+    r = collect_journal({"enabled": True, "journal": j, "maxAgeMins": 60})
+    assert any(i["id"] == "journal-stale" for i in r["items"])
+    j2 = EventJournal(durable=False)
+    j2.publish("s", {"ts": time.time() * 1000})
+    r2 = collect_journal({"enabled": True, "journal": j2, "maxAgeMins": 60})
+    assert r2["status"] == "ok" and "1 message(s)" in r2["summary"]
+
+
+def test_custom_collector_threshold_and_flags():
+    r = run_custom_collector({"id": "disk", "command": "echo 85", "warnThreshold": "80",
+                              "criticalThreshold": "95"})
+    assert r["items"][0]["severity"] == "warn"
+    r2 = run_custom_collector({"id": "disk", "command": "echo 97", "warnThreshold": "80",
+                               "criticalThreshold": "95"})
+    assert r2["items"][0]["severity"] == "critical"
+    assert r2["items"][0]["category"] == "needs_owner"
+    r3 = run_custom_collector({"id": "log", "command": "echo bad", "warnIfOutput": True})
+    assert r3["items"][0]["id"] == "custom-log-output"
+    r4 = run_custom_collector({"id": "hb", "command": "true", "warnIfNoOutput": True})
+    assert r4["items"][0]["id"] == "custom-hb-no-output"
+
+
+# -- aggregator --------------------------------------------------------------
+
+def _min_cfg(tmp_path, **collectors):
+    return {
+        "enabled": True,
+        "outputPath": str(tmp_path / "sitrep.json"),
+        "previousPath": str(tmp_path / "sitrep-previous.json"),
+        "collectors": collectors,
+        "customCollectors": [],
+        "summaryMaxChars": 2000,
+    }
+
+
+def test_generate_sitrep_schema_and_categories(tmp_path):
+    now = time.time()
+    gp = tmp_path / "goals.json"
+    gp.write_text(json.dumps({"goals": [
+        {"id": "g1", "title": "X", "zone": "red", "status": "approved"}]}))
+    cfg = _min_cfg(tmp_path, goals={"enabled": True, "goalsPath": str(gp)})
+    rep = generate_sitrep(cfg)
+    assert rep["version"] == 1 and rep["generated"].endswith("Z")
+    assert rep["health"]["overall"] == "warn"
+    assert rep["health"]["details"]["goals"] == "warn"
+    assert rep["health"]["details"]["errors"] == "disabled"
+    assert len(rep["categories"]["needs_owner"]) == 1
+    assert "need owner attention" in rep["summary"]
+
+
+def test_sitrep_delta_tracking(tmp_path):
+    cfg = _min_cfg(tmp_path)
+    rep1 = generate_sitrep(cfg)
+    write_sitrep(rep1, cfg["outputPath"], cfg["previousPath"])
+    gp = tmp_path / "goals.json"
+    gp.write_text(json.dumps({"goals": [
+        {"id": "g1", "title": "X", "zone": "red", "status": "approved"}]}))
+    cfg2 = _min_cfg(tmp_path, goals={"enabled": True, "goalsPath": str(gp)})
+    cfg2["previousPath"] = cfg["outputPath"]  # delta vs last written
+    rep2 = generate_sitrep(cfg2)
+    assert rep2["delta"]["new_items"] == 1
+    assert rep2["delta"]["previous_generated"] == rep1["generated"]
+
+
+def test_sitrep_all_nominal_summary(tmp_path):
+    rep = generate_sitrep(_min_cfg(tmp_path))
+    assert rep["summary"] == "All systems nominal."
+    assert rep["health"]["overall"] == "ok"
+
+
+# -- anomaly detection -------------------------------------------------------
+
+def test_metric_history_slope(tmp_path):
+    clock = FakeClock()
+    h = MetricHistory(str(tmp_path / "m.jsonl"), clock=clock)
+    for i in range(10):
+        h.record("x", 100.0 + i * 50.0, ts=clock.t + i * 3600.0)
+    slope = h.slope_per_hour("x")
+    assert slope == pytest.approx(50.0, rel=1e-6)
+
+
+def test_metric_history_persists(tmp_path):
+    p = str(tmp_path / "m.jsonl")
+    h = MetricHistory(p)
+    h.record("y", 1.0, ts=1.0)
+    h.record("y", 2.0, ts=3600.0)
+    h2 = MetricHistory(p)
+    assert len(h2.series("y")) == 2
+
+
+def test_directory_growth_detection(tmp_path):
+    clock = FakeClock()
+    h = MetricHistory(str(tmp_path / "m.jsonl"), clock=clock)
+    det = AnomalyDetector(h, clock=clock)
+    d = tmp_path / "grow"
+    d.mkdir()
+    # simulate 200 MB/h growth: fabricated history below the current (0 B)
+    # sample so the trend continues through the real measurement
+    name = f"dirsize:{d}"
+    for i in range(5):
+        h.record(name, -(5 - i) * 2e8, ts=clock.t - (5 - i) * 3600.0)
+    it = det.check_directory_growth(str(d), warn_mb_per_hour=100.0)
+    assert it is not None and "growing" in it["title"]
+
+
+def test_declining_metric_detection(tmp_path):
+    clock = FakeClock()
+    h = MetricHistory(str(tmp_path / "m.jsonl"), clock=clock)
+    det = AnomalyDetector(h, clock=clock)
+    it = None
+    for i in range(8):
+        clock.t += 3600.0
+        it = det.check_declining_metric("throughput", 1000.0 - i * 100.0, 50.0)
+    assert it is not None and "declining" in it["title"]
+
+
+def test_failure_correlation():
+    clock = FakeClock()
+    j = EventJournal(durable=False)
+    for i in range(4):
+        j.publish("s", {"ts": clock.t * 1000, "canonicalType": "tool.call.failed",
+                        "agent": f"a{i % 2}"})
+    det = AnomalyDetector(MetricHistory("/nonexistent/never.jsonl"), clock=clock)
+    it = det.correlate_failures(j, window_s=600.0, threshold=3)
+    assert it is not None and it["severity"] == "critical"
+    assert "2 agent(s)" in it["title"]
+
+
+def test_bootstrap_integrity(tmp_path):
+    (tmp_path / "governance").mkdir()
+    (tmp_path / "governance" / "trust.json").write_text("{not json")
+    det = AnomalyDetector(MetricHistory(str(tmp_path / "m.jsonl")))
+    items = det.check_bootstrap_integrity(str(tmp_path))
+    assert len(items) == 1 and "corrupt" in items[0]["title"]
+
+
+# -- plugin ------------------------------------------------------------------
+
+def test_plugin_run_once_and_command(tmp_path, monkeypatch):
+    monkeypatch.setenv("HOME", str(tmp_path))
+    bus = HookBus()
+    api = PluginApi(id="openclaw-leuko",
+                    plugin_config={"collectors": {"systemd_timers": {"enabled": False},
+                                                  "gpu_health": {"enabled": False}}},
+                    logger=NullLogger(), config={}, bus=bus)
+    p = LeukoPlugin(workspace=str(tmp_path), journal=EventJournal(durable=False))
+    p.register(api)
+    out = api.commands["sitrep"]()
+    assert "Sitrep" in out["text"]
+    assert (tmp_path / ".openclaw" / "sitrep" / "sitrep.json").exists()
+    rep = api.gateway_methods["leuko.report"]()
+    assert rep["version"] == 1
+    assert "leuko_anomaly" in rep["collectors"]
+
+
+def test_resolve_config_merges_collectors(tmp_path):
+    cfg = resolve_config({"collectors": {"goals": {"enabled": True, "goalsPath": "/x"}},
+                          "intervalMinutes": 5}, home=str(tmp_path))
+    assert cfg["collectors"]["goals"]["enabled"] is True
+    assert cfg["collectors"]["systemd_timers"]["enabled"] is True
+    assert cfg["intervalMinutes"] == 5
