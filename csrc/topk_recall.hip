@@ -34,17 +34,19 @@
 
 #define BM 256
 #define BN 256
-#define BK 64
+#define BK 32
+#define NBUF 3
 #define TK_THREADS 512
 #define TOPK_MAX 32
 #define QCAP 2048
 #define AS1 __attribute__((address_space(1)))
 #define AS3 __attribute__((address_space(3)))
 
-// LDS tile addressing: row-major [rows][BK] bf16, 128 B per row = 8 slots
-// of 16 B. Swizzle: slot' = slot ^ ((row>>1)&7).
+// LDS tile addressing: row-major [rows][BK] bf16, 64 B per row = 4 slots
+// of 16 B. Swizzle: slot' = slot ^ ((row>>2)&3) -> each 16-lane
+// ds_read_b128 group covers 16 distinct bank-quads: conflict-free.
 DEVINL uint32_t lds_off_bytes(uint32_t row, uint32_t slot) {
-  return (row * 8u + (slot ^ ((row >> 1u) & 7u))) * 16u;
+  return (row * 4u + (slot ^ ((row >> 2u) & 3u))) * 16u;
 }
 
 // Stage a [rows x BK] tile into LDS via global_load_lds. Destination is
@@ -54,14 +56,14 @@ DEVINL uint32_t lds_off_bytes(uint32_t row, uint32_t slot) {
 DEVINL void stage_tile(const bf16* __restrict__ src, long long ld,
                        long long row0, long long row_max, int k0,
                        bf16* lds_base, int tile_rows) {
-  int n_pieces = tile_rows * 8;  // 16 B pieces
+  int n_pieces = tile_rows * 4;  // 16 B pieces
   int w = wave_id();
   int lane = lane_id();
   for (int piece0 = w * WAVE; piece0 < n_pieces; piece0 += TK_THREADS) {
     int piece = piece0 + lane;
-    uint32_t r = piece >> 3;
-    uint32_t slot = piece & 7;
-    uint32_t src_slot = slot ^ ((r >> 1u) & 7u);
+    uint32_t r = piece >> 2;
+    uint32_t slot = piece & 3;
+    uint32_t src_slot = slot ^ ((r >> 2u) & 3u);
     long long gr = row0 + r;
     if (gr >= row_max) gr = row_max - 1;  // clamp: garbage filtered later
     const bf16* p = src + gr * ld + k0 + src_slot * 8;
@@ -77,13 +79,14 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
                    int nq, int nx, int D, int k, int n_swaths,
                    float* __restrict__ cand_scores,
                    int32_t* __restrict__ cand_ids) {
-  // LDS holds ONLY the X double-buffer: Q fragments are read directly
-  // from global memory. The whole Q tensor (4096 x 1024 bf16 = 8 MB) is
-  // L3-resident and shared by every block, so direct a-frag loads (64 B
-  // coalesced segments) cost ~36 GB/s/CU from cache — while halving the
-  // LDS-DMA staging traffic, which is this kernel's measured bound.
-  __shared__ bf16 lds_all[2 * BN * BK];
-#define XS(buf) (lds_all + (buf) * BN * BK)
+  // Triple-buffered Q+X K-tiles: staging runs TWO K-steps ahead of
+  // compute, synchronized with counted s_waitcnt vmcnt(4) + raw
+  // s_barrier. (__syncthreads() makes the compiler drain the whole
+  // global_load_lds queue at every barrier — the ~20% stall the guide
+  // documents — and a 1-deep double buffer forces vmcnt(0) anyway.)
+  __shared__ bf16 lds_all[NBUF * (BM + BN) * BK];
+#define QS(buf) (lds_all + (buf) * BM * BK)
+#define XS(buf) (lds_all + NBUF * BM * BK + (buf) * BN * BK)
   __shared__ float row_min[BM];
   __shared__ int row_min_slot[BM];
   __shared__ float q_score[QCAP];
@@ -117,54 +120,71 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   // candidate slice this block owns: [qb][swath][BM][k]
   size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
 
-  // per-lane Q row pointers for the 8 m-fragments (row clamped once)
-  const bf16* qrow[8];
-#pragma unroll
-  for (int m = 0; m < 8; ++m) {
-    long long gr = row0 + wm * 128 + m * 16 + lrow;
-    if (gr >= nq) gr = nq - 1;  // garbage rows filtered in the push phase
-    qrow[m] = Q + gr * D;
-  }
-
   for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
     f32x4 acc[8][4] = {};
+    // prologue: 2 K-tiles in flight (each stage = 4 glds per lane:
+    // 2 for the Q tile + 2 for the X tile)
+    stage_tile(Q, D, row0, nq, 0, QS(0), BM);
     stage_tile(X, D, x0, (long long)nx, 0, XS(0), BN);
+    stage_tile(Q, D, row0, nq, BK, QS(1), BM);
+    stage_tile(X, D, x0, (long long)nx, BK, XS(1), BN);
+    for (int kt = 0; kt < nk; ++kt) {
+      int cur = kt % NBUF;
+      // stage(kt) landed when only stage(kt+1)'s 4 glds are outstanding
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      // raw barrier: (a) stage(kt) visible block-wide, (b) every wave is
+      // done computing kt-1, so its buffer ((kt+2) % NBUF) is free.
+      // Safe without the __syncthreads() fence: all LDS reads of kt-1
+      // were consumed by mfma issue before this point.
+      __builtin_amdgcn_s_barrier();
+      if (kt + 2 < nk) {
+        int pre = (kt + 2) % NBUF;
+        stage_tile(Q, D, row0, nq, (kt + 2) * BK, QS(pre), BM);
+        stage_tile(X, D, x0, (long long)nx, (kt + 2) * BK, XS(pre), BN);
+      }
+      // Fragment loads as ONE inline-asm block: a plain ds_read after
+      // global_load_lds makes the memory legalizer insert s_waitcnt
+      // vmcnt(0) (it cannot prove the read targets an already-landed
+      // buffer), which drains the 2-deep prefetch every K-step — the
+      // exact stall the guide's "inline-asm K-loop" path removes. The
+      // swizzle term (lrow>>2)&3 is invariant across fragments (row
+      // deltas are multiples of 16), so fragments sit at base + n*1024.
+      uint32_t base_x = (uint32_t)(size_t)XS(cur)
+                        + lds_off_bytes(wn * 64 + lrow, kgrp);
+      uint32_t base_q = (uint32_t)(size_t)QS(cur)
+                        + lds_off_bytes(wm * 128 + lrow, kgrp);
+      bf16x8 xf[4], qf[8];
+      asm volatile(
+          "ds_read_b128 %0, %12\n\t"
+          "ds_read_b128 %1, %12 offset:1024\n\t"
+          "ds_read_b128 %2, %12 offset:2048\n\t"
+          "ds_read_b128 %3, %12 offset:3072\n\t"
+          "ds_read_b128 %4, %13\n\t"
+          "ds_read_b128 %5, %13 offset:1024\n\t"
+          "ds_read_b128 %6, %13 offset:2048\n\t"
+          "ds_read_b128 %7, %13 offset:3072\n\t"
+          "ds_read_b128 %8, %13 offset:4096\n\t"
+          "ds_read_b128 %9, %13 offset:5120\n\t"
+          "ds_read_b128 %10, %13 offset:6144\n\t"
+          "ds_read_b128 %11, %13 offset:7168\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          // early-clobber: ds_read destinations land asynchronously and
+          // must never alias the (still-live) address inputs
+          : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
+            "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
+            "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7])
+          : "v"(base_x), "v"(base_q));
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[m], xf[n], acc[m][n], 0, 0, 0);
+    }
+    // drain remaining staging and make the last compute visible before
+    // the push phase reuses LDS-adjacent state
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    for (int kt = 0; kt < nk; ++kt) {
-      int cur = kt & 1, nxt = cur ^ 1;
-      if (kt + 1 < nk) {
-        stage_tile(X, D, x0, (long long)nx, (kt + 1) * BK, XS(nxt), BN);
-      }
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {  // two 32-deep mfma steps per BK
-        int slot = ks * 4 + kgrp;
-        int qoff = kt * BK + ks * 32 + kgrp * 8;
-        bf16x8 xf[4];
-#pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          uint32_t r = wn * 64 + n * 16 + lrow;
-          xf[n] = *(const bf16x8*)((const char*)XS(cur) + lds_off_bytes(r, slot));
-        }
-        // m-fragments in halves of 4: keeps peak live registers at
-        // acc(128) + xf(16) + qf(16) instead of + qf(32)
-#pragma unroll
-        for (int mh = 0; mh < 2; ++mh) {
-          bf16x8 qf[4];
-#pragma unroll
-          for (int mi = 0; mi < 4; ++mi)
-            qf[mi] = *(const bf16x8*)(qrow[mh * 4 + mi] + qoff);
-#pragma unroll
-          for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-            for (int n = 0; n < 4; ++n)
-              acc[mh * 4 + mi][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  qf[mi], xf[n], acc[mh * 4 + mi][n], 0, 0, 0);
-        }
-      }
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      __syncthreads();
-    }
 
     // ---- streaming top-k from the accumulators -------------------------
     // lane holds acc[m][n][r] at row = wm*128 + m*16 + (lane>>4)*4 + r,
