@@ -1,0 +1,184 @@
+"""Brainplex tests: scanner (parse/discovery/agent shapes), configurator
+(trust heuristics, generated configs), writer (never-overwrite, backup,
+entries/allow merge), CLI flow (init, dry-run)."""
+
+import json
+import os
+
+import pytest
+
+from vainplex_openclaw_amd.brainplex import (
+    build_trust_defaults,
+    compute_trust_score,
+    extract_agents,
+    find_config,
+    generate_configs,
+    parse_args,
+    parse_config,
+    run_init,
+    scan,
+    update_openclaw_config,
+    write_configs,
+)
+
+
+# -- scanner -----------------------------------------------------------------
+
+def test_parse_config_json5_tolerance():
+    raw = """{
+      // comment
+      "agents": [ {"id": "main"}, ], /* block */
+    }"""
+    cfg = parse_config(raw)
+    assert cfg["agents"][0]["id"] == "main"
+
+
+def test_find_config_walk_up_and_nested(tmp_path):
+    deep = tmp_path / "a" / "b" / "c"
+    deep.mkdir(parents=True)
+    (tmp_path / "a" / ".openclaw").mkdir()
+    (tmp_path / "a" / ".openclaw" / "openclaw.json").write_text("{}")
+    assert find_config(str(deep), home=str(tmp_path)) == str(
+        tmp_path / "a" / ".openclaw" / "openclaw.json"
+    )
+    (tmp_path / "a" / "b" / "openclaw.json").write_text("{}")
+    assert find_config(str(deep), home=str(tmp_path)) == str(tmp_path / "a" / "b" / "openclaw.json")
+
+
+def test_find_config_home_fallback(tmp_path):
+    (tmp_path / ".openclaw").mkdir()
+    (tmp_path / ".openclaw" / "openclaw.json").write_text("{}")
+    lonely = tmp_path / "elsewhere"
+    lonely.mkdir()
+    got = find_config(str(lonely), home=str(tmp_path))
+    # walk-up from tmp may find nothing above; home fallback applies
+    assert got is None or got.endswith("openclaw.json")
+
+
+@pytest.mark.parametrize("agents,expected", [
+    ([{"id": "main"}, {"name": "forge"}, "cerberus"], ["main", "forge", "cerberus"]),
+    ({"list": [{"id": "main"}]}, ["main"]),
+    ({"definitions": [{"name": "viola"}]}, ["viola"]),
+    ({"main": {}, "forge": {}, "defaults": {}}, ["main", "forge"]),
+    (None, []),
+])
+def test_extract_agents_four_shapes(agents, expected):
+    assert extract_agents({"agents": agents} if agents is not None else {}) == expected
+
+
+# -- configurator ------------------------------------------------------------
+
+@pytest.mark.parametrize("name,score", [
+    ("*", 10), ("admin-bot", 70), ("ROOT", 70), ("main", 60), ("the-main-one", 60),
+    ("reviewer", 50), ("cerberus", 50), ("forge", 45), ("builder", 45), ("misc", 40),
+])
+def test_trust_score_heuristics(name, score):
+    assert compute_trust_score(name) == score
+
+
+def test_build_trust_defaults_always_wildcard():
+    d = build_trust_defaults(["main", "forge"])
+    assert d == {"main": 60, "forge": 45, "*": 10}
+
+
+def test_generate_configs_core_and_full():
+    core = generate_configs(["main"], "UTC", full=False)
+    ids = [c["pluginId"] for c in core]
+    assert "openclaw-governance" in ids and "openclaw-membrane" in ids
+    assert "openclaw-knowledge-engine" not in ids
+    full = generate_configs(["main"], "UTC", full=True)
+    assert "openclaw-knowledge-engine" in [c["pluginId"] for c in full]
+    gov = next(c for c in core if c["pluginId"] == "openclaw-governance")
+    assert gov["config"]["trust"]["defaults"]["main"] == 60
+    assert gov["config"]["nightMode"] == {"enabled": True, "start": "23:00", "end": "06:00"}
+    mem = next(c for c in core if c["pluginId"] == "openclaw-membrane")
+    assert mem["config"]["buffer_size"] == 10
+    assert mem["config"]["retrieve_min_salience"] == 0.1
+    assert mem["config"]["retrieve_max_sensitivity"] == "medium"
+
+
+# -- writer ------------------------------------------------------------------
+
+def test_write_configs_never_overwrites(tmp_path):
+    configs = [{"pluginId": "p1", "config": {"a": 1}}]
+    r1 = write_configs(configs, home=str(tmp_path))
+    assert r1["written"] == ["p1"]
+    path = tmp_path / ".openclaw" / "plugins" / "p1" / "config.json"
+    path.write_text('{"custom": true}')
+    r2 = write_configs(configs, home=str(tmp_path))
+    assert r2["skipped"] == ["p1"]
+    assert json.loads(path.read_text()) == {"custom": True}
+
+
+def test_update_openclaw_config_merge_and_backup(tmp_path):
+    p = tmp_path / "openclaw.json"
+    original = {"agents": [{"id": "main"}],
+                "plugins": {"entries": {"existing": {"enabled": False}}, "allow": ["existing"]}}
+    p.write_text(json.dumps(original))
+    r = update_openclaw_config(str(p), original, ["new-plugin", "existing"])
+    assert r["added_entries"] == ["new-plugin"]
+    assert r["added_allow"] == ["new-plugin"]
+    assert r["backed_up"] and (tmp_path / "openclaw.json.bak").exists()
+    merged = json.loads(p.read_text())
+    assert merged["plugins"]["entries"]["existing"] == {"enabled": False}  # preserved
+    assert merged["plugins"]["entries"]["new-plugin"] == {"enabled": True}
+
+
+def test_update_openclaw_config_noop(tmp_path):
+    p = tmp_path / "openclaw.json"
+    cfg = {"plugins": {"entries": {"x": {"enabled": True}}, "allow": ["x"]}}
+    p.write_text(json.dumps(cfg))
+    r = update_openclaw_config(str(p), cfg, ["x"])
+    assert not r["updated"] and not (tmp_path / "openclaw.json.bak").exists()
+
+
+# -- CLI ---------------------------------------------------------------------
+
+def test_parse_args_flags():
+    o = parse_args(["init", "--full", "--dry-run", "--config", "/x", "--verbose"])
+    assert o["full"] and o["dry_run"] and o["config_path"] == "/x" and o["verbose"]
+    with pytest.raises(SystemExit):
+        parse_args(["--bogus"])
+
+
+def test_run_init_end_to_end(tmp_path):
+    ws = tmp_path / "ws"
+    ws.mkdir()
+    (ws / "openclaw.json").write_text(json.dumps({"agents": [{"id": "main"}, {"id": "forge"}]}))
+    msgs = []
+    result = run_init({"full": True, "dry_run": False, "config_path": None},
+                      start_dir=str(ws), home=str(tmp_path), echo=msgs.append)
+    assert not result["install"]["failed"]
+    assert len(result["install"]["installed"]) == 6  # 5 core + KE
+    assert set(result["written"]["written"]) >= {"openclaw-governance", "openclaw-cortex"}
+    merged = json.loads((ws / "openclaw.json").read_text())
+    assert merged["plugins"]["entries"]["openclaw-governance"] == {"enabled": True}
+    assert "openclaw-membrane" in merged["plugins"]["allow"]
+    # generated governance config has heuristic trust for scanned agents
+    gov = json.loads((tmp_path / ".openclaw" / "plugins" / "openclaw-governance" /
+                      "config.json").read_text())
+    assert gov["trust"]["defaults"] == {"main": 60, "forge": 45, "*": 10}
+
+
+def test_run_init_dry_run_writes_nothing(tmp_path):
+    ws = tmp_path / "ws"
+    ws.mkdir()
+    (ws / "openclaw.json").write_text(json.dumps({"agents": []}))
+    before = (ws / "openclaw.json").read_text()
+    run_init({"full": False, "dry_run": True, "config_path": None},
+             start_dir=str(ws), home=str(tmp_path), echo=lambda *a: None)
+    assert (ws / "openclaw.json").read_text() == before
+    assert not (tmp_path / ".openclaw" / "plugins").exists()
+
+
+def test_run_init_idempotent(tmp_path):
+    ws = tmp_path / "ws"
+    ws.mkdir()
+    (ws / "openclaw.json").write_text(json.dumps({"agents": [{"id": "main"}]}))
+    run_init({"full": False, "dry_run": False, "config_path": None},
+             start_dir=str(ws), home=str(tmp_path), echo=lambda *a: None)
+    first = (ws / "openclaw.json").read_text()
+    r2 = run_init({"full": False, "dry_run": False, "config_path": None},
+                  start_dir=str(ws), home=str(tmp_path), echo=lambda *a: None)
+    assert (ws / "openclaw.json").read_text() == first
+    assert r2["written"]["written"] == []  # all kept
