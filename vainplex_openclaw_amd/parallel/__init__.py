@@ -1,0 +1,5 @@
+"""Distributed (DP over RCCL/xGMI) collective helpers."""
+
+from . import collectives
+
+__all__ = ["collectives"]

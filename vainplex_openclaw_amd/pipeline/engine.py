@@ -39,6 +39,7 @@ import torch
 
 from ..ops import gpu as g
 from ..ops import pattern_sets
+from ..parallel import collectives as coll
 from .synth import SynthBatch
 
 
@@ -131,37 +132,24 @@ class FirewallPipeline:
         # 4. classifier head (fused sigmoid)
         logits = g.gemm_nt(feats, self.head, bias=self.head_bias, act=1)
 
-        # 5. Membrane recall (full index across ranks)
+        # 5. Membrane recall (full index across ranks; parallel/collectives)
         if self.world_size > 1 and torch.distributed.is_initialized():
-            q_all = torch.empty(
-                self.world_size * B, cfg.dim, dtype=torch.bfloat16, device=self.device
-            )
-            torch.distributed.all_gather_into_tensor(q_all, feats)
+            q_all = coll.allgather_queries(feats, self.world_size)
             scores, ids = g.topk_recall(q_all, self.index, cfg.topk)
-            ids = ids + self.rank * cfg.index_size  # globalize shard ids
-            # gather every rank's candidates, keep my rows, merge
-            cand_s = torch.empty(self.world_size, q_all.shape[0], cfg.topk, device=self.device)
-            cand_i = torch.empty(
-                self.world_size, q_all.shape[0], cfg.topk, dtype=torch.int32, device=self.device
+            ids = coll.globalize_ids(ids, self.rank, cfg.index_size)
+            recall_scores, recall_ids = coll.merge_topk_candidates(
+                scores, ids, self.rank, B, self.world_size, cfg.topk
             )
-            torch.distributed.all_gather_into_tensor(cand_s, scores.unsqueeze(0).contiguous())
-            torch.distributed.all_gather_into_tensor(cand_i, ids.unsqueeze(0).contiguous())
-            my0 = self.rank * B
-            mine_s = cand_s[:, my0 : my0 + B].permute(1, 0, 2).reshape(B, -1)
-            mine_i = cand_i[:, my0 : my0 + B].permute(1, 0, 2).reshape(B, -1)
-            top = torch.topk(mine_s, cfg.topk, dim=1)
-            recall_scores = top.values
-            recall_ids = torch.gather(mine_i, 1, top.indices)
         else:
             recall_scores, recall_ids = g.topk_recall(feats, self.index, cfg.topk)
 
         # salience reinforcement + decay (Membrane recall semantics)
-        flat_local = recall_ids.reshape(-1)
         if self.world_size > 1:
-            base = self.rank * cfg.index_size
-            local_mask = (flat_local >= base) & (flat_local < base + cfg.index_size)
-            flat_local = flat_local[local_mask] - base
+            flat_local = coll.local_shard_ids(
+                recall_ids.reshape(-1), self.rank, cfg.index_size
+            )
         else:
+            flat_local = recall_ids.reshape(-1)
             flat_local = flat_local[flat_local >= 0]  # -1 = unfilled slot
         self.salience.mul_(0.9999)
         self.salience.index_add_(
@@ -192,8 +180,7 @@ class FirewallPipeline:
         root = g.merkle_root(leaves)
 
         if self.world_size > 1 and torch.distributed.is_initialized():
-            roots = torch.empty(self.world_size, 32, dtype=torch.uint8, device=self.device)
-            torch.distributed.all_gather_into_tensor(roots, root.unsqueeze(0))
+            roots = coll.allgather_roots(root, self.world_size)
             root = g.merkle_root(roots)
 
         self.batch_seq += 1
