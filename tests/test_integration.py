@@ -1,0 +1,121 @@
+"""Full-suite integration: one Gateway, all seven plugins, a scripted
+conversation driven through the hook bus.
+
+Mirrors the reference's data flow (suite README.md:60-106):
+message -> Governance (gate) -> Membrane (recall/inject) -> agent ->
+Cortex + KE + Membrane-ingest; all components -> EventStore -> Leuko.
+"""
+
+import json
+import os
+
+import pytest
+
+from vainplex_openclaw_amd.core.api import NullLogger
+from vainplex_openclaw_amd.core.gateway import Gateway
+from vainplex_openclaw_amd.cortex.hooks import create_plugin as create_cortex
+from vainplex_openclaw_amd.eventstore import EventJournal
+from vainplex_openclaw_amd.eventstore.plugin import create_plugin as create_eventstore
+from vainplex_openclaw_amd.governance.plugin import create_plugin as create_governance
+from vainplex_openclaw_amd.knowledge.hooks import create_plugin as create_knowledge
+from vainplex_openclaw_amd.leuko.plugin import create_plugin as create_leuko
+from vainplex_openclaw_amd.membrane.hooks import create_plugin as create_membrane
+
+
+@pytest.fixture()
+def suite(tmp_path, monkeypatch):
+    monkeypatch.setenv("HOME", str(tmp_path))  # isolate ~/.openclaw
+    ws = str(tmp_path / "workspace")
+    os.makedirs(ws, exist_ok=True)
+    gw = Gateway(config={"agents": [{"id": "main"}, {"id": "forge"}]},
+                 logger=NullLogger(), home=str(tmp_path))
+    journal = EventJournal(durable=False)
+    gw.load(create_governance(workspace=ws), {})
+    gw.load(create_cortex(workspace=ws), {})
+    gw.load(create_knowledge(workspace=ws), {})
+    gw.load(create_membrane(workspace=ws), {})
+    gw.load(create_eventstore(journal=journal), {})
+    gw.load(create_leuko(workspace=ws, journal=journal), {})
+    return gw, journal, ws
+
+
+def _msg(gw, content, agent="main"):
+    return gw.bus.emit("message_received", {
+        "content": content, "from": "user",
+        "ctx": {"agentId": agent, "sessionKey": f"{agent}:test:1"},
+    })
+
+
+def test_suite_loads_all_plugins(suite):
+    gw, journal, ws = suite
+    assert set(gw.plugins) == {
+        "openclaw-governance", "openclaw-cortex", "openclaw-knowledge-engine",
+        "openclaw-membrane", "nats-eventstore", "openclaw-leuko",
+    }
+
+
+def test_message_flow_fans_out_to_all_plugins(suite):
+    gw, journal, ws = suite
+    gw.bus.emit("session_start", {"sessionId": "s1", "ctx": {"sessionKey": "main:test:1"}})
+    _msg(gw, "We decided to use postgres for the storage layer.")
+    _msg(gw, "Contact ada@example.org at Acme Corp. about the migration")
+    ev = _msg(gw, "what did we decide about storage?")
+
+    # Membrane injected the earlier memory
+    assert "postgres" in ev.get("membrane_context", "")
+    # Cortex tracked a thread (threads.json persisted per message)
+    cortex = gw.plugins["openclaw-cortex"]
+    ws_state = cortex.hooks.ws(ws)
+    assert ws_state.threads.threads or ws_state.decisions.decisions
+    # KE extracted entities
+    ke = gw.plugins["openclaw-knowledge-engine"]
+    assert any(e.type == "email" for e in ke.hooks.entities.values())
+    # EventStore captured envelopes for every message
+    types = [e["canonicalType"] for _s, e in journal.replay()]
+    assert types.count("message.in.received") == 3
+    assert "session.started" in types
+
+
+def test_tool_call_gate_and_audit_flow(suite):
+    gw, journal, ws = suite
+    ev = gw.bus.emit("before_tool_call", {
+        "toolName": "exec", "params": {"command": "ls /tmp"},
+        "ctx": {"agentId": "main", "sessionKey": "main:test:1", "toolCallId": "t1"},
+    })
+    assert not ev.get("block")  # benign command passes
+    # eventstore saw the tool call request
+    assert any(e["canonicalType"] == "tool.call.requested" for _s, e in journal.replay())
+
+
+def test_redaction_protects_outbound(suite):
+    gw, journal, ws = suite
+    ev = gw.bus.emit("message_sending", {
+        "content": "here is the key sk-abcdefghijklmnopqrst1234",
+        "to": "user", "ctx": {"agentId": "main", "sessionKey": "main:test:1"},
+    })
+    out = ev.get("content", "")
+    assert "sk-abcdefghijklmnopqrst1234" not in out
+    assert "[REDACTED:" in out
+
+
+def test_leuko_sees_journal_and_reports(suite):
+    gw, journal, ws = suite
+    _msg(gw, "hello")
+    leuko = gw.plugins["openclaw-leuko"]
+    report = leuko.run_once()
+    assert report["version"] == 1
+    assert "nats" in report["collectors"]
+    # journal collector enabled via injected journal
+    assert report["health"]["details"]["nats"] in ("ok", "warn")
+
+
+def test_gateway_stop_flushes_everything(suite):
+    gw, journal, ws = suite
+    gw.bus.emit("session_start", {"sessionId": "s1", "ctx": {"sessionKey": "main:test:1"}})
+    _msg(gw, "We decided to ship v2 on friday. I will prepare the notes.")
+    gw.bus.emit("gateway_stop", {})
+    # KE fact store flushed to disk
+    assert os.path.exists(os.path.join(ws, "facts.json"))
+    # membrane store flushed
+    membrane = gw.plugins["openclaw-membrane"]
+    assert membrane.engine.stats["ingested"] >= 1

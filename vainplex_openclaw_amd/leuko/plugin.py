@@ -127,8 +127,11 @@ class LeukoPlugin:
         self.detector = AnomalyDetector(MetricHistory(hist_path, clock=self._clock),
                                         clock=self._clock)
         if self.journal is not None:
-            self.config["collectors"].setdefault("nats", {})["journal"] = self.journal
-            self.config["collectors"]["nats"].setdefault("enabled", True)
+            # in-process journal injected: the event-backbone collector is
+            # live by construction (sitrep's `nats` collector needed a CLI)
+            ncfg = self.config["collectors"].setdefault("nats", {})
+            ncfg["journal"] = self.journal
+            ncfg["enabled"] = True
 
         api.register_service({
             "id": self.id,
