@@ -27,6 +27,8 @@ __global__ void gemm_nt_bf16_kernel(const __bf16*, const __bf16*, float*, __bf16
                                     const float*, int, int, int, int, int);
 __global__ void topk_recall_kernel(const __bf16*, const __bf16*, int, int, int, int,
                                    int, float*, int32_t*);
+__global__ void topk_recall_fp8_kernel(const uint8_t*, const uint8_t*, int, int, int,
+                                       int, int, float*, int32_t*);
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
                                   float*, int32_t*);
 __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigned long long*,
@@ -184,6 +186,42 @@ std::vector<torch::Tensor> topk_recall(torch::Tensor Q, torch::Tensor X, int64_t
   return {out_s, out_i};
 }
 
+std::vector<torch::Tensor> topk_recall_fp8(torch::Tensor Q8, torch::Tensor X8,
+                                           int64_t k, int64_t n_swaths) {
+  // stage-1 fp8 scan of the two-stage recall: inputs are e4m3 bytes
+  // (viewed as uint8), pre-scaled x8; scores rank candidates only.
+  CHECK_GPU(Q8); CHECK_CONTIG(Q8); CHECK_GPU(X8); CHECK_CONTIG(X8);
+  TORCH_CHECK(Q8.dtype() == torch::kUInt8 && X8.dtype() == torch::kUInt8,
+              "fp8 operands passed as uint8 views");
+  int nq = Q8.size(0), D = Q8.size(1);
+  long long nx = X8.size(0);
+  TORCH_CHECK(X8.size(1) == D && D % 64 == 0, "D must be a multiple of 64");
+  TORCH_CHECK(k >= 1 && k <= 64, "k in [1,64]");
+  int n_qblocks = (nq + 255) / 256;
+  TORCH_CHECK(n_swaths >= 1 && (long long)n_swaths * k <= 1024,
+              "n_swaths * k must be <= 1024");
+  auto f32opts = torch::dtype(torch::kFloat32).device(Q8.device());
+  auto i32opts = torch::dtype(torch::kInt32).device(Q8.device());
+  auto cand_s = torch::full({(long long)n_qblocks * n_swaths * 256 * k}, -1e30,
+                            f32opts);
+  auto cand_i = torch::full({(long long)n_qblocks * n_swaths * 256 * k}, -1,
+                            i32opts);
+  dim3 grid((unsigned)(n_qblocks * n_swaths));
+  hipLaunchKernelGGL(topk_recall_fp8_kernel, grid, dim3(512), 0, cur_stream(),
+                     Q8.data_ptr<uint8_t>(), X8.data_ptr<uint8_t>(), nq, (int)nx,
+                     D, (int)k, (int)n_swaths, cand_s.data_ptr<float>(),
+                     cand_i.data_ptr<int32_t>());
+  auto out_s = torch::empty({nq, k}, f32opts);
+  auto out_i = torch::empty({nq, k}, i32opts);
+  int waves_per_block = 4;
+  int blocks = (nq + waves_per_block - 1) / waves_per_block;
+  hipLaunchKernelGGL(topk_merge_kernel, dim3(blocks), dim3(waves_per_block * 64), 0,
+                     cur_stream(), cand_s.data_ptr<float>(), cand_i.data_ptr<int32_t>(),
+                     nq, (int)k, (int)n_swaths, out_s.data_ptr<float>(),
+                     out_i.data_ptr<int32_t>());
+  return {out_s, out_i};
+}
+
 std::vector<torch::Tensor> firewall_verdict(
     torch::Tensor inj_hits, torch::Tensor red_hits, torch::Tensor logits,
     torch::Tensor agent_idx, torch::Tensor agent_trust, torch::Tensor tool_risk,
@@ -253,6 +291,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("bias") = c10::nullopt,
         py::arg("act") = 0, py::arg("out_bf16") = false);
   m.def("topk_recall", &topk_recall, "Fused cosine top-k recall");
+  m.def("topk_recall_fp8", &topk_recall_fp8, "fp8 stage-1 scan of two-stage recall");
   m.def("firewall_verdict", &firewall_verdict, "Fused verdict/risk/trust-delta");
   m.def("trust_recompute", &trust_recompute, "Agent trust score recompute");
   m.def("audit_pack", &audit_pack, "Pack 64-byte audit records");

@@ -309,3 +309,203 @@ extern "C" __global__ void topk_merge_kernel(
     }
   }
 }
+
+// ===========================================================================
+// fp8 scan variant: stage-1 of the two-stage recall. Index + queries are
+// OCP e4m3 (pre-scaled x8 so typical unit-vector components sit in the
+// normal range); scores rank candidates only — stage-2 rescores the
+// overfetched top-k2 exactly in bf16 (ops/gpu.py topk_recall_two_stage).
+//
+// Same geometry and byte layout as the bf16 kernel: BK_F8 = 64 fp8
+// elements = the SAME 64 B rows / 4 x 16 B slots / swizzle, so
+// stage_tile and the asm ds_read block carry over. Each staged tile
+// feeds TWO v_mfma_f32_16x16x32_fp8_fp8 sub-steps; a lane's 16 B
+// fragment holds the i64 operands for both k-halves of its slot pair
+// (sub-step s reads slots {2s, 2s+1}; lane kgrp selects slot s*2 +
+// (kgrp>>1), i64 half kgrp&1).
+// ===========================================================================
+
+typedef long long i64x2 __attribute__((ext_vector_type(2)));
+#define BK_F8 64
+
+DEVINL void stage_tile8(const uint8_t* __restrict__ src, long long ld,
+                        long long row0, long long row_max, int k0,
+                        bf16* lds_base, int tile_rows) {
+  // identical piece layout: 64 B rows = 4 x 16 B swizzled slots
+  stage_tile((const bf16*)src, ld / 2, row0, row_max, k0 / 2, lds_base, tile_rows);
+}
+
+extern "C" __global__ void __launch_bounds__(TK_THREADS)
+topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X,
+                       int nq, int nx, int D, int k, int n_swaths,
+                       float* __restrict__ cand_scores,
+                       int32_t* __restrict__ cand_ids) {
+  __shared__ bf16 lds_all[NBUF * (BM + BN) * (BK_F8 / 2)];
+#define QS8(buf) (lds_all + (buf) * BM * (BK_F8 / 2))
+#define XS8(buf) (lds_all + NBUF * BM * (BK_F8 / 2) + (buf) * BN * (BK_F8 / 2))
+  __shared__ float row_min[BM];
+  __shared__ int row_min_slot[BM];
+  __shared__ float q_score[QCAP];
+  __shared__ uint32_t q_meta[QCAP];
+  __shared__ int q_count;
+  __shared__ int q_overflow;
+
+  int S = n_swaths;
+  int qb = blockIdx.x / S;
+  int swath = blockIdx.x % S;
+  long long row0 = (long long)qb * BM;
+
+  long long per = ((long long)nx + S - 1) / S;
+  per = ((per + BN - 1) / BN) * BN;
+  long long x_begin = (long long)swath * per;
+  long long x_end = min((long long)nx, x_begin + per);
+
+  for (int i = threadIdx.x; i < BM; i += blockDim.x) {
+    row_min[i] = -1e30f;
+    row_min_slot[i] = 0;
+  }
+  if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
+  __syncthreads();
+
+  int wid = wave_id();
+  int wm = wid >> 2, wn = wid & 3;
+  int lane = lane_id();
+  int lrow = lane & 15;
+  int kgrp = lane >> 4;
+  int half = kgrp & 1;            // i64 half within the 16 B fragment
+  int nk = D / BK_F8;
+  size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
+
+  for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
+    f32x4 acc[8][4] = {};
+    stage_tile8(Q, D, row0, nq, 0, QS8(0), BM);
+    stage_tile8(X, D, x0, (long long)nx, 0, XS8(0), BN);
+    stage_tile8(Q, D, row0, nq, BK_F8, QS8(1), BM);
+    stage_tile8(X, D, x0, (long long)nx, BK_F8, XS8(1), BN);
+    for (int kt = 0; kt < nk; ++kt) {
+      int cur = kt % NBUF;
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (kt + 2 < nk) {
+        int pre = (kt + 2) % NBUF;
+        stage_tile8(Q, D, row0, nq, (kt + 2) * BK_F8, QS8(pre), BM);
+        stage_tile8(X, D, x0, (long long)nx, (kt + 2) * BK_F8, XS8(pre), BN);
+      }
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {  // k-halves 0..31, 32..63
+        uint32_t slot = sub * 2 + (kgrp >> 1);
+        uint32_t base_x = (uint32_t)(size_t)XS8(cur)
+                          + lds_off_bytes(wn * 64 + lrow, slot);
+        uint32_t base_q = (uint32_t)(size_t)QS8(cur)
+                          + lds_off_bytes(wm * 128 + lrow, slot);
+        bf16x8 xf[4], qf[8];
+        asm volatile(
+            "ds_read_b128 %0, %12\n\t"
+            "ds_read_b128 %1, %12 offset:1024\n\t"
+            "ds_read_b128 %2, %12 offset:2048\n\t"
+            "ds_read_b128 %3, %12 offset:3072\n\t"
+            "ds_read_b128 %4, %13\n\t"
+            "ds_read_b128 %5, %13 offset:1024\n\t"
+            "ds_read_b128 %6, %13 offset:2048\n\t"
+            "ds_read_b128 %7, %13 offset:3072\n\t"
+            "ds_read_b128 %8, %13 offset:4096\n\t"
+            "ds_read_b128 %9, %13 offset:5120\n\t"
+            "ds_read_b128 %10, %13 offset:6144\n\t"
+            "ds_read_b128 %11, %13 offset:7168\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
+              "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
+              "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7])
+            : "v"(base_x), "v"(base_q));
+        long long xv[4], qv[8];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          i64x2 t = __builtin_bit_cast(i64x2, xf[n]);
+          xv[n] = half ? t.y : t.x;
+        }
+#pragma unroll
+        for (int m = 0; m < 8; ++m) {
+          i64x2 t = __builtin_bit_cast(i64x2, qf[m]);
+          qv[m] = half ? t.y : t.x;
+        }
+#pragma unroll
+        for (int m = 0; m < 8; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                qv[m], xv[n], acc[m][n], 0, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    // ---- streaming top-k (identical to the bf16 kernel) ----------------
+    unsigned long long pend0 = ~0ull, pend1 = ~0ull;
+    for (int round = 0; ; ++round) {
+      unsigned long long still0 = 0ull, still1 = 0ull;
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int vi = (m & 3) * 16 + n * 4 + r;
+            unsigned long long bit = 1ull << vi;
+            if (!(((m < 4) ? pend0 : pend1) & bit)) continue;
+            int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+            if ((row0 + row) >= nq) continue;
+            long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+            if (col >= x_end) continue;
+            float v = acc[m][n][r];
+            if (!(v > row_min[row])) continue;
+            int idx = atomicAdd(&q_count, 1);
+            if (idx < QCAP) {
+              q_score[idx] = v;
+              q_meta[idx] = (uint32_t(row) << 16) | uint32_t(col - x0);
+            } else {
+              if (m < 4) still0 |= bit; else still1 |= bit;
+              atomicExch(&q_overflow, 1);
+            }
+          }
+      __syncthreads();
+      int total = min(q_count, QCAP);
+      if (threadIdx.x < BM) {
+        int my_row = threadIdx.x;
+        float rmin = row_min[my_row];
+        int rslot = row_min_slot[my_row];
+        float* cs = cand_scores + cbase + (size_t)my_row * k;
+        int32_t* ci = cand_ids + cbase + (size_t)my_row * k;
+        bool touched = false;
+        for (int i = 0; i < total; ++i) {
+          uint32_t meta = q_meta[i];
+          if (int(meta >> 16) != my_row) continue;
+          float v = q_score[i];
+          if (v > rmin) {
+            cs[rslot] = v;
+            ci[rslot] = int32_t(x0 + (meta & 0xFFFFu));
+            float mn = cs[0];
+            int ms = 0;
+            for (int j = 1; j < k; ++j)
+              if (cs[j] < mn) { mn = cs[j]; ms = j; }
+            rmin = mn;
+            rslot = ms;
+            touched = true;
+          }
+        }
+        if (touched) {
+          row_min[my_row] = rmin;
+          row_min_slot[my_row] = rslot;
+        }
+      }
+      __syncthreads();
+      int of = q_overflow;
+      __syncthreads();
+      if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
+      pend0 = still0;
+      pend1 = still1;
+      if (!of) break;
+      __syncthreads();
+    }
+    __syncthreads();
+  }
+}

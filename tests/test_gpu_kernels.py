@@ -204,3 +204,25 @@ def test_audit_pack_roundtrip():
     root = g.merkle_root(leaves)
     ref = g.reference_merkle_root([bytes(raw[i]) for i in range(B_)])
     assert bytes(root.cpu().numpy()) == ref
+
+
+@pytest.mark.gpu
+def test_topk_recall_two_stage_vs_torch():
+    """fp8 scan + exact bf16 rescore must match the fp32 reference at
+    least as well as the pure-bf16 kernel (final scores are exact)."""
+    torch.manual_seed(5)
+    nq, nx, D, k = 512, 16384, 1024, 16
+    Q = torch.nn.functional.normalize(torch.randn(nq, D, device="cuda"), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(nx, D, device="cuda"), dim=1).bfloat16()
+    X8 = g.to_fp8_bytes(X)
+    scores, ids = g.topk_recall_two_stage(Q, X, X8, k)
+    ref = torch.topk(Q.float() @ X.float().T, k, dim=1)
+    ids_np = ids.cpu().numpy()
+    ref_ids = ref.indices.cpu().numpy()
+    ref_vals = ref.values.cpu().numpy()
+    got_vals = scores.cpu().numpy()
+    for q in range(nq):
+        inter = len(set(ids_np[q]) & set(ref_ids[q]))
+        assert inter >= k - 2, f"q={q}: only {inter}/{k} overlap"
+        assert abs(got_vals[q][0] - ref_vals[q][0]) < 2e-2
+        assert all(got_vals[q][i] >= got_vals[q][i + 1] - 1e-6 for i in range(k - 1))

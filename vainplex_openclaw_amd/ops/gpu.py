@@ -185,6 +185,51 @@ def topk_recall(Q: torch.Tensor, X: torch.Tensor, k: int, n_swaths: int = 0) -> 
     return s, i
 
 
+def to_fp8_bytes(x: torch.Tensor, scale: float = 8.0) -> torch.Tensor:
+    """bf16 -> e4m3 bytes, pre-scaled so typical unit-vector components
+    (~N(0, 1/sqrt(D))) land in e4m3's normal range."""
+    return (x.float() * scale).to(torch.float8_e4m3fn).view(torch.uint8)
+
+
+def topk_recall_fp8(Q8: torch.Tensor, X8: torch.Tensor, k: int, n_swaths: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Stage-1 fp8 scan: candidate ids ranked by e4m3 cosine (csrc
+    topk_recall_fp8_kernel). Inputs are uint8 views of e4m3 bytes."""
+    if n_swaths <= 0:
+        qblocks = (Q8.shape[0] + 255) // 256
+        want = max(1, 256 // max(qblocks, 1))
+        n_swaths = max(8, (want // 8) * 8)
+        n_swaths = min(n_swaths, (1024 // max(k, 1)) // 8 * 8)
+        n_swaths = min(n_swaths, max(1, X8.shape[0] // 256))
+        n_swaths = max(1, n_swaths)
+    s, i = ext().topk_recall_fp8(Q8, X8, k, n_swaths)
+    return s, i
+
+
+def topk_recall_two_stage(
+    Q: torch.Tensor,
+    X: torch.Tensor,
+    X8: torch.Tensor,
+    k: int,
+    overfetch: int = 4,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Two-stage exact-rescore recall: fp8 scan of the full index for
+    k*overfetch candidates (half the staged bytes of the bf16 scan = the
+    measured GLDS transport bound), then exact bf16/fp32 rescore of the
+    candidates and final top-k. Final scores are EXACT cosines; stage-1
+    only has to keep the true top-k inside the candidate set (error
+    sigma ~0.003 vs candidate margins ~10x that).
+    """
+    k2 = min(64, max(k * overfetch, k))
+    Q8 = to_fp8_bytes(Q)
+    _s8, ids8 = topk_recall_fp8(Q8, X8, k2)
+    ids = ids8.long().clamp_min(0)  # -1 slots -> row 0 (rescored, never top)
+    cand = X[ids]  # [nq, k2, D] bf16 gather
+    exact = torch.einsum("qd,qkd->qk", Q.float(), cand.float())
+    exact = torch.where(ids8 < 0, torch.full_like(exact, -1e30), exact)
+    top = torch.topk(exact, k, dim=1)
+    return top.values, torch.gather(ids8, 1, top.indices)
+
+
 # -- firewall tail ----------------------------------------------------------
 
 def firewall_verdict(

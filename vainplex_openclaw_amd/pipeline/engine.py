@@ -52,6 +52,7 @@ class PipelineConfig:
     n_agents: int = 64
     index_size: int = 6_250_000  # per-GPU shard (8 GPUs x 6.25M = 50M)
     topk: int = 16
+    recall_fp8: bool = True  # two-stage fp8 scan + exact bf16 rescore
     inj_threshold: float = 0.9
     seed: int = 1234
     families: tuple = ("redaction", "injection", "claims", "entity")
@@ -85,6 +86,17 @@ class FirewallPipeline:
                 x = torch.nn.functional.normalize(x, dim=1)
                 self.index[i : i + n] = x.bfloat16()
                 del x
+
+            # fp8 (e4m3) copy of the index for the stage-1 scan: half the
+            # staged bytes at the measured GLDS transport bound
+            self.index8 = None
+            if cfg.recall_fp8:
+                self.index8 = torch.empty(
+                    cfg.index_size, cfg.dim, dtype=torch.uint8, device=self.device
+                )
+                for i in range(0, cfg.index_size, chunk):
+                    n = min(chunk, cfg.index_size - i)
+                    self.index8[i : i + n] = g.to_fp8_bytes(self.index[i : i + n])
 
             # salience state: recall strength + decay (Membrane semantics)
             self.salience = torch.ones(cfg.index_size, device=self.device)
@@ -133,15 +145,20 @@ class FirewallPipeline:
         logits = g.gemm_nt(feats, self.head, bias=self.head_bias, act=1)
 
         # 5. Membrane recall (full index across ranks; parallel/collectives)
+        def local_recall(queries):
+            if self.index8 is not None:
+                return g.topk_recall_two_stage(queries, self.index, self.index8, cfg.topk)
+            return g.topk_recall(queries, self.index, cfg.topk)
+
         if self.world_size > 1 and torch.distributed.is_initialized():
             q_all = coll.allgather_queries(feats, self.world_size)
-            scores, ids = g.topk_recall(q_all, self.index, cfg.topk)
-            ids = coll.globalize_ids(ids, self.rank, cfg.index_size)
+            scores, ids = local_recall(q_all)
+            ids = coll.globalize_ids(ids.to(torch.int32), self.rank, cfg.index_size)
             recall_scores, recall_ids = coll.merge_topk_candidates(
                 scores, ids, self.rank, B, self.world_size, cfg.topk
             )
         else:
-            recall_scores, recall_ids = g.topk_recall(feats, self.index, cfg.topk)
+            recall_scores, recall_ids = local_recall(feats)
 
         # salience reinforcement + decay (Membrane recall semantics)
         if self.world_size > 1:
