@@ -46,6 +46,8 @@ def main() -> int:
     ap.add_argument("--index", type=int, default=TOTAL_INDEX, help="TOTAL index rows across ranks")
     ap.add_argument("--topk", type=int, default=16)
     ap.add_argument("--pool", type=int, default=4, help="pre-generated batch pool size")
+    ap.add_argument("--profile", action="store_true",
+                    help="after the timed region, run 3 instrumented steps and print per-stage ms (stderr)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -130,6 +132,19 @@ def main() -> int:
                 "topk": args.topk,
             },
         }))
+    if args.profile:
+        # hipEvent per-stage timing, outside the timed region (the
+        # profiler synchronizes per step)
+        pipe.profiler.enabled = torch.cuda.is_available()
+        for i in range(3):
+            b, s2 = pool[i % len(pool)]
+            pipe.step(b, staged=s2)
+            pipe.profiler.commit()
+        if rank == 0:
+            print("stage ms/step: " + json.dumps(
+                {k: round(v, 3) for k, v in pipe.profiler.summary().items()}
+            ), file=sys.stderr)
+
     if distributed:
         torch.distributed.destroy_process_group()
     return 0

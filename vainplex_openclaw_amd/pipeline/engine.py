@@ -58,13 +58,50 @@ class PipelineConfig:
     families: tuple = ("redaction", "injection", "claims", "entity")
 
 
+class StageProfiler:
+    """hipEvent (torch.cuda.Event) timing per pipeline stage (SURVEY.md §5:
+    keep the reference's µs self-timing; add device-side stage timing).
+    Enabled via FirewallPipeline(..., profile=True); `summary()` returns
+    average ms per stage over recorded steps."""
+
+    def __init__(self, enabled: bool = False):
+        self.enabled = enabled and torch.cuda.is_available()
+        self._events: List = []
+        self.totals: Dict[str, float] = {}
+        self.steps = 0
+
+    def mark(self, name: str) -> None:
+        if not self.enabled:
+            return
+        ev = torch.cuda.Event(enable_timing=True)
+        ev.record()
+        self._events.append((name, ev))
+
+    def commit(self) -> None:
+        """Call after a step (outside the hot loop) to fold event deltas."""
+        if not self.enabled or len(self._events) < 2:
+            self._events = []
+            return
+        torch.cuda.synchronize()
+        for (n0, e0), (_n1, e1) in zip(self._events, self._events[1:]):
+            self.totals[n0] = self.totals.get(n0, 0.0) + e0.elapsed_time(e1)
+        self._events = []
+        self.steps += 1
+
+    def summary(self) -> Dict[str, float]:
+        if not self.steps:
+            return {}
+        return {k: v / self.steps for k, v in self.totals.items()}
+
+
 class FirewallPipeline:
-    def __init__(self, cfg: PipelineConfig, device: str = "cuda:0", world_size: int = 1, rank: int = 0):
+    def __init__(self, cfg: PipelineConfig, device: str = "cuda:0", world_size: int = 1, rank: int = 0, profile: bool = False):
         self.cfg = cfg
         self.device = torch.device(device)
         self.world_size = world_size
         self.rank = rank
         self.batch_seq = 0
+        self.profiler = StageProfiler(profile)
         torch.manual_seed(cfg.seed + rank)
 
         with torch.cuda.device(self.device):
@@ -135,15 +172,20 @@ class FirewallPipeline:
         agent_idx, tool_risk = s["agent_idx"], s["tool_risk"]
         B = offsets.numel() - 1
 
+        prof = self.profiler
+        prof.mark("dfa_scan")
         # 2. pattern scans
         hits = {fam: g.dfa_scan(bytes_t, offsets, fam) for fam in cfg.families}
 
+        prof.mark("encoder")
         # 3. encoder
         feats = g.encode_messages(bytes_t, offsets, self.embed, normalize=True)
 
+        prof.mark("classifier")
         # 4. classifier head (fused sigmoid)
         logits = g.gemm_nt(feats, self.head, bias=self.head_bias, act=1)
 
+        prof.mark("recall")
         # 5. Membrane recall (full index across ranks; parallel/collectives)
         def local_recall(queries):
             if self.index8 is not None:
@@ -173,6 +215,7 @@ class FirewallPipeline:
             0, flat_local.long(), torch.full((flat_local.numel(),), 0.01, device=self.device)
         )
 
+        prof.mark("verdict_trust")
         # 6. verdict + trust
         hour = time.localtime().tm_hour
         freq = torch.bincount(agent_idx.long(), minlength=cfg.n_agents).to(torch.int32)
@@ -184,6 +227,7 @@ class FirewallPipeline:
         )
         g.trust_recompute(self.trust_state, sdelta, vdelta)
 
+        prof.mark("audit_merkle")
         # 7. audit Merkle
         inj_score = logits.max(dim=1).values
         records = g.audit_pack(
@@ -200,6 +244,7 @@ class FirewallPipeline:
             roots = coll.allgather_roots(root, self.world_size)
             root = g.merkle_root(roots)
 
+        prof.mark("end")
         self.batch_seq += 1
         if self.audit_sink is not None:
             self.audit_sink(records, root)
