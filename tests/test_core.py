@@ -106,3 +106,21 @@ def test_gateway_lifecycle():
     assert gw.gateway_method("demo.status") == {"ok": True}
     gw.stop()
     assert not plugin.started
+
+
+def test_plugin_manifests():
+    """openclaw.plugin.json parity (reference ships one per package)."""
+    from vainplex_openclaw_amd.core.manifests import MANIFESTS, get_manifest, write_manifest
+    import json as _json
+    import tempfile
+
+    assert set(MANIFESTS) == {
+        "openclaw-governance", "openclaw-cortex", "openclaw-knowledge-engine",
+        "nats-eventstore", "openclaw-membrane", "openclaw-leuko",
+    }
+    m = get_manifest("openclaw-governance")
+    assert m["configSchema"]["properties"]["enabled"]["default"] is True
+    with tempfile.TemporaryDirectory() as d:
+        p = write_manifest("openclaw-cortex", d)
+        data = _json.loads(open(p).read())
+        assert data["id"] == "openclaw-cortex" and data["version"]
