@@ -89,6 +89,12 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
 #define XS(buf) (lds_all + NBUF * BM * BK + (buf) * BN * BK)
   __shared__ float row_min[BM];
   __shared__ int row_min_slot[BM];
+  // top-k VALUES live in LDS: the drain's insert+min-rescan is then a
+  // pure-LDS dependent chain instead of serial L2 round trips (measured
+  // ~2.4 us per insert through global memory). Ids go straight to
+  // global (write-only); scores are written out once per block at the
+  // end of the sweep.
+  __shared__ float topk_vals[BM][TOPK_MAX];
   __shared__ float q_score[QCAP];
   __shared__ uint32_t q_meta[QCAP];  // (row<<16) | col_in_tile
   __shared__ int q_count;
@@ -108,6 +114,8 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
     row_min[i] = -1e30f;
     row_min_slot[i] = 0;
   }
+  for (int i = threadIdx.x; i < BM * TOPK_MAX; i += blockDim.x)
+    topk_vals[i / TOPK_MAX][i % TOPK_MAX] = -1e30f;
   if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
   __syncthreads();
 
@@ -226,7 +234,7 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
         int my_row = threadIdx.x;
         float rmin = row_min[my_row];
         int rslot = row_min_slot[my_row];
-        float* cs = cand_scores + cbase + (size_t)my_row * k;
+        float* vals = topk_vals[my_row];
         int32_t* ci = cand_ids + cbase + (size_t)my_row * k;
         bool touched = false;
         for (int i = 0; i < total; ++i) {
@@ -234,12 +242,12 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
           if (int(meta >> 16) != my_row) continue;
           float v = q_score[i];
           if (v > rmin) {
-            cs[rslot] = v;
-            ci[rslot] = int32_t(x0 + (meta & 0xFFFFu));
-            float mn = cs[0];
+            vals[rslot] = v;
+            ci[rslot] = int32_t(x0 + (meta & 0xFFFFu));  // write-only
+            float mn = vals[0];
             int ms = 0;
             for (int j = 1; j < k; ++j)
-              if (cs[j] < mn) { mn = cs[j]; ms = j; }
+              if (vals[j] < mn) { mn = vals[j]; ms = j; }
             rmin = mn;
             rslot = ms;
             touched = true;
@@ -261,6 +269,9 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
     }
     __syncthreads();
   }
+  // scores out: one pass from LDS (ids were streamed during drains)
+  for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
+    cand_scores[cbase + (size_t)(i / k) * k + (i % k)] = topk_vals[i / k][i % k];
 }
 
 // Merge per-swath candidates into final [nq][k] (one wave per query).
@@ -327,6 +338,7 @@ extern "C" __global__ void topk_merge_kernel(
 
 typedef long long i64x2 __attribute__((ext_vector_type(2)));
 #define BK_F8 64
+#define NBUF_F8 3
 
 DEVINL void stage_tile8(const uint8_t* __restrict__ src, long long ld,
                         long long row0, long long row_max, int k0,
@@ -340,11 +352,12 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
                        int nq, int nx, int D, int k, int n_swaths,
                        float* __restrict__ cand_scores,
                        int32_t* __restrict__ cand_ids) {
-  __shared__ bf16 lds_all[NBUF * (BM + BN) * (BK_F8 / 2)];
+  __shared__ bf16 lds_all[NBUF_F8 * (BM + BN) * (BK_F8 / 2)];
 #define QS8(buf) (lds_all + (buf) * BM * (BK_F8 / 2))
-#define XS8(buf) (lds_all + NBUF * BM * (BK_F8 / 2) + (buf) * BN * (BK_F8 / 2))
+#define XS8(buf) (lds_all + NBUF_F8 * BM * (BK_F8 / 2) + (buf) * BN * (BK_F8 / 2))
   __shared__ float row_min[BM];
   __shared__ int row_min_slot[BM];
+  __shared__ float topk_vals[BM][TOPK_MAX];  // LDS candidate values
   __shared__ float q_score[QCAP];
   __shared__ uint32_t q_meta[QCAP];
   __shared__ int q_count;
@@ -364,6 +377,8 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
     row_min[i] = -1e30f;
     row_min_slot[i] = 0;
   }
+  for (int i = threadIdx.x; i < BM * TOPK_MAX; i += blockDim.x)
+    topk_vals[i / TOPK_MAX][i % TOPK_MAX] = -1e30f;
   if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
   __syncthreads();
 
@@ -378,16 +393,17 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
 
   for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
     f32x4 acc[8][4] = {};
-    stage_tile8(Q, D, row0, nq, 0, QS8(0), BM);
-    stage_tile8(X, D, x0, (long long)nx, 0, XS8(0), BN);
-    stage_tile8(Q, D, row0, nq, BK_F8, QS8(1), BM);
-    stage_tile8(X, D, x0, (long long)nx, BK_F8, XS8(1), BN);
+    for (int p = 0; p < 2 && p < nk; ++p) {
+      stage_tile8(Q, D, row0, nq, p * BK_F8, QS8(p), BM);
+      stage_tile8(X, D, x0, (long long)nx, p * BK_F8, XS8(p), BN);
+    }
     for (int kt = 0; kt < nk; ++kt) {
-      int cur = kt % NBUF;
+      int cur = kt % NBUF_F8;
+      // stage(kt) landed when only stage(kt+1)'s 4 glds are outstanding
       asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       __builtin_amdgcn_s_barrier();
       if (kt + 2 < nk) {
-        int pre = (kt + 2) % NBUF;
+        int pre = (kt + 2) % NBUF_F8;
         stage_tile8(Q, D, row0, nq, (kt + 2) * BK_F8, QS8(pre), BM);
         stage_tile8(X, D, x0, (long long)nx, (kt + 2) * BK_F8, XS8(pre), BN);
       }
@@ -473,7 +489,7 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
         int my_row = threadIdx.x;
         float rmin = row_min[my_row];
         int rslot = row_min_slot[my_row];
-        float* cs = cand_scores + cbase + (size_t)my_row * k;
+        float* vals = topk_vals[my_row];
         int32_t* ci = cand_ids + cbase + (size_t)my_row * k;
         bool touched = false;
         for (int i = 0; i < total; ++i) {
@@ -481,12 +497,12 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
           if (int(meta >> 16) != my_row) continue;
           float v = q_score[i];
           if (v > rmin) {
-            cs[rslot] = v;
-            ci[rslot] = int32_t(x0 + (meta & 0xFFFFu));
-            float mn = cs[0];
+            vals[rslot] = v;
+            ci[rslot] = int32_t(x0 + (meta & 0xFFFFu));  // write-only
+            float mn = vals[0];
             int ms = 0;
             for (int j = 1; j < k; ++j)
-              if (cs[j] < mn) { mn = cs[j]; ms = j; }
+              if (vals[j] < mn) { mn = vals[j]; ms = j; }
             rmin = mn;
             rslot = ms;
             touched = true;
@@ -508,4 +524,6 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
     }
     __syncthreads();
   }
+  for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
+    cand_scores[cbase + (size_t)(i / k) * k + (i % k)] = topk_vals[i / k][i % k];
 }
