@@ -196,7 +196,7 @@ std::vector<torch::Tensor> topk_recall_fp8(torch::Tensor Q8, torch::Tensor X8,
   int nq = Q8.size(0), D = Q8.size(1);
   long long nx = X8.size(0);
   TORCH_CHECK(X8.size(1) == D && D % 64 == 0, "D must be a multiple of 64");
-  TORCH_CHECK(k >= 1 && k <= 64, "k in [1,64]");
+  TORCH_CHECK(k >= 1 && k <= 32, "k in [1,32]");  // TOPK_MAX LDS bound
   int n_qblocks = (nq + 255) / 256;
   TORCH_CHECK(n_swaths >= 1 && (long long)n_swaths * k <= 1024,
               "n_swaths * k must be <= 1024");

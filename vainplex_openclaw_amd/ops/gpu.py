@@ -210,7 +210,7 @@ def topk_recall_two_stage(
     X: torch.Tensor,
     X8: torch.Tensor,
     k: int,
-    overfetch: int = 4,
+    overfetch: int = 2,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Two-stage exact-rescore recall: fp8 scan of the full index for
     k*overfetch candidates (half the staged bytes of the bf16 scan = the
@@ -219,7 +219,7 @@ def topk_recall_two_stage(
     only has to keep the true top-k inside the candidate set (error
     sigma ~0.003 vs candidate margins ~10x that).
     """
-    k2 = min(64, max(k * overfetch, k))
+    k2 = min(32, max(k * overfetch, k))  # TOPK_MAX LDS bound in the scan kernel
     Q8 = to_fp8_bytes(Q)
     _s8, ids8 = topk_recall_fp8(Q8, X8, k2)
     ids = ids8.long().clamp_min(0)  # -1 slots -> row 0 (rescored, never top)
