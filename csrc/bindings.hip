@@ -291,6 +291,30 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("bias") = c10::nullopt,
         py::arg("act") = 0, py::arg("out_bf16") = false);
   m.def("topk_recall", &topk_recall, "Fused cosine top-k recall");
+  m.def("topk_scan_only", [](torch::Tensor Q, torch::Tensor X, int64_t n_swaths, bool fp8) {
+    // perf diagnosis: run the scan loop with the top-k phase skipped
+    CHECK_GPU(Q); CHECK_GPU(X);
+    int nq = Q.size(0), D = Q.size(1);
+    long long nx = X.size(0);
+    int n_qblocks = (nq + 255) / 256;
+    auto f32opts = torch::dtype(torch::kFloat32).device(Q.device());
+    auto i32opts = torch::dtype(torch::kInt32).device(Q.device());
+    auto cand_s = torch::zeros({(long long)n_qblocks * n_swaths * 256}, f32opts);
+    auto cand_i = torch::zeros({1}, i32opts);
+    dim3 grid((unsigned)(n_qblocks * n_swaths));
+    if (fp8) {
+      hipLaunchKernelGGL(topk_recall_fp8_kernel, grid, dim3(512), 0, cur_stream(),
+                         Q.data_ptr<uint8_t>(), X.data_ptr<uint8_t>(), nq, (int)nx,
+                         D, -1, (int)n_swaths, cand_s.data_ptr<float>(),
+                         cand_i.data_ptr<int32_t>());
+    } else {
+      hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(512), 0, cur_stream(),
+                         reinterpret_cast<const __bf16*>(Q.data_ptr()),
+                         reinterpret_cast<const __bf16*>(X.data_ptr()), nq, (int)nx,
+                         D, -1, (int)n_swaths, cand_s.data_ptr<float>(),
+                         cand_i.data_ptr<int32_t>());
+    }
+  }, "scan-only diagnosis");
   m.def("topk_recall_fp8", &topk_recall_fp8, "fp8 stage-1 scan of two-stage recall");
   m.def("firewall_verdict", &firewall_verdict, "Fused verdict/risk/trust-delta");
   m.def("trust_recompute", &trust_recompute, "Agent trust score recompute");

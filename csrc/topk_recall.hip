@@ -194,6 +194,10 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
+    if (k < 0) {  // scan-only diagnosis mode: keep acc live, skip top-k
+      if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
     // ---- streaming top-k from the accumulators -------------------------
     // lane holds acc[m][n][r] at row = wm*128 + m*16 + (lane>>4)*4 + r,
     //                         col = wn*64  + n*16 + (lane&15)
@@ -455,6 +459,10 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
+    if (k < 0) {
+      if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
     // ---- streaming top-k (identical to the bf16 kernel) ----------------
     unsigned long long pend0 = ~0ull, pend1 = ~0ull;
     for (int round = 0; ; ++round) {

@@ -61,5 +61,30 @@ def main():
         print(f"{name:24s} {ms:9.2f} ms   {work_tf/ms*1000:7.1f} TF/s   index {gb:.1f} GB")
 
 
+
+
+def scan_only_bench():
+    import torch, time
+    from vainplex_openclaw_amd.ops import gpu as g
+    torch.manual_seed(0)
+    dev = "cuda:0"
+    rows, nq, dim = 4_194_304, 4096, 1024
+    Q = torch.nn.functional.normalize(torch.randn(nq, dim, device=dev), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(rows, dim, device=dev), dim=1).bfloat16()
+    X8 = g.to_fp8_bytes(X); Q8 = g.to_fp8_bytes(Q)
+    for name, fn in [
+        ("scan-only bf16 S=16", lambda: g.ext().topk_scan_only(Q, X, 16, False)),
+        ("scan-only fp8  S=16", lambda: g.ext().topk_scan_only(Q8, X8, 16, True)),
+        ("scan-only bf16 S=8",  lambda: g.ext().topk_scan_only(Q, X, 8, False)),
+        ("scan-only bf16 S=32", lambda: g.ext().topk_scan_only(Q, X, 32, False)),
+    ]:
+        ms = timed(fn, iters=3, warmup=1)
+        print(f"{name:24s} {ms:9.2f} ms   {2*nq*rows*dim/1e12/ms*1000:7.1f} TF/s")
+
+
 if __name__ == "__main__":
-    main()
+    import sys as _sys
+    if "--scan-only" in _sys.argv:
+        scan_only_bench()
+    else:
+        main()
