@@ -192,3 +192,52 @@ def test_output_validator_trust_proportional():
     assert ov.validate("all good here", trust_score=10)["verdict"] == "pass"
     assert more_restrictive("flag", "block") == "block"
     assert more_restrictive("pass", "flag") == "flag"
+
+
+# -- LLM validator (llm-validator.ts) ----------------------------------------
+
+def test_llm_validator_external_detection():
+    from vainplex_openclaw_amd.governance.llm_validator import is_external_comm
+
+    assert is_external_comm(channel="twitter")
+    assert is_external_comm(tool_name="send_email")
+    assert is_external_comm(command="bird tweet hello world")
+    assert not is_external_comm(channel="internal-slack", command="ls -la")
+
+
+def test_llm_validator_cache_ttl_and_verdicts():
+    from vainplex_openclaw_amd.governance.llm_validator import LlmValidator
+
+    calls = []
+
+    def fake_llm(prompt):
+        calls.append(prompt)
+        return '{"verdict": "flag", "reason": "unverifiable claim"}'
+
+    t = [1000.0]
+    v = LlmValidator(fake_llm, clock=lambda: t[0])
+    facts = [{"subject": "system", "predicate": "status", "value": "online"}]
+    r1 = v.validate("our system has 1M users", facts, True)
+    assert r1["verdict"] == "flag" and not r1["cached"]
+    r2 = v.validate("our system has 1M users", facts, True)
+    assert r2["cached"] and len(calls) == 1  # djb2 cache hit
+    t[0] += 301.0
+    r3 = v.validate("our system has 1M users", facts, True)
+    assert not r3["cached"] and len(calls) == 2  # TTL expired
+    # non-external never hits the LLM
+    assert v.validate("internal note", facts, False)["verdict"] == "pass"
+    assert len(calls) == 2
+
+
+def test_llm_validator_fail_open_and_bad_json():
+    from vainplex_openclaw_amd.governance.llm_validator import LlmValidator
+
+    v = LlmValidator(lambda p: "not json")
+    assert v.validate("x", [], True)["verdict"] == "pass"
+
+    def boom(p):
+        raise OSError("llm down")
+
+    v2 = LlmValidator(boom)
+    r = v2.validate("y", [], True)
+    assert r["verdict"] == "pass" and "llm-error" in r["reason"]
