@@ -99,6 +99,12 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   __shared__ uint32_t q_meta[QCAP];  // (row<<16) | col_in_tile
   __shared__ int q_count;
   __shared__ int q_overflow;
+  // conservative per-half-block threshold minima (rows 0-127 / 128-255):
+  // a wave whose 128 accumulator values all fall below its half's min
+  // skips the entire push scan for the tile (steady-state pushes are
+  // rare; the scan's fixed cost measured ~half of kernel time)
+  __shared__ float block_tmin[2];
+  __shared__ int tmin_dirty;
 
   int S = n_swaths;
   int qb = blockIdx.x / S;      // same-swath groups land on one XCD
@@ -116,7 +122,10 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   }
   for (int i = threadIdx.x; i < BM * TOPK_MAX; i += blockDim.x)
     topk_vals[i / TOPK_MAX][i % TOPK_MAX] = -1e30f;
-  if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
+  if (threadIdx.x == 0) {
+    q_count = 0; q_overflow = 0; tmin_dirty = 0;
+    block_tmin[0] = -1e30f; block_tmin[1] = -1e30f;
+  }
   __syncthreads();
 
   int wid = wave_id();              // 0..7
@@ -203,9 +212,24 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
     //                         col = wn*64  + n*16 + (lane&15)
     // Rounds: push survivors into the queue; drain by row-owner threads;
     // repeat if the queue overflowed (only plausible on the first tiles).
+    // wave early-out: max over this lane's 128 values (registers only),
+    // reduced across the wave, vs the conservative min of the wave's
+    // row thresholds — steady-state tiles push nothing and skip the
+    // whole scan
+    float vmax = -1e30f;
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) vmax = fmaxf(vmax, acc[m][n][r]);
+    for (int off = 32; off; off >>= 1) vmax = fmaxf(vmax, __shfl_down(vmax, off));
+    vmax = __shfl(vmax, 0);
+    bool wave_skip = !(vmax > block_tmin[wm]);
     unsigned long long pend0 = ~0ull, pend1 = ~0ull;  // 128 pending bits
     for (int round = 0; ; ++round) {
       unsigned long long still0 = 0ull, still1 = 0ull;
+      if (!wave_skip)
 #pragma unroll
       for (int m = 0; m < 8; ++m)
 #pragma unroll
@@ -231,6 +255,7 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
             }
           }
       __syncthreads();
+      if (q_count == 0) { __syncthreads(); break; }  // uniform: no pushes
       // drain: thread t < BM owns row t; candidate state is in GLOBAL
       // memory (block-private slice, L2-hot)
       int total = min(q_count, QCAP);
@@ -260,6 +285,7 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
         if (touched) {
           row_min[my_row] = rmin;
           row_min_slot[my_row] = rslot;
+          tmin_dirty = 1;  // any-writer, same value
         }
       }
       __syncthreads();
@@ -272,6 +298,16 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
       __syncthreads();  // reset visible before next round's pushes
     }
     __syncthreads();
+    if (tmin_dirty) {
+      if (threadIdx.x < 2) {
+        float mn = row_min[threadIdx.x * 128];
+        for (int j = 1; j < 128; ++j)
+          mn = fminf(mn, row_min[threadIdx.x * 128 + j]);
+        block_tmin[threadIdx.x] = mn;
+      }
+      __syncthreads();  // tmin visible before the next tile's skip test
+      if (threadIdx.x == 0) tmin_dirty = 0;
+    }
   }
   // scores out: one pass from LDS (ids were streamed during drains)
   for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
@@ -366,6 +402,8 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
   __shared__ uint32_t q_meta[QCAP];
   __shared__ int q_count;
   __shared__ int q_overflow;
+  __shared__ float block_tmin[2];  // see bf16 kernel comment
+  __shared__ int tmin_dirty;
 
   int S = n_swaths;
   int qb = blockIdx.x / S;
@@ -383,7 +421,10 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
   }
   for (int i = threadIdx.x; i < BM * TOPK_MAX; i += blockDim.x)
     topk_vals[i / TOPK_MAX][i % TOPK_MAX] = -1e30f;
-  if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
+  if (threadIdx.x == 0) {
+    q_count = 0; q_overflow = 0; tmin_dirty = 0;
+    block_tmin[0] = -1e30f; block_tmin[1] = -1e30f;
+  }
   __syncthreads();
 
   int wid = wave_id();
@@ -464,9 +505,24 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
       continue;
     }
     // ---- streaming top-k (identical to the bf16 kernel) ----------------
+    // wave early-out: max over this lane's 128 values (registers only),
+    // reduced across the wave, vs the conservative min of the wave's
+    // row thresholds — steady-state tiles push nothing and skip the
+    // whole scan
+    float vmax = -1e30f;
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) vmax = fmaxf(vmax, acc[m][n][r]);
+    for (int off = 32; off; off >>= 1) vmax = fmaxf(vmax, __shfl_down(vmax, off));
+    vmax = __shfl(vmax, 0);
+    bool wave_skip = !(vmax > block_tmin[wm]);
     unsigned long long pend0 = ~0ull, pend1 = ~0ull;
     for (int round = 0; ; ++round) {
       unsigned long long still0 = 0ull, still1 = 0ull;
+      if (!wave_skip)
 #pragma unroll
       for (int m = 0; m < 8; ++m)
 #pragma unroll
@@ -492,6 +548,7 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
             }
           }
       __syncthreads();
+      if (q_count == 0) { __syncthreads(); break; }  // uniform: no pushes
       int total = min(q_count, QCAP);
       if (threadIdx.x < BM) {
         int my_row = threadIdx.x;
@@ -519,6 +576,7 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
         if (touched) {
           row_min[my_row] = rmin;
           row_min_slot[my_row] = rslot;
+          tmin_dirty = 1;  // any-writer, same value
         }
       }
       __syncthreads();
@@ -531,6 +589,16 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
       __syncthreads();
     }
     __syncthreads();
+    if (tmin_dirty) {
+      if (threadIdx.x < 2) {
+        float mn = row_min[threadIdx.x * 128];
+        for (int j = 1; j < 128; ++j)
+          mn = fminf(mn, row_min[threadIdx.x * 128 + j]);
+        block_tmin[threadIdx.x] = mn;
+      }
+      __syncthreads();  // tmin visible before the next tile's skip test
+      if (threadIdx.x == 0) tmin_dirty = 0;
+    }
   }
   for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
     cand_scores[cbase + (size_t)(i / k) * k + (i % k)] = topk_vals[i / k][i % k];
