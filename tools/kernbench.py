@@ -49,7 +49,6 @@ def main():
         configs += [
             ("bf16 k=32", lambda: g.topk_recall(Q, X, 32)),
             ("fp8  k=32", lambda: g.topk_recall_fp8(Q8, X8, 32)),
-            ("fp8  k=64", lambda: g.topk_recall_fp8(Q8, X8, 64)),
             ("two-stage k=16 (of=4)", lambda: g.topk_recall_two_stage(Q, X, X8, 16)),
             ("two-stage k=16 (of=2)", lambda: g.topk_recall_two_stage(Q, X, X8, 16, overfetch=2)),
         ]
