@@ -169,9 +169,8 @@ std::vector<torch::Tensor> topk_recall(torch::Tensor Q, torch::Tensor X, int64_t
   // flat grid: block f -> (qb = f / S, swath = f % S); S a multiple of 8
   // keeps each co-sweeping same-swath group on one XCD
   dim3 grid((unsigned)(n_qblocks * n_swaths));
-  // 256 threads = the v5 kernel's 4-wave 2x2 grid (TK256 in topk_recall.hip):
-  // one wave per SIMD unlocks the 512-register (VGPR+AGPR) budget
-  hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(256), 0, cur_stream(),
+  // 512 threads = the kernel's 8-wave 2x4 grid (TK_THREADS in topk_recall.hip)
+  hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(512), 0, cur_stream(),
                      reinterpret_cast<const __bf16*>(Q.data_ptr()),
                      reinterpret_cast<const __bf16*>(X.data_ptr()), nq, (int)nx, D,
                      (int)k, (int)n_swaths, cand_s.data_ptr<float>(),

@@ -53,13 +53,13 @@ DEVINL uint32_t lds_off_bytes(uint32_t row, uint32_t slot) {
 // linear in piece index (the hardware adds lane*16 to the wave-uniform
 // base); the swizzle is applied on the SOURCE slot (involution with the
 // read side).
-DEVINL void stage_tile_s(const bf16* __restrict__ src, long long ld,
-                         long long row0, long long row_max, int k0,
-                         bf16* lds_base, int tile_rows, int stride) {
+DEVINL void stage_tile(const bf16* __restrict__ src, long long ld,
+                       long long row0, long long row_max, int k0,
+                       bf16* lds_base, int tile_rows) {
   int n_pieces = tile_rows * 4;  // 16 B pieces
   int w = wave_id();
   int lane = lane_id();
-  for (int piece0 = w * WAVE; piece0 < n_pieces; piece0 += stride) {
+  for (int piece0 = w * WAVE; piece0 < n_pieces; piece0 += TK_THREADS) {
     int piece = piece0 + lane;
     uint32_t r = piece >> 2;
     uint32_t slot = piece & 3;
@@ -74,46 +74,35 @@ DEVINL void stage_tile_s(const bf16* __restrict__ src, long long ld,
   }
 }
 
-DEVINL void stage_tile(const bf16* __restrict__ src, long long ld,
-                       long long row0, long long row_max, int k0,
-                       bf16* lds_base, int tile_rows) {
-  stage_tile_s(src, ld, row0, row_max, k0, lds_base, tile_rows, TK_THREADS);
-}
-
-#define TK256 256
-
-extern "C" __global__ void __launch_bounds__(TK256)
-__attribute__((amdgpu_waves_per_eu(1, 1)))
+extern "C" __global__ void __launch_bounds__(TK_THREADS)
 topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
                    int nq, int nx, int D, int k, int n_swaths,
                    float* __restrict__ cand_scores,
                    int32_t* __restrict__ cand_ids) {
-  // v5: 256 threads = 4 waves = ONE wave per SIMD -> a 512-VGPR budget
-  // per wave. The 512-thread variant capped at 256 VGPRs and spilled ~88
-  // registers that the push phase reloaded as a serial scratch chain
-  // every x-tile (~half of kernel time at k=16). Wave grid 2x2, wave
-  // tile 128x128 = 8x8 fragments = 256 accumulator VGPRs, spill-free.
-  //
   // Triple-buffered Q+X K-tiles: staging runs TWO K-steps ahead of
-  // compute, synchronized with counted s_waitcnt vmcnt(8) + raw
-  // s_barrier (a __syncthreads() would make the compiler drain the whole
-  // global_load_lds queue; a plain ds_read would make the legalizer
-  // insert vmcnt(0) — the fragment reads live in one inline-asm block).
+  // compute, synchronized with counted s_waitcnt vmcnt(4) + raw
+  // s_barrier. (__syncthreads() makes the compiler drain the whole
+  // global_load_lds queue at every barrier — the ~20% stall the guide
+  // documents — and a 1-deep double buffer forces vmcnt(0) anyway.)
   __shared__ bf16 lds_all[NBUF * (BM + BN) * BK];
 #define QS(buf) (lds_all + (buf) * BM * BK)
 #define XS(buf) (lds_all + NBUF * BM * BK + (buf) * BN * BK)
   __shared__ float row_min[BM];
   __shared__ int row_min_slot[BM];
-  // top-k VALUES live in LDS: the drain's insert+min-rescan is a pure-
-  // LDS chain instead of serial L2 round trips (~2.4us/insert measured).
-  // Ids go straight to global (write-only); scores written out at the
+  // top-k VALUES live in LDS: the drain's insert+min-rescan is then a
+  // pure-LDS dependent chain instead of serial L2 round trips (measured
+  // ~2.4 us per insert through global memory). Ids go straight to
+  // global (write-only); scores are written out once per block at the
   // end of the sweep.
   __shared__ float topk_vals[BM][TOPK_MAX];
   __shared__ float q_score[QCAP];
   __shared__ uint32_t q_meta[QCAP];  // (row<<16) | col_in_tile
   __shared__ int q_count;
   __shared__ int q_overflow;
-  // conservative per-half-block threshold minima (rows 0-127 / 128-255)
+  // conservative per-half-block threshold minima (rows 0-127 / 128-255):
+  // a wave whose 128 accumulator values all fall below its half's min
+  // skips the entire push scan for the tile (steady-state pushes are
+  // rare; the scan's fixed cost measured ~half of kernel time)
   __shared__ float block_tmin[2];
   __shared__ int tmin_dirty;
 
@@ -139,8 +128,8 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   }
   __syncthreads();
 
-  int wid = wave_id();              // 0..3
-  int wm = wid >> 1, wn = wid & 1;  // 2 x 2 wave grid, tile 128x128
+  int wid = wave_id();              // 0..7
+  int wm = wid >> 2, wn = wid & 3;  // 2 x 4 wave grid
   int lane = lane_id();
   int lrow = lane & 15;
   int kgrp = lane >> 4;             // 0..3 -> 16 B k-group within 32 elems
@@ -149,96 +138,110 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
 
   for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
-    f32x4 acc[8][8] = {};
-    // prologue: 2 K-tiles in flight (each stage pair = 8 glds per lane)
-    stage_tile_s(Q, D, row0, nq, 0, QS(0), BM, TK256);
-    stage_tile_s(X, D, x0, (long long)nx, 0, XS(0), BN, TK256);
-    stage_tile_s(Q, D, row0, nq, BK, QS(1), BM, TK256);
-    stage_tile_s(X, D, x0, (long long)nx, BK, XS(1), BN, TK256);
+    f32x4 acc[8][4] = {};
+    // prologue: 2 K-tiles in flight (each stage = 4 glds per lane:
+    // 2 for the Q tile + 2 for the X tile)
+    stage_tile(Q, D, row0, nq, 0, QS(0), BM);
+    stage_tile(X, D, x0, (long long)nx, 0, XS(0), BN);
+    stage_tile(Q, D, row0, nq, BK, QS(1), BM);
+    stage_tile(X, D, x0, (long long)nx, BK, XS(1), BN);
     for (int kt = 0; kt < nk; ++kt) {
       int cur = kt % NBUF;
-      // stage(kt) landed when only stage(kt+1)'s 8 glds are outstanding
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      // stage(kt) landed when only stage(kt+1)'s 4 glds are outstanding
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      // raw barrier: (a) stage(kt) visible block-wide, (b) every wave is
+      // done computing kt-1, so its buffer ((kt+2) % NBUF) is free.
+      // Safe without the __syncthreads() fence: all LDS reads of kt-1
+      // were consumed by mfma issue before this point.
       __builtin_amdgcn_s_barrier();
       if (kt + 2 < nk) {
         int pre = (kt + 2) % NBUF;
-        stage_tile_s(Q, D, row0, nq, (kt + 2) * BK, QS(pre), BM, TK256);
-        stage_tile_s(X, D, x0, (long long)nx, (kt + 2) * BK, XS(pre), BN, TK256);
+        stage_tile(Q, D, row0, nq, (kt + 2) * BK, QS(pre), BM);
+        stage_tile(X, D, x0, (long long)nx, (kt + 2) * BK, XS(pre), BN);
       }
-      // fragment reads: one asm block (see v4 note); swizzle term is
-      // fragment-invariant, so fragments sit at base + i*1024
+      // Fragment loads as ONE inline-asm block: a plain ds_read after
+      // global_load_lds makes the memory legalizer insert s_waitcnt
+      // vmcnt(0) (it cannot prove the read targets an already-landed
+      // buffer), which drains the 2-deep prefetch every K-step — the
+      // exact stall the guide's "inline-asm K-loop" path removes. The
+      // swizzle term (lrow>>2)&3 is invariant across fragments (row
+      // deltas are multiples of 16), so fragments sit at base + n*1024.
       uint32_t base_x = (uint32_t)(size_t)XS(cur)
-                        + lds_off_bytes(wn * 128 + lrow, kgrp);
+                        + lds_off_bytes(wn * 64 + lrow, kgrp);
       uint32_t base_q = (uint32_t)(size_t)QS(cur)
                         + lds_off_bytes(wm * 128 + lrow, kgrp);
-      bf16x8 xf[8], qf[8];
+      bf16x8 xf[4], qf[8];
       asm volatile(
-          "ds_read_b128 %0, %16\n\t"
-          "ds_read_b128 %1, %16 offset:1024\n\t"
-          "ds_read_b128 %2, %16 offset:2048\n\t"
-          "ds_read_b128 %3, %16 offset:3072\n\t"
-          "ds_read_b128 %4, %16 offset:4096\n\t"
-          "ds_read_b128 %5, %16 offset:5120\n\t"
-          "ds_read_b128 %6, %16 offset:6144\n\t"
-          "ds_read_b128 %7, %16 offset:7168\n\t"
-          "ds_read_b128 %8, %17\n\t"
-          "ds_read_b128 %9, %17 offset:1024\n\t"
-          "ds_read_b128 %10, %17 offset:2048\n\t"
-          "ds_read_b128 %11, %17 offset:3072\n\t"
-          "ds_read_b128 %12, %17 offset:4096\n\t"
-          "ds_read_b128 %13, %17 offset:5120\n\t"
-          "ds_read_b128 %14, %17 offset:6144\n\t"
-          "ds_read_b128 %15, %17 offset:7168\n\t"
+          "ds_read_b128 %0, %12\n\t"
+          "ds_read_b128 %1, %12 offset:1024\n\t"
+          "ds_read_b128 %2, %12 offset:2048\n\t"
+          "ds_read_b128 %3, %12 offset:3072\n\t"
+          "ds_read_b128 %4, %13\n\t"
+          "ds_read_b128 %5, %13 offset:1024\n\t"
+          "ds_read_b128 %6, %13 offset:2048\n\t"
+          "ds_read_b128 %7, %13 offset:3072\n\t"
+          "ds_read_b128 %8, %13 offset:4096\n\t"
+          "ds_read_b128 %9, %13 offset:5120\n\t"
+          "ds_read_b128 %10, %13 offset:6144\n\t"
+          "ds_read_b128 %11, %13 offset:7168\n\t"
           "s_waitcnt lgkmcnt(0)"
+          // early-clobber: ds_read destinations land asynchronously and
+          // must never alias the (still-live) address inputs
           : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
-            "=&v"(xf[4]), "=&v"(xf[5]), "=&v"(xf[6]), "=&v"(xf[7]),
             "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
             "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7])
           : "v"(base_x), "v"(base_q));
 #pragma unroll
       for (int m = 0; m < 8; ++m)
 #pragma unroll
-        for (int n = 0; n < 8; ++n)
+        for (int n = 0; n < 4; ++n)
           acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               qf[m], xf[n], acc[m][n], 0, 0, 0);
     }
+    // drain remaining staging and make the last compute visible before
+    // the push phase reuses LDS-adjacent state
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
-    if (k < 0) {  // scan-only diagnosis mode
+    if (k < 0) {  // scan-only diagnosis mode: keep acc live, skip top-k
       if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
       continue;
     }
     // ---- streaming top-k from the accumulators -------------------------
     // lane holds acc[m][n][r] at row = wm*128 + m*16 + (lane>>4)*4 + r,
-    //                         col = wn*128 + n*16 + (lane&15)
+    //                         col = wn*64  + n*16 + (lane&15)
+    // Rounds: push survivors into the queue; drain by row-owner threads;
+    // repeat if the queue overflowed (only plausible on the first tiles).
+    // wave early-out: max over this lane's 128 values (registers only),
+    // reduced across the wave, vs the conservative min of the wave's
+    // row thresholds — steady-state tiles push nothing and skip the
+    // whole scan
     float vmax = -1e30f;
 #pragma unroll
     for (int m = 0; m < 8; ++m)
 #pragma unroll
-      for (int n = 0; n < 8; ++n)
+      for (int n = 0; n < 4; ++n)
 #pragma unroll
         for (int r = 0; r < 4; ++r) vmax = fmaxf(vmax, acc[m][n][r]);
     for (int off = 32; off; off >>= 1) vmax = fmaxf(vmax, __shfl_down(vmax, off));
     vmax = __shfl(vmax, 0);
     bool wave_skip = !(vmax > block_tmin[wm]);
-    // 256 pending bits: vi = m*32 + n*4 + r
-    unsigned long long pend[4] = {~0ull, ~0ull, ~0ull, ~0ull};
+    unsigned long long pend0 = ~0ull, pend1 = ~0ull;  // 128 pending bits
     for (int round = 0; ; ++round) {
-      unsigned long long still[4] = {0ull, 0ull, 0ull, 0ull};
+      unsigned long long still0 = 0ull, still1 = 0ull;
       if (!wave_skip)
 #pragma unroll
       for (int m = 0; m < 8; ++m)
 #pragma unroll
-        for (int n = 0; n < 8; ++n)
+        for (int n = 0; n < 4; ++n)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            int vi = m * 32 + n * 4 + r;
-            unsigned long long bit = 1ull << (vi & 63);
-            if (!(pend[vi >> 6] & bit)) continue;
+            int vi = (m & 3) * 16 + n * 4 + r;
+            unsigned long long bit = 1ull << vi;
+            if (!(((m < 4) ? pend0 : pend1) & bit)) continue;
             int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
             if ((row0 + row) >= nq) continue;
-            long long col = x0 + wn * 128 + n * 16 + (lane & 15);
+            long long col = x0 + wn * 64 + n * 16 + (lane & 15);
             if (col >= x_end) continue;
             float v = acc[m][n][r];
             if (!(v > row_min[row])) continue;
@@ -247,15 +250,16 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
               q_score[idx] = v;
               q_meta[idx] = (uint32_t(row) << 16) | uint32_t(col - x0);
             } else {
-              still[vi >> 6] |= bit;
+              if (m < 4) still0 |= bit; else still1 |= bit;
               atomicExch(&q_overflow, 1);
             }
           }
       __syncthreads();
       if (q_count == 0) { __syncthreads(); break; }  // uniform: no pushes
-      // drain: thread t owns row t (blockDim == BM)
+      // drain: thread t < BM owns row t; candidate state is in GLOBAL
+      // memory (block-private slice, L2-hot)
       int total = min(q_count, QCAP);
-      {
+      if (threadIdx.x < BM) {
         int my_row = threadIdx.x;
         float rmin = row_min[my_row];
         int rslot = row_min_slot[my_row];
@@ -288,8 +292,8 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
       int of = q_overflow;
       __syncthreads();  // all reads of q_overflow done before the reset
       if (threadIdx.x == 0) { q_count = 0; q_overflow = 0; }
-      pend[0] = still[0]; pend[1] = still[1];
-      pend[2] = still[2]; pend[3] = still[3];
+      pend0 = still0;
+      pend1 = still1;
       if (!of) break;   // of is uniform (LDS): no divergence
       __syncthreads();  // reset visible before next round's pushes
     }
