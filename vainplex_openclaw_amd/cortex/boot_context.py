@@ -13,7 +13,7 @@ import datetime as _dt
 import os
 import time
 from dataclasses import dataclass
-from typing import Any, Dict, List, Optional
+from typing import List, Optional
 
 from .storage import is_file_older_than, load_json, load_text, reboot_dir, save_text
 

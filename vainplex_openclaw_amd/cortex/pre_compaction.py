@@ -12,7 +12,7 @@ import datetime as _dt
 import os
 import time
 from collections import deque
-from typing import Any, Deque, Dict, List, Optional, Tuple
+from typing import Any, Deque, Dict, List, Tuple
 
 from .boot_context import BootContextGenerator
 from .storage import reboot_dir, save_text

@@ -13,7 +13,7 @@ from __future__ import annotations
 import hashlib
 import time
 import uuid
-from typing import Any, Callable, Dict, List, Optional
+from typing import Any, Dict, List, Optional
 
 from .events import SCHEMA_VERSION
 from .mappings import EXTRA_EMITTERS, HOOK_MAPPINGS, HookMapping

@@ -8,7 +8,7 @@ events; llm_input/llm_output payloads are redacted to lengths/counts only.
 
 from __future__ import annotations
 
-from typing import Any, Callable, Dict, List, Optional
+from typing import Callable, Dict, List, Optional
 
 
 class HookMapping:

@@ -24,7 +24,7 @@ import os
 import threading
 import time
 import uuid
-from typing import Any, Callable, Dict, List, Optional, Sequence
+from typing import Any, Dict, List, Optional, Sequence
 
 from ..core.api import PluginLogger, NullLogger
 from .audit_redactor import create_redactor

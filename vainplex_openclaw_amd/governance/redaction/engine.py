@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import json
 import time
-from typing import Any, Dict, Optional, Set
+from typing import Any, Dict, Set
 
 from .registry import PatternRegistry
 from .vault import RedactionVault

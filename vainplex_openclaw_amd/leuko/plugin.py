@@ -13,7 +13,7 @@ import threading
 import time
 from typing import Any, Dict, Optional
 
-from ..core.api import NullLogger, PluginApi
+from ..core.api import PluginApi
 from ..core.config import load_plugin_config
 from .aggregator import generate_sitrep, write_sitrep
 from .anomaly import AnomalyDetector, MetricHistory

@@ -15,7 +15,7 @@ from __future__ import annotations
 import time
 from typing import Any, Dict, Optional
 
-from ..core.api import NullLogger, PluginApi, PluginLogger
+from ..core.api import PluginApi, PluginLogger
 from ..core.config import load_plugin_config
 from .engine import DEFAULT_CONFIG, MembraneEngine
 

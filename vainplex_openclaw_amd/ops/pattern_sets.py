@@ -16,7 +16,7 @@ family gets its own DFA + u64 hit mask).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Set, Tuple
+from typing import Dict, List, Set, Tuple
 
 import hashlib
 import os
