@@ -170,29 +170,40 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
                         + lds_off_bytes(wn * 64 + lrow, kgrp);
       uint32_t base_q = (uint32_t)(size_t)QS(cur)
                         + lds_off_bytes(wm * 128 + lrow, kgrp);
+      // two 6-read asm blocks (xf[4]+qf[0..1], then qf[2..7]) halve the
+      // peak early-clobber register footprint vs one 12-read block; the
+      // first mfma quads overlap the second block's LDS latency
       bf16x8 xf[4], qf[8];
       asm volatile(
-          "ds_read_b128 %0, %12\n\t"
-          "ds_read_b128 %1, %12 offset:1024\n\t"
-          "ds_read_b128 %2, %12 offset:2048\n\t"
-          "ds_read_b128 %3, %12 offset:3072\n\t"
-          "ds_read_b128 %4, %13\n\t"
-          "ds_read_b128 %5, %13 offset:1024\n\t"
-          "ds_read_b128 %6, %13 offset:2048\n\t"
-          "ds_read_b128 %7, %13 offset:3072\n\t"
-          "ds_read_b128 %8, %13 offset:4096\n\t"
-          "ds_read_b128 %9, %13 offset:5120\n\t"
-          "ds_read_b128 %10, %13 offset:6144\n\t"
-          "ds_read_b128 %11, %13 offset:7168\n\t"
+          "ds_read_b128 %0, %6\n\t"
+          "ds_read_b128 %1, %6 offset:1024\n\t"
+          "ds_read_b128 %2, %6 offset:2048\n\t"
+          "ds_read_b128 %3, %6 offset:3072\n\t"
+          "ds_read_b128 %4, %7\n\t"
+          "ds_read_b128 %5, %7 offset:1024\n\t"
           "s_waitcnt lgkmcnt(0)"
-          // early-clobber: ds_read destinations land asynchronously and
-          // must never alias the (still-live) address inputs
           : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
-            "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
-            "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7])
+            "=&v"(qf[0]), "=&v"(qf[1])
           : "v"(base_x), "v"(base_q));
+      asm volatile(
+          "ds_read_b128 %0, %6 offset:2048\n\t"
+          "ds_read_b128 %1, %6 offset:3072\n\t"
+          "ds_read_b128 %2, %6 offset:4096\n\t"
+          "ds_read_b128 %3, %6 offset:5120\n\t"
+          "ds_read_b128 %4, %6 offset:6144\n\t"
+          "ds_read_b128 %5, %6 offset:7168\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(qf[2]), "=&v"(qf[3]), "=&v"(qf[4]), "=&v"(qf[5]),
+            "=&v"(qf[6]), "=&v"(qf[7])
+          : "v"(base_q));
 #pragma unroll
-      for (int m = 0; m < 8; ++m)
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[m], xf[n], acc[m][n], 0, 0, 0);
+#pragma unroll
+      for (int m = 2; m < 8; ++m)
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
