@@ -26,9 +26,10 @@ __global__ void encode_messages_kernel(const uint8_t*, const int32_t*, const __b
 __global__ void gemm_nt_bf16_kernel(const __bf16*, const __bf16*, float*, __bf16*,
                                     const float*, int, int, int, int, int);
 __global__ void topk_recall_kernel(const __bf16*, const __bf16*, int, int, int, int,
-                                   int, float*, int32_t*);
+                                   int, float*, int32_t*, const float*, int32_t*, int);
 __global__ void topk_recall_fp8_kernel(const uint8_t*, const uint8_t*, int, int, int,
-                                       int, int, float*, int32_t*);
+                                       int, int, float*, int32_t*, const float*,
+                                       int32_t*, int);
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
                                   float*, int32_t*);
 __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigned long long*,
@@ -174,7 +175,7 @@ std::vector<torch::Tensor> topk_recall(torch::Tensor Q, torch::Tensor X, int64_t
                      reinterpret_cast<const __bf16*>(Q.data_ptr()),
                      reinterpret_cast<const __bf16*>(X.data_ptr()), nq, (int)nx, D,
                      (int)k, (int)n_swaths, cand_s.data_ptr<float>(),
-                     cand_i.data_ptr<int32_t>());
+                     cand_i.data_ptr<int32_t>(), nullptr, nullptr, 0);
   auto out_s = torch::empty({nq, k}, f32opts);
   auto out_i = torch::empty({nq, k}, i32opts);
   int waves_per_block = 4;
@@ -210,7 +211,7 @@ std::vector<torch::Tensor> topk_recall_fp8(torch::Tensor Q8, torch::Tensor X8,
   hipLaunchKernelGGL(topk_recall_fp8_kernel, grid, dim3(512), 0, cur_stream(),
                      Q8.data_ptr<uint8_t>(), X8.data_ptr<uint8_t>(), nq, (int)nx,
                      D, (int)k, (int)n_swaths, cand_s.data_ptr<float>(),
-                     cand_i.data_ptr<int32_t>());
+                     cand_i.data_ptr<int32_t>(), nullptr, nullptr, 0);
   auto out_s = torch::empty({nq, k}, f32opts);
   auto out_i = torch::empty({nq, k}, i32opts);
   int waves_per_block = 4;
@@ -282,6 +283,52 @@ torch::Tensor audit_pack(torch::Tensor verdict, torch::Tensor risk, torch::Tenso
   return out;
 }
 
+
+std::vector<torch::Tensor> topk_scan_threshold(torch::Tensor Q, torch::Tensor X,
+                                               torch::Tensor theta, int64_t cap,
+                                               int64_t n_swaths, bool fp8) {
+  // threshold-scan mode: append every score > theta[q] to a per-query
+  // candidate buffer (no top-k maintenance in-kernel)
+  CHECK_GPU(Q); CHECK_CONTIG(Q); CHECK_GPU(X); CHECK_CONTIG(X);
+  CHECK_GPU(theta); CHECK_CONTIG(theta);
+  TORCH_CHECK(theta.dtype() == torch::kFloat32);
+  int nq = Q.size(0), D = Q.size(1);
+  long long nx = X.size(0);
+  TORCH_CHECK(theta.numel() == nq);
+  TORCH_CHECK(X.size(1) == D && D % 64 == 0);
+  TORCH_CHECK(cap >= 32 && cap <= 4096);
+  int n_qblocks = (nq + 255) / 256;
+  if (n_swaths <= 0) {
+    int want = 256 / (n_qblocks > 0 ? n_qblocks : 1);
+    n_swaths = want >= 8 ? (want / 8) * 8 : 8;
+    long long max_s = nx / 256; if (max_s < 1) max_s = 1;
+    if (n_swaths > max_s) n_swaths = max_s;
+  }
+  auto f32opts = torch::dtype(torch::kFloat32).device(Q.device());
+  auto i32opts = torch::dtype(torch::kInt32).device(Q.device());
+  auto cand_s = torch::full({(long long)nq, cap}, -1e30, f32opts);
+  auto cand_i = torch::full({(long long)nq, cap}, -1, i32opts);
+  auto counts = torch::zeros({(long long)nq}, i32opts);
+  dim3 grid((unsigned)(n_qblocks * n_swaths));
+  if (fp8) {
+    TORCH_CHECK(Q.dtype() == torch::kUInt8 && X.dtype() == torch::kUInt8);
+    hipLaunchKernelGGL(topk_recall_fp8_kernel, grid, dim3(512), 0, cur_stream(),
+                       Q.data_ptr<uint8_t>(), X.data_ptr<uint8_t>(), nq, (int)nx,
+                       D, 1, (int)n_swaths, cand_s.data_ptr<float>(),
+                       cand_i.data_ptr<int32_t>(), theta.data_ptr<float>(),
+                       counts.data_ptr<int32_t>(), (int)cap);
+  } else {
+    TORCH_CHECK(Q.dtype() == torch::kBFloat16 && X.dtype() == torch::kBFloat16);
+    hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(512), 0, cur_stream(),
+                       reinterpret_cast<const __bf16*>(Q.data_ptr()),
+                       reinterpret_cast<const __bf16*>(X.data_ptr()), nq, (int)nx,
+                       D, 1, (int)n_swaths, cand_s.data_ptr<float>(),
+                       cand_i.data_ptr<int32_t>(), theta.data_ptr<float>(),
+                       counts.data_ptr<int32_t>(), (int)cap);
+  }
+  return {cand_s, cand_i, counts};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sha256_leaves", &sha256_leaves, "Batched SHA-256 leaf digests");
   m.def("merkle_root", &merkle_root_gpu, "Merkle root over leaf digests");
@@ -306,16 +353,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       hipLaunchKernelGGL(topk_recall_fp8_kernel, grid, dim3(512), 0, cur_stream(),
                          Q.data_ptr<uint8_t>(), X.data_ptr<uint8_t>(), nq, (int)nx,
                          D, -1, (int)n_swaths, cand_s.data_ptr<float>(),
-                         cand_i.data_ptr<int32_t>());
+                         cand_i.data_ptr<int32_t>(), nullptr, nullptr, 0);
     } else {
       hipLaunchKernelGGL(topk_recall_kernel, grid, dim3(512), 0, cur_stream(),
                          reinterpret_cast<const __bf16*>(Q.data_ptr()),
                          reinterpret_cast<const __bf16*>(X.data_ptr()), nq, (int)nx,
                          D, -1, (int)n_swaths, cand_s.data_ptr<float>(),
-                         cand_i.data_ptr<int32_t>());
+                         cand_i.data_ptr<int32_t>(), nullptr, nullptr, 0);
     }
   }, "scan-only diagnosis");
   m.def("topk_recall_fp8", &topk_recall_fp8, "fp8 stage-1 scan of two-stage recall");
+  m.def("topk_scan_threshold", &topk_scan_threshold,
+        "threshold-scan candidate collection (bf16 or fp8)");
   m.def("firewall_verdict", &firewall_verdict, "Fused verdict/risk/trust-delta");
   m.def("trust_recompute", &trust_recompute, "Agent trust score recompute");
   m.def("audit_pack", &audit_pack, "Pack 64-byte audit records");

@@ -78,7 +78,13 @@ extern "C" __global__ void __launch_bounds__(TK_THREADS)
 topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
                    int nq, int nx, int D, int k, int n_swaths,
                    float* __restrict__ cand_scores,
-                   int32_t* __restrict__ cand_ids) {
+                   int32_t* __restrict__ cand_ids,
+                   // threshold mode (theta != nullptr): append every score
+                   // > theta[row] to a per-query global candidate buffer —
+                   // no queue, no drain, no per-tile barriers. cand_scores/
+                   // cand_ids are then [nq][cap]; tc_n[nq] counts appends.
+                   const float* __restrict__ theta,
+                   int32_t* __restrict__ tc_n, int cap) {
   // Triple-buffered Q+X K-tiles: staging runs TWO K-steps ahead of
   // compute, synchronized with counted s_waitcnt vmcnt(4) + raw
   // s_barrier. (__syncthreads() makes the compiler drain the whole
@@ -117,7 +123,8 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
   long long x_end = min((long long)nx, x_begin + per);
 
   for (int i = threadIdx.x; i < BM; i += blockDim.x) {
-    row_min[i] = -1e30f;
+    // threshold mode reuses row_min as the per-row fixed threshold
+    row_min[i] = (theta != nullptr && row0 + i < nq) ? theta[row0 + i] : -1e30f;
     row_min_slot[i] = 0;
   }
   for (int i = threadIdx.x; i < BM * TOPK_MAX; i += blockDim.x)
@@ -216,6 +223,30 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
 
     if (k < 0) {  // scan-only diagnosis mode: keep acc live, skip top-k
       if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
+    if (theta != nullptr) {
+      // threshold mode: fixed per-row thresholds (LDS), rare global
+      // atomic appends, zero epilogue barriers
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+            float v = acc[m][n][r];
+            if (!(v > row_min[row])) continue;
+            long long grow = row0 + row;
+            if (grow >= nq) continue;
+            long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+            if (col >= x_end) continue;
+            int pos = atomicAdd(&tc_n[grow], 1);
+            if (pos < cap) {
+              cand_scores[grow * cap + pos] = v;
+              cand_ids[grow * cap + pos] = int32_t(col);
+            }
+          }
       continue;
     }
     // ---- streaming top-k from the accumulators -------------------------
@@ -320,9 +351,11 @@ topk_recall_kernel(const bf16* __restrict__ Q, const bf16* __restrict__ X,
       if (threadIdx.x == 0) tmin_dirty = 0;
     }
   }
-  // scores out: one pass from LDS (ids were streamed during drains)
-  for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
-    cand_scores[cbase + (size_t)(i / k) * k + (i % k)] = topk_vals[i / k][i % k];
+  // scores out: one pass from LDS (ids were streamed during drains);
+  // threshold mode wrote its buffers inline
+  if (theta == nullptr && k > 0)
+    for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
+      cand_scores[cbase + (size_t)(i / k) * k + (i % k)] = topk_vals[i / k][i % k];
 }
 
 // Merge per-swath candidates into final [nq][k] (one wave per query).
@@ -400,9 +433,15 @@ DEVINL void stage_tile8(const uint8_t* __restrict__ src, long long ld,
 
 extern "C" __global__ void __launch_bounds__(TK_THREADS)
 topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X,
-                       int nq, int nx, int D, int k, int n_swaths,
-                       float* __restrict__ cand_scores,
-                       int32_t* __restrict__ cand_ids) {
+                   int nq, int nx, int D, int k, int n_swaths,
+                   float* __restrict__ cand_scores,
+                   int32_t* __restrict__ cand_ids,
+                   // threshold mode (theta != nullptr): append every score
+                   // > theta[row] to a per-query global candidate buffer —
+                   // no queue, no drain, no per-tile barriers. cand_scores/
+                   // cand_ids are then [nq][cap]; tc_n[nq] counts appends.
+                   const float* __restrict__ theta,
+                   int32_t* __restrict__ tc_n, int cap) {
   __shared__ bf16 lds_all[NBUF_F8 * (BM + BN) * (BK_F8 / 2)];
 #define QS8(buf) (lds_all + (buf) * BM * (BK_F8 / 2))
 #define XS8(buf) (lds_all + NBUF_F8 * BM * (BK_F8 / 2) + (buf) * BN * (BK_F8 / 2))
@@ -427,7 +466,8 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
   long long x_end = min((long long)nx, x_begin + per);
 
   for (int i = threadIdx.x; i < BM; i += blockDim.x) {
-    row_min[i] = -1e30f;
+    // threshold mode reuses row_min as the per-row fixed threshold
+    row_min[i] = (theta != nullptr && row0 + i < nq) ? theta[row0 + i] : -1e30f;
     row_min_slot[i] = 0;
   }
   for (int i = threadIdx.x; i < BM * TOPK_MAX; i += blockDim.x)
@@ -513,6 +553,30 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
 
     if (k < 0) {
       if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
+    if (theta != nullptr) {
+      // threshold mode: fixed per-row thresholds (LDS), rare global
+      // atomic appends, zero epilogue barriers
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+            float v = acc[m][n][r];
+            if (!(v > row_min[row])) continue;
+            long long grow = row0 + row;
+            if (grow >= nq) continue;
+            long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+            if (col >= x_end) continue;
+            int pos = atomicAdd(&tc_n[grow], 1);
+            if (pos < cap) {
+              cand_scores[grow * cap + pos] = v;
+              cand_ids[grow * cap + pos] = int32_t(col);
+            }
+          }
       continue;
     }
     // ---- streaming top-k (identical to the bf16 kernel) ----------------
@@ -611,6 +675,7 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
       if (threadIdx.x == 0) tmin_dirty = 0;
     }
   }
-  for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
-    cand_scores[cbase + (size_t)(i / k) * k + (i % k)] = topk_vals[i / k][i % k];
+  if (theta == nullptr && k > 0)
+    for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
+      cand_scores[cbase + (size_t)(i / k) * k + (i % k)] = topk_vals[i / k][i % k];
 }

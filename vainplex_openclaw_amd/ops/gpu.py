@@ -268,3 +268,93 @@ def audit_pack(
         verdict, risk, inj_hits, red_hits, agent_idx, agent_trust, inj_score,
         ts_ms, msg_id0, batch_seq,
     )
+
+
+def topk_recall_threshold(
+    Q: torch.Tensor,
+    X: torch.Tensor,
+    k: int,
+    X8: Optional[torch.Tensor] = None,
+    sample_rows: int = 131072,
+    target_candidates: int = 128,
+    cap: int = 1024,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Threshold-scan recall: per-query score thresholds estimated from a
+    sampled pre-pass (Gaussian tail extrapolation), then a fixed-threshold
+    scan whose epilogue is just rare global appends — no in-kernel top-k
+    maintenance, so the scan runs at its K-loop rate regardless of k.
+
+    With X8 (e4m3 view) the scan runs in fp8 and the top candidates are
+    exact-rescored in bf16; final scores are exact either way. Queries
+    whose candidate buffer underflowed (<k survivors: threshold too high
+    for non-Gaussian score tails) fall back to the direct kernel.
+    """
+    nq, D = Q.shape
+    nx = X.shape[0]
+    m = min(nx, sample_rows)
+    use_fp8 = X8 is not None
+
+    # 1. per-query score statistics from a sample (bf16 matmul)
+    sample = torch.matmul(Q, X[:m].T).float()  # [nq, m]
+    mu = sample.mean(dim=1)
+    sigma = sample.std(dim=1).clamp_min(1e-6)
+    # Gaussian tail: P(score > theta) = C/nx
+    p = min(0.5, max(target_candidates / max(nx, 1), 1e-12))
+    try:
+        from scipy.stats import norm
+
+        z = float(norm.ppf(1.0 - p))
+    except ImportError:  # pragma: no cover
+        import math
+
+        # Beasley-Springer-Moro style rough inverse via bisection
+        lo, hi = 0.0, 9.0
+        for _ in range(60):
+            mid = (lo + hi) / 2
+            if 0.5 * math.erfc(mid / math.sqrt(2)) > p:
+                lo = mid
+            else:
+                hi = mid
+        z = (lo + hi) / 2
+    theta = (mu + sigma * z).contiguous()
+
+    # 2. fixed-threshold scan
+    if use_fp8:
+        Q8 = to_fp8_bytes(Q)
+        # thresholds are in fp8-score units: inputs scaled x8 each -> x64
+        cs, ci, counts = ext().topk_scan_threshold(Q8, X8, (theta * 64.0).contiguous(),
+                                                   cap, 0, True)
+    else:
+        cs, ci, counts = ext().topk_scan_threshold(Q, X, theta, cap, 0, False)
+
+    # 3. mask unfilled/overflowed slots
+    slot = torch.arange(cap, device=Q.device).unsqueeze(0)
+    valid = slot < counts.clamp_max(cap).unsqueeze(1)
+    cs = torch.where(valid, cs, torch.full_like(cs, -1e30))
+
+    # 4. select: top candidates by scan score
+    sel = min(cap, max(2 * k, 32))
+    top = torch.topk(cs, sel, dim=1)
+    ids = torch.gather(ci, 1, top.indices)
+
+    if use_fp8:
+        # exact bf16 rescore of the selected candidates
+        gidx = ids.long().clamp_min(0)
+        cand = X[gidx]  # [nq, sel, D]
+        exact = torch.einsum("qd,qkd->qk", Q.float(), cand.float())
+        exact = torch.where(ids < 0, torch.full_like(exact, -1e30), exact)
+        fin = torch.topk(exact, k, dim=1)
+        out_s = fin.values
+        out_i = torch.gather(ids, 1, fin.indices)
+    else:
+        out_s = top.values[:, :k]
+        out_i = ids[:, :k]
+
+    # 5. fallback for underflowed queries (non-Gaussian tails / theta high)
+    bad = (counts < k) | (counts > cap)
+    if bool(bad.any()):
+        rows = bad.nonzero(as_tuple=True)[0]
+        fb_s, fb_i = topk_recall(Q[rows].contiguous(), X, k)
+        out_s[rows] = fb_s
+        out_i[rows] = fb_i
+    return out_s, out_i
