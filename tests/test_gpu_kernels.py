@@ -226,3 +226,23 @@ def test_topk_recall_two_stage_vs_torch():
         assert inter >= k - 2, f"q={q}: only {inter}/{k} overlap"
         assert abs(got_vals[q][0] - ref_vals[q][0]) < 2e-2
         assert all(got_vals[q][i] >= got_vals[q][i + 1] - 1e-6 for i in range(k - 1))
+
+
+@pytest.mark.gpu
+def test_topk_recall_threshold_bf16_and_fp8():
+    """Threshold-scan path (both dtypes) vs fp32 reference."""
+    torch.manual_seed(11)
+    nq, nx, D, k = 512, 32768, 1024, 16
+    Q = torch.nn.functional.normalize(torch.randn(nq, D, device="cuda"), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(nx, D, device="cuda"), dim=1).bfloat16()
+    ref = torch.topk(Q.float() @ X.float().T, k, dim=1)
+    ref_ids = ref.indices.cpu().numpy()
+    for X8 in (None, g.to_fp8_bytes(X)):
+        scores, ids = g.topk_recall_threshold(Q, X, k, X8=X8)
+        ids_np = ids.cpu().numpy()
+        vals = scores.cpu().numpy()
+        for q in range(nq):
+            inter = len(set(ids_np[q]) & set(ref_ids[q]))
+            assert inter >= k - 2, f"fp8={X8 is not None} q={q}: {inter}/{k}"
+            assert abs(vals[q][0] - ref.values[q, 0].item()) < 2e-2
+            assert all(vals[q][i] >= vals[q][i + 1] - 1e-6 for i in range(k - 1))

@@ -85,5 +85,26 @@ if __name__ == "__main__":
     import sys as _sys
     if "--scan-only" in _sys.argv:
         scan_only_bench()
+    elif "--threshold" in _sys.argv:
+        threshold_bench()
     else:
         main()
+
+
+def threshold_bench():
+    import torch
+    from vainplex_openclaw_amd.ops import gpu as g
+    torch.manual_seed(0)
+    dev = "cuda:0"
+    rows, nq, dim = 4_194_304, 4096, 1024
+    Q = torch.nn.functional.normalize(torch.randn(nq, dim, device=dev), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(rows, dim, device=dev), dim=1).bfloat16()
+    X8 = g.to_fp8_bytes(X)
+    for name, fn in [
+        ("thresh bf16 k=16", lambda: g.topk_recall_threshold(Q, X, 16)),
+        ("thresh fp8  k=16", lambda: g.topk_recall_threshold(Q, X, 16, X8=X8)),
+        ("thresh bf16 k=32", lambda: g.topk_recall_threshold(Q, X, 32)),
+        ("direct bf16 k=16", lambda: g.topk_recall(Q, X, 16)),
+    ]:
+        ms = timed(fn, iters=2, warmup=1)
+        print(f"{name:24s} {ms:9.2f} ms   {2*nq*rows*dim/1e12/ms*1000:7.1f} TF/s")
