@@ -81,14 +81,6 @@ def scan_only_bench():
         print(f"{name:24s} {ms:9.2f} ms   {2*nq*rows*dim/1e12/ms*1000:7.1f} TF/s")
 
 
-if __name__ == "__main__":
-    import sys as _sys
-    if "--scan-only" in _sys.argv:
-        scan_only_bench()
-    elif "--threshold" in _sys.argv:
-        threshold_bench()
-    else:
-        main()
 
 
 def threshold_bench():
@@ -108,3 +100,13 @@ def threshold_bench():
     ]:
         ms = timed(fn, iters=2, warmup=1)
         print(f"{name:24s} {ms:9.2f} ms   {2*nq*rows*dim/1e12/ms*1000:7.1f} TF/s")
+
+
+if __name__ == "__main__":
+    import sys as _sys
+    if "--scan-only" in _sys.argv:
+        scan_only_bench()
+    elif "--threshold" in _sys.argv:
+        threshold_bench()
+    else:
+        main()

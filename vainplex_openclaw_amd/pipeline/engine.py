@@ -52,9 +52,12 @@ class PipelineConfig:
     n_agents: int = 64
     index_size: int = 6_250_000  # per-GPU shard (8 GPUs x 6.25M = 50M)
     topk: int = 16
-    recall_fp8: bool = False  # opt-in: two-stage fp8 scan + exact rescore
-    # (bf16 direct currently faster: the top-k phase cost scales with k,
-    #  eating the fp8 byte savings at the k2=32 overfetch width)
+    # recall modes: "direct" = bf16 streaming top-k kernel;
+    # "two_stage" = fp8 scan (k2=32) + exact rescore;
+    # "threshold" = Gaussian-tail threshold scan + select (+ fp8 rescore
+    # when recall_fp8) — no in-kernel top-k maintenance
+    recall_mode: str = "threshold"
+    recall_fp8: bool = True
     inj_threshold: float = 0.9
     seed: int = 1234
     families: tuple = ("redaction", "injection", "claims", "entity")
@@ -129,7 +132,7 @@ class FirewallPipeline:
             # fp8 (e4m3) copy of the index for the stage-1 scan: half the
             # staged bytes at the measured GLDS transport bound
             self.index8 = None
-            if cfg.recall_fp8:
+            if cfg.recall_fp8 and cfg.recall_mode in ("two_stage", "threshold"):
                 self.index8 = torch.empty(
                     cfg.index_size, cfg.dim, dtype=torch.uint8, device=self.device
                 )
@@ -190,7 +193,11 @@ class FirewallPipeline:
         prof.mark("recall")
         # 5. Membrane recall (full index across ranks; parallel/collectives)
         def local_recall(queries):
-            if self.index8 is not None:
+            if cfg.recall_mode == "threshold":
+                return g.topk_recall_threshold(
+                    queries, self.index, cfg.topk, X8=self.index8
+                )
+            if cfg.recall_mode == "two_stage" and self.index8 is not None:
                 return g.topk_recall_two_stage(queries, self.index, self.index8, cfg.topk)
             return g.topk_recall(queries, self.index, cfg.topk)
 
