@@ -212,3 +212,28 @@ def test_extract_agent_ids_shapes():
     assert extract_agent_ids({"agents": [{"id": "x"}, {"name": "y"}]}) == ["x", "y"]
     assert extract_agent_ids({"agents": {"m": {}, "n": {}}}) == ["m", "n"]
     assert extract_agent_ids({"defaultAgent": "main"}) == ["main"]
+
+
+def test_before_agent_start_context_injection(tmp_path, monkeypatch):
+    """Context injection at priority 5 + ERC-8004 config nesting
+    (reference hooks.ts:445-498, config.ts:330-334)."""
+    monkeypatch.setenv("HOME", str(tmp_path))
+    from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+    from vainplex_openclaw_amd.governance.plugin import create_plugin
+    from vainplex_openclaw_amd.governance.hooks import resolve_erc8004_config
+
+    bus = HookBus()
+    api = PluginApi(id="openclaw-governance", plugin_config={}, logger=NullLogger(),
+                    config={}, bus=bus)
+    p = create_plugin(workspace=str(tmp_path))
+    p.register(api)
+    ev = bus.emit("before_agent_start", {"ctx": {"agentId": "main", "sessionKey": "m:1"}})
+    ctx = ev.get("prependContext", "")
+    assert "[Governance] Agent: main" in ctx and "Session:" in ctx
+
+    # erc8004 nesting: top-level and under agentFirewall both resolve
+    assert resolve_erc8004_config({"erc8004": {"enabled": True}})["enabled"]
+    assert resolve_erc8004_config(
+        {"agentFirewall": {"erc8004": {"enabled": True, "agentMapping": {"main": 7}}}}
+    )["agentMapping"] == {"main": 7}
+    assert not resolve_erc8004_config({})["enabled"]

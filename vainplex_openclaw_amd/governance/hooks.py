@@ -79,6 +79,14 @@ class GovernanceHooks:
         self.response_gate = ResponseGate(config.get("responseGate"))
         self.tool_call_log: Dict[str, List[Dict[str, str]]] = {}
         self._recent_agent_ctx: Dict[str, Dict[str, str]] = {}
+        # ERC-8004 config lives top-level or nested under agentFirewall
+        # (reference config.ts:205,330-334)
+        self.erc8004_config = resolve_erc8004_config(config)
+        self.erc8004_provider = None
+        if self.erc8004_config.get("enabled"):
+            from .security.erc8004 import ERC8004Provider
+
+            self.erc8004_provider = ERC8004Provider()
 
     # -- handlers ----------------------------------------------------------
     def before_tool_call(self, ev: Dict[str, Any]) -> Optional[Dict[str, Any]]:
@@ -180,6 +188,39 @@ class GovernanceHooks:
         except Exception:
             pass
 
+    def before_agent_start(self, ev: Dict[str, Any]) -> Optional[Dict[str, Any]]:
+        """Context injection + ERC-8004 reputation log (hooks.ts:445-498)."""
+        try:
+            ctx = ev.get("ctx") or {}
+            agent_id = ctx.get("agentId") or ev.get("agentId") or "main"
+            session = ctx.get("sessionKey") or ctx.get("sessionId") or f"agent:{agent_id}"
+            agent_rec = self.engine.trust_manager.get(agent_id)
+            sess_rec = self.engine.session_trust.get(session, agent_id)
+            cfg = self.erc8004_config
+            if cfg.get("enabled") and self.erc8004_provider is not None:
+                on_chain = cfg.get("agentMapping", {}).get(agent_id)
+                if isinstance(on_chain, int):
+                    try:
+                        rep = self.erc8004_provider.lookup_reputation(on_chain)
+                        if rep:
+                            self.logger.info(
+                                "[firewall] ERC-8004 reputation for agent %s: %s (score=%s)",
+                                agent_id, rep.get("tier"), rep.get("reputationScore"),
+                            )
+                    except Exception as exc:  # fail-open
+                        self.logger.warn("[firewall] ERC-8004 lookup failed: %s", exc)
+            status = self.engine.status()
+            context = (
+                f"\n[Governance] Agent: {agent_id} "
+                f"({agent_rec.get('score')}/{agent_rec.get('tier')}) | "
+                f"Session: {sess_rec.get('score')}/{sess_rec.get('tier')} | "
+                f"Policies: {len(status.get('policies', []))}"
+            )
+            return {"prependContext": context}
+        except Exception as exc:
+            self.logger.error("[governance] Error in before_agent_start: %s", exc)
+            return None
+
     def session_start(self, ev: Dict[str, Any]) -> None:
         agent_id = resolve_agent_id(ev)
         session = ev.get("sessionKey") or ev.get("sessionId") or f"agent:{agent_id}"
@@ -212,6 +253,25 @@ class GovernanceHooks:
         return self.engine.trust_manager.snapshot()
 
 
+def resolve_erc8004_config(config: Dict[str, Any]) -> Dict[str, Any]:
+    """Top-level `erc8004` or nested `agentFirewall.erc8004`
+    (reference config.ts:204-217, 330-334)."""
+    af = config.get("agentFirewall")
+    raw = None
+    if isinstance(af, dict) and isinstance(af.get("erc8004"), dict):
+        raw = af["erc8004"]
+    elif isinstance(config.get("erc8004"), dict):
+        raw = config["erc8004"]
+    if not isinstance(raw, dict):
+        return {"enabled": False, "agentMapping": {}}
+    return {
+        "enabled": bool(raw.get("enabled", False)),
+        "rpcUrl": raw.get("rpcUrl"),
+        "identityRegistryAddress": raw.get("identityRegistryAddress"),
+        "agentMapping": raw.get("agentMapping") if isinstance(raw.get("agentMapping"), dict) else {},
+    }
+
+
 def register_governance_hooks(
     api: PluginApi,
     engine: GovernanceEngine,
@@ -224,6 +284,7 @@ def register_governance_hooks(
     api.on("before_message_write", h.before_message_write, priority=1000)
     api.on("after_tool_call", h.after_tool_call, priority=900)
     api.on("message_received", h.totp_intercept, priority=1000)
+    api.on("before_agent_start", h.before_agent_start, priority=5)
     api.on("session_start", h.session_start, priority=1)
     api.on("session_end", h.session_end, priority=999)
     api.on("gateway_start", lambda ev: None, priority=1)
