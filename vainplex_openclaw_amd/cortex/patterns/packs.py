@@ -74,7 +74,7 @@ PACKS: Dict[str, dict] = {
     "es": {
         "name": "Español",
         "patterns": {
-            "decision": [r"(?:decidido|decisión|acordado|acordamos|el plan es|vamos a hacer|enfoque:)"],
+            "decision": [r"(?:decidido|decidimos|decisión|acordado|acordamos|el plan es|vamos a hacer|enfoque:)"],
             "close": [r"(?:^|\s)(?:está |todo )?(?:hecho|listo|resuelto|arreglado|cerrado|terminado)(?:\s|[.!]|$)", r"(?:^|\s)funciona(?:\s|[.!]|$)", r"✅"],
             "wait": [r"(?:esperando a|esperando por|bloqueado por|necesita primero|en espera)"],
             "topic": [r"(?:volviendo a|ahora sobre|respecto a|hablemos de|miremos)\s+(?:el\s+|la\s+|los\s+)?(\w[\w\s-]{3,40})"],
@@ -131,7 +131,7 @@ PACKS: Dict[str, dict] = {
     "pt": {
         "name": "Português",
         "patterns": {
-            "decision": [r"(?:decidido|decisão|combinado|acordamos|o plano é|vamos fazer|abordagem:)"],
+            "decision": [r"(?:decidido|decidimos|decisão|combinado|acordamos|o plano é|vamos fazer|abordagem:)"],
             "close": [r"(?:^|\s)(?:está |tudo )?(?:feito|pronto|resolvido|corrigido|fechado|concluído)(?:\s|[.!]|$)", r"(?:^|\s)funciona(?:\s|[.!]|$)", r"✅"],
             "wait": [r"(?:esperando por|aguardando|bloqueado por|precisa primeiro|em espera)"],
             "topic": [r"(?:voltando a|agora sobre|sobre o|falemos de|vejamos)\s+(?:o\s+|a\s+|os\s+)?(\w[\w\s-]{3,40})"],
