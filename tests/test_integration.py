@@ -119,3 +119,14 @@ def test_gateway_stop_flushes_everything(suite):
     # membrane store flushed
     membrane = gw.plugins["openclaw-membrane"]
     assert membrane.engine.stats["ingested"] >= 1
+
+
+def test_suite_cli_demo(tmp_path, monkeypatch, capsys):
+    """`python -m vainplex_openclaw_amd --demo` end-to-end."""
+    monkeypatch.setenv("HOME", str(tmp_path))
+    from vainplex_openclaw_amd.__main__ import main
+
+    rc = main(["--demo", "--workspace", str(tmp_path / "ws")])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "membrane_context" in out and "postgres" in out
