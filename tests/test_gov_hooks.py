@@ -237,3 +237,81 @@ def test_before_agent_start_context_injection(tmp_path, monkeypatch):
         {"agentFirewall": {"erc8004": {"enabled": True, "agentMapping": {"main": 7}}}}
     )["agentMapping"] == {"main": 7}
     assert not resolve_erc8004_config({})["enabled"]
+
+
+def test_rate_limiter_through_full_stack(workspace):
+    """Rate Limiter (15/min; 2x for trusted) triggers through the hook
+    stack after enough calls (builtin-policies.ts:152-196)."""
+    gw = Gateway(config={"agents": ["spammy"]})
+    plugin = GovernancePlugin(workspace)
+    gw.load(plugin, plugin_config={
+        "builtinPolicies": {"rateLimiter": True}, "workspace": workspace,
+    })
+    gw.start()
+    try:
+        blocked = 0
+        for i in range(25):
+            ev = gw.emit("before_tool_call", {
+                "agentId": "spammy", "sessionKey": "agent:spammy",
+                "toolName": "exec", "params": {"command": f"echo {i}"},
+            })
+            if ev.get("block"):
+                blocked += 1
+        assert blocked >= 5  # beyond 15/min everything denies
+    finally:
+        gw.stop()
+
+
+def test_message_sending_redaction_layer(workspace):
+    """Layer-2 outbound redaction fires on message_sending through the
+    full plugin stack (redaction/hooks.ts:127-141)."""
+    gw = Gateway(config={"agents": ["a1"]})
+    plugin = GovernancePlugin(workspace)
+    gw.load(plugin, plugin_config={"workspace": workspace})
+    gw.start()
+    try:
+        ev = gw.emit("message_sending", {
+            "content": "token is ghp_" + "z" * 36,
+            "agentId": "a1", "sessionKey": "agent:a1", "to": "user",
+        })
+        assert "ghp_" + "z" * 36 not in ev.get("content", "")
+    finally:
+        gw.stop()
+
+
+def test_trust_violation_feedback_on_deny(workspace):
+    """Deny verdicts record violations + session signal (engine.ts:248-263)
+    except for night-mode denials (covered in test_audit_engine)."""
+    gw = Gateway(config={"agents": ["risky"]})
+    plugin = GovernancePlugin(workspace)
+    gw.load(plugin, plugin_config={
+        "builtinPolicies": {"credentialGuard": True}, "workspace": workspace,
+    })
+    gw.start()
+    try:
+        before = plugin.engine.trust_manager.get("risky")["signals"]["violationCount"]
+        gw.emit("before_tool_call", {
+            "agentId": "risky", "sessionKey": "agent:risky",
+            "toolName": "read", "params": {"file_path": "~/.aws/credentials"},
+        })
+        after = plugin.engine.trust_manager.get("risky")["signals"]["violationCount"]
+        assert after == before + 1
+    finally:
+        gw.stop()
+
+
+def test_session_end_clears_toolcall_log(workspace):
+    gw = Gateway(config={"agents": ["a1"]})
+    plugin = GovernancePlugin(workspace)
+    gw.load(plugin, plugin_config={"workspace": workspace})
+    gw.start()
+    try:
+        gw.emit("after_tool_call", {
+            "agentId": "a1", "sessionKey": "agent:a1", "toolName": "read", "result": "x",
+        })
+        assert plugin.hooks.tool_call_log
+        gw.emit("session_end", {"agentId": "a1", "sessionKey": "agent:a1",
+                                "ctx": {"sessionKey": "agent:a1"}})
+        assert not plugin.hooks.tool_call_log.get("agent:a1")
+    finally:
+        gw.stop()
