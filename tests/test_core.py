@@ -124,3 +124,54 @@ def test_plugin_manifests():
         p = write_manifest("openclaw-cortex", d)
         data = _json.loads(open(p).read())
         assert data["id"] == "openclaw-cortex" and data["version"]
+
+
+def test_atomic_write_and_backup(tmp_path):
+    """tmp+rename atomicity + .bak backups (writer.ts:14-49 parity)."""
+    from vainplex_openclaw_amd.utils.storage import (
+        atomic_write_json, backup_then_write, read_json,
+    )
+
+    p = tmp_path / "x.json"
+    atomic_write_json(str(p), {"a": 1})
+    assert read_json(str(p)) == {"a": 1}
+    assert not list(tmp_path.glob(".x.json.tmp*"))  # no tmp litter
+    bak = backup_then_write(str(p), '{"a": 2}')
+    assert bak and read_json(bak) == {"a": 1}
+    assert read_json(str(p)) == {"a": 2}
+    # corrupt file -> default
+    p.write_text("{nope")
+    assert read_json(str(p), default="d") == "d"
+
+
+def test_debounced_saver_coalesces(tmp_path):
+    import time as _t
+    from vainplex_openclaw_amd.utils.storage import DebouncedSaver
+
+    calls = []
+    s = DebouncedSaver(lambda: calls.append(1), delay=0.05)
+    for _ in range(10):
+        s.mark_dirty()
+    _t.sleep(0.15)
+    assert len(calls) == 1  # coalesced
+    s.flush()
+    assert len(calls) == 1  # nothing dirty -> no extra save
+    s.mark_dirty()
+    s.flush()  # explicit flush beats the timer
+    assert len(calls) == 2
+    s.close()
+
+
+def test_interval_flusher_dirty_flag():
+    from vainplex_openclaw_amd.utils.storage import IntervalFlusher
+
+    calls = []
+    f = IntervalFlusher(lambda: calls.append(1), interval=99)
+    f.flush()
+    assert calls == []  # not dirty
+    f.mark_dirty()
+    f.flush()
+    assert calls == [1]
+    f.mark_dirty()
+    f.stop()  # stop flushes
+    assert calls == [1, 1]

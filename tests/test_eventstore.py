@@ -276,3 +276,38 @@ def test_journal_feeds_trace_analyzer_source():
     events = src.fetch()
     assert len(events) >= 2
     assert any(e.type == "tool_result" or "tool" in e.type for e in events)
+
+
+def test_journal_retention_bytes_and_age():
+    """maxBytes + maxAgeHours retention (config.ts limits semantics)."""
+    t = [1000.0]
+    j = EventJournal(durable=False, max_bytes=400, clock=lambda: t[0])
+    for i in range(20):
+        j.publish("s", {"ts": t[0] * 1000, "pad": "x" * 20, "i": i})
+    assert len(j) < 20  # byte cap evicted oldest
+    kept = [e["i"] for _s, e in j.replay()]
+    assert kept == sorted(kept) and kept[-1] == 19
+
+    j2 = EventJournal(durable=False, max_age_hours=1, clock=lambda: t[0])
+    j2.publish("s", {"ts": (t[0] - 7200) * 1000, "i": "old"})
+    j2.publish("s", {"ts": t[0] * 1000, "i": "new"})
+    j2.publish("s", {"ts": t[0] * 1000, "i": "new2"})  # triggers eviction pass
+    ids = [e["i"] for _s, e in j2.replay()]
+    assert "old" not in ids and "new" in ids
+
+
+def test_journal_replay_limit_and_fetch_range():
+    j = EventJournal(durable=False)
+    for i in range(10):
+        j.publish("s", {"ts": i * 1000.0, "i": i})
+    assert len(list(j.replay(limit=3))) == 3
+    got = j.fetch_range(2.0, 5.0)
+    assert [e["i"] for e in got] == [2, 3, 4, 5]
+
+
+def test_envelope_visibility_default_and_values():
+    env = build_envelope("run.started", "a", "s", {})
+    assert env["visibility"] == "internal"
+    from vainplex_openclaw_amd.eventstore.events import VISIBILITIES
+
+    assert VISIBILITIES == ("public", "internal", "confidential", "secret")
