@@ -208,3 +208,33 @@ def test_trace_to_facts_bridge(workspace):
     n = apply_report_to_registry(report, reg)
     assert n == 1
     assert reg.lookup("nginx", "state")["value"] == "error"
+
+
+SIGNAL_I18N_SAMPLES = {
+    "de": ("nein, das ist falsch", "das hilft nicht, frustrierend"),
+    "es": ("no, eso está mal", "inútil, esto no ayuda"),
+    "fr": ("non, c'est faux", "inutile, ça n'aide pas"),
+    "it": ("no, è sbagliato", "inutile, non aiuta"),
+    "pt": ("não, está errado", "inútil, isso não ajuda"),
+    "ru": ("нет, это неверно", "бесполезно, это не помогает"),
+    "ja": ("違います、やり直して", "役に立たない"),
+    "ko": ("아니요, 틀렸습니다", "쓸모없어요"),
+    "zh": ("不对，这是错的", "没用，这没有帮助"),
+}
+
+
+@pytest.mark.parametrize("lang", sorted(SIGNAL_I18N_SAMPLES))
+def test_correction_and_dissatisfied_i18n(lang):
+    """Localized phrase packs fire in all 10 languages
+    (signals/lang/signal-lang-*.ts parity)."""
+    corr, dissat = SIGNAL_I18N_SAMPLES[lang]
+    events = [
+        ev("msg.out", content="here is the result"),
+        ev("msg.in", content=corr),
+        ev("msg.out", content="another answer"),
+        ev("msg.in", content=dissat),
+    ]
+    chains = reconstruct_chains(events)
+    kinds = {f.signal_type for f in detect_all_signals(chains, ["correction", "dissatisfied"])}
+    assert "correction" in kinds, lang
+    assert "dissatisfied" in kinds, lang
