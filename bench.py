@@ -46,6 +46,9 @@ def main() -> int:
     ap.add_argument("--index", type=int, default=TOTAL_INDEX, help="TOTAL index rows across ranks")
     ap.add_argument("--topk", type=int, default=16)
     ap.add_argument("--pool", type=int, default=4, help="pre-generated batch pool size")
+    ap.add_argument("--recall-mode", default="threshold",
+                    choices=["threshold", "two_stage", "direct"])
+    ap.add_argument("--no-fp8", action="store_true")
     ap.add_argument("--profile", action="store_true",
                     help="after the timed region, run 3 instrumented steps and print per-stage ms (stderr)")
     args = ap.parse_args()
@@ -64,6 +67,7 @@ def main() -> int:
     shard = (shard // 128) * 128
     cfg = PipelineConfig(
         batch=args.batch, dim=args.dim, index_size=shard, topk=args.topk,
+        recall_mode=args.recall_mode, recall_fp8=not args.no_fp8,
     )
     t0 = time.time()
     pipe = FirewallPipeline(cfg, device=device, world_size=world, rank=rank)
