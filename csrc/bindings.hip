@@ -187,16 +187,6 @@ std::vector<torch::Tensor> topk_recall(torch::Tensor Q, torch::Tensor X, int64_t
   return {out_s, out_i};
 }
 
-
-// fp8 X-direct loads use u32 per-lane offsets within a swath: keep each
-// swath under 4 GB of index bytes (raise S in multiples of 8).
-static int64_t min_swaths_for_u32(long long nx, int D, int64_t s) {
-  const long long limit = 0xE0000000LL;  // 3.5 GiB headroom
-  if (s < 1) s = 1;
-  while (((nx + s - 1) / s + 255) * (long long)D > limit) s += 8;
-  return s;
-}
-
 std::vector<torch::Tensor> topk_recall_fp8(torch::Tensor Q8, torch::Tensor X8,
                                            int64_t k, int64_t n_swaths) {
   // stage-1 fp8 scan of the two-stage recall: inputs are e4m3 bytes
@@ -209,7 +199,6 @@ std::vector<torch::Tensor> topk_recall_fp8(torch::Tensor Q8, torch::Tensor X8,
   TORCH_CHECK(X8.size(1) == D && D % 64 == 0, "D must be a multiple of 64");
   TORCH_CHECK(k >= 1 && k <= 32, "k in [1,32]");  // TOPK_MAX LDS bound
   int n_qblocks = (nq + 255) / 256;
-  n_swaths = min_swaths_for_u32(nx, D, n_swaths);
   TORCH_CHECK(n_swaths >= 1 && (long long)n_swaths * k <= 1024,
               "n_swaths * k must be <= 1024");
   auto f32opts = torch::dtype(torch::kFloat32).device(Q8.device());
@@ -315,7 +304,6 @@ std::vector<torch::Tensor> topk_scan_threshold(torch::Tensor Q, torch::Tensor X,
     long long max_s = nx / 256; if (max_s < 1) max_s = 1;
     if (n_swaths > max_s) n_swaths = max_s;
   }
-  if (fp8) n_swaths = min_swaths_for_u32(nx, D, n_swaths);
   auto f32opts = torch::dtype(torch::kFloat32).device(Q.device());
   auto i32opts = torch::dtype(torch::kInt32).device(Q.device());
   auto cand_s = torch::full({(long long)nq, cap}, -1e30, f32opts);

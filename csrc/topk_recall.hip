@@ -442,13 +442,9 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
                    // cand_ids are then [nq][cap]; tc_n[nq] counts appends.
                    const float* __restrict__ theta,
                    int32_t* __restrict__ tc_n, int cap) {
-  // LDS holds ONLY Q k-tiles: X fragments come straight from global
-  // memory via saddr-form dwordx2 loads (issue order X-after-stage +
-  // counted vmcnt keeps the Q GLDS prefetch in flight). This halves the
-  // traffic on the ~6.4 TB/s global_load_lds transport path — the
-  // measured bound — and Q's re-reads are L3 hits (8 MB resident).
-  __shared__ bf16 lds_all[NBUF_F8 * BM * (BK_F8 / 2)];
+  __shared__ bf16 lds_all[NBUF_F8 * (BM + BN) * (BK_F8 / 2)];
 #define QS8(buf) (lds_all + (buf) * BM * (BK_F8 / 2))
+#define XS8(buf) (lds_all + NBUF_F8 * BM * (BK_F8 / 2) + (buf) * BN * (BK_F8 / 2))
   __shared__ float row_min[BM];
   __shared__ int row_min_slot[BM];
   __shared__ float topk_vals[BM][TOPK_MAX];  // LDS candidate values
@@ -491,92 +487,54 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
   int nk = D / BK_F8;
   size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
 
-  // X swath base for saddr-form direct loads: per-lane 32-bit offsets
-  // cover the swath (the wrapper keeps swaths under 4 GB)
-  const uint8_t* xbase = X + x_begin * D;
-
   for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
     f32x4 acc[8][4] = {};
-    for (int p = 0; p < 2 && p < nk; ++p)
+    for (int p = 0; p < 2 && p < nk; ++p) {
       stage_tile8(Q, D, row0, nq, p * BK_F8, QS8(p), BM);
-    // per-lane X row byte offsets for this tile (clamped into the swath;
-    // out-of-range columns produce garbage scores filtered by the
-    // epilogues' col bound, same as the staged path)
-    uint32_t xoff[4];
-#pragma unroll
-    for (int n = 0; n < 4; ++n) {
-      long long gcol = x0 + wn * 64 + n * 16 + lrow;
-      if (gcol >= x_end) gcol = x_end - 1;
-      xoff[n] = (uint32_t)((gcol - x_begin) * (long long)D);
+      stage_tile8(X, D, x0, (long long)nx, p * BK_F8, XS8(p), BN);
     }
     for (int kt = 0; kt < nk; ++kt) {
       int cur = kt % NBUF_F8;
-      // stage(kt) landed at the previous kt's sub-0 wait; this counted
-      // wait only has to cover the first iterations
-      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+      // stage(kt) landed when only stage(kt+1)'s 4 glds are outstanding
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       __builtin_amdgcn_s_barrier();
-      // Both sub-steps' X fragments issued BEFORE the next Q stage:
-      // vmcnt retires IN ISSUE ORDER, so sub-0's vmcnt(6) proves
-      // [stage(kt+1), X0..3] landed while [X4..7, stage(kt+2)] ride,
-      // and sub-1's vmcnt(2) proves X4..7 landed while stage(kt+2)
-      // rides into the next iteration.
-      uint32_t kb0 = (uint32_t)(kt * BK_F8) + (uint32_t)kgrp * 8u;
-      long long xv[8];
-      asm volatile(
-          "global_load_dwordx2 %0, %8, %16\n\t"
-          "global_load_dwordx2 %1, %9, %16\n\t"
-          "global_load_dwordx2 %2, %10, %16\n\t"
-          "global_load_dwordx2 %3, %11, %16\n\t"
-          "global_load_dwordx2 %4, %12, %16\n\t"
-          "global_load_dwordx2 %5, %13, %16\n\t"
-          "global_load_dwordx2 %6, %14, %16\n\t"
-          "global_load_dwordx2 %7, %15, %16"
-          : "=&v"(xv[0]), "=&v"(xv[1]), "=&v"(xv[2]), "=&v"(xv[3]),
-            "=&v"(xv[4]), "=&v"(xv[5]), "=&v"(xv[6]), "=&v"(xv[7])
-          : "v"(xoff[0] + kb0), "v"(xoff[1] + kb0),
-            "v"(xoff[2] + kb0), "v"(xoff[3] + kb0),
-            "v"(xoff[0] + kb0 + 32u), "v"(xoff[1] + kb0 + 32u),
-            "v"(xoff[2] + kb0 + 32u), "v"(xoff[3] + kb0 + 32u),
-            "s"(xbase));
       if (kt + 2 < nk) {
         int pre = (kt + 2) % NBUF_F8;
         stage_tile8(Q, D, row0, nq, (kt + 2) * BK_F8, QS8(pre), BM);
+        stage_tile8(X, D, x0, (long long)nx, (kt + 2) * BK_F8, XS8(pre), BN);
       }
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {  // k-halves 0..31, 32..63
         uint32_t slot = sub * 2 + (kgrp >> 1);
+        uint32_t base_x = (uint32_t)(size_t)XS8(cur)
+                          + lds_off_bytes(wn * 64 + lrow, slot);
         uint32_t base_q = (uint32_t)(size_t)QS8(cur)
                           + lds_off_bytes(wm * 128 + lrow, slot);
-        bf16x8 qf[8];
-        if (sub == 0)
-          asm volatile(
-              "ds_read_b128 %0, %8\n\t"
-              "ds_read_b128 %1, %8 offset:1024\n\t"
-              "ds_read_b128 %2, %8 offset:2048\n\t"
-              "ds_read_b128 %3, %8 offset:3072\n\t"
-              "ds_read_b128 %4, %8 offset:4096\n\t"
-              "ds_read_b128 %5, %8 offset:5120\n\t"
-              "ds_read_b128 %6, %8 offset:6144\n\t"
-              "ds_read_b128 %7, %8 offset:7168\n\t"
-              "s_waitcnt vmcnt(6) lgkmcnt(0)"
-              : "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
-                "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7])
-              : "v"(base_q));
-        else
-          asm volatile(
-              "ds_read_b128 %0, %8\n\t"
-              "ds_read_b128 %1, %8 offset:1024\n\t"
-              "ds_read_b128 %2, %8 offset:2048\n\t"
-              "ds_read_b128 %3, %8 offset:3072\n\t"
-              "ds_read_b128 %4, %8 offset:4096\n\t"
-              "ds_read_b128 %5, %8 offset:5120\n\t"
-              "ds_read_b128 %6, %8 offset:6144\n\t"
-              "ds_read_b128 %7, %8 offset:7168\n\t"
-              "s_waitcnt vmcnt(2) lgkmcnt(0)"
-              : "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
-                "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7])
-              : "v"(base_q));
-        long long qv[8];
+        bf16x8 xf[4], qf[8];
+        asm volatile(
+            "ds_read_b128 %0, %12\n\t"
+            "ds_read_b128 %1, %12 offset:1024\n\t"
+            "ds_read_b128 %2, %12 offset:2048\n\t"
+            "ds_read_b128 %3, %12 offset:3072\n\t"
+            "ds_read_b128 %4, %13\n\t"
+            "ds_read_b128 %5, %13 offset:1024\n\t"
+            "ds_read_b128 %6, %13 offset:2048\n\t"
+            "ds_read_b128 %7, %13 offset:3072\n\t"
+            "ds_read_b128 %8, %13 offset:4096\n\t"
+            "ds_read_b128 %9, %13 offset:5120\n\t"
+            "ds_read_b128 %10, %13 offset:6144\n\t"
+            "ds_read_b128 %11, %13 offset:7168\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
+              "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
+              "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7])
+            : "v"(base_x), "v"(base_q));
+        long long xv[4], qv[8];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          i64x2 t = __builtin_bit_cast(i64x2, xf[n]);
+          xv[n] = half ? t.y : t.x;
+        }
 #pragma unroll
         for (int m = 0; m < 8; ++m) {
           i64x2 t = __builtin_bit_cast(i64x2, qf[m]);
@@ -587,7 +545,7 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
 #pragma unroll
           for (int n = 0; n < 4; ++n)
             acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-                qv[m], xv[sub * 4 + n], acc[m][n], 0, 0, 0);
+                qv[m], xv[n], acc[m][n], 0, 0, 0);
       }
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
