@@ -275,3 +275,29 @@ def test_topk_recall_threshold_clustered_fallback():
         assert abs(vals[q][0] - ref_vals[q][0]) < 2e-2
         assert (ids_np[q] >= 0).all()
         assert all(vals[q][i] >= vals[q][i + 1] - 1e-6 for i in range(k - 1))
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("nq,nx,k", [
+    (1, 300, 1),        # single query, tiny index
+    (100, 511, 3),      # nothing aligns
+    (257, 4097, 32),    # one past the tile boundaries, max k
+    (512, 256, 16),     # index smaller than a swath wants
+    (300, 16384, 7),    # odd everything
+])
+def test_topk_recall_odd_shapes(nq, nx, k):
+    """Boundary handling: nq not a multiple of BM=256, nx not a multiple
+    of BN=256, k across [1,32]."""
+    torch.manual_seed(nq + nx + k)
+    D = 1024
+    Q = torch.nn.functional.normalize(torch.randn(nq, D, device="cuda"), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(nx, D, device="cuda"), dim=1).bfloat16()
+    scores, ids = g.topk_recall(Q, X, k)
+    ref = torch.topk(Q.float() @ X.float().T, k, dim=1)
+    ids_np, ref_ids = ids.cpu().numpy(), ref.indices.cpu().numpy()
+    vals, ref_vals = scores.cpu().numpy(), ref.values.cpu().numpy()
+    for q in range(nq):
+        assert (ids_np[q] >= 0).all() and (ids_np[q] < nx).all()
+        inter = len(set(ids_np[q]) & set(ref_ids[q]))
+        assert inter >= max(1, k - 2), f"q={q}: {inter}/{k}"
+        assert abs(vals[q][0] - ref_vals[q][0]) < 2e-2
