@@ -130,3 +130,34 @@ def test_suite_cli_demo(tmp_path, monkeypatch, capsys):
     assert rc == 0
     out = capsys.readouterr().out
     assert "membrane_context" in out and "postgres" in out
+
+
+def test_cortex_trace_analyzer_wired_to_journal(tmp_path, monkeypatch):
+    """Cortex registers the trace analyzer over the shared journal and
+    cortex.analyze produces a report from real hook traffic."""
+    monkeypatch.setenv("HOME", str(tmp_path))
+    from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+    from vainplex_openclaw_amd.cortex.hooks import create_plugin as mk_cortex
+    from vainplex_openclaw_amd.eventstore import EventJournal
+    from vainplex_openclaw_amd.eventstore.plugin import create_plugin as mk_es
+
+    ws = str(tmp_path / "ws")
+    bus = HookBus()
+    journal = EventJournal(durable=False)
+    for plugin, pid in ((mk_es(journal=journal), "nats-eventstore"),
+                        (mk_cortex(workspace=ws, journal=journal), "openclaw-cortex")):
+        api = PluginApi(id=pid, plugin_config={}, logger=NullLogger(), config={}, bus=bus)
+        plugin.register(api)
+        if pid == "openclaw-cortex":
+            cortex_api = api
+    # drive a doom loop through the hooks -> journal -> analyzer
+    for i in range(4):
+        bus.emit("before_tool_call", {"toolName": "exec", "params": {"command": "make build"},
+                                      "ctx": {"sessionKey": "main:t:1", "toolCallId": f"t{i}"}})
+        bus.emit("after_tool_call", {"toolName": "exec", "params": {"command": "make build"},
+                                     "error": "exit 1",
+                                     "ctx": {"sessionKey": "main:t:1", "toolCallId": f"t{i}"}})
+    report = cortex_api.gateway_methods["cortex.analyze"]()
+    assert report["eventsAnalyzed"] >= 8
+    assert any(f["signalType"] in ("doom_loop", "repeat_fail", "tool_fail")
+               for f in report["findings"])

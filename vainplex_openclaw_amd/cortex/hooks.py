@@ -168,9 +168,13 @@ class CortexPlugin:
     description = "Conversation intelligence: thread/decision/commitment tracking, boot context"
     version = "0.1.0"
 
-    def __init__(self, workspace: Optional[str] = None):
+    def __init__(self, workspace: Optional[str] = None, journal=None, call_llm=None):
         self.workspace = workspace
         self.hooks: Optional[CortexHooks] = None
+        self.journal = journal
+        self.call_llm = call_llm
+        self.analyzer = None
+        self._analyzer_timer = None
 
     def register(self, api: PluginApi) -> None:
         config = load_plugin_config(self.id, fallback=api.plugin_config)
@@ -189,7 +193,56 @@ class CortexPlugin:
         api.register_command("cortex.decisions", h.tool_decisions)
         api.register_command("cortex.commitments", h.tool_commitments)
         api.register_command("cortex.search", h.tool_search)
+        self._register_trace_analyzer(api, config, workspace)
+
+    def _register_trace_analyzer(self, api: PluginApi, config, workspace: str) -> None:
+        """Conditional trace-analyzer registration + interval timer
+        (reference cortex hooks.ts:242-253, trace-analyzer/hooks.ts)."""
+        ta_cfg = config.get("traceAnalyzer") or {}
+        if not ta_cfg.get("enabled", self.journal is not None):
+            return
+        from .trace.analyzer import AnalyzerConfig, JournalTraceSource, TraceAnalyzer, create_nats_source
+
+        source = None
+        if self.journal is not None:
+            source = JournalTraceSource(self.journal)
+        else:
+            source = create_nats_source(ta_cfg.get("natsUrl"))
+        if source is None:
+            api.logger.info("[cortex] trace analyzer: no event source available")
+            return
+        self.analyzer = TraceAnalyzer(
+            workspace, source,
+            AnalyzerConfig(
+                detectors=ta_cfg.get("detectors"),
+                min_confidence=ta_cfg.get("minConfidence", 0.0),
+            ),
+            call_llm=self.call_llm,
+        )
+        api.register_command("cortexanalyze", lambda *a, **kw: self.analyzer.run())
+        api.register_gateway_method("cortex.analyze", lambda *a, **kw: self.analyzer.run())
+
+        interval_min = float(ta_cfg.get("intervalMinutes", 0))
+        if interval_min > 0:
+            import threading
+
+            def schedule():
+                def fire():
+                    try:
+                        self.analyzer.run()
+                    except Exception:
+                        pass
+                    schedule()
+
+                self._analyzer_timer = threading.Timer(interval_min * 60.0, fire)
+                self._analyzer_timer.daemon = True
+                self._analyzer_timer.start()
+
+            schedule()
+            api.on("gateway_stop",
+                   lambda ev: self._analyzer_timer and self._analyzer_timer.cancel(),
+                   priority=999)
 
 
-def create_plugin(workspace: Optional[str] = None) -> CortexPlugin:
-    return CortexPlugin(workspace)
+def create_plugin(workspace: Optional[str] = None, journal=None, call_llm=None) -> CortexPlugin:
+    return CortexPlugin(workspace, journal=journal, call_llm=call_llm)
