@@ -241,3 +241,51 @@ def test_llm_validator_fail_open_and_bad_json():
     v2 = LlmValidator(boom)
     r = v2.validate("y", [], True)
     assert r["verdict"] == "pass" and "llm-error" in r["reason"]
+
+
+# -- full config resolution (config.ts) --------------------------------------
+
+def test_resolve_governance_config_defaults():
+    from vainplex_openclaw_amd.governance.config import resolve_config
+
+    cfg = resolve_config({})
+    assert cfg["failMode"] == "open" and cfg["timezone"] == "UTC"
+    assert cfg["trust"]["defaults"] == {"main": 60, "*": 10}
+    assert cfg["trust"]["decay"] == {"enabled": True, "inactivityDays": 30, "rate": 0.95}
+    assert cfg["trust"]["sessionTrust"]["seedFactor"] == 0.7
+    assert cfg["trust"]["sessionTrust"]["signals"]["credentialViolation"] == -10
+    assert cfg["audit"] == {"enabled": True, "retentionDays": 90,
+                            "redactPatterns": [], "level": "standard"}
+    assert set(cfg["outputValidation"]["detectors"]) == {
+        "system_state", "entity_name", "existence",
+        "operational_status", "self_referential"}
+    assert cfg["outputValidation"]["llmValidator"]["externalChannels"] == [
+        "twitter", "linkedin", "email"]
+    assert cfg["approval2fa"] is None  # needs enabled + totpSecret
+    assert not cfg["erc8004"]["enabled"]
+
+
+def test_resolve_governance_config_overrides_and_gates():
+    from vainplex_openclaw_amd.governance.config import resolve_config
+
+    cfg = resolve_config({
+        "failMode": "closed",
+        "trust": {"defaults": {"boss": 80, "bad": "nope"},
+                  "decay": {"rate": 0.9}},
+        "audit": {"level": "bogus", "retentionDays": 7},
+        "approval2fa": {"enabled": True, "totpSecret": "JBSWY3DP"},
+        "agentFirewall": {"erc8004": {"enabled": True, "agentMapping": {"main": 3}}},
+        "builtinPolicies": {"nightMode": True},
+    })
+    assert cfg["failMode"] == "closed"
+    assert cfg["trust"]["defaults"] == {"boss": 80}  # non-numeric dropped
+    assert cfg["trust"]["decay"]["rate"] == 0.9
+    assert cfg["trust"]["decay"]["inactivityDays"] == 30  # default kept
+    assert cfg["audit"]["level"] == "standard"  # invalid -> default
+    assert cfg["audit"]["retentionDays"] == 7
+    assert cfg["approval2fa"]["totpSecret"] == "JBSWY3DP"
+    assert cfg["approval2fa"]["sessionApprovalMinutes"] == 10
+    assert cfg["erc8004"]["enabled"] and cfg["erc8004"]["agentMapping"] == {"main": 3}
+    assert cfg["builtinPolicies"]["nightMode"] is True
+    # 2fa without secret stays None
+    assert resolve_config({"approval2fa": {"enabled": True}})["approval2fa"] is None
