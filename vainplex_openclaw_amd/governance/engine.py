@@ -53,8 +53,11 @@ class GovernanceEngine:
         self.trust_manager = TrustManager(
             TrustConfig.from_dict(self.config.get("trust")), workspace, self.logger, clock=clock
         )
+        # sessionTrust nests under trust (config.ts resolveTrust); accept
+        # the top-level spelling as a fallback
+        st_cfg = (self.config.get("trust") or {}).get("sessionTrust") or self.config.get("sessionTrust")
         self.session_trust = SessionTrustManager(
-            SessionTrustConfig.from_dict(self.config.get("sessionTrust")), self.trust_manager, clock=clock
+            SessionTrustConfig.from_dict(st_cfg), self.trust_manager, clock=clock
         )
         self.cross_agent = CrossAgentManager(self.trust_manager, clock=clock)
         self.frequency = FrequencyTracker(clock=clock)

@@ -62,7 +62,9 @@ class GovernancePlugin:
         self.approval: Optional[Approval2FA] = None
 
     def register(self, api: PluginApi) -> None:
-        config = load_plugin_config(self.id, fallback=api.plugin_config)
+        from .config import resolve_config
+
+        config = resolve_config(load_plugin_config(self.id, fallback=api.plugin_config))
         workspace = self.workspace or config.get("workspace") or "."
         engine = GovernanceEngine(config, workspace, api.logger)
         engine.set_known_agents(extract_agent_ids(api.config))

@@ -88,17 +88,25 @@ class TrustConfig:
 
     @classmethod
     def from_dict(cls, d: Optional[Dict[str, Any]]) -> "TrustConfig":
+        """Accepts both the resolved reference shape (config.ts resolveTrust:
+        defaults / decay.{inactivityDays,rate} / persistIntervalSeconds /
+        maxHistoryPerAgent) and the flat internal spelling."""
         d = d or {}
+        defaults = d.get("initialScores") or d.get("defaults") or {}
+        default_score = d.get("defaultScore")
+        if default_score is None:
+            default_score = defaults.get("*", 40)
+        decay = d.get("decay") if isinstance(d.get("decay"), dict) else {}
         return cls(
-            default_score=float(d.get("defaultScore", 40)),
-            initial_scores=dict(d.get("initialScores", {})),
-            weights=dict(d.get("weights", {})),
-            decay_after_days=float(d.get("decayAfterDays", 30)),
-            decay_factor=float(d.get("decayFactor", 0.95)),
-            tier_locks=dict(d.get("tierLocks", {})),
-            tier_floors=dict(d.get("tierFloors", {})),
-            history_limit=int(d.get("historyLimit", 50)),
-            flush_interval=float(d.get("flushIntervalSeconds", 5)),
+            default_score=float(default_score),
+            initial_scores={k: float(v) for k, v in defaults.items() if k != "*"},
+            weights=dict(d.get("weights") or {}),
+            decay_after_days=float(d.get("decayAfterDays", decay.get("inactivityDays", 30))),
+            decay_factor=float(d.get("decayFactor", decay.get("rate", 0.95))),
+            tier_locks=dict(d.get("tierLocks") or {}),
+            tier_floors=dict(d.get("tierFloors") or {}),
+            history_limit=int(d.get("historyLimit", d.get("maxHistoryPerAgent", 50))),
+            flush_interval=float(d.get("flushIntervalSeconds", d.get("persistIntervalSeconds", 5))),
         )
 
 
