@@ -338,3 +338,45 @@ def test_hook_manager_ignores_blank_and_disabled(tmp_path):
     assert hm.entities == {}
     hm.on_message({"content": "   "})
     assert hm.entities == {}
+
+
+def test_http_client_post_loopback():
+    """http-client.ts parity: JSON POST round trip against a loopback
+    server (no egress)."""
+    import http.server
+    import threading
+
+    from vainplex_openclaw_amd.knowledge.http_client import HttpError, http_post
+
+    received = {}
+
+    class H(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            n = int(self.headers.get("Content-Length", 0))
+            received["body"] = json.loads(self.rfile.read(n))
+            received["ct"] = self.headers.get("Content-Type")
+            if self.path == "/fail":
+                self.send_response(500)
+                self.end_headers()
+                self.wfile.write(b"boom")
+                return
+            self.send_response(200)
+            self.end_headers()
+            self.wfile.write(b'{"ok": true}')
+
+        def log_message(self, *a):
+            pass
+
+    srv = http.server.HTTPServer(("127.0.0.1", 0), H)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        port = srv.server_address[1]
+        out = http_post(f"http://127.0.0.1:{port}/x", {"a": 1})
+        assert json.loads(out) == {"ok": True}
+        assert received["body"] == {"a": 1} and received["ct"] == "application/json"
+        with pytest.raises(HttpError) as ei:
+            http_post(f"http://127.0.0.1:{port}/fail", {})
+        assert ei.value.status == 500
+    finally:
+        srv.shutdown()
