@@ -49,6 +49,9 @@ def main() -> int:
     ap.add_argument("--recall-mode", default="threshold",
                     choices=["threshold", "two_stage", "direct"])
     ap.add_argument("--no-fp8", action="store_true")
+    ap.add_argument("--no-audit-sink", action="store_true",
+                    help="disable the async audit writer + event journal (on by default: "
+                    "the timed path includes audit record D2H + JSONL manifest + journal envelopes)")
     ap.add_argument("--profile", action="store_true",
                     help="after the timed region, run 3 instrumented steps and print per-stage ms (stderr)")
     args = ap.parse_args()
@@ -73,6 +76,32 @@ def main() -> int:
     pipe = FirewallPipeline(cfg, device=device, world_size=world, rank=rank)
     torch.cuda.synchronize()
     setup_s = time.time() - t0
+
+    # BASELINE config #5: the full pipeline INCLUDING the event-store leg —
+    # async audit writer (binary records + Merkle manifest JSONL) and a
+    # per-batch envelope into the embedded journal
+    writer = None
+    journal = None
+    if not args.no_audit_sink:
+        import tempfile
+
+        from vainplex_openclaw_amd.eventstore import EventJournal
+        from vainplex_openclaw_amd.pipeline.engine import AsyncAuditWriter
+
+        audit_dir = tempfile.mkdtemp(prefix=f"bench-audit-r{rank}-")
+        writer = AsyncAuditWriter(audit_dir)
+        journal = EventJournal(durable=False)
+
+        def sink(records, root):
+            writer(records, root)
+            journal.publish(
+                f"openclaw.events.rank{rank}.batch",
+                {"ts": int(time.time() * 1000), "type": "batch.audited",
+                 "agent": f"rank{rank}", "session": "bench",
+                 "payload": {"count": int(records.shape[0])}},
+            )
+
+        pipe.audit_sink = sink
 
     # pre-generated, pre-staged synthetic batch pool (rotated; every step
     # still runs the full pipeline on real message bytes)
@@ -148,6 +177,13 @@ def main() -> int:
             print("stage ms/step: " + json.dumps(
                 {k: round(v, 3) for k, v in pipe.profiler.summary().items()}
             ), file=sys.stderr)
+
+    if writer is not None:
+        writer.close()
+        if rank == 0:
+            replayed = sum(1 for _ in journal.replay())
+            print(json.dumps({"audit_batches_written": writer.batches_written,
+                              "journal_events_replayed": replayed}), file=sys.stderr)
 
     if distributed:
         torch.distributed.destroy_process_group()
