@@ -175,3 +175,25 @@ def test_interval_flusher_dirty_flag():
     f.mark_dirty()
     f.stop()  # stop flushes
     assert calls == [1, 1]
+
+
+def test_async_audit_writer_cpu(tmp_path):
+    """AsyncAuditWriter: binary records + Merkle manifest JSONL
+    (device-agnostic; GPU bench attaches it as the audit sink)."""
+    import json as _json
+    import torch
+
+    from vainplex_openclaw_amd.pipeline.engine import AsyncAuditWriter
+
+    w = AsyncAuditWriter(str(tmp_path))
+    recs = torch.arange(128, dtype=torch.uint8).reshape(2, 64)
+    root = torch.full((32,), 7, dtype=torch.uint8)
+    w(recs, root)
+    w(recs, root)
+    w.close()
+    assert w.batches_written == 2
+    assert (tmp_path / "audit-records.bin").stat().st_size == 256
+    lines = (tmp_path / "audit-manifest.jsonl").read_text().strip().split("\n")
+    assert len(lines) == 2
+    m = _json.loads(lines[0])
+    assert m["count"] == 2 and m["root"] == "07" * 32

@@ -310,7 +310,8 @@ class AsyncAuditWriter:
                 self._q = []
             with open(bin_path, "ab") as bf, open(manifest, "a") as mf:
                 for rec, root, ts in items:
-                    torch.cuda.synchronize()
+                    if torch.cuda.is_available():
+                        torch.cuda.synchronize()  # non_blocking D2H complete
                     raw = rec.numpy().tobytes()
                     bf.write(raw)
                     mf.write(
