@@ -53,6 +53,23 @@ def test_matches_thread_overlap():
     assert not matches_thread(t, "completely unrelated text")
 
 
+def test_matches_thread_reference_cases():
+    """Mirrors reference thread-tracker.test.ts matching block."""
+    t = {"title": "database migration plan"}
+    # 2+ title words in text -> match; 1 word -> no match
+    assert matches_thread(t, "the database migration starts tomorrow")
+    assert not matches_thread(t, "the database is slow")
+    # case-insensitive
+    assert matches_thread(t, "DATABASE MIGRATION done")
+    # words <3 chars ignored on both sides
+    assert not matches_thread({"title": "go to it"}, "go to it now yes")
+    # custom min_overlap
+    assert matches_thread(t, "the database is slow", min_overlap=1)
+    # degenerate inputs never match
+    assert not matches_thread({"title": ""}, "anything at all")
+    assert not matches_thread(t, "")
+
+
 def test_thread_tracker_lifecycle(workspace):
     tt = ThreadTracker(workspace)
     tt.process_message("Let's talk about the payment gateway integration", "user")
