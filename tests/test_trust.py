@@ -139,3 +139,144 @@ def test_agent_id_resolution():
     assert resolve_agent_id({"sessionKey": "agent:main"}) == "main"
     assert resolve_agent_id({}) == "unresolved"
     assert resolve_agent_id({}, {"metadata": {"agentId": "meta"}}) == "meta"
+
+
+# -- reference trust-manager.test.ts mirrors ------------------------------
+
+def _write_store(workspace, agents):
+    import datetime
+    path = os.path.join(workspace, "governance", "trust.json")
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    with open(path, "w") as fh:
+        json.dump({"version": 1, "updated": "2020-01-01T00:00:00Z", "agents": agents}, fh)
+    return path
+
+
+def _agent_record(aid, score, ts, signals=None, **extra):
+    rec = {
+        "agentId": aid, "score": score,
+        "tier": score_to_tier(score),
+        "signals": {"successCount": 0, "violationCount": 0, "ageDays": 0,
+                    "cleanStreak": 0, "manualAdjustment": 0, **(signals or {})},
+        "history": [], "lastEvaluation": ts, "created": ts, "lastActivity": ts,
+    }
+    rec.update(extra)
+    return rec
+
+
+def test_set_score_manually(workspace):
+    tm = TrustManager(TrustConfig(default_score=10), workspace)
+    tm.set_score("test", 75)
+    a = tm.get("test")
+    assert a["score"] == 75
+    assert a["tier"] == "trusted"
+    # survives a recalculate
+    tm.record_success("test")
+    assert tm.score("test") == pytest.approx(75.1)
+
+
+def test_set_score_maps_all_tier_ranges(workspace):
+    tm = TrustManager(TrustConfig(), workspace)
+    for score, tier in [(5, "untrusted"), (25, "restricted"), (45, "standard"),
+                        (65, "trusted"), (85, "elevated")]:
+        tm.set_score(f"t{score}", score)
+        assert tm.tier(f"t{score}") == tier
+
+
+def test_runtime_lock_unlock_tier(workspace):
+    tm = TrustManager(TrustConfig(default_score=10), workspace)
+    tm.lock_tier("test", "elevated")
+    a = tm.get("test")
+    assert a["tier"] == "elevated" and a["locked"] == "elevated"
+    tm.record_success("test")          # recalc keeps the locked tier
+    assert tm.tier("test") == "elevated"
+    tm.unlock_tier("test")
+    assert "locked" not in tm.get("test")
+    assert tm.tier("test") == score_to_tier(tm.score("test"))
+
+
+def test_set_floor_binds_now_and_on_decay(workspace):
+    tm = TrustManager(TrustConfig(default_score=10), workspace)
+    tm.set_floor("test", 30)
+    a = tm.get("test")
+    assert a["floor"] == 30 and a["score"] == 30
+
+    # floor respected while decaying (reference: 50*0.95=47.5 -> floor 48)
+    stale = "2020-01-01T00:00:00Z"
+    _write_store(workspace, {"floored": _agent_record("floored", 50, stale, floor=48)})
+    t = [1_900_000_000.0]
+    tm2 = TrustManager(TrustConfig(), workspace, clock=lambda: t[0])
+    tm2.load()
+    assert tm2.get("floored")["score"] == 48
+
+
+def test_history_trim_and_reset(workspace):
+    tm = TrustManager(TrustConfig(history_limit=5), workspace)
+    for _ in range(10):
+        tm.record_success("test")
+    assert len(tm.get("test")["history"]) <= 5
+    tm.reset_history("test")
+    assert tm.get("test")["history"] == []
+    assert tm.get("test")["signals"]["successCount"] == 0
+
+
+def test_get_store(workspace):
+    tm = TrustManager(TrustConfig(), workspace)
+    tm.get("main")
+    store = tm.get_store()
+    assert store["version"] == 1 and "main" in store["agents"]
+
+
+def test_unknown_agent_removed_on_load_with_warning(workspace):
+    from vainplex_openclaw_amd.core.api import PluginLogger
+
+    _write_store(workspace, {
+        "unknown": _agent_record("unknown", 20, "2020-01-01T00:00:00Z",
+                                 signals={"successCount": 340, "violationCount": 32}),
+        "main": _agent_record("main", 60, "2020-01-01T00:00:00Z"),
+    })
+    warnings = []
+    logger = PluginLogger("t", lambda lvl, msg: warnings.append(msg) if lvl == "warn" else None)
+    tm = TrustManager(TrustConfig(), workspace, logger=logger)
+    tm.load()
+    assert "unknown" not in tm.get_store()["agents"]
+    assert "main" in tm.get_store()["agents"]
+    assert any("Trust migration" in w for w in warnings)
+    assert any("340 successes" in w for w in warnings)
+
+
+def test_fresh_agent_migration_on_load(workspace):
+    # fresh agent (no activity, manualAdjustment=0) gets its score
+    # backfilled into manualAdjustment; active agents are left alone
+    _write_store(workspace, {
+        "fresh": _agent_record("fresh", 60, "2020-01-01T00:00:00Z"),
+        "active": _agent_record("active", 55, "2020-01-01T00:00:00Z",
+                                signals={"successCount": 100, "manualAdjustment": 45}),
+    })
+    t = [1_577_836_800.0 + 3600]  # 1h after the store timestamp: no decay
+    tm = TrustManager(TrustConfig(), workspace, clock=lambda: t[0])
+    tm.load()
+    assert tm.get_store()["agents"]["fresh"]["signals"]["manualAdjustment"] == 60
+    assert tm.get_store()["agents"]["active"]["signals"]["manualAdjustment"] == 45
+    # and the migrated score survives a recalculate
+    tm.record_success("fresh")
+    assert tm.score("fresh") == pytest.approx(60.1, abs=0.2)
+
+
+def test_wildcard_default_for_unknown_agents(workspace):
+    cfg = TrustConfig.from_dict({"initialScores": {"*": 33, "vip": 80}})
+    tm = TrustManager(cfg, workspace)
+    assert tm.score("random-agent") == 33
+    assert tm.score("vip") == 80
+
+
+def test_age_days_refreshed_on_load(workspace):
+    _write_store(workspace, {
+        "aged": _agent_record("aged", 40, "2020-01-01T00:00:00Z",
+                              signals={"manualAdjustment": 40}),
+    })
+    # 2020-01-01 + 10 days
+    t = [1_577_836_800.0 + 10 * 86400]
+    tm = TrustManager(TrustConfig(), workspace, clock=lambda: t[0])
+    tm.load()
+    assert tm.get_store()["agents"]["aged"]["signals"]["ageDays"] == 10

@@ -136,9 +136,24 @@ class TrustManager:
         data = read_json(self.file_path)
         if isinstance(data, dict) and isinstance(data.get("agents"), dict):
             self.store = data
+            self._drop_unknown_agent()
             self._apply_decay()
             self._migrate_fresh_agents()
             self._refresh_age_days()
+
+    def _drop_unknown_agent(self) -> None:
+        """Bug 3 (trust-manager.ts): activity recorded before agent-id
+        resolution landed under the literal id 'unknown' — drop it on load
+        so its stats never pollute a real agent."""
+        ghost = self.store["agents"].pop("unknown", None)
+        if ghost is not None:
+            s = ghost.get("signals", {})
+            self.logger.warn(
+                "[trust] Trust migration: removing 'unknown' agent "
+                "(%s successes, %s violations) — activity predates agent-id resolution",
+                s.get("successCount", 0), s.get("violationCount", 0),
+            )
+            self.flusher.mark_dirty()
 
     def start(self) -> None:
         self.load()
@@ -242,8 +257,12 @@ class TrustManager:
         floor = self.config.tier_floors.get(aid)
         if floor in TIER_FLOOR_SCORE:
             new = max(new, TIER_FLOOR_SCORE[floor])
+        # per-agent numeric floor / runtime tier lock (trust-manager.ts
+        # setFloor/lockTier) — the floor also binds during decay
+        if isinstance(agent.get("floor"), (int, float)):
+            new = max(new, float(agent["floor"]))
         agent["score"] = round(new, 4)
-        agent["tier"] = score_to_tier(new)
+        agent["tier"] = agent["locked"] if agent.get("locked") else score_to_tier(new)
         agent["lastEvaluation"] = self._iso()
         agent["lastActivity"] = agent["lastEvaluation"]
         self._record_event(agent, kind, reason, new - old)
@@ -269,6 +288,42 @@ class TrustManager:
         a = self.get(agent_id)
         a["signals"]["manualAdjustment"] = a["signals"].get("manualAdjustment", 0) + delta
         self._recalculate(a, "manual", reason)
+
+    def set_score(self, agent_id: str, score: float, reason: str = "manual set") -> None:
+        """Pin the score exactly (trust-manager.ts setScore): shifts
+        manualAdjustment so the formula lands on `score`."""
+        a = self.get(agent_id)
+        target = clamp(score, 0, 100)
+        current = compute_score(a["signals"], self.config.weights)
+        a["signals"]["manualAdjustment"] = a["signals"].get("manualAdjustment", 0) + (target - current)
+        self._recalculate(a, "manual", reason)
+
+    def lock_tier(self, agent_id: str, tier: str) -> None:
+        a = self.get(agent_id)
+        a["locked"] = tier
+        a["tier"] = tier
+        self.flusher.mark_dirty()
+
+    def unlock_tier(self, agent_id: str) -> None:
+        a = self.get(agent_id)
+        a.pop("locked", None)
+        a["tier"] = score_to_tier(a.get("score", 0))
+        self.flusher.mark_dirty()
+
+    def set_floor(self, agent_id: str, floor: float) -> None:
+        """Numeric per-agent floor; binds immediately and during decay."""
+        a = self.get(agent_id)
+        a["floor"] = float(floor)
+        self._recalculate(a, "manual", f"floor set to {floor}")
+
+    def reset_history(self, agent_id: str) -> None:
+        a = self.get(agent_id)
+        a["history"] = []
+        a["signals"]["successCount"] = 0
+        self.flusher.mark_dirty()
+
+    def get_store(self) -> Dict[str, Any]:
+        return self.store
 
     def snapshot(self) -> Dict[str, Any]:
         return {
