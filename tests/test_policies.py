@@ -229,3 +229,66 @@ def test_policy_index_by_hook_and_agent():
     assert any(p["id"] == "builtin-credential-guard" for p in idx.for_hook("before_tool_call"))
     assert idx.by_id("builtin-rate-limiter") is not None
     assert "builtin-credential-guard" in [p["id"] for p in idx.by_agent["*"]]
+
+
+# -- policy-evaluator.test.ts mirrors ------------------------------------
+
+def test_channel_scope():
+    ev = PolicyEvaluator()
+    policy = {"id": "p", "scope": {"channels": ["slack"]},
+              "rules": [{"id": "r", "conditions": [], "effect": {"action": "deny"}}]}
+    assert ev.evaluate(ctx_base(channel="slack"), [policy], {"level": "low", "score": 0})["action"] == "deny"
+    assert ev.evaluate(ctx_base(channel="email"), [policy], {"level": "low", "score": 0})["action"] == "allow"
+    # no channel in ctx -> scoped policy skipped
+    assert ev.evaluate(ctx_base(), [policy], {"level": "low", "score": 0})["action"] == "allow"
+
+
+def test_max_trust_rule_guard():
+    ev = PolicyEvaluator()
+    policy = {"id": "p", "priority": 1, "scope": {}, "rules": [
+        {"id": "r", "maxTrust": "standard",
+         "conditions": [{"type": "tool", "name": "exec"}], "effect": {"action": "deny"}}]}
+    low = ctx_base()  # session untrusted <= standard: rule applies
+    assert ev.evaluate(low, [policy], {"level": "low", "score": 0})["action"] == "deny"
+    high = ctx_base(trust={"agent": {"score": 70, "tier": "trusted"},
+                           "session": {"score": 65, "tier": "trusted"}})
+    assert ev.evaluate(high, [policy], {"level": "low", "score": 0})["action"] == "allow"
+
+
+def test_first_match_within_policy():
+    ev = PolicyEvaluator()
+    policy = {"id": "p", "scope": {}, "rules": [
+        {"id": "r1", "conditions": [{"type": "tool", "name": "exec"}], "effect": {"action": "audit"}},
+        {"id": "r2", "conditions": [], "effect": {"action": "deny"}},
+    ]}
+    res = ev.evaluate(ctx_base(), [policy], {"level": "low", "score": 0})
+    # r1 matched first -> audit, r2 never consulted
+    assert res["action"] == "allow"
+    assert res["matches"][0]["ruleId"] == "r1"
+    # different tool -> falls through to r2
+    res2 = ev.evaluate(ctx_base(toolName="write"), [policy], {"level": "low", "score": 0})
+    assert res2["action"] == "deny" and res2["matches"][0]["ruleId"] == "r2"
+
+
+def test_deny_wins_across_policies():
+    ev = PolicyEvaluator()
+    allow_p = {"id": "a", "priority": 100, "scope": {},
+               "rules": [{"id": "r", "conditions": [], "effect": {"action": "allow"}}]}
+    deny_p = {"id": "d", "priority": 1, "scope": {},
+              "rules": [{"id": "r", "conditions": [], "effect": {"action": "deny",
+                         "reason": "blocked"}}]}
+    res = ev.evaluate(ctx_base(), [allow_p, deny_p], {"level": "low", "score": 0})
+    assert res["action"] == "deny" and res["reason"] == "blocked"
+
+
+def test_controls_propagate_into_matches():
+    """Bug 4: ISO/SOC controls on the policy ride along on each match."""
+    ev = PolicyEvaluator()
+    with_controls = {"id": "p", "scope": {}, "controls": ["SOC2-CC6.1", "A.5.15"],
+                     "rules": [{"id": "r", "conditions": [], "effect": {"action": "audit"}}]}
+    res = ev.evaluate(ctx_base(), [with_controls], {"level": "low", "score": 0})
+    assert res["matches"][0]["controls"] == ["SOC2-CC6.1", "A.5.15"]
+    without = {"id": "p2", "scope": {},
+               "rules": [{"id": "r", "conditions": [], "effect": {"action": "audit"}}]}
+    res2 = ev.evaluate(ctx_base(), [without], {"level": "low", "score": 0})
+    assert res2["matches"][0]["controls"] == []
