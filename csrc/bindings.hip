@@ -30,6 +30,9 @@ __global__ void topk_recall_kernel(const __bf16*, const __bf16*, int, int, int, 
 __global__ void topk_recall_fp8_kernel(const uint8_t*, const uint8_t*, int, int, int,
                                        int, int, float*, int32_t*, const float*,
                                        int32_t*, int);
+__global__ void topk_scan_mx_kernel(const uint8_t*, const uint8_t*, int, int, int,
+                                    int, int, float*, int32_t*, const float*,
+                                    int32_t*, int);
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
                                   float*, int32_t*);
 __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigned long long*,
@@ -286,7 +289,7 @@ torch::Tensor audit_pack(torch::Tensor verdict, torch::Tensor risk, torch::Tenso
 
 std::vector<torch::Tensor> topk_scan_threshold(torch::Tensor Q, torch::Tensor X,
                                                torch::Tensor theta, int64_t cap,
-                                               int64_t n_swaths, bool fp8) {
+                                               int64_t n_swaths, bool fp8, bool mx) {
   // threshold-scan mode: append every score > theta[q] to a per-query
   // candidate buffer (no top-k maintenance in-kernel)
   CHECK_GPU(Q); CHECK_CONTIG(Q); CHECK_GPU(X); CHECK_CONTIG(X);
@@ -310,7 +313,16 @@ std::vector<torch::Tensor> topk_scan_threshold(torch::Tensor Q, torch::Tensor X,
   auto cand_i = torch::full({(long long)nq, cap}, -1, i32opts);
   auto counts = torch::zeros({(long long)nq}, i32opts);
   dim3 grid((unsigned)(n_qblocks * n_swaths));
-  if (fp8) {
+  if (mx) {
+    // MX-scaled x128 scan: same e4m3 bytes, unit block scales
+    TORCH_CHECK(Q.dtype() == torch::kUInt8 && X.dtype() == torch::kUInt8);
+    TORCH_CHECK(D % 128 == 0, "MX scan needs D % 128 == 0");
+    hipLaunchKernelGGL(topk_scan_mx_kernel, grid, dim3(512), 0, cur_stream(),
+                       Q.data_ptr<uint8_t>(), X.data_ptr<uint8_t>(), nq, (int)nx,
+                       D, 1, (int)n_swaths, cand_s.data_ptr<float>(),
+                       cand_i.data_ptr<int32_t>(), theta.data_ptr<float>(),
+                       counts.data_ptr<int32_t>(), (int)cap);
+  } else if (fp8) {
     TORCH_CHECK(Q.dtype() == torch::kUInt8 && X.dtype() == torch::kUInt8);
     hipLaunchKernelGGL(topk_recall_fp8_kernel, grid, dim3(512), 0, cur_stream(),
                        Q.data_ptr<uint8_t>(), X.data_ptr<uint8_t>(), nq, (int)nx,
@@ -338,7 +350,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("bias") = c10::nullopt,
         py::arg("act") = 0, py::arg("out_bf16") = false);
   m.def("topk_recall", &topk_recall, "Fused cosine top-k recall");
-  m.def("topk_scan_only", [](torch::Tensor Q, torch::Tensor X, int64_t n_swaths, bool fp8) {
+  m.def("topk_scan_only", [](torch::Tensor Q, torch::Tensor X, int64_t n_swaths, bool fp8,
+                             bool mx) {
     // perf diagnosis: run the scan loop with the top-k phase skipped
     CHECK_GPU(Q); CHECK_GPU(X);
     int nq = Q.size(0), D = Q.size(1);
@@ -349,7 +362,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     auto cand_s = torch::zeros({(long long)n_qblocks * n_swaths * 256}, f32opts);
     auto cand_i = torch::zeros({1}, i32opts);
     dim3 grid((unsigned)(n_qblocks * n_swaths));
-    if (fp8) {
+    if (mx) {
+      hipLaunchKernelGGL(topk_scan_mx_kernel, grid, dim3(512), 0, cur_stream(),
+                         Q.data_ptr<uint8_t>(), X.data_ptr<uint8_t>(), nq, (int)nx,
+                         D, -1, (int)n_swaths, cand_s.data_ptr<float>(),
+                         cand_i.data_ptr<int32_t>(), nullptr, nullptr, 0);
+    } else if (fp8) {
       hipLaunchKernelGGL(topk_recall_fp8_kernel, grid, dim3(512), 0, cur_stream(),
                          Q.data_ptr<uint8_t>(), X.data_ptr<uint8_t>(), nq, (int)nx,
                          D, -1, (int)n_swaths, cand_s.data_ptr<float>(),
@@ -361,10 +379,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                          D, -1, (int)n_swaths, cand_s.data_ptr<float>(),
                          cand_i.data_ptr<int32_t>(), nullptr, nullptr, 0);
     }
-  }, "scan-only diagnosis");
+  }, "scan-only diagnosis", py::arg("Q"), py::arg("X"), py::arg("n_swaths"),
+     py::arg("fp8") = false, py::arg("mx") = false);
   m.def("topk_recall_fp8", &topk_recall_fp8, "fp8 stage-1 scan of two-stage recall");
   m.def("topk_scan_threshold", &topk_scan_threshold,
-        "threshold-scan candidate collection (bf16 or fp8)");
+        "threshold-scan candidate collection (bf16, fp8 or MX-fp8)",
+        py::arg("Q"), py::arg("X"), py::arg("theta"), py::arg("cap"),
+        py::arg("n_swaths"), py::arg("fp8") = false, py::arg("mx") = false);
   m.def("firewall_verdict", &firewall_verdict, "Fused verdict/risk/trust-delta");
   m.def("trust_recompute", &trust_recompute, "Agent trust score recompute");
   m.def("audit_pack", &audit_pack, "Pack 64-byte audit records");

@@ -679,3 +679,193 @@ topk_recall_fp8_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict_
     for (int i = threadIdx.x; i < BM * k; i += blockDim.x)
       cand_scores[cbase + (size_t)(i / k) * k + (i % k)] = topk_vals[i / k][i % k];
 }
+
+// ===========================================================================
+// MX-scaled threshold scan: the fp8 scan rebuilt on
+// v_mfma_scale_f32_16x16x128_f8f6f4 with unit e8m0 scales (0x7F = x1), so
+// it consumes the SAME pre-scaled e4m3 index bytes. K=128 per issue is
+// 2.25x the non-scaled fp8 MFMA rate (4661 vs 2075 TF/s peak), and each
+// staged byte feeds exactly one ds_read instead of two — the fp8 scan
+// measured at 67% of its MFMA issue peak, so issue pressure, not bytes,
+// was the next wall.
+//
+// Staging is unchanged (64 B rows / 4 swizzled 16 B slots, stage_tile8,
+// counted-vmcnt GLDS pipeline) but tiles are consumed in PAIRS: a lane's
+// 32 B operand (k = kgrp*32 + [0,32)) is source slots {s0, s0+1} of tile
+// (pair, kgrp>>1) with s0 = (kgrp&1)*2. The swizzle term ((row>>2)&3) is
+// lrow-only (fragment row deltas are multiples of 16), so each matrix
+// needs two base addresses (lo/hi slot) and fragments sit at +1024 asm
+// offsets, exactly like the bf16 kernel's read blocks. NBUF_MX=4 x
+// (BM+BN) x 64 B = 128 KB LDS (threshold mode carries no top-k state);
+// prefetch is one pair (= two tiles, 8 per-lane glds) deep:
+// s_waitcnt vmcnt(8). Threshold mode + the k<0 scan-only diagnostic.
+// ===========================================================================
+
+#define NBUF_MX 4
+typedef int v8i_mx __attribute__((ext_vector_type(8)));
+
+DEVINL v8i_mx mx_frag(bf16x8 lo, bf16x8 hi) {
+  int4 l = __builtin_bit_cast(int4, lo), h = __builtin_bit_cast(int4, hi);
+  v8i_mx f;
+  f[0] = l.x; f[1] = l.y; f[2] = l.z; f[3] = l.w;
+  f[4] = h.x; f[5] = h.y; f[6] = h.z; f[7] = h.w;
+  return f;
+}
+
+extern "C" __global__ void __launch_bounds__(TK_THREADS)
+topk_scan_mx_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X,
+                    int nq, int nx, int D, int k, int n_swaths,
+                    float* __restrict__ cand_scores,
+                    int32_t* __restrict__ cand_ids,
+                    const float* __restrict__ theta,
+                    int32_t* __restrict__ tc_n, int cap) {
+  __shared__ bf16 lds_mx[NBUF_MX * (BM + BN) * BK];
+#define QSM(buf) (lds_mx + (buf) * BM * BK)
+#define XSM(buf) (lds_mx + NBUF_MX * BM * BK + (buf) * BN * BK)
+  __shared__ float row_min[BM];
+
+  int S = n_swaths;
+  int qb = blockIdx.x / S;
+  int swath = blockIdx.x % S;
+  long long row0 = (long long)qb * BM;
+
+  long long per = ((long long)nx + S - 1) / S;
+  per = ((per + BN - 1) / BN) * BN;
+  long long x_begin = (long long)swath * per;
+  long long x_end = min((long long)nx, x_begin + per);
+
+  for (int i = threadIdx.x; i < BM; i += blockDim.x)
+    row_min[i] = (theta != nullptr && row0 + i < nq) ? theta[row0 + i] : -1e30f;
+  __syncthreads();
+
+  int wid = wave_id();
+  int wm = wid >> 2, wn = wid & 3;
+  int lane = lane_id();
+  int lrow = lane & 15;
+  int kgrp = lane >> 4;             // k-block (of 32) within the 128 pair
+  int th = kgrp >> 1;               // which tile of the pair holds it
+  int s0 = (kgrp & 1) * 2;          // source slot pair base within that tile
+  int np = D / (2 * BK_F8);         // K pairs of 128
+  size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
+  const int unit_scale = 0x7F7F7F7F;  // e8m0 biased-127 exponent = x1.0
+
+  for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
+    f32x4 acc[8][4] = {};
+    // prologue: two PAIRS (four tiles) in flight = 16 per-lane glds
+    for (int t = 0; t < 4 && t < 2 * np; ++t) {
+      stage_tile8(Q, D, row0, nq, t * BK_F8, QSM(t & 3), BM);
+      stage_tile8(X, D, x0, (long long)nx, t * BK_F8, XSM(t & 3), BN);
+    }
+    for (int p = 0; p < np; ++p) {
+      // pair p landed when at most pair p+1's 8 glds are outstanding;
+      // on the final pair nothing else is in flight, so count to zero
+      if (p + 1 < np)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (2 * p + 4 < 2 * np) {
+        stage_tile8(Q, D, row0, nq, (2 * p + 4) * BK_F8, QSM((2 * p + 4) & 3), BM);
+        stage_tile8(X, D, x0, (long long)nx, (2 * p + 4) * BK_F8, XSM((2 * p + 4) & 3), BN);
+        stage_tile8(Q, D, row0, nq, (2 * p + 5) * BK_F8, QSM((2 * p + 5) & 3), BM);
+        stage_tile8(X, D, x0, (long long)nx, (2 * p + 5) * BK_F8, XSM((2 * p + 5) & 3), BN);
+      }
+      int buf = (2 * p + th) & 3;
+      uint32_t r3 = ((uint32_t)lrow >> 2u) & 3u;
+      uint32_t off_lo = ((uint32_t)s0 ^ r3) * 16u;
+      uint32_t off_hi = (((uint32_t)s0 + 1u) ^ r3) * 16u;
+      uint32_t xrow = (uint32_t)(wn * 64 + lrow) * 64u;
+      uint32_t qrow = (uint32_t)(wm * 128 + lrow) * 64u;
+      uint32_t xlo = (uint32_t)(size_t)XSM(buf) + xrow + off_lo;
+      uint32_t xhi = (uint32_t)(size_t)XSM(buf) + xrow + off_hi;
+      uint32_t qlo = (uint32_t)(size_t)QSM(buf) + qrow + off_lo;
+      uint32_t qhi = (uint32_t)(size_t)QSM(buf) + qrow + off_hi;
+      bf16x8 xl[4], xh[4], ql[8], qh[8];
+      asm volatile(
+          "ds_read_b128 %0, %8\n\t"
+          "ds_read_b128 %1, %8 offset:1024\n\t"
+          "ds_read_b128 %2, %8 offset:2048\n\t"
+          "ds_read_b128 %3, %8 offset:3072\n\t"
+          "ds_read_b128 %4, %9\n\t"
+          "ds_read_b128 %5, %9 offset:1024\n\t"
+          "ds_read_b128 %6, %9 offset:2048\n\t"
+          "ds_read_b128 %7, %9 offset:3072\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(xl[0]), "=&v"(xl[1]), "=&v"(xl[2]), "=&v"(xl[3]),
+            "=&v"(xh[0]), "=&v"(xh[1]), "=&v"(xh[2]), "=&v"(xh[3])
+          : "v"(xlo), "v"(xhi));
+      asm volatile(
+          "ds_read_b128 %0, %8\n\t"
+          "ds_read_b128 %1, %8 offset:1024\n\t"
+          "ds_read_b128 %2, %8 offset:2048\n\t"
+          "ds_read_b128 %3, %8 offset:3072\n\t"
+          "ds_read_b128 %4, %9\n\t"
+          "ds_read_b128 %5, %9 offset:1024\n\t"
+          "ds_read_b128 %6, %9 offset:2048\n\t"
+          "ds_read_b128 %7, %9 offset:3072\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(ql[0]), "=&v"(ql[1]), "=&v"(ql[2]), "=&v"(ql[3]),
+            "=&v"(qh[0]), "=&v"(qh[1]), "=&v"(qh[2]), "=&v"(qh[3])
+          : "v"(qlo), "v"(qhi));
+      v8i_mx xv[4];
+#pragma unroll
+      for (int n = 0; n < 4; ++n) xv[n] = mx_frag(xl[n], xh[n]);
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        v8i_mx qv = mx_frag(ql[m], qh[m]);
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              qv, xv[n], acc[m][n], 0, 0, 0, unit_scale, 0, unit_scale);
+      }
+      asm volatile(
+          "ds_read_b128 %0, %8 offset:4096\n\t"
+          "ds_read_b128 %1, %8 offset:5120\n\t"
+          "ds_read_b128 %2, %8 offset:6144\n\t"
+          "ds_read_b128 %3, %8 offset:7168\n\t"
+          "ds_read_b128 %4, %9 offset:4096\n\t"
+          "ds_read_b128 %5, %9 offset:5120\n\t"
+          "ds_read_b128 %6, %9 offset:6144\n\t"
+          "ds_read_b128 %7, %9 offset:7168\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(ql[4]), "=&v"(ql[5]), "=&v"(ql[6]), "=&v"(ql[7]),
+            "=&v"(qh[4]), "=&v"(qh[5]), "=&v"(qh[6]), "=&v"(qh[7])
+          : "v"(qlo), "v"(qhi));
+#pragma unroll
+      for (int m = 4; m < 8; ++m) {
+        v8i_mx qv = mx_frag(ql[m], qh[m]);
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              qv, xv[n], acc[m][n], 0, 0, 0, unit_scale, 0, unit_scale);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    if (k < 0) {  // scan-only diagnosis mode
+      if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
+    // threshold appends (identical to the fp8 kernel's threshold branch)
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+          float v = acc[m][n][r];
+          if (!(v > row_min[row])) continue;
+          long long grow = row0 + row;
+          if (grow >= nq) continue;
+          long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+          if (col >= x_end) continue;
+          int pos = atomicAdd(&tc_n[grow], 1);
+          if (pos < cap) {
+            cand_scores[grow * cap + pos] = v;
+            cand_ids[grow * cap + pos] = int32_t(col);
+          }
+        }
+  }
+}

@@ -237,15 +237,44 @@ def test_topk_recall_threshold_bf16_and_fp8():
     X = torch.nn.functional.normalize(torch.randn(nx, D, device="cuda"), dim=1).bfloat16()
     ref = torch.topk(Q.float() @ X.float().T, k, dim=1)
     ref_ids = ref.indices.cpu().numpy()
-    for X8 in (None, g.to_fp8_bytes(X)):
-        scores, ids = g.topk_recall_threshold(Q, X, k, X8=X8)
+    X8 = g.to_fp8_bytes(X)
+    for label, kwargs in [("bf16", {}), ("fp8", {"X8": X8, "mx": False}),
+                          ("mx", {"X8": X8, "mx": True})]:
+        scores, ids = g.topk_recall_threshold(Q, X, k, **kwargs)
         ids_np = ids.cpu().numpy()
         vals = scores.cpu().numpy()
         for q in range(nq):
             inter = len(set(ids_np[q]) & set(ref_ids[q]))
-            assert inter >= k - 2, f"fp8={X8 is not None} q={q}: {inter}/{k}"
+            assert inter >= k - 2, f"{label} q={q}: {inter}/{k}"
             assert abs(vals[q][0] - ref.values[q, 0].item()) < 2e-2
             assert all(vals[q][i] >= vals[q][i + 1] - 1e-6 for i in range(k - 1))
+
+
+@pytest.mark.gpu
+def test_mx_scan_matches_fp8_scan():
+    """Raw MX x128 scan vs the non-scaled fp8 scan on identical inputs and
+    thresholds: same arithmetic up to summation order, so the candidate
+    sets must agree except at the exact threshold boundary."""
+    torch.manual_seed(17)
+    nq, nx, D = 512, 65536, 1024
+    Q = torch.nn.functional.normalize(torch.randn(nq, D, device="cuda"), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(nx, D, device="cuda"), dim=1).bfloat16()
+    Q8, X8 = g.to_fp8_bytes(Q), g.to_fp8_bytes(X)
+    sample = torch.matmul(Q, X[:16384].T).float()
+    theta = ((sample.mean(1) + 3.3 * sample.std(1)) * 64.0).contiguous()
+    cap = 2048
+    ext = g.ext()
+    s_f8, i_f8, n_f8 = ext.topk_scan_threshold(Q8, X8, theta, cap, 0, True, False)
+    s_mx, i_mx, n_mx = ext.topk_scan_threshold(Q8, X8, theta, cap, 0, True, True)
+    n_f8, n_mx = n_f8.cpu(), n_mx.cpu()
+    boundary_flips = 0
+    for q in range(nq):
+        a = {int(v) for v in i_f8[q, : min(int(n_f8[q]), cap)].cpu()}
+        b = {int(v) for v in i_mx[q, : min(int(n_mx[q]), cap)].cpu()}
+        boundary_flips += len(a ^ b)
+        assert len(a ^ b) <= max(4, len(a | b) // 8), f"q={q}: {len(a^b)} diffs"
+    # overall the two scans agree almost everywhere
+    assert boundary_flips <= nq  # avg <=1 flip per query
 
 
 @pytest.mark.gpu

@@ -74,6 +74,8 @@ def scan_only_bench():
     for name, fn in [
         ("scan-only bf16 S=16", lambda: g.ext().topk_scan_only(Q, X, 16, False)),
         ("scan-only fp8  S=16", lambda: g.ext().topk_scan_only(Q8, X8, 16, True)),
+        ("scan-only MX   S=16", lambda: g.ext().topk_scan_only(Q8, X8, 16, True, True)),
+        ("scan-only MX   S=8",  lambda: g.ext().topk_scan_only(Q8, X8, 8, True, True)),
         ("scan-only bf16 S=8",  lambda: g.ext().topk_scan_only(Q, X, 8, False)),
         ("scan-only bf16 S=32", lambda: g.ext().topk_scan_only(Q, X, 32, False)),
     ]:
@@ -94,7 +96,8 @@ def threshold_bench():
     X8 = g.to_fp8_bytes(X)
     for name, fn in [
         ("thresh bf16 k=16", lambda: g.topk_recall_threshold(Q, X, 16)),
-        ("thresh fp8  k=16", lambda: g.topk_recall_threshold(Q, X, 16, X8=X8)),
+        ("thresh fp8  k=16", lambda: g.topk_recall_threshold(Q, X, 16, X8=X8, mx=False)),
+        ("thresh MX   k=16", lambda: g.topk_recall_threshold(Q, X, 16, X8=X8)),
         ("thresh bf16 k=32", lambda: g.topk_recall_threshold(Q, X, 32)),
         ("direct bf16 k=16", lambda: g.topk_recall(Q, X, 16)),
     ]:

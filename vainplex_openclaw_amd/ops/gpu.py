@@ -278,6 +278,7 @@ def topk_recall_threshold(
     sample_rows: int = 131072,
     target_candidates: int = 128,
     cap: int = 1024,
+    mx: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Threshold-scan recall: per-query score thresholds estimated from a
     sampled pre-pass (Gaussian tail extrapolation), then a fixed-threshold
@@ -285,14 +286,18 @@ def topk_recall_threshold(
     maintenance, so the scan runs at its K-loop rate regardless of k.
 
     With X8 (e4m3 view) the scan runs in fp8 and the top candidates are
-    exact-rescored in bf16; final scores are exact either way. Queries
-    whose candidate buffer underflowed (<k survivors: threshold too high
-    for non-Gaussian score tails) fall back to the direct kernel.
+    exact-rescored in bf16; final scores are exact either way. `mx` (the
+    default, needs D % 128 == 0) runs the fp8 scan on the MX-scaled
+    16x16x128 MFMA at unit block scales — same bytes, 2.25x the issue
+    rate (csrc topk_scan_mx_kernel). Queries whose candidate buffer
+    underflowed (<k survivors: threshold too high for non-Gaussian score
+    tails) fall back to the direct kernel.
     """
     nq, D = Q.shape
     nx = X.shape[0]
     m = min(nx, sample_rows)
     use_fp8 = X8 is not None
+    use_mx = mx and use_fp8 and D % 128 == 0
 
     # 1. per-query score statistics from a sample (bf16 matmul)
     sample = torch.matmul(Q, X[:m].T).float()  # [nq, m]
@@ -323,9 +328,9 @@ def topk_recall_threshold(
         Q8 = to_fp8_bytes(Q)
         # thresholds are in fp8-score units: inputs scaled x8 each -> x64
         cs, ci, counts = ext().topk_scan_threshold(Q8, X8, (theta * 64.0).contiguous(),
-                                                   cap, 0, True)
+                                                   cap, 0, True, use_mx)
     else:
-        cs, ci, counts = ext().topk_scan_threshold(Q, X, theta, cap, 0, False)
+        cs, ci, counts = ext().topk_scan_threshold(Q, X, theta, cap, 0, False, False)
 
     # 3. mask unfilled/overflowed slots
     slot = torch.arange(cap, device=Q.device).unsqueeze(0)
