@@ -764,12 +764,6 @@ topk_scan_mx_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
-      if (2 * p + 4 < 2 * np) {
-        stage_tile8(Q, D, row0, nq, (2 * p + 4) * BK_F8, QSM((2 * p + 4) & 3), BM);
-        stage_tile8(X, D, x0, (long long)nx, (2 * p + 4) * BK_F8, XSM((2 * p + 4) & 3), BN);
-        stage_tile8(Q, D, row0, nq, (2 * p + 5) * BK_F8, QSM((2 * p + 5) & 3), BM);
-        stage_tile8(X, D, x0, (long long)nx, (2 * p + 5) * BK_F8, XSM((2 * p + 5) & 3), BN);
-      }
       int buf = (2 * p + th) & 3;
       uint32_t r3 = ((uint32_t)lrow >> 2u) & 3u;
       uint32_t off_lo = ((uint32_t)s0 ^ r3) * 16u;
@@ -807,17 +801,6 @@ topk_scan_mx_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X
           : "=&v"(ql[0]), "=&v"(ql[1]), "=&v"(ql[2]), "=&v"(ql[3]),
             "=&v"(qh[0]), "=&v"(qh[1]), "=&v"(qh[2]), "=&v"(qh[3])
           : "v"(qlo), "v"(qhi));
-      v8i_mx xv[4];
-#pragma unroll
-      for (int n = 0; n < 4; ++n) xv[n] = mx_frag(xl[n], xh[n]);
-#pragma unroll
-      for (int m = 0; m < 4; ++m) {
-        v8i_mx qv = mx_frag(ql[m], qh[m]);
-#pragma unroll
-        for (int n = 0; n < 4; ++n)
-          acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-              qv, xv[n], acc[m][n], 0, 0, 0, unit_scale, 0, unit_scale);
-      }
       asm volatile(
           "ds_read_b128 %0, %8 offset:4096\n\t"
           "ds_read_b128 %1, %8 offset:5120\n\t"
@@ -831,8 +814,21 @@ topk_scan_mx_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X
           : "=&v"(ql[4]), "=&v"(ql[5]), "=&v"(ql[6]), "=&v"(ql[7]),
             "=&v"(qh[4]), "=&v"(qh[5]), "=&v"(qh[6]), "=&v"(qh[7])
           : "v"(qlo), "v"(qhi));
+      // every wave now holds pair p in registers; after this barrier the
+      // pair's buffers are dead and pair p+2 can stage into them while
+      // the MFMAs run (tiles t and t+4 share a buffer: NBUF_MX = 4)
+      __builtin_amdgcn_s_barrier();
+      if (2 * p + 4 < 2 * np) {
+        stage_tile8(Q, D, row0, nq, (2 * p + 4) * BK_F8, QSM((2 * p + 4) & 3), BM);
+        stage_tile8(X, D, x0, (long long)nx, (2 * p + 4) * BK_F8, XSM((2 * p + 4) & 3), BN);
+        stage_tile8(Q, D, row0, nq, (2 * p + 5) * BK_F8, QSM((2 * p + 5) & 3), BM);
+        stage_tile8(X, D, x0, (long long)nx, (2 * p + 5) * BK_F8, XSM((2 * p + 5) & 3), BN);
+      }
+      v8i_mx xv[4];
 #pragma unroll
-      for (int m = 4; m < 8; ++m) {
+      for (int n = 0; n < 4; ++n) xv[n] = mx_frag(xl[n], xh[n]);
+#pragma unroll
+      for (int m = 0; m < 8; ++m) {
         v8i_mx qv = mx_frag(ql[m], qh[m]);
 #pragma unroll
         for (int n = 0; n < 4; ++n)
