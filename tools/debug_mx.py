@@ -1,0 +1,53 @@
+"""Localize MX scan numeric errors: exact fp8-decoded reference vs the
+raw threshold-scan scores, per D (np=1 isolates the pipeline) and by
+lane-coordinate groupings (row%16, col%16, col-block) to identify which
+fragment mapping is off."""
+
+import torch
+
+from vainplex_openclaw_amd.ops import gpu as g
+
+
+def run(D, mode, nq=256, nx=256, S=8):
+    torch.manual_seed(3)
+    Q = torch.nn.functional.normalize(torch.randn(nq, D, device="cuda"), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(nx, D, device="cuda"), dim=1).bfloat16()
+    Q8, X8 = g.to_fp8_bytes(Q), g.to_fp8_bytes(X)
+    Qd = Q8.view(torch.float8_e4m3fn).float()
+    Xd = X8.view(torch.float8_e4m3fn).float()
+    ref = Qd @ Xd.T  # includes the x8 input scales -> x64 scores
+    theta = torch.full((nq,), -1e9, device="cuda")
+    cs, ci, counts = g.ext().topk_scan_threshold(Q8, X8, theta, 4096, S,
+                                                 True, mode == "mx")
+    assert int(counts.min()) == nx and int(counts.max()) == nx, counts
+    S = torch.full((nq, nx), float("nan"), device="cuda")
+    rows = torch.arange(nq, device="cuda").unsqueeze(1).expand(nq, nx)
+    S[rows.reshape(-1), ci[:, :nx].reshape(-1).long()] = cs[:, :nx].reshape(-1)
+    err = (S - ref).abs()
+    bad = err > 0.05 * ref.abs().clamp_min(1.0)
+    print(f"{mode} D={D} nq={nq} nx={nx} S={s_used(S, nq, nx)}: "
+          f"maxerr={err.max().item():.4f} badfrac={bad.float().mean().item():.3f}")
+    if bad.any():
+        # localize: error rate by row%16, by col%16, by col 16-block, row 16-block
+        for name, idx, n in [("row%16", rows % 16, 16),
+                             ("col%16", torch.arange(nx, device="cuda").unsqueeze(0).expand(nq, nx) % 16, 16),
+                             ("colblk", torch.arange(nx, device="cuda").unsqueeze(0).expand(nq, nx) // 16, 16),
+                             ("rowblk", rows // 16, 16)]:
+            rates = [bad[idx == i].float().mean().item() for i in range(n)]
+            print(f"  {name}: " + " ".join(f"{r:.2f}" for r in rates))
+
+
+def s_used(S, nq, nx):
+    return S
+
+
+if __name__ == "__main__":
+    # multi-x-tile per swath (nx=1024, S=1 -> 4 sequential tiles)
+    run(1024, "fp8", nq=256, nx=1024, S=1)
+    run(1024, "mx", nq=256, nx=1024, S=1)
+    # multiple q-blocks
+    run(1024, "mx", nq=512, nx=256, S=8)
+    # the failing test's shape with default swath heuristic
+    run(1024, "fp8", nq=512, nx=4096, S=0)
+    run(1024, "mx", nq=512, nx=4096, S=0)
+    run(1024, "mx", nq=256, nx=4096, S=16)

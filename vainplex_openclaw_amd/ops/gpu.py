@@ -342,18 +342,15 @@ def topk_recall_threshold(
     top = torch.topk(cs, sel, dim=1)
     ids = torch.gather(ci, 1, top.indices)
 
-    if use_fp8:
-        # exact bf16 rescore of the selected candidates
-        gidx = ids.long().clamp_min(0)
-        cand = X[gidx]  # [nq, sel, D]
-        exact = torch.einsum("qd,qkd->qk", Q.float(), cand.float())
-        exact = torch.where(ids < 0, torch.full_like(exact, -1e30), exact)
-        fin = torch.topk(exact, k, dim=1)
-        out_s = fin.values
-        out_i = torch.gather(ids, 1, fin.indices)
-    else:
-        out_s = top.values[:, :k]
-        out_i = ids[:, :k]
+    # exact fp32 rescore of the selected candidates (both dtypes: the
+    # bf16 scan's near-ties otherwise reorder the top-k at bf16 precision)
+    gidx = ids.long().clamp_min(0)
+    cand = X[gidx]  # [nq, sel, D]
+    exact = torch.einsum("qd,qkd->qk", Q.float(), cand.float())
+    exact = torch.where(ids < 0, torch.full_like(exact, -1e30), exact)
+    fin = torch.topk(exact, k, dim=1)
+    out_s = fin.values
+    out_i = torch.gather(ids, 1, fin.indices)
 
     # 5. fallback for underflowed queries (non-Gaussian tails / theta high)
     bad = (counts < k) | (counts > cap)
