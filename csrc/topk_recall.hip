@@ -865,3 +865,210 @@ topk_scan_mx_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X
         }
   }
 }
+
+// ===========================================================================
+// MX fp8xfp4 threshold scan: Q stays e4m3 (cbsz=0, unit scales); X is
+// MXFP4 — e2m1 nibbles (two per byte, low nibble = even k) with one
+// e8m0 scale per 32-element block (blgp=4). The scan is at the GLDS
+// transport ceiling, so halving the X bytes (64 B per 128-k row + 32 B
+// of scales per full row) buys what no MFMA-side change can.
+//
+// Geometry: Q staging identical to topk_scan_mx_kernel. X4 rows are
+// D/2 bytes; one 128-k PAIR of X is a single 64 B-row tile (4 swizzled
+// slots), so a lane's B fragment is ONE ds_read_b128 at slot kgrp —
+// the bf16 kernel's X read pattern. Scales are staged once per x-tile
+// (BN x D/32 bytes, linear rows) and read per pair as ds_read_u8.
+// Per-pair staging = 6 per-lane glds (Q 2x2 + X4 2): s_waitcnt vmcnt(6).
+// ===========================================================================
+
+#define MX4_SMAX 64  // max D/32 scale bytes per row (D <= 2048)
+
+DEVINL void stage_scale_rows(const uint8_t* __restrict__ src, int sb,
+                             long long row0, long long row_max,
+                             uint8_t* lds_base, int tile_rows) {
+  // rows of sb bytes (sb % 16 == 0), staged linearly (no swizzle: the
+  // consumers are byte reads)
+  int per_row = sb / 16;
+  int n_pieces = tile_rows * per_row;
+  int w = wave_id();
+  int lane = lane_id();
+  for (int piece0 = w * WAVE; piece0 < n_pieces; piece0 += TK_THREADS) {
+    int piece = piece0 + lane;
+    long long gr = row0 + piece / per_row;
+    if (gr >= row_max) gr = row_max - 1;
+    const uint8_t* p = src + gr * sb + (long long)(piece % per_row) * 16;
+    int piece0_u = __builtin_amdgcn_readfirstlane(piece0);
+    auto ldst = (AS3 char*)lds_base + piece0_u * 16;
+    __builtin_amdgcn_global_load_lds((const AS1 void*)p, (AS3 void*)ldst, 16, 0, 0);
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(TK_THREADS)
+topk_scan_mx4_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X4,
+                     const uint8_t* __restrict__ XS,
+                     int nq, int nx, int D, int k, int n_swaths,
+                     float* __restrict__ cand_scores,
+                     int32_t* __restrict__ cand_ids,
+                     const float* __restrict__ theta,
+                     int32_t* __restrict__ tc_n, int cap) {
+  __shared__ bf16 lds_q4[NBUF_MX * BM * BK];          // 64 KB
+  __shared__ bf16 lds_x4[2 * BN * BK];                // 32 KB (2 pair-bufs)
+  __shared__ uint8_t lds_xs[BN * MX4_SMAX];           // 16 KB scale sheet
+  __shared__ float row_min[BM];
+#define QS4(buf) (lds_q4 + (buf) * BM * BK)
+#define XS4(buf) (lds_x4 + (buf) * BN * BK)
+
+  int S = n_swaths;
+  int qb = blockIdx.x / S;
+  int swath = blockIdx.x % S;
+  long long row0 = (long long)qb * BM;
+
+  long long per = ((long long)nx + S - 1) / S;
+  per = ((per + BN - 1) / BN) * BN;
+  long long x_begin = (long long)swath * per;
+  long long x_end = min((long long)nx, x_begin + per);
+
+  for (int i = threadIdx.x; i < BM; i += blockDim.x)
+    row_min[i] = (theta != nullptr && row0 + i < nq) ? theta[row0 + i] : -1e30f;
+  __syncthreads();
+
+  int wid = wave_id();
+  int wm = wid >> 2, wn = wid & 3;
+  int lane = lane_id();
+  int lrow = lane & 15;
+  int kgrp = lane >> 4;
+  int th = kgrp >> 1;
+  int s0 = (kgrp & 1) * 2;
+  int np = D / (2 * BK_F8);
+  int sb = D / 32;                  // scale bytes per X row
+  size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
+  const int unit_scale = 0x7F7F7F7F;
+  long long x4_ld = D / 4;          // X4 row length in bf16 units (D/2 bytes)
+
+  for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
+    f32x4 acc[8][4] = {};
+    // prologue: scale sheet + two pairs in flight (1 + 2x6 glds)
+    stage_scale_rows(XS, sb, x0, (long long)nx, lds_xs, BN);
+    for (int pp = 0; pp < 2 && pp < np; ++pp) {
+      stage_tile8(Q, D, row0, nq, (2 * pp) * BK_F8, QS4((2 * pp) & 3), BM);
+      stage_tile8(Q, D, row0, nq, (2 * pp + 1) * BK_F8, QS4((2 * pp + 1) & 3), BM);
+      stage_tile((const bf16*)X4, x4_ld, x0, (long long)nx, pp * 32, XS4(pp & 1), BN);
+    }
+    for (int p = 0; p < np; ++p) {
+      if (p + 1 < np)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      int buf = (2 * p + th) & 3;
+      uint32_t r3 = ((uint32_t)lrow >> 2u) & 3u;
+      uint32_t off_lo = ((uint32_t)s0 ^ r3) * 16u;
+      uint32_t off_hi = (((uint32_t)s0 + 1u) ^ r3) * 16u;
+      uint32_t qrow = (uint32_t)(wm * 128 + lrow) * 64u;
+      uint32_t qlo = (uint32_t)(size_t)QS4(buf) + qrow + off_lo;
+      uint32_t qhi = (uint32_t)(size_t)QS4(buf) + qrow + off_hi;
+      // X fragment: one b128 at slot kgrp of the pair-row
+      uint32_t xrow_base = (uint32_t)(wn * 64 + lrow);
+      uint32_t xaddr = (uint32_t)(size_t)XS4(p & 1)
+                       + lds_off_bytes(xrow_base, (uint32_t)kgrp);
+      // per-n scale byte: lds_xs[row * sb + p*4 + kgrp]
+      uint32_t saddr = (uint32_t)(size_t)lds_xs + xrow_base * (uint32_t)sb
+                       + (uint32_t)(p * 4 + kgrp);
+      uint32_t sstride = 16u * (uint32_t)sb;
+      bf16x8 xf[4], ql[8], qh[8];
+      uint32_t sv[4];
+      asm volatile(
+          "ds_read_b128 %0, %8\n\t"
+          "ds_read_b128 %1, %8 offset:1024\n\t"
+          "ds_read_b128 %2, %8 offset:2048\n\t"
+          "ds_read_b128 %3, %8 offset:3072\n\t"
+          "ds_read_u8 %4, %9\n\t"
+          "ds_read_u8 %5, %10\n\t"
+          "ds_read_u8 %6, %11\n\t"
+          "ds_read_u8 %7, %12\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
+            "=&v"(sv[0]), "=&v"(sv[1]), "=&v"(sv[2]), "=&v"(sv[3])
+          : "v"(xaddr), "v"(saddr), "v"(saddr + sstride),
+            "v"(saddr + 2 * sstride), "v"(saddr + 3 * sstride));
+      asm volatile(
+          "ds_read_b128 %0, %8\n\t"
+          "ds_read_b128 %1, %8 offset:1024\n\t"
+          "ds_read_b128 %2, %8 offset:2048\n\t"
+          "ds_read_b128 %3, %8 offset:3072\n\t"
+          "ds_read_b128 %4, %9\n\t"
+          "ds_read_b128 %5, %9 offset:1024\n\t"
+          "ds_read_b128 %6, %9 offset:2048\n\t"
+          "ds_read_b128 %7, %9 offset:3072\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(ql[0]), "=&v"(ql[1]), "=&v"(ql[2]), "=&v"(ql[3]),
+            "=&v"(qh[0]), "=&v"(qh[1]), "=&v"(qh[2]), "=&v"(qh[3])
+          : "v"(qlo), "v"(qhi));
+      asm volatile(
+          "ds_read_b128 %0, %8 offset:4096\n\t"
+          "ds_read_b128 %1, %8 offset:5120\n\t"
+          "ds_read_b128 %2, %8 offset:6144\n\t"
+          "ds_read_b128 %3, %8 offset:7168\n\t"
+          "ds_read_b128 %4, %9 offset:4096\n\t"
+          "ds_read_b128 %5, %9 offset:5120\n\t"
+          "ds_read_b128 %6, %9 offset:6144\n\t"
+          "ds_read_b128 %7, %9 offset:7168\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(ql[4]), "=&v"(ql[5]), "=&v"(ql[6]), "=&v"(ql[7]),
+            "=&v"(qh[4]), "=&v"(qh[5]), "=&v"(qh[6]), "=&v"(qh[7])
+          : "v"(qlo), "v"(qhi));
+      // all of pair p is in registers: free its buffers for pair p+2
+      __builtin_amdgcn_s_barrier();
+      if (2 * p + 4 < 2 * np) {
+        stage_tile8(Q, D, row0, nq, (2 * p + 4) * BK_F8, QS4((2 * p + 4) & 3), BM);
+        stage_tile8(Q, D, row0, nq, (2 * p + 5) * BK_F8, QS4((2 * p + 5) & 3), BM);
+        stage_tile((const bf16*)X4, x4_ld, x0, (long long)nx, (p + 2) * 32,
+                   XS4(p & 1), BN);
+      }
+      v8i_mx xv[4];
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        int4 l = __builtin_bit_cast(int4, xf[n]);
+        v8i_mx f;
+        f[0] = l.x; f[1] = l.y; f[2] = l.z; f[3] = l.w;
+        f[4] = 0; f[5] = 0; f[6] = 0; f[7] = 0;
+        xv[n] = f;
+      }
+#pragma unroll
+      for (int m = 0; m < 8; ++m) {
+        v8i_mx qv = mx_frag(ql[m], qh[m]);
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              qv, xv[n], acc[m][n], 0 /*A fp8*/, 4 /*B fp4*/,
+              0, unit_scale, 0, (int)sv[n]);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    if (k < 0) {
+      if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+          float v = acc[m][n][r];
+          if (!(v > row_min[row])) continue;
+          long long grow = row0 + row;
+          if (grow >= nq) continue;
+          long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+          if (col >= x_end) continue;
+          int pos = atomicAdd(&tc_n[grow], 1);
+          if (pos < cap) {
+            cand_scores[grow * cap + pos] = v;
+            cand_ids[grow * cap + pos] = int32_t(col);
+          }
+        }
+  }
+}

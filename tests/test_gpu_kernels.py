@@ -330,3 +330,40 @@ def test_topk_recall_odd_shapes(nq, nx, k):
         inter = len(set(ids_np[q]) & set(ref_ids[q]))
         assert inter >= max(1, k - 2), f"q={q}: {inter}/{k}"
         assert abs(vals[q][0] - ref_vals[q][0]) < 2e-2
+
+
+def test_to_fp4_mx_roundtrip_cpu():
+    """MXFP4 quantizer (pure torch, CPU): decode matches within e2m1
+    half-gap bounds; direction preserved (cos > 0.98)."""
+    torch.manual_seed(0)
+    X = torch.nn.functional.normalize(torch.randn(64, 128), dim=1).bfloat16()
+    X4, XS = g.to_fp4_mx(X)
+    assert X4.shape == (64, 64) and XS.shape == (64, 4)
+    GRID = torch.tensor([0., .5, 1., 1.5, 2., 3., 4., 6.])
+    codes = torch.stack([(X4 & 0xF).long(), (X4 >> 4).long()], dim=2).reshape(64, 128)
+    dec = GRID[codes & 7] * torch.where(codes >= 8, -1.0, 1.0) * \
+        torch.exp2(XS.float() - 127.0).repeat_interleave(32, dim=1)
+    err = (dec - X.float()).abs()
+    blk_max = X.float().view(64, 4, 32).abs().amax(2).repeat_interleave(32, 1)
+    assert (err <= blk_max / 3 + 1e-6).all()
+    cos = torch.nn.functional.cosine_similarity(dec, X.float(), dim=1)
+    assert cos.min() > 0.98
+
+
+@pytest.mark.gpu
+def test_topk_recall_threshold_fp4():
+    """MXFP4-X threshold scan + exact rescore vs fp32 reference."""
+    torch.manual_seed(11)
+    nq, nx, D, k = 512, 32768, 1024, 16
+    Q = torch.nn.functional.normalize(torch.randn(nq, D, device="cuda"), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(nx, D, device="cuda"), dim=1).bfloat16()
+    ref = torch.topk(Q.float() @ X.float().T, k, dim=1)
+    ref_ids = ref.indices.cpu().numpy()
+    X4 = g.to_fp4_mx(X)
+    scores, ids = g.topk_recall_threshold(Q, X, k, X4=X4)
+    ids_np = ids.cpu().numpy()
+    vals = scores.cpu().numpy()
+    for q in range(nq):
+        inter = len(set(ids_np[q]) & set(ref_ids[q]))
+        assert inter >= k - 2, f"fp4 q={q}: {inter}/{k}"
+        assert abs(vals[q][0] - ref.values[q, 0].item()) < 2e-2

@@ -94,10 +94,12 @@ def threshold_bench():
     Q = torch.nn.functional.normalize(torch.randn(nq, dim, device=dev), dim=1).bfloat16()
     X = torch.nn.functional.normalize(torch.randn(rows, dim, device=dev), dim=1).bfloat16()
     X8 = g.to_fp8_bytes(X)
+    X4 = g.to_fp4_mx(X)
     for name, fn in [
         ("thresh bf16 k=16", lambda: g.topk_recall_threshold(Q, X, 16)),
         ("thresh fp8  k=16", lambda: g.topk_recall_threshold(Q, X, 16, X8=X8, mx=False)),
         ("thresh MX   k=16", lambda: g.topk_recall_threshold(Q, X, 16, X8=X8)),
+        ("thresh fp4  k=16", lambda: g.topk_recall_threshold(Q, X, 16, X4=X4)),
         ("thresh bf16 k=32", lambda: g.topk_recall_threshold(Q, X, 32)),
         ("direct bf16 k=16", lambda: g.topk_recall(Q, X, 16)),
     ]:

@@ -45,6 +45,27 @@ __global__ void mx_probe_kernel(const uint8_t* A, const uint8_t* B, float* D,
   for (int r = 0; r < 4; ++r) D[(kg * 4 + r) * 16 + (lane & 15)] = c[r];
 }
 
+typedef int v4i __attribute__((ext_vector_type(4)));
+
+// mixed A=fp8, B=fp4 (blgp=4): B bytes hold two e2m1 nibbles (assumed
+// low nibble = even k); per-lane B scale byte applies to the lane's own
+// 32-elem block (B_scales[col][kg], passed in the lane's scale VGPR).
+__global__ void mx4_probe_kernel(const uint8_t* A, const uint8_t* B4,
+                                 const uint8_t* BS, float* D) {
+  int lane = threadIdx.x & 63;
+  int row = lane & 15;
+  int kg = lane >> 4;
+  v8i av = *(const v8i*)(A + row * 128 + kg * 32);
+  v4i b4 = *(const v4i*)(B4 + row * 64 + kg * 16);  // 32 fp4 = 16 B
+  v8i bv = {b4.x, b4.y, b4.z, b4.w, 0, 0, 0, 0};
+  int sb = BS[row * 4 + kg];  // e8m0 for this lane's block, low byte
+  f32x4 c = {};
+  c = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+      av, bv, c, 0 /*A fp8*/, 4 /*B fp4*/, 0, 0x7F7F7F7F, 0, sb);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) D[(kg * 4 + r) * 16 + (lane & 15)] = c[r];
+}
+
 // e4m3 OCP encode (round-to-nearest-even on the mantissa, no inf)
 static uint8_t f32_to_e4m3(float f) {
   if (f == 0.0f) return 0;
@@ -118,6 +139,56 @@ int main() {
       for (int i = 0; i < 4; ++i)
         printf("  d[0][%d]=%f ref=%f\n", i, hd[i], ref[i] * cs.mult);
     }
+  }
+
+  // ---- fp4 B case: e2m1 nibbles + per-block e8m0 scales ----
+  static const float E2M1[8] = {0.f, .5f, 1.f, 1.5f, 2.f, 3.f, 4.f, 6.f};
+  std::vector<uint8_t> hb4(M * 64), hbs(M * 4);
+  std::vector<float> bdec(M * K);
+  for (int j = 0; j < M; ++j)
+    for (int blk = 0; blk < 4; ++blk) {
+      int e = (rand() % 5) - 2;  // block scale 2^e, e in [-2,2]
+      hbs[j * 4 + blk] = (uint8_t)(e + 127);
+      for (int t = 0; t < 32; ++t) {
+        int code = rand() % 16;  // sign bit 3 + magnitude 0..7
+        float mag = E2M1[code & 7];
+        float val = ((code & 8) ? -mag : mag) * exp2f((float)e);
+        bdec[j * K + blk * 32 + t] = val;
+        int byte_idx = j * 64 + blk * 16 + t / 2;
+        if (t % 2 == 0)
+          hb4[byte_idx] = (uint8_t)code;          // low nibble = even k
+        else
+          hb4[byte_idx] |= (uint8_t)(code << 4);  // high nibble = odd k
+      }
+    }
+  std::vector<double> ref4(M * M, 0.0);
+  for (int i = 0; i < M; ++i)
+    for (int j = 0; j < M; ++j) {
+      double s = 0;
+      for (int k = 0; k < K; ++k)
+        s += (double)e4m3_to_f32(ha[i * K + k]) * (double)bdec[j * K + k];
+      ref4[i * M + j] = s;
+    }
+  uint8_t *db4, *dbs;
+  HIP_CHECK(hipMalloc(&db4, M * 64));
+  HIP_CHECK(hipMalloc(&dbs, M * 4));
+  HIP_CHECK(hipMemcpy(db4, hb4.data(), M * 64, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(dbs, hbs.data(), M * 4, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemset(dd, 0, M * M * sizeof(float)));
+  hipLaunchKernelGGL(mx4_probe_kernel, dim3(1), dim3(64), 0, 0, da, db4, dbs, dd);
+  HIP_CHECK(hipDeviceSynchronize());
+  std::vector<float> hd4(M * M);
+  HIP_CHECK(hipMemcpy(hd4.data(), dd, M * M * sizeof(float), hipMemcpyDeviceToHost));
+  double maxerr4 = 0;
+  for (int i = 0; i < M * M; ++i)
+    maxerr4 = fmax(maxerr4, fabs(hd4[i] - ref4[i]));
+  bool ok4 = maxerr4 < 1e-2;
+  printf("%-28s maxerr=%.6f  %s\n", "B fp4 + block scales", maxerr4,
+         ok4 ? "PASS" : "FAIL");
+  if (!ok4) {
+    ++fails;
+    for (int i = 0; i < 4; ++i)
+      printf("  d[0][%d]=%f ref=%f\n", i, hd4[i], ref4[i]);
   }
   return fails ? 1 : 0;
 }

@@ -191,6 +191,32 @@ def to_fp8_bytes(x: torch.Tensor, scale: float = 8.0) -> torch.Tensor:
     return (x.float() * scale).to(torch.float8_e4m3fn).view(torch.uint8)
 
 
+_E2M1_MIDPOINTS = (0.25, 0.75, 1.25, 1.75, 2.5, 3.5, 5.0)
+
+
+def to_fp4_mx(x: torch.Tensor, chunk_rows: int = 1_048_576) -> Tuple[torch.Tensor, torch.Tensor]:
+    """bf16 [N, D] -> MXFP4: packed e2m1 nibbles [N, D/2] (low nibble =
+    even k) + per-32-block e8m0 scales [N, D/32]. The e8m0 exponent is
+    chosen so each block's max |v| maps onto e2m1's top code (6.0)."""
+    N, D = x.shape
+    assert D % 32 == 0
+    mids = torch.tensor(_E2M1_MIDPOINTS, device=x.device)
+    out4 = torch.empty((N, D // 2), dtype=torch.uint8, device=x.device)
+    outs = torch.empty((N, D // 32), dtype=torch.uint8, device=x.device)
+    for r0 in range(0, N, chunk_rows):
+        v = x[r0:r0 + chunk_rows].float().view(-1, D // 32, 32)
+        amax = v.abs().amax(dim=2, keepdim=True)
+        e = torch.where(amax > 0, (amax / 6.0).log2().ceil(), torch.zeros_like(amax))
+        e = e.clamp(-127, 127)
+        outs[r0:r0 + chunk_rows] = (e.squeeze(2) + 127).to(torch.uint8)
+        y = v * torch.exp2(-e)
+        code = torch.bucketize(y.abs().contiguous(), mids).to(torch.uint8)
+        code |= (y < 0).to(torch.uint8) << 3
+        code = code.view(-1, D)
+        out4[r0:r0 + chunk_rows] = code[:, 0::2] | (code[:, 1::2] << 4)
+    return out4, outs
+
+
 def topk_recall_fp8(Q8: torch.Tensor, X8: torch.Tensor, k: int, n_swaths: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:
     """Stage-1 fp8 scan: candidate ids ranked by e4m3 cosine (csrc
     topk_recall_fp8_kernel). Inputs are uint8 views of e4m3 bytes."""
@@ -279,6 +305,7 @@ def topk_recall_threshold(
     target_candidates: int = 128,
     cap: int = 1024,
     mx: bool = True,
+    X4: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Threshold-scan recall: per-query score thresholds estimated from a
     sampled pre-pass (Gaussian tail extrapolation), then a fixed-threshold
@@ -296,8 +323,9 @@ def topk_recall_threshold(
     nq, D = Q.shape
     nx = X.shape[0]
     m = min(nx, sample_rows)
-    use_fp8 = X8 is not None
-    use_mx = mx and use_fp8 and D % 128 == 0
+    use_fp4 = X4 is not None and D % 128 == 0 and D <= 2048
+    use_fp8 = X8 is not None or use_fp4
+    use_mx = mx and X8 is not None and D % 128 == 0
 
     # 1. per-query score statistics from a sample (bf16 matmul)
     sample = torch.matmul(Q, X[:m].T).float()  # [nq, m]
@@ -324,7 +352,12 @@ def topk_recall_threshold(
     theta = (mu + sigma * z).contiguous()
 
     # 2. fixed-threshold scan
-    if use_fp8:
+    if use_fp4:
+        # Q carries the only static scale (x8); X scales are per-block
+        Q8 = to_fp8_bytes(Q)
+        cs, ci, counts = ext().topk_scan_threshold_fp4(
+            Q8, X4[0], X4[1], (theta * 8.0).contiguous(), cap, 0)
+    elif use_fp8:
         Q8 = to_fp8_bytes(Q)
         # thresholds are in fp8-score units: inputs scaled x8 each -> x64
         cs, ci, counts = ext().topk_scan_threshold(Q8, X8, (theta * 64.0).contiguous(),
