@@ -141,54 +141,65 @@ int main() {
     }
   }
 
-  // ---- fp4 B case: e2m1 nibbles + per-block e8m0 scales ----
+  // ---- fp4 B layout search: one packed byte buffer, FOUR candidate
+  // nibble->k interpretations checked on the host; unit scales first to
+  // isolate the data layout from scale semantics, then per-block scales
+  // under every candidate.
   static const float E2M1[8] = {0.f, .5f, 1.f, 1.5f, 2.f, 3.f, 4.f, 6.f};
   std::vector<uint8_t> hb4(M * 64), hbs(M * 4);
-  std::vector<float> bdec(M * K);
+  for (auto& v : hb4) v = (uint8_t)(rand() & 0xFF);
   for (int j = 0; j < M; ++j)
-    for (int blk = 0; blk < 4; ++blk) {
-      int e = (rand() % 5) - 2;  // block scale 2^e, e in [-2,2]
-      hbs[j * 4 + blk] = (uint8_t)(e + 127);
-      for (int t = 0; t < 32; ++t) {
-        int code = rand() % 16;  // sign bit 3 + magnitude 0..7
-        float mag = E2M1[code & 7];
-        float val = ((code & 8) ? -mag : mag) * exp2f((float)e);
-        bdec[j * K + blk * 32 + t] = val;
-        int byte_idx = j * 64 + blk * 16 + t / 2;
-        if (t % 2 == 0)
-          hb4[byte_idx] = (uint8_t)code;          // low nibble = even k
-        else
-          hb4[byte_idx] |= (uint8_t)(code << 4);  // high nibble = odd k
-      }
+    for (int blk = 0; blk < 4; ++blk)
+      hbs[j * 4 + blk] = (uint8_t)(127 + (rand() % 5) - 2);
+
+  // decode candidate `var`: value of k-element t (in [0,32)) of block blk
+  auto dec = [&](int var, int j, int blk, int t) -> float {
+    int b = j * 64 + blk * 16;  // 16 bytes of this block
+    int code;
+    switch (var) {
+      case 0: code = (hb4[b + t / 2] >> ((t % 2) * 4)) & 0xF; break;  // lo=even
+      case 1: code = (hb4[b + t / 2] >> (((t + 1) % 2) * 4)) & 0xF; break;  // hi=even
+      case 2: code = (hb4[b + (t % 16)] >> ((t / 16) * 4)) & 0xF; break;  // planes lo=k0..15
+      default: code = (hb4[b + (t % 16)] >> ((1 - t / 16) * 4)) & 0xF; break;  // planes hi=k0..15
     }
-  std::vector<double> ref4(M * M, 0.0);
-  for (int i = 0; i < M; ++i)
-    for (int j = 0; j < M; ++j) {
-      double s = 0;
-      for (int k = 0; k < K; ++k)
-        s += (double)e4m3_to_f32(ha[i * K + k]) * (double)bdec[j * K + k];
-      ref4[i * M + j] = s;
-    }
+    float mag = E2M1[code & 7];
+    return (code & 8) ? -mag : mag;
+  };
+
   uint8_t *db4, *dbs;
   HIP_CHECK(hipMalloc(&db4, M * 64));
   HIP_CHECK(hipMalloc(&dbs, M * 4));
   HIP_CHECK(hipMemcpy(db4, hb4.data(), M * 64, hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(dbs, hbs.data(), M * 4, hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemset(dd, 0, M * M * sizeof(float)));
-  hipLaunchKernelGGL(mx4_probe_kernel, dim3(1), dim3(64), 0, 0, da, db4, dbs, dd);
-  HIP_CHECK(hipDeviceSynchronize());
   std::vector<float> hd4(M * M);
-  HIP_CHECK(hipMemcpy(hd4.data(), dd, M * M * sizeof(float), hipMemcpyDeviceToHost));
-  double maxerr4 = 0;
-  for (int i = 0; i < M * M; ++i)
-    maxerr4 = fmax(maxerr4, fabs(hd4[i] - ref4[i]));
-  bool ok4 = maxerr4 < 1e-2;
-  printf("%-28s maxerr=%.6f  %s\n", "B fp4 + block scales", maxerr4,
-         ok4 ? "PASS" : "FAIL");
-  if (!ok4) {
-    ++fails;
-    for (int i = 0; i < 4; ++i)
-      printf("  d[0][%d]=%f ref=%f\n", i, hd4[i], ref4[i]);
+  int matched = -1;
+  for (int pass = 0; pass < 2; ++pass) {
+    bool unit = (pass == 0);
+    std::vector<uint8_t> scales = hbs;
+    if (unit) std::fill(scales.begin(), scales.end(), (uint8_t)127);
+    HIP_CHECK(hipMemcpy(dbs, scales.data(), M * 4, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemset(dd, 0, M * M * sizeof(float)));
+    hipLaunchKernelGGL(mx4_probe_kernel, dim3(1), dim3(64), 0, 0, da, db4, dbs, dd);
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipMemcpy(hd4.data(), dd, M * M * sizeof(float), hipMemcpyDeviceToHost));
+    for (int var = 0; var < 4; ++var) {
+      double maxerr = 0;
+      for (int i = 0; i < M; ++i)
+        for (int j = 0; j < M; ++j) {
+          double s = 0;
+          for (int blk = 0; blk < 4; ++blk) {
+            double e = unit ? 1.0 : exp2((double)scales[j * 4 + blk] - 127.0);
+            for (int t = 0; t < 32; ++t)
+              s += (double)e4m3_to_f32(ha[i * K + blk * 32 + t]) *
+                   (double)dec(var, j, blk, t) * e;
+          }
+          maxerr = fmax(maxerr, fabs(hd4[i * M + j] - s));
+        }
+      bool ok = maxerr < 1e-2;
+      printf("fp4 %s layout-v%d           maxerr=%.6f  %s\n",
+             unit ? "unit  " : "scaled", var, maxerr, ok ? "PASS" : "FAIL");
+      if (ok && unit) matched = var;
+    }
   }
+  if (matched < 0) ++fails;
   return fails ? 1 : 0;
 }

@@ -194,17 +194,35 @@ def to_fp8_bytes(x: torch.Tensor, scale: float = 8.0) -> torch.Tensor:
 _E2M1_MIDPOINTS = (0.25, 0.75, 1.25, 1.75, 2.5, 3.5, 5.0)
 
 
+def fp4_perm128() -> torch.Tensor:
+    """Element order the v_mfma_scale..f8f6f4 fp4 B operand expects, per
+    128-k chunk (probed on hardware, tools/probe_fp4map.hip): LDS slot g
+    (= lane kgrp g) holds k in {base, base+32} + [0,16) with base =
+    (g&1)*64 + (g>>1)*16, packed two-per-byte (low nibble = even
+    position). A lane's 32 elements share ONE e8m0 scale, so scale
+    groups are 32 consecutive elements in THIS order."""
+    base = torch.arange(16)
+    order = []
+    for g in range(4):
+        s = (g & 1) * 64 + (g >> 1) * 16
+        order.append(s + base)
+        order.append(s + 32 + base)
+    return torch.cat(order)
+
+
 def to_fp4_mx(x: torch.Tensor, chunk_rows: int = 1_048_576) -> Tuple[torch.Tensor, torch.Tensor]:
-    """bf16 [N, D] -> MXFP4: packed e2m1 nibbles [N, D/2] (low nibble =
-    even k) + per-32-block e8m0 scales [N, D/32]. The e8m0 exponent is
-    chosen so each block's max |v| maps onto e2m1's top code (6.0)."""
+    """bf16 [N, D] -> MXFP4 in the hardware fragment order: packed e2m1
+    nibbles [N, D/2] + per-lane-group e8m0 scales [N, D/32]. The e8m0
+    exponent maps each group's max |v| onto e2m1's top code (6.0)."""
     N, D = x.shape
-    assert D % 32 == 0
+    assert D % 128 == 0
     mids = torch.tensor(_E2M1_MIDPOINTS, device=x.device)
+    p128 = fp4_perm128().to(x.device)
+    perm = (torch.arange(0, D, 128, device=x.device).unsqueeze(1) + p128.unsqueeze(0)).reshape(-1)
     out4 = torch.empty((N, D // 2), dtype=torch.uint8, device=x.device)
     outs = torch.empty((N, D // 32), dtype=torch.uint8, device=x.device)
     for r0 in range(0, N, chunk_rows):
-        v = x[r0:r0 + chunk_rows].float().view(-1, D // 32, 32)
+        v = x[r0:r0 + chunk_rows].float()[:, perm].view(-1, D // 32, 32)
         amax = v.abs().amax(dim=2, keepdim=True)
         e = torch.where(amax > 0, (amax / 6.0).log2().ceil(), torch.zeros_like(amax))
         e = e.clamp(-127, 127)
