@@ -50,6 +50,7 @@ def main() -> int:
                     choices=["threshold", "two_stage", "direct"])
     ap.add_argument("--no-fp8", action="store_true")
     ap.add_argument("--no-mx", action="store_true")
+    ap.add_argument("--no-fp4", action="store_true")
     ap.add_argument("--no-audit-sink", action="store_true",
                     help="disable the async audit writer + event journal (on by default: "
                     "the timed path includes audit record D2H + JSONL manifest + journal envelopes)")
@@ -72,7 +73,7 @@ def main() -> int:
     cfg = PipelineConfig(
         batch=args.batch, dim=args.dim, index_size=shard, topk=args.topk,
         recall_mode=args.recall_mode, recall_fp8=not args.no_fp8,
-        recall_mx=not args.no_mx,
+        recall_mx=not args.no_mx, recall_fp4=not args.no_fp4,
     )
     t0 = time.time()
     pipe = FirewallPipeline(cfg, device=device, world_size=world, rank=rank)

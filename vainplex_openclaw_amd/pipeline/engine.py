@@ -61,6 +61,9 @@ class PipelineConfig:
     # MX-scaled x128 scan for the fp8 threshold path (same bytes, higher
     # MFMA issue rate — csrc topk_scan_mx_kernel); needs dim % 128 == 0
     recall_mx: bool = True
+    # MXFP4 X operand for the threshold scan (half the fp8 index bytes;
+    # csrc topk_scan_mx4_kernel + the probed fragment permutation)
+    recall_fp4: bool = True
     inj_threshold: float = 0.9
     seed: int = 1234
     families: tuple = ("redaction", "injection", "claims", "entity")
@@ -132,10 +135,23 @@ class FirewallPipeline:
                 self.index[i : i + n] = x.bfloat16()
                 del x
 
-            # fp8 (e4m3) copy of the index for the stage-1 scan: half the
-            # staged bytes at the measured GLDS transport bound
+            # low-precision scan copies of the index. fp4 (MXFP4, half
+            # the fp8 bytes) is the default threshold-scan operand; the
+            # e4m3 copy serves the two_stage mode and the fp8/MX scans.
             self.index8 = None
-            if cfg.recall_fp8 and cfg.recall_mode in ("two_stage", "threshold"):
+            self.index4 = None
+            if cfg.recall_fp4 and cfg.recall_mode == "threshold" and cfg.dim % 128 == 0:
+                x4 = torch.empty(cfg.index_size, cfg.dim // 2, dtype=torch.uint8,
+                                 device=self.device)
+                xs = torch.empty(cfg.index_size, cfg.dim // 32, dtype=torch.uint8,
+                                 device=self.device)
+                for i in range(0, cfg.index_size, chunk):
+                    n = min(chunk, cfg.index_size - i)
+                    c4, cs = g.to_fp4_mx(self.index[i : i + n])
+                    x4[i : i + n] = c4
+                    xs[i : i + n] = cs
+                self.index4 = (x4, xs)
+            elif cfg.recall_fp8 and cfg.recall_mode in ("two_stage", "threshold"):
                 self.index8 = torch.empty(
                     cfg.index_size, cfg.dim, dtype=torch.uint8, device=self.device
                 )
@@ -198,7 +214,8 @@ class FirewallPipeline:
         def local_recall(queries):
             if cfg.recall_mode == "threshold":
                 return g.topk_recall_threshold(
-                    queries, self.index, cfg.topk, X8=self.index8, mx=cfg.recall_mx
+                    queries, self.index, cfg.topk, X8=self.index8,
+                    mx=cfg.recall_mx, X4=self.index4
                 )
             if cfg.recall_mode == "two_stage" and self.index8 is not None:
                 return g.topk_recall_two_stage(queries, self.index, self.index8, cfg.topk)
