@@ -36,6 +36,9 @@ __global__ void topk_scan_mx_kernel(const uint8_t*, const uint8_t*, int, int, in
 __global__ void topk_scan_mx4_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
                                      int, int, int, int, int, float*, int32_t*,
                                      const float*, int32_t*, int);
+__global__ void topk_scan_fp4_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
+                                     const uint8_t*, int, int, int, int, int,
+                                     float*, int32_t*, const float*, int32_t*, int);
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
                                   float*, int32_t*);
 __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigned long long*,
@@ -290,6 +293,42 @@ torch::Tensor audit_pack(torch::Tensor verdict, torch::Tensor risk, torch::Tenso
 }
 
 
+std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
+    torch::Tensor Q4, torch::Tensor QS, torch::Tensor X4, torch::Tensor XS,
+    torch::Tensor theta, int64_t cap, int64_t n_swaths) {
+  // both operands MXFP4 (csrc topk_scan_fp4_kernel); scores at x1
+  CHECK_GPU(Q4); CHECK_CONTIG(Q4); CHECK_GPU(QS); CHECK_CONTIG(QS);
+  CHECK_GPU(X4); CHECK_CONTIG(X4); CHECK_GPU(XS); CHECK_CONTIG(XS);
+  CHECK_GPU(theta); CHECK_CONTIG(theta);
+  int nq = Q4.size(0), D = (int)Q4.size(1) * 2;
+  long long nx = X4.size(0);
+  TORCH_CHECK(D % 128 == 0 && D / 32 <= 64);
+  TORCH_CHECK(QS.size(0) == nq && QS.size(1) == D / 32);
+  TORCH_CHECK(X4.size(1) == D / 2 && XS.size(0) == nx && XS.size(1) == D / 32);
+  TORCH_CHECK(theta.numel() == nq && cap >= 32 && cap <= 4096);
+  int n_qblocks = (nq + 255) / 256;
+  if (n_swaths <= 0) {
+    int want = 256 / (n_qblocks > 0 ? n_qblocks : 1);
+    n_swaths = want >= 8 ? (want / 8) * 8 : 8;
+    long long max_s = nx / 256; if (max_s < 1) max_s = 1;
+    if (n_swaths > max_s) n_swaths = max_s;
+  }
+  auto f32opts = torch::dtype(torch::kFloat32).device(Q4.device());
+  auto i32opts = torch::dtype(torch::kInt32).device(Q4.device());
+  auto cand_s = torch::full({(long long)nq, cap}, -1e30, f32opts);
+  auto cand_i = torch::full({(long long)nq, cap}, -1, i32opts);
+  auto counts = torch::zeros({(long long)nq}, i32opts);
+  dim3 grid((unsigned)(n_qblocks * n_swaths));
+  hipLaunchKernelGGL(topk_scan_fp4_kernel, grid, dim3(512), 0, cur_stream(),
+                     Q4.data_ptr<uint8_t>(), QS.data_ptr<uint8_t>(),
+                     X4.data_ptr<uint8_t>(), XS.data_ptr<uint8_t>(),
+                     nq, (int)nx, D, 1, (int)n_swaths,
+                     cand_s.data_ptr<float>(), cand_i.data_ptr<int32_t>(),
+                     theta.data_ptr<float>(), counts.data_ptr<int32_t>(),
+                     (int)cap);
+  return {cand_s, cand_i, counts};
+}
+
 std::vector<torch::Tensor> topk_scan_threshold(torch::Tensor Q, torch::Tensor X,
                                                torch::Tensor theta, int64_t cap,
                                                int64_t n_swaths, bool fp8, bool mx) {
@@ -426,6 +465,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                        (int)cap);
     return std::vector<torch::Tensor>{cand_s, cand_i, counts};
   }, "MXFP4-X threshold scan");
+  m.def("topk_scan_threshold_fp4x4", &topk_scan_threshold_fp4x4,
+        "full-MXFP4 threshold scan (both operands e2m1 + e8m0 group scales)");
   m.def("firewall_verdict", &firewall_verdict, "Fused verdict/risk/trust-delta");
   m.def("trust_recompute", &trust_recompute, "Agent trust score recompute");
   m.def("audit_pack", &audit_pack, "Pack 64-byte audit records");

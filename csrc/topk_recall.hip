@@ -1072,3 +1072,180 @@ topk_scan_mx4_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ 
         }
   }
 }
+
+// ===========================================================================
+// Full-MXFP4 threshold scan: BOTH operands e2m1 with per-lane-group
+// e8m0 scales (cbsz=4, blgp=4). Per-pair staging drops to 4 per-lane
+// glds (Q4 16 KB + X4 16 KB), Q fragment reads halve to one b128 per
+// m-tile, and register pressure falls ~70 VGPRs vs the fp8-Q variant.
+// The A-operand fp4 layout was probed identical to B
+// (tools/probe_fp4map.hip): the same fragment permutation and scale
+// grouping apply, so queries go through the same to_fp4_mx quantizer.
+// Scores carry NO static prescale (the e8m0 block scales hold the
+// magnitudes): theta is used at x1.
+// ===========================================================================
+
+extern "C" __global__ void __launch_bounds__(TK_THREADS)
+topk_scan_fp4_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__ QS,
+                     const uint8_t* __restrict__ X4, const uint8_t* __restrict__ XS,
+                     int nq, int nx, int D, int k, int n_swaths,
+                     float* __restrict__ cand_scores,
+                     int32_t* __restrict__ cand_ids,
+                     const float* __restrict__ theta,
+                     int32_t* __restrict__ tc_n, int cap) {
+  __shared__ bf16 lds_q[2 * BM * BK];       // 2 pair-bufs x 16 KB
+  __shared__ bf16 lds_x[2 * BN * BK];
+  __shared__ uint8_t lds_qs[BM * MX4_SMAX];
+  __shared__ uint8_t lds_xs2[BN * MX4_SMAX];
+  __shared__ float row_min[BM];
+#define QP4(buf) (lds_q + (buf) * BM * BK)
+#define XP4(buf) (lds_x + (buf) * BN * BK)
+
+  int S = n_swaths;
+  int qb = blockIdx.x / S;
+  int swath = blockIdx.x % S;
+  long long row0 = (long long)qb * BM;
+
+  long long per = ((long long)nx + S - 1) / S;
+  per = ((per + BN - 1) / BN) * BN;
+  long long x_begin = (long long)swath * per;
+  long long x_end = min((long long)nx, x_begin + per);
+
+  for (int i = threadIdx.x; i < BM; i += blockDim.x)
+    row_min[i] = (theta != nullptr && row0 + i < nq) ? theta[row0 + i] : -1e30f;
+  __syncthreads();
+
+  int wid = wave_id();
+  int wm = wid >> 2, wn = wid & 3;
+  int lane = lane_id();
+  int lrow = lane & 15;
+  int kgrp = lane >> 4;
+  int np = D / (2 * BK_F8);
+  int sb = D / 32;
+  size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
+  long long p4_ld = D / 4;  // packed row length in bf16 units
+
+  for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
+    f32x4 acc[8][4] = {};
+    stage_scale_rows(QS, sb, row0, (long long)nq, lds_qs, BM);
+    stage_scale_rows(XS, sb, x0, (long long)nx, lds_xs2, BN);
+    for (int pp = 0; pp < 2 && pp < np; ++pp) {
+      stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, pp * 32, QP4(pp & 1), BM);
+      stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, pp * 32, XP4(pp & 1), BN);
+    }
+    for (int p = 0; p < np; ++p) {
+      if (p + 1 < np)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      uint32_t xrow_base = (uint32_t)(wn * 64 + lrow);
+      uint32_t qrow_base = (uint32_t)(wm * 128 + lrow);
+      uint32_t xaddr = (uint32_t)(size_t)XP4(p & 1)
+                       + lds_off_bytes(xrow_base, (uint32_t)kgrp);
+      uint32_t qaddr = (uint32_t)(size_t)QP4(p & 1)
+                       + lds_off_bytes(qrow_base, (uint32_t)kgrp);
+      uint32_t sst = 16u * (uint32_t)sb;
+      uint32_t xs_a = (uint32_t)(size_t)lds_xs2 + xrow_base * (uint32_t)sb
+                      + (uint32_t)(p * 4 + kgrp);
+      uint32_t qs_a = (uint32_t)(size_t)lds_qs + qrow_base * (uint32_t)sb
+                      + (uint32_t)(p * 4 + kgrp);
+      bf16x8 xf[4], qf[8];
+      uint32_t xs_v[4], qs_v[8];
+      asm volatile(
+          "ds_read_b128 %0, %8\n\t"
+          "ds_read_b128 %1, %8 offset:1024\n\t"
+          "ds_read_b128 %2, %8 offset:2048\n\t"
+          "ds_read_b128 %3, %8 offset:3072\n\t"
+          "ds_read_u8 %4, %9\n\t"
+          "ds_read_u8 %5, %10\n\t"
+          "ds_read_u8 %6, %11\n\t"
+          "ds_read_u8 %7, %12\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
+            "=&v"(xs_v[0]), "=&v"(xs_v[1]), "=&v"(xs_v[2]), "=&v"(xs_v[3])
+          : "v"(xaddr), "v"(xs_a), "v"(xs_a + sst), "v"(xs_a + 2 * sst),
+            "v"(xs_a + 3 * sst));
+      asm volatile(
+          "ds_read_b128 %0, %16\n\t"
+          "ds_read_b128 %1, %16 offset:1024\n\t"
+          "ds_read_b128 %2, %16 offset:2048\n\t"
+          "ds_read_b128 %3, %16 offset:3072\n\t"
+          "ds_read_b128 %4, %16 offset:4096\n\t"
+          "ds_read_b128 %5, %16 offset:5120\n\t"
+          "ds_read_b128 %6, %16 offset:6144\n\t"
+          "ds_read_b128 %7, %16 offset:7168\n\t"
+          "ds_read_u8 %8, %17\n\t"
+          "ds_read_u8 %9, %18\n\t"
+          "ds_read_u8 %10, %19\n\t"
+          "ds_read_u8 %11, %20\n\t"
+          "ds_read_u8 %12, %21\n\t"
+          "ds_read_u8 %13, %22\n\t"
+          "ds_read_u8 %14, %23\n\t"
+          "ds_read_u8 %15, %24\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
+            "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7]),
+            "=&v"(qs_v[0]), "=&v"(qs_v[1]), "=&v"(qs_v[2]), "=&v"(qs_v[3]),
+            "=&v"(qs_v[4]), "=&v"(qs_v[5]), "=&v"(qs_v[6]), "=&v"(qs_v[7])
+          : "v"(qaddr), "v"(qs_a), "v"(qs_a + sst), "v"(qs_a + 2 * sst),
+            "v"(qs_a + 3 * sst), "v"(qs_a + 4 * sst), "v"(qs_a + 5 * sst),
+            "v"(qs_a + 6 * sst), "v"(qs_a + 7 * sst));
+      __builtin_amdgcn_s_barrier();
+      if (2 * p + 4 < 2 * np) {
+        stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, (p + 2) * 32,
+                   QP4(p & 1), BM);
+        stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, (p + 2) * 32,
+                   XP4(p & 1), BN);
+      }
+      v8i_mx xv[4], qv[8];
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        int4 l = __builtin_bit_cast(int4, xf[n]);
+        v8i_mx f; f[0] = l.x; f[1] = l.y; f[2] = l.z; f[3] = l.w;
+        f[4] = 0; f[5] = 0; f[6] = 0; f[7] = 0;
+        xv[n] = f;
+      }
+#pragma unroll
+      for (int m = 0; m < 8; ++m) {
+        int4 l = __builtin_bit_cast(int4, qf[m]);
+        v8i_mx f; f[0] = l.x; f[1] = l.y; f[2] = l.z; f[3] = l.w;
+        f[4] = 0; f[5] = 0; f[6] = 0; f[7] = 0;
+        qv[m] = f;
+      }
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              qv[m], xv[n], acc[m][n], 4 /*A fp4*/, 4 /*B fp4*/,
+              0, (int)qs_v[m], 0, (int)xs_v[n]);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    if (k < 0) {
+      if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+          float v = acc[m][n][r];
+          if (!(v > row_min[row])) continue;
+          long long grow = row0 + row;
+          if (grow >= nq) continue;
+          long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+          if (col >= x_end) continue;
+          int pos = atomicAdd(&tc_n[grow], 1);
+          if (pos < cap) {
+            cand_scores[grow * cap + pos] = v;
+            cand_ids[grow * cap + pos] = int32_t(col);
+          }
+        }
+  }
+}

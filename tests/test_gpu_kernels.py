@@ -374,3 +374,22 @@ def test_topk_recall_threshold_fp4():
         inter = len(set(ids_np[q]) & set(ref_ids[q]))
         assert inter >= k - 2, f"fp4 q={q}: {inter}/{k}"
         assert abs(vals[q][0] - ref.values[q, 0].item()) < 2e-2
+
+
+@pytest.mark.gpu
+def test_topk_recall_threshold_fp4x4():
+    """Full-MXFP4 (both operands) threshold scan + exact rescore."""
+    torch.manual_seed(11)
+    nq, nx, D, k = 512, 32768, 1024, 16
+    Q = torch.nn.functional.normalize(torch.randn(nq, D, device="cuda"), dim=1).bfloat16()
+    X = torch.nn.functional.normalize(torch.randn(nx, D, device="cuda"), dim=1).bfloat16()
+    ref = torch.topk(Q.float() @ X.float().T, k, dim=1)
+    ref_ids = ref.indices.cpu().numpy()
+    X4 = g.to_fp4_mx(X)
+    scores, ids = g.topk_recall_threshold(Q, X, k, X4=X4, q4=True)
+    ids_np = ids.cpu().numpy()
+    vals = scores.cpu().numpy()
+    for q in range(nq):
+        inter = len(set(ids_np[q]) & set(ref_ids[q]))
+        assert inter >= k - 2, f"fp4x4 q={q}: {inter}/{k}"
+        assert abs(vals[q][0] - ref.values[q, 0].item()) < 2e-2

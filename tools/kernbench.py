@@ -99,7 +99,8 @@ def threshold_bench():
         ("thresh bf16 k=16", lambda: g.topk_recall_threshold(Q, X, 16)),
         ("thresh fp8  k=16", lambda: g.topk_recall_threshold(Q, X, 16, X8=X8, mx=False)),
         ("thresh MX   k=16", lambda: g.topk_recall_threshold(Q, X, 16, X8=X8)),
-        ("thresh fp4  k=16", lambda: g.topk_recall_threshold(Q, X, 16, X4=X4)),
+        ("thresh fp4  k=16", lambda: g.topk_recall_threshold(Q, X, 16, X4=X4, q4=False)),
+        ("thresh fp4x4 k=16", lambda: g.topk_recall_threshold(Q, X, 16, X4=X4)),
         ("thresh bf16 k=32", lambda: g.topk_recall_threshold(Q, X, 32)),
         ("direct bf16 k=16", lambda: g.topk_recall(Q, X, 16)),
     ]:

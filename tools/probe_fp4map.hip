@@ -60,6 +60,26 @@ __global__ void scale_map_kernel(const uint8_t* A, float* D, int kg_target,
 
 static uint8_t fp8_one = 0x38;  // e4m3 1.0
 
+// A-side fp4 map: A one-hot fp4 nibble (cbsz=4), B fp8 bit-plane COLS
+// (B[col][k] = bit col of k). D[0][c]/D[0][7] recovers the fed k.
+__global__ void map_a_kernel(const uint8_t* B, float* D, int byte_pos, int nib,
+                             int kg_target, int code) {
+  int lane = threadIdx.x & 63;
+  int row = lane & 15;
+  int kg = lane >> 4;
+  uint8_t abytes[16] = {};
+  if (kg == kg_target)
+    abytes[byte_pos] = (uint8_t)(nib ? (code << 4) : code);
+  v4i a4 = *(const v4i*)abytes;
+  v8i av = {a4.x, a4.y, a4.z, a4.w, 0, 0, 0, 0};
+  v8i bv = *(const v8i*)(B + row * 128 + kg * 32);
+  f32x4 c = {};
+  c = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+      av, bv, c, 4 /*A fp4*/, 0 /*B fp8*/, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) D[(kg * 4 + r) * 16 + (lane & 15)] = c[r];
+}
+
 int main() {
   const int M = 16, K = 128;
   std::vector<uint8_t> ha(M * K, 0);
@@ -90,6 +110,24 @@ int main() {
             if (fabsf(hd[r * M + 0] / mag - 1.0f) < 0.25f) kk |= 1 << r;
         }
         printf("kg%d byte%02d nib%d -> k=%3d mag=%.3f\n", kg, bp, nib, kk, mag);
+      }
+
+  printf("== A-side fp4 position map (kg0/kg1, nib0/1, bytes 0/1/7/8) ==\n");
+  for (int kg = 0; kg < 2; ++kg)
+    for (int nib = 0; nib < 2; ++nib)
+      for (int bp : {0, 1, 7, 8}) {
+        HIP_CHECK(hipMemset(dd, 0, M * M * sizeof(float)));
+        hipLaunchKernelGGL(map_a_kernel, dim3(1), dim3(64), 0, 0, da, dd, bp, nib, kg, 2);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(hd.data(), dd, M * M * sizeof(float), hipMemcpyDeviceToHost));
+        float mag = hd[0 * M + 7];  // row 0, col 7 (all-ones B col)
+        int kk = -1;
+        if (fabsf(mag) > 1e-6) {
+          kk = 0;
+          for (int r = 0; r < 7; ++r)
+            if (fabsf(hd[0 * M + r] / mag - 1.0f) < 0.25f) kk |= 1 << r;
+        }
+        printf("A kg%d byte%02d nib%d -> k=%3d mag=%.3f\n", kg, bp, nib, kk, mag);
       }
 
   printf("== scale-byte map: all-ones data in kg, scale byte sb = x2 ==\n");

@@ -324,6 +324,7 @@ def topk_recall_threshold(
     cap: int = 1024,
     mx: bool = True,
     X4: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
+    q4: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Threshold-scan recall: per-query score thresholds estimated from a
     sampled pre-pass (Gaussian tail extrapolation), then a fixed-threshold
@@ -370,7 +371,12 @@ def topk_recall_threshold(
     theta = (mu + sigma * z).contiguous()
 
     # 2. fixed-threshold scan
-    if use_fp4:
+    if use_fp4 and q4:
+        # both operands MXFP4: scores carry no static prescale
+        Q4, QS = to_fp4_mx(Q)
+        cs, ci, counts = ext().topk_scan_threshold_fp4x4(
+            Q4, QS, X4[0], X4[1], theta.contiguous(), cap, 0)
+    elif use_fp4:
         # Q carries the only static scale (x8); X scales are per-block
         Q8 = to_fp8_bytes(Q)
         cs, ci, counts = ext().topk_scan_threshold_fp4(
