@@ -886,20 +886,25 @@ topk_scan_mx_kernel(const uint8_t* __restrict__ Q, const uint8_t* __restrict__ X
 DEVINL void stage_scale_rows(const uint8_t* __restrict__ src, int sb,
                              long long row0, long long row_max,
                              uint8_t* lds_base, int tile_rows) {
-  // rows of sb bytes (sb % 16 == 0), staged linearly (no swizzle: the
-  // consumers are byte reads)
-  int per_row = sb / 16;
-  int n_pieces = tile_rows * per_row;
+  // scale sheet: tile_rows x sb bytes, staged as LINEAR bytes (no
+  // swizzle: the consumers are byte reads). row0*sb is 16-aligned
+  // (row0 is a multiple of 256); the trailing piece clamps to the last
+  // aligned 16 B inside the buffer — garbage scales only ever land on
+  // rows the epilogue bound-checks away.
+  long long base = row0 * (long long)sb;
+  long long last = ((row_max * (long long)sb) >> 4) * 16 - 16;
+  if (last < 0) last = 0;
+  int n_pieces = (tile_rows * sb + 15) / 16;
   int w = wave_id();
   int lane = lane_id();
   for (int piece0 = w * WAVE; piece0 < n_pieces; piece0 += TK_THREADS) {
     int piece = piece0 + lane;
-    long long gr = row0 + piece / per_row;
-    if (gr >= row_max) gr = row_max - 1;
-    const uint8_t* p = src + gr * sb + (long long)(piece % per_row) * 16;
+    long long off = base + (long long)piece * 16;
+    if (off > last) off = last;
     int piece0_u = __builtin_amdgcn_readfirstlane(piece0);
     auto ldst = (AS3 char*)lds_base + piece0_u * 16;
-    __builtin_amdgcn_global_load_lds((const AS1 void*)p, (AS3 void*)ldst, 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((const AS1 void*)(src + off),
+                                     (AS3 void*)ldst, 16, 0, 0);
   }
 }
 

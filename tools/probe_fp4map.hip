@@ -60,6 +60,27 @@ __global__ void scale_map_kernel(const uint8_t* A, float* D, int kg_target,
 
 static uint8_t fp8_one = 0x38;  // e4m3 1.0
 
+// A-side fp4 SCALE semantics: A all-ones fp4 in kg_target, scale_a byte
+// sbyte = 0x80 (x2), B fp8 bit-plane cols: bit sums show which k doubled
+__global__ void scale_a_map_kernel(const uint8_t* B, float* D, int kg_target,
+                                   int sbyte) {
+  int lane = threadIdx.x & 63;
+  int row = lane & 15;
+  int kg = lane >> 4;
+  uint8_t abytes[16];
+#pragma unroll
+  for (int i = 0; i < 16; ++i) abytes[i] = (kg == kg_target) ? 0x22 : 0;
+  v4i a4 = *(const v4i*)abytes;
+  v8i av = {a4.x, a4.y, a4.z, a4.w, 0, 0, 0, 0};
+  v8i bv = *(const v8i*)(B + row * 128 + kg * 32);
+  int sa = (0x7F7F7F7F & ~(0xFF << (8 * sbyte))) | (0x80 << (8 * sbyte));
+  f32x4 c = {};
+  c = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+      av, bv, c, 4, 0, 0, sa, 0, 0x7F7F7F7F);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) D[(kg * 4 + r) * 16 + (lane & 15)] = c[r];
+}
+
 // A-side fp4 map: A one-hot fp4 nibble (cbsz=4), B fp8 bit-plane COLS
 // (B[col][k] = bit col of k). D[0][c]/D[0][7] recovers the fed k.
 __global__ void map_a_kernel(const uint8_t* B, float* D, int byte_pos, int nib,
@@ -129,6 +150,18 @@ int main() {
         }
         printf("A kg%d byte%02d nib%d -> k=%3d mag=%.3f\n", kg, bp, nib, kk, mag);
       }
+
+  printf("== A-side scale-byte map (x2 on byte sb; row0 bit-plane cols) ==\n");
+  for (int kg = 0; kg < 2; ++kg)
+    for (int sbk = 0; sbk < 4; ++sbk) {
+      HIP_CHECK(hipMemset(dd, 0, M * M * sizeof(float)));
+      hipLaunchKernelGGL(scale_a_map_kernel, dim3(1), dim3(64), 0, 0, da, dd, kg, sbk);
+      HIP_CHECK(hipDeviceSynchronize());
+      HIP_CHECK(hipMemcpy(hd.data(), dd, M * M * sizeof(float), hipMemcpyDeviceToHost));
+      printf("A kg%d sb%d: tot=%5.1f bits:", kg, sbk, hd[0 * M + 7]);
+      for (int r = 0; r < 7; ++r) printf(" %5.1f", hd[0 * M + r]);
+      printf("\n");
+    }
 
   printf("== scale-byte map: all-ones data in kg, scale byte sb = x2 ==\n");
   printf("   (for each (kg, sb): bit-plane sums s[r] = sum scale(k)*bit_r(k))\n");
