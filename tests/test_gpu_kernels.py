@@ -367,7 +367,7 @@ def test_topk_recall_threshold_fp4():
     ref = torch.topk(Q.float() @ X.float().T, k, dim=1)
     ref_ids = ref.indices.cpu().numpy()
     X4 = g.to_fp4_mx(X)
-    scores, ids = g.topk_recall_threshold(Q, X, k, X4=X4)
+    scores, ids = g.topk_recall_threshold(Q, X, k, X4=X4, q4=False)
     ids_np = ids.cpu().numpy()
     vals = scores.cpu().numpy()
     for q in range(nq):

@@ -368,6 +368,11 @@ def topk_recall_threshold(
             else:
                 hi = mid
         z = (lo + hi) / 2
+    # fp4 scan scores carry quantization noise (~0.2 sigma): lower theta
+    # so near-threshold true candidates still land in the buffer, and
+    # widen the rescore window below (select is by NOISY scan score).
+    if use_fp4:
+        z = z - 0.25
     theta = (mu + sigma * z).contiguous()
 
     # 2. fixed-threshold scan
@@ -394,8 +399,9 @@ def topk_recall_threshold(
     valid = slot < counts.clamp_max(cap).unsqueeze(1)
     cs = torch.where(valid, cs, torch.full_like(cs, -1e30))
 
-    # 4. select: top candidates by scan score
-    sel = min(cap, max(2 * k, 32))
+    # 4. select: top candidates by scan score (wider for the noisier
+    # fp4 scan; the exact rescore makes overfetch nearly free)
+    sel = min(cap, max(4 * k, 128)) if use_fp4 else min(cap, max(2 * k, 32))
     top = torch.topk(cs, sel, dim=1)
     ids = torch.gather(ci, 1, top.indices)
 
