@@ -1098,8 +1098,8 @@ topk_scan_fp4_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__
                      int32_t* __restrict__ cand_ids,
                      const float* __restrict__ theta,
                      int32_t* __restrict__ tc_n, int cap) {
-  __shared__ bf16 lds_q[2 * BM * BK];       // 2 pair-bufs x 16 KB
-  __shared__ bf16 lds_x[2 * BN * BK];
+  __shared__ bf16 lds_q[3 * BM * BK];       // 3 pair-bufs x 16 KB
+  __shared__ bf16 lds_x[3 * BN * BK];
   __shared__ uint8_t lds_qs[BM * MX4_SMAX];
   __shared__ uint8_t lds_xs2[BN * MX4_SMAX];
   __shared__ float row_min[BM];
@@ -1135,8 +1135,8 @@ topk_scan_fp4_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__
     stage_scale_rows(QS, sb, row0, (long long)nq, lds_qs, BM);
     stage_scale_rows(XS, sb, x0, (long long)nx, lds_xs2, BN);
     for (int pp = 0; pp < 2 && pp < np; ++pp) {
-      stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, pp * 32, QP4(pp & 1), BM);
-      stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, pp * 32, XP4(pp & 1), BN);
+      stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, pp * 32, QP4(pp), BM);
+      stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, pp * 32, XP4(pp), BN);
     }
     for (int p = 0; p < np; ++p) {
       if (p + 1 < np)
@@ -1144,11 +1144,20 @@ topk_scan_fp4_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
+      // three pair-buffers: the stage target (p+2)%3 is disjoint from
+      // both live pairs, so one barrier per pair suffices (the barrier
+      // orders everyone's pair p-1 reads before its buffer is rewritten)
+      if (2 * p + 4 < 2 * np) {
+        stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, (p + 2) * 32,
+                   QP4((p + 2) % 3), BM);
+        stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, (p + 2) * 32,
+                   XP4((p + 2) % 3), BN);
+      }
       uint32_t xrow_base = (uint32_t)(wn * 64 + lrow);
       uint32_t qrow_base = (uint32_t)(wm * 128 + lrow);
-      uint32_t xaddr = (uint32_t)(size_t)XP4(p & 1)
+      uint32_t xaddr = (uint32_t)(size_t)XP4(p % 3)
                        + lds_off_bytes(xrow_base, (uint32_t)kgrp);
-      uint32_t qaddr = (uint32_t)(size_t)QP4(p & 1)
+      uint32_t qaddr = (uint32_t)(size_t)QP4(p % 3)
                        + lds_off_bytes(qrow_base, (uint32_t)kgrp);
       uint32_t sst = 16u * (uint32_t)sb;
       uint32_t xs_a = (uint32_t)(size_t)lds_xs2 + xrow_base * (uint32_t)sb
@@ -1196,13 +1205,6 @@ topk_scan_fp4_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__
           : "v"(qaddr), "v"(qs_a), "v"(qs_a + sst), "v"(qs_a + 2 * sst),
             "v"(qs_a + 3 * sst), "v"(qs_a + 4 * sst), "v"(qs_a + 5 * sst),
             "v"(qs_a + 6 * sst), "v"(qs_a + 7 * sst));
-      __builtin_amdgcn_s_barrier();
-      if (2 * p + 4 < 2 * np) {
-        stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, (p + 2) * 32,
-                   QP4(p & 1), BM);
-        stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, (p + 2) * 32,
-                   XP4(p & 1), BN);
-      }
       v8i_mx xv[4], qv[8];
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
