@@ -302,7 +302,7 @@ std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
   CHECK_GPU(theta); CHECK_CONTIG(theta);
   int nq = Q4.size(0), D = (int)Q4.size(1) * 2;
   long long nx = X4.size(0);
-  TORCH_CHECK(D % 128 == 0 && D / 32 <= 64);
+  TORCH_CHECK(D % 128 == 0 && D / 32 <= 32, "fp4x4 kernel caps at D=1024");
   TORCH_CHECK(QS.size(0) == nq && QS.size(1) == D / 32);
   TORCH_CHECK(X4.size(1) == D / 2 && XS.size(0) == nx && XS.size(1) == D / 32);
   TORCH_CHECK(theta.numel() == nq && cap >= 32 && cap <= 4096);

@@ -1098,10 +1098,14 @@ topk_scan_fp4_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__
                      int32_t* __restrict__ cand_ids,
                      const float* __restrict__ theta,
                      int32_t* __restrict__ tc_n, int cap) {
-  __shared__ bf16 lds_q[3 * BM * BK];       // 3 pair-bufs x 16 KB
-  __shared__ bf16 lds_x[3 * BN * BK];
-  __shared__ uint8_t lds_qs[BM * MX4_SMAX];
-  __shared__ uint8_t lds_xs2[BN * MX4_SMAX];
+  // 4 pair-bufs x 16 KB each side: staging runs THREE pairs ahead
+  // (stage target (p+3)%4 is disjoint from pairs p..p+2). Scale sheets
+  // shrink to 32 B/row, capping this kernel at D = 1024 (the binding
+  // routes larger D to the fp8xfp4 kernel).
+  __shared__ bf16 lds_q[4 * BM * BK];
+  __shared__ bf16 lds_x[4 * BN * BK];
+  __shared__ uint8_t lds_qs[BM * 32];
+  __shared__ uint8_t lds_xs2[BN * 32];
   __shared__ float row_min[BM];
 #define QP4(buf) (lds_q + (buf) * BM * BK)
 #define XP4(buf) (lds_x + (buf) * BN * BK)
@@ -1134,30 +1138,32 @@ topk_scan_fp4_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__
     f32x4 acc[8][4] = {};
     stage_scale_rows(QS, sb, row0, (long long)nq, lds_qs, BM);
     stage_scale_rows(XS, sb, x0, (long long)nx, lds_xs2, BN);
-    for (int pp = 0; pp < 2 && pp < np; ++pp) {
+    for (int pp = 0; pp < 3 && pp < np; ++pp) {
       stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, pp * 32, QP4(pp), BM);
       stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, pp * 32, XP4(pp), BN);
     }
     for (int p = 0; p < np; ++p) {
-      if (p + 1 < np)
+      // pairs p+1, p+2 may still be in flight (4 glds each)
+      if (p + 2 < np)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else if (p + 1 < np)
         asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
-      // three pair-buffers: the stage target (p+2)%3 is disjoint from
-      // both live pairs, so one barrier per pair suffices (the barrier
-      // orders everyone's pair p-1 reads before its buffer is rewritten)
-      if (2 * p + 4 < 2 * np) {
-        stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, (p + 2) * 32,
-                   QP4((p + 2) % 3), BM);
-        stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, (p + 2) * 32,
-                   XP4((p + 2) % 3), BN);
+      // stage three pairs ahead: target (p+3)%4 touches only the buffer
+      // consumed at pair p-1, so the single barrier suffices
+      if (2 * p + 6 < 2 * np) {
+        stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, (p + 3) * 32,
+                   QP4((p + 3) & 3), BM);
+        stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, (p + 3) * 32,
+                   XP4((p + 3) & 3), BN);
       }
       uint32_t xrow_base = (uint32_t)(wn * 64 + lrow);
       uint32_t qrow_base = (uint32_t)(wm * 128 + lrow);
-      uint32_t xaddr = (uint32_t)(size_t)XP4(p % 3)
+      uint32_t xaddr = (uint32_t)(size_t)XP4(p & 3)
                        + lds_off_bytes(xrow_base, (uint32_t)kgrp);
-      uint32_t qaddr = (uint32_t)(size_t)QP4(p % 3)
+      uint32_t qaddr = (uint32_t)(size_t)QP4(p & 3)
                        + lds_off_bytes(qrow_base, (uint32_t)kgrp);
       uint32_t sst = 16u * (uint32_t)sb;
       uint32_t xs_a = (uint32_t)(size_t)lds_xs2 + xrow_base * (uint32_t)sb

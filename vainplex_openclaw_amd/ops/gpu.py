@@ -376,7 +376,7 @@ def topk_recall_threshold(
     theta = (mu + sigma * z).contiguous()
 
     # 2. fixed-threshold scan
-    if use_fp4 and q4:
+    if use_fp4 and q4 and D <= 1024:
         # both operands MXFP4: scores carry no static prescale
         Q4, QS = to_fp4_mx(Q)
         cs, ci, counts = ext().topk_scan_threshold_fp4x4(
