@@ -32,6 +32,7 @@ class RedactionVault:
         self.clock = clock
         self._entries: Dict[str, Dict] = {}  # full hash -> entry
         self._hash_index: Dict[str, List[str]] = {}  # hash8 -> [full hashes]
+        self._by_placeholder: Dict[str, str] = {}  # placeholder -> full hash
 
     def store(self, original: str, category: str) -> str:
         full = _sha256(original)
@@ -61,6 +62,7 @@ class RedactionVault:
         bucket = self._hash_index.setdefault(h8, [])
         if full not in bucket:
             bucket.append(full)
+        self._by_placeholder[placeholder] = full
         return placeholder
 
     def resolve(self, text: str) -> str:
@@ -68,20 +70,20 @@ class RedactionVault:
         now_ms = self.clock() * 1000
 
         def sub(m: "re.Match[str]") -> str:
-            slice_ = m.group(1)
-            for entry in self._entries.values():
-                if entry["hash"].startswith(slice_) and entry["expiresAt"] > now_ms:
-                    if entry["placeholder"] == m.group(0):
-                        return entry["original"]
+            full = self._by_placeholder.get(m.group(0))
+            entry = self._entries.get(full) if full else None
+            if entry and entry["expiresAt"] > now_ms:
+                return entry["original"]
             return m.group(0)
 
         return PLACEHOLDER_RX.sub(sub, text)
 
     def lookup(self, placeholder: str) -> Optional[str]:
         now_ms = self.clock() * 1000
-        for entry in self._entries.values():
-            if entry["placeholder"] == placeholder and entry["expiresAt"] > now_ms:
-                return entry["original"]
+        full = self._by_placeholder.get(placeholder)
+        entry = self._entries.get(full) if full else None
+        if entry and entry["expiresAt"] > now_ms:
+            return entry["original"]
         return None
 
     def evict_expired(self) -> int:
@@ -92,12 +94,16 @@ class RedactionVault:
             bucket = self._hash_index.get(e["hash"][:8])
             if bucket and h in bucket:
                 bucket.remove(h)
+            self._by_placeholder.pop(e["placeholder"], None)
         return len(expired)
 
     @property
     def size(self) -> int:
-        return len(self._entries)
+        """Non-expired entries only (vault.ts size semantics)."""
+        now_ms = self.clock() * 1000
+        return sum(1 for e in self._entries.values() if e["expiresAt"] > now_ms)
 
     def clear(self) -> None:
         self._entries.clear()
         self._hash_index.clear()
+        self._by_placeholder.clear()
