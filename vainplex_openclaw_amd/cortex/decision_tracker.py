@@ -102,6 +102,13 @@ class DecisionTracker:
     def recent(self, n: int = 10) -> List[Dict[str, Any]]:
         return self.decisions[-n:]
 
+    def recent_within(self, days: float = 7.0, limit: int = 10) -> List[Dict[str, Any]]:
+        """Decisions from the last `days`, newest last, capped at `limit`
+        (decision-tracker.ts getRecentDecisions)."""
+        cutoff = (self._now_dt() - _dt.timedelta(days=days)).date().isoformat()
+        recent = [d for d in self.decisions if str(d.get("date", "")) >= cutoff]
+        return recent[-limit:]
+
     def persist(self) -> None:
         if not self.writeable:
             return
