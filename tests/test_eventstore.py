@@ -311,3 +311,80 @@ def test_envelope_visibility_default_and_values():
     from vainplex_openclaw_amd.eventstore.events import VISIBILITIES
 
     assert VISIBILITIES == ("public", "internal", "confidential", "secret")
+
+
+# -- hook-mappings.test.ts per-hook table ---------------------------------
+
+HOOK_TABLE = [
+    # (hook, event, canonical type, legacy type, payload key checks)
+    ("message_received", {"content": "hi", "channel": "slack"},
+     "message.in.received", "msg.in", {}),
+    ("message_sending", {"content": "out"}, "message.out.sending", "msg.sending", {}),
+    ("message_sent", {"content": "sent"}, "message.out.sent", "msg.out", {}),
+    ("before_tool_call", {"toolName": "exec", "params": {"c": "ls"}},
+     "tool.call.requested", "tool.call", {"toolName": "exec"}),
+    ("after_tool_call", {"toolName": "exec", "result": "ok", "durationMs": 5},
+     "tool.call.executed", "tool.result", {"durationMs": 5}),
+    ("after_tool_call", {"toolName": "exec", "error": "boom"},
+     "tool.call.failed", "tool.result", {"error": "boom"}),
+    ("before_agent_start", {"prompt": "go"}, "run.started", "run.start",
+     {"prompt": "go"}),
+    ("agent_end", {"success": True, "durationMs": 9, "messages": [1, 2]},
+     "run.ended", "run.end", {"messageCount": 2, "success": True}),
+    ("llm_input", {"systemPrompt": "spsp", "prompt": "pp", "historyMessages": [1]},
+     "model.input.observed", "llm.input", {}),
+    ("llm_output", {"assistantTexts": ["aaaa", "bb"]},
+     "model.output.observed", "llm.output", {}),
+    ("before_compaction", {"messageCount": 10, "compactingCount": 4, "tokenCount": 99},
+     "session.compaction.started", "session.compaction_start", {"messageCount": 10}),
+    ("after_compaction", {"messageCount": 6, "compactedCount": 4, "tokenCount": 50},
+     "session.compaction.ended", "session.compaction_end", {"compactedCount": 4}),
+    ("before_reset", {"reason": "manual"}, "session.reset", None, {"reason": "manual"}),
+    ("session_start", {"sessionId": "s1"}, "session.started", "session.start",
+     {"sessionId": "s1"}),
+    ("session_end", {"sessionId": "s1", "messageCount": 3, "durationMs": 7},
+     "session.ended", "session.end", {"messageCount": 3}),
+    ("gateway_start", {"port": 8080}, "gateway.started", "gateway.start", {"port": 8080}),
+    ("gateway_stop", {"reason": "shutdown"}, "gateway.stopped", "gateway.stop", {}),
+]
+
+
+@pytest.mark.parametrize("hook,event,ctype,legacy,payload_checks", HOOK_TABLE,
+                         ids=[f"{h}-{t}" for h, _, t, _, _ in HOOK_TABLE])
+def test_hook_mapping_table(hook, event, ctype, legacy, payload_checks):
+    mapping = next(m for m in HOOK_MAPPINGS
+                   if m.hook_name == hook and m.resolve_type(event, {}) == ctype)
+    assert mapping.legacy_type == legacy
+    payload = mapping.mapper(event, {})
+    for key, want in payload_checks.items():
+        assert payload.get(key) == want, (key, payload)
+
+
+def test_llm_mappings_privacy_lengths_only():
+    m_in = next(m for m in HOOK_MAPPINGS if m.hook_name == "llm_input")
+    p = m_in.mapper({"systemPrompt": "abcd", "prompt": "xy",
+                      "historyMessages": [1, 2, 3]}, {})
+    dumped = json.dumps(p)
+    assert "abcd" not in dumped and "xy" not in dumped
+    assert p.get("systemPromptLength") == 4 or p.get("systemPromptChars") == 4
+    m_out = next(m for m in HOOK_MAPPINGS if m.hook_name == "llm_output")
+    p2 = m_out.mapper({"assistantTexts": ["hello", "hi!"]}, {})
+    assert "hello" not in json.dumps(p2)
+    # graceful on missing assistantTexts
+    p3 = m_out.mapper({}, {})
+    assert isinstance(p3, dict)
+    p4 = m_out.mapper({"assistantTexts": None}, {})
+    assert isinstance(p4, dict)
+
+
+def test_gateway_events_marked_system():
+    for hook in ("gateway_start", "gateway_stop"):
+        m = next(m for m in HOOK_MAPPINGS if m.hook_name == hook)
+        assert m.system_event is True
+
+
+def test_run_failed_only_on_failure():
+    ee = EXTRA_EMITTERS[0]
+    assert ee.hook_name == "agent_end" and ee.legacy_type == "run.error"
+    assert ee.condition({"success": False})
+    assert not ee.condition({"success": True})
