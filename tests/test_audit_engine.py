@@ -289,3 +289,116 @@ def test_resolve_governance_config_overrides_and_gates():
     assert cfg["builtinPolicies"]["nightMode"] is True
     # 2fa without secret stays None
     assert resolve_config({"approval2fa": {"enabled": True}})["approval2fa"] is None
+
+
+# -- engine.test.ts mirrors ------------------------------------------------
+
+def _engine(workspace, **cfg):
+    base = {"trust": {"enabled": True, "defaultScore": 40}}
+    base.update(cfg)
+    return GovernanceEngine(base, workspace)
+
+
+def test_engine_start_stop_and_status(workspace):
+    e = _engine(workspace)
+    e.start()
+    try:
+        e.evaluate(e.build_context("before_tool_call", "a1", tool_name="read"))
+        st = e.status()
+        assert st["stats"]["evaluations"] == 1
+        assert isinstance(st["policies"], list)
+        assert "a1" in st["agents"]
+        assert "audit" in st and "crossAgent" in st
+    finally:
+        e.stop()
+
+
+def test_engine_allow_when_no_policies(workspace):
+    e = _engine(workspace)
+    e.start()
+    try:
+        v = e.evaluate(e.build_context("before_tool_call", "a1", tool_name="read"))
+        assert v["action"] == "allow"
+    finally:
+        e.stop()
+
+
+def test_engine_record_outcomes_success_only_on_record(workspace):
+    """Bug 3: evaluate() allow must NOT bump successCount; only an
+    explicit record_outcome(success=True) does."""
+    e = _engine(workspace)
+    e.start()
+    try:
+        e.evaluate(e.build_context("before_tool_call", "a1", tool_name="read"))
+        assert e.trust_manager.get("a1")["signals"]["successCount"] == 0
+        e.record_outcome("a1", "agent:a1", True)
+        assert e.trust_manager.get("a1")["signals"]["successCount"] == 1
+    finally:
+        e.stop()
+
+
+def test_engine_denial_increments_violation_but_not_night(workspace):
+    e = _engine(workspace, builtinPolicies={"credentialGuard": True})
+    e.start()
+    try:
+        e.evaluate(e.build_context("before_tool_call", "a1", tool_name="read",
+                                   tool_params={"file_path": "/app/.env"}))
+        assert e.trust_manager.get("a1")["signals"]["violationCount"] == 1
+    finally:
+        e.stop()
+
+
+def test_engine_denial_audit_includes_reason_and_policy_controls(workspace):
+    """Bug 2 + Bug 4: the audit record carries the aggregated reason and
+    controls derived from the MATCHED POLICY (plus the deny baseline),
+    not from the hook name."""
+    e = _engine(workspace, builtinPolicies={"credentialGuard": True})
+    e.start()
+    try:
+        v = e.evaluate(e.build_context("before_tool_call", "a1", tool_name="read",
+                                       tool_params={"file_path": "/x/.env"}))
+        assert v["action"] == "deny" and v["reason"]
+        e.audit_trail.flush()
+        rec = e.audit_trail.query({"verdict": "deny"})[0]
+        assert rec["reason"] == v["reason"]
+        assert "A.5.24" in rec["controls"] and "A.5.28" in rec["controls"]
+        # the credential guard's own ISO tags ride along
+        policy_controls = [c for c in rec["controls"] if c not in ("A.5.24", "A.5.28")]
+        assert policy_controls, rec["controls"]
+        # allow records carry a reason too
+        e.evaluate(e.build_context("before_tool_call", "a1", tool_name="read",
+                                   tool_params={"file_path": "/x/notes.md"}))
+        e.audit_trail.flush()
+        allows = e.audit_trail.query({"verdict": "allow"})
+        assert allows and allows[0]["reason"]
+    finally:
+        e.stop()
+
+
+def test_engine_evaluation_stats_running_average(workspace):
+    e = _engine(workspace)
+    e.start()
+    try:
+        for _ in range(3):
+            e.evaluate(e.build_context("before_tool_call", "a1", tool_name="read"))
+        assert e.stats["evaluations"] == 3
+        assert e.stats["allows"] == 3
+        assert e.stats["avgEvaluationUs"] > 0
+    finally:
+        e.stop()
+
+
+def test_engine_known_agents_auto_registered_and_kept(workspace):
+    e = _engine(workspace, trust={"enabled": True, "defaults": {"main": 60, "*": 25}})
+    e.start()
+    try:
+        e.set_known_agents(["main", "forge"])
+        assert e.trust_manager.score("main") == 60
+        assert e.trust_manager.score("forge") == 25  # wildcard default
+        # unspecified agent also gets the wildcard
+        assert e.trust_manager.score("random") == 25
+        # removing an agent keeps its trust data
+        e.set_known_agents(["main"])
+        assert "forge" in e.trust_manager.get_store()["agents"]
+    finally:
+        e.stop()
