@@ -211,3 +211,38 @@ def test_llm_enhance_parse_and_merge(workspace):
     tt.apply_llm_analysis(analysis)
     assert any(t["title"] == "vector search rollout" for t in tt.get_threads())
     assert parse_analysis("not json") is None
+
+
+def test_thread_tracker_dedupe_mood_and_decisions(workspace):
+    tt = ThreadTracker(workspace)
+    tt.process_message("let's talk about the billing system rework", "user")
+    tt.process_message("more thoughts regarding the billing system rework", "user")
+    assert len(tt.get_threads()) == 1  # same topic -> no duplicate
+    # decision appended to the matching thread
+    tt.process_message("we decided the billing system rework ships in Q2", "user")
+    th = tt.get_threads()[0]
+    assert th.get("decisions"), th
+    # mood propagates to matching threads and the session
+    tt.process_message("the billing system rework is broken and annoying", "user")
+    assert tt.session_mood == "frustrated"
+    assert tt.get_threads()[0].get("mood") == "frustrated"
+    assert tt.events_processed == 4
+    # empty content skipped
+    tt.process_message("", "user")
+    assert tt.events_processed == 4
+
+
+def test_thread_tracker_loads_existing_state(workspace):
+    tt = ThreadTracker(workspace)
+    tt.process_message("let's talk about the observability dashboard", "user")
+    tt.flush()
+    tt2 = ThreadTracker(workspace)
+    assert any("observability" in t["title"] for t in tt2.get_threads())
+    assert tt2.session_mood in ("neutral", "exploratory", "productive")
+
+
+def test_extract_signals_families():
+    s = extract_signals("we decided to use rust. the migration is done. "
+                        "waiting for the security review. let's talk about budgets.")
+    assert s["decisions"] and s["closures"] and s["waits"] and s["topics"]
+    assert extract_signals("nothing interesting here")["decisions"] == []
