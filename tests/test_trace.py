@@ -65,9 +65,13 @@ def test_chain_reconstruction_splits():
     # different agent bucket
     events.append(ev("msg.in", agent="other", content="x"))
     chains = reconstruct_chains(events)
-    sessions = [(c.agent, len(c.events)) for c in chains]
-    assert len(chains) == 4
-    assert ("other", 1) in sessions
+    # 1-event chains ("later" alone, "other" agent, post-lifecycle tail)
+    # are dropped per the reference's min-2-events rule
+    assert len(chains) == 2
+    assert all(len(c.events) >= 2 for c in chains)
+    # with the filter disabled the old buckets reappear
+    raw = reconstruct_chains(events, min_events=1)
+    assert ("other", 1) in [(c.agent, len(c.events)) for c in raw]
     # dedupe by id
     dup = events[0]
     chains2 = reconstruct_chains(events + [dup])
@@ -138,7 +142,8 @@ def test_hallucination_and_unverified():
 
 
 def test_redactor_strips_credentials():
-    events = [ev("msg.in", content="token ghp_" + "f" * 36 + " here")]
+    events = [ev("msg.in", content="token ghp_" + "f" * 36 + " here"),
+              ev("msg.out", content="noted")]
     chains = reconstruct_chains(events)
     red = ChainRedactor().redact_chain(chains[0])
     assert "ghp_" + "f" * 36 not in str(red)
@@ -238,3 +243,58 @@ def test_correction_and_dissatisfied_i18n(lang):
     kinds = {f.signal_type for f in detect_all_signals(chains, ["correction", "dissatisfied"])}
     assert "correction" in kinds, lang
     assert "dissatisfied" in kinds, lang
+
+
+def test_chain_run_boundary_split_and_metadata():
+    """run.end -> run.start with >5 min gap splits; <5 min does not;
+    boundary_type and type_counts mirror chain-reconstructor.ts."""
+    events = [
+        ev("msg.in", content="a"),
+        ev("run.ended", gap_ms=1000),
+        ev("run.started", gap_ms=6 * 60 * 1000),  # >5 min after run end
+        ev("msg.out", content="b", gap_ms=1000),
+    ]
+    chains = reconstruct_chains(events)
+    assert len(chains) == 2
+    assert chains[0].type_counts == {"msg.in": 1, "run.ended": 1}
+    assert chains[0].boundary_type == "gap"
+
+    # <5 min: stays one chain
+    events2 = [
+        ev("msg.in", content="a"),
+        ev("run.ended", gap_ms=1000),
+        ev("run.started", gap_ms=2 * 60 * 1000),
+        ev("msg.out", content="b", gap_ms=1000),
+    ]
+    assert len(reconstruct_chains(events2)) == 1
+
+    # lifecycle boundary marks the chain
+    events3 = [
+        ev("session.start"),
+        ev("msg.in", content="x", gap_ms=100),
+        ev("msg.out", content="y", gap_ms=100),
+    ]
+    c3 = reconstruct_chains(events3)
+    assert c3[0].boundary_type == "lifecycle"
+
+
+def test_chain_configurable_gap_and_cap():
+    events = [ev("msg.in", content="a"), ev("msg.out", content="b", gap_ms=10 * 60 * 1000),
+              ev("msg.in", content="c", gap_ms=10 * 60 * 1000),
+              ev("msg.out", content="d", gap_ms=10 * 60 * 1000)]
+    assert len(reconstruct_chains(events)) == 1  # default 30-min gap
+    small = reconstruct_chains(events, gap_minutes=5.0)
+    assert len(small) == 0 or all(len(c.events) >= 2 for c in small)
+    # cap: 5 events with max_events=2 -> 2-event chains
+    burst = [ev("msg.in", content=str(i), gap_ms=10) for i in range(6)]
+    capped = reconstruct_chains(burst, max_events=2)
+    assert all(len(c.events) == 2 for c in capped)
+    assert sum(len(c.events) for c in capped) == 6
+
+
+def test_chain_deterministic_ids_and_empty():
+    assert reconstruct_chains([]) == []
+    events = [ev("msg.in", content="a"), ev("msg.out", content="b", gap_ms=50)]
+    a = reconstruct_chains(list(events))
+    b = reconstruct_chains(list(events))
+    assert a[0].id == b[0].id and len(a[0].id) == 16
