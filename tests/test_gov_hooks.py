@@ -315,3 +315,47 @@ def test_session_end_clears_toolcall_log(workspace):
         assert not plugin.hooks.tool_call_log.get("agent:a1")
     finally:
         gw.stop()
+
+
+def test_2fa_reference_behaviors():
+    """approval-2fa.test.ts mirrors: invalid codes, per-session batches,
+    attempt rate limiting, notify with all commands, pending edges."""
+    t = [1_700_000_000.0]
+    notified = []
+    ap = Approval2FA(clock=lambda: t[0], notify=lambda b: notified.append(b),
+                     max_attempts=3, cooldown_s=60)
+    # no_pending edge: resolving with nothing queued
+    assert ap.try_resolve_any(totp_at(ap.secret, t[0])) == []
+    assert not ap.has_pending_batch()
+    # different agents/sessions -> separate batches; same session -> one
+    ap.request("agent:a", "a", "deploy prod")
+    ap.request("agent:a", "a", "push tags")
+    ap.request("agent:b", "b", "rm -rf data")
+    assert ap.has_pending_batch("agent:a") and ap.has_pending_batch("agent:b")
+    assert len(notified) == 2  # one notification per batch
+    batch_a = next(b for b in notified if b["sessionKey"] == "agent:a")
+    assert [r["reason"] for r in batch_a["requests"]][:1] == ["deploy prod"]
+    # invalid code rejected; 3 bad attempts trip the cooldown
+    t[0] += 40
+    assert ap.try_resolve_any("000001") == []
+    assert ap.try_resolve_any("000002") == []
+    assert ap.try_resolve_any("999999") == []
+    assert ap.in_attempt_cooldown()
+    good = totp_at(ap.secret, t[0])
+    assert ap.try_resolve_any(good) == []  # even a valid code during cooldown
+    t[0] += 61
+    good2 = totp_at(ap.secret, t[0])
+    resolved = ap.try_resolve_any(good2)
+    assert len(resolved) == 3 and all(r["status"] == "approved" for r in resolved)
+    assert not ap.has_pending_batch()
+
+
+def test_2fa_timeout_denies_all():
+    t = [0.0]
+    ap = Approval2FA(clock=lambda: t[0], timeout_s=30)
+    ap.request("agent:a", "a", "one")
+    ap.request("agent:b", "b", "two")
+    t[0] = 31.0
+    expired = ap.expire_stale()
+    assert len(expired) == 2 and all(r["status"] == "expired" for r in expired)
+    assert ap.pending_requests() == []

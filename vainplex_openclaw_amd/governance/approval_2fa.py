@@ -78,12 +78,16 @@ class Approval2FA:
         clock=time.time,
         timeout_s: float = DEFAULT_TIMEOUT_S,
         cooldown_s: float = DEFAULT_COOLDOWN_S,
+        max_attempts: int = 5,
     ):
         self.secret = secret or generate_secret()
         self.notify = notify
         self.clock = clock
         self.timeout_s = timeout_s
         self.cooldown_s = cooldown_s
+        self.max_attempts = max_attempts
+        self._failed_attempts = 0
+        self._attempt_cooldown_until = 0.0
         self._lock = threading.Lock()
         self._pending: Dict[str, Dict[str, Any]] = {}  # batchId -> batch
         self._used_counters: set = set()
@@ -156,9 +160,17 @@ class Approval2FA:
     def try_resolve_any(self, code: str, approve: bool = True) -> List[Dict[str, Any]]:
         """A valid TOTP code resolves ALL pending batches' requests."""
         now = self.clock()
+        # rate limiting: too many bad codes trips a cooldown
+        if now < self._attempt_cooldown_until:
+            return []
         counter = verify_totp(self.secret, code, now)
         if counter is None:
+            self._failed_attempts += 1
+            if self._failed_attempts >= self.max_attempts:
+                self._attempt_cooldown_until = now + self.cooldown_s
+                self._failed_attempts = 0
             return []
+        self._failed_attempts = 0
         with self._lock:
             if counter in self._used_counters:  # replay protection
                 return []
@@ -188,6 +200,17 @@ class Approval2FA:
                             expired.append(req)
                     del self._pending[bid]
         return expired
+
+    def in_attempt_cooldown(self) -> bool:
+        return self.clock() < self._attempt_cooldown_until
+
+    def has_pending_batch(self, session_key: Optional[str] = None) -> bool:
+        with self._lock:
+            return any(
+                (session_key is None or b["sessionKey"] == session_key)
+                and any(r["status"] == "pending" for r in b["requests"])
+                for b in self._pending.values()
+            )
 
     def pending_requests(self) -> List[Dict[str, Any]]:
         with self._lock:
