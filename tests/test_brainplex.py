@@ -182,3 +182,72 @@ def test_run_init_idempotent(tmp_path):
                   start_dir=str(ws), home=str(tmp_path), echo=lambda *a: None)
     assert (ws / "openclaw.json").read_text() == first
     assert r2["written"]["written"] == []  # all kept
+
+
+# -- writer.test.ts mirrors ------------------------------------------------
+
+def test_atomic_write_no_tmp_left_and_indentation(tmp_path):
+    from vainplex_openclaw_amd.utils.storage import atomic_write_text
+
+    path = tmp_path / "deep" / "nested" / "cfg.json"
+    atomic_write_text(str(path), json.dumps({"a": {"b": 1}}, indent=2) + "\n")
+    assert path.is_file()  # parents created recursively
+    text = path.read_text()
+    assert '  "a"' in text  # 2-space indentation
+    leftovers = [f for f in os.listdir(path.parent) if ".tmp" in f or f.endswith("~")]
+    assert leftovers == []
+
+
+def test_backup_then_write_semantics(tmp_path):
+    from vainplex_openclaw_amd.utils.storage import backup_then_write
+
+    path = tmp_path / "openclaw.json"
+    # non-existent file: no backup created
+    bak = backup_then_write(str(path), "{}\n")
+    assert bak is None
+    assert path.read_text() == "{}\n"
+    # existing file: .bak holds the OLD content
+    bak2 = backup_then_write(str(path), '{"new": 1}\n')
+    assert bak2 and bak2.endswith(".bak")
+    assert open(bak2).read() == "{}\n"
+    assert json.loads(path.read_text()) == {"new": 1}
+
+
+def test_write_configs_dry_run_writes_nothing(tmp_path):
+    from vainplex_openclaw_amd.brainplex.writer import write_configs
+
+    res = write_configs([{"pluginId": "p1", "config": {"x": 1}}],
+                        home=str(tmp_path), dry_run=True)
+    assert res["written"] == ["p1"]
+    assert not (tmp_path / ".openclaw" / "plugins" / "p1" / "config.json").exists()
+
+
+def test_update_config_preserves_existing_and_originals(tmp_path):
+    from vainplex_openclaw_amd.brainplex.writer import update_openclaw_config
+
+    path = tmp_path / "openclaw.json"
+    original = {"plugins": {"entries": {"keep": {"enabled": False}},
+                            "allow": ["keep"]}, "other": {"setting": 1}}
+    path.write_text(json.dumps(original))
+    res = update_openclaw_config(str(path), original, ["keep", "new-plugin"])
+    cfg = res["config"]
+    # existing entry untouched, new one added, unrelated keys preserved
+    assert cfg["plugins"]["entries"]["keep"] == {"enabled": False}
+    assert cfg["plugins"]["entries"]["new-plugin"] == {"enabled": True}
+    assert cfg["plugins"]["allow"] == ["keep", "new-plugin"]
+    assert cfg["other"] == {"setting": 1}
+    assert res["added_entries"] == ["new-plugin"]
+    # the ORIGINAL dict was not mutated
+    assert "new-plugin" not in original["plugins"]["entries"]
+    # backup got created with old content
+    assert res["backed_up"]
+
+
+def test_update_config_dry_run_no_write(tmp_path):
+    from vainplex_openclaw_amd.brainplex.writer import update_openclaw_config
+
+    path = tmp_path / "openclaw.json"
+    path.write_text("{}")
+    res = update_openclaw_config(str(path), {}, ["p"], dry_run=True)
+    assert res["updated"] and not res["backed_up"]
+    assert path.read_text() == "{}"
