@@ -142,3 +142,84 @@ def test_custom_pattern_and_redos_rejection():
     assert "custom-ticket" in ids
     assert "custom-bad" not in ids
     assert any(m.pattern.id == "custom-ticket" for m in reg.find_matches("see TICKET-1234"))
+
+
+# -- redaction/hooks.test.ts mirrors --------------------------------------
+
+def _state(cfg=None):
+    from vainplex_openclaw_amd.governance.redaction.hooks import RedactionState
+    return RedactionState(cfg or {"enabled": True})
+
+
+def test_hooks_vault_resolution_roundtrip():
+    st = _state()
+    # a credential redacted at layer 1 ...
+    out = st.on_tool_result_persist({"toolName": "exec",
+                                     "result": "key ghp_" + "f" * 36})
+    ph = out["result"].split("key ")[1]
+    assert ph.startswith("[REDACTED:credential:")
+    # ... resolves back when used in a later tool call
+    res = st.on_before_tool_call({"toolName": "exec",
+                                  "params": {"cmd": f"use {ph} now", "n": 1}})
+    assert res["params"]["cmd"] == "use ghp_" + "f" * 36 + " now"
+    assert res["params"]["n"] == 1
+    # nested structures resolve too
+    res2 = st.on_before_tool_call({"params": {"a": [f"x {ph}"], "b": {"c": ph}}})
+    assert "ghp_" in res2["params"]["a"][0] and "ghp_" in res2["params"]["b"]["c"]
+
+
+def test_hooks_block_on_unresolvable_placeholder():
+    st = _state()
+    res = st.on_before_tool_call({"params": {"cmd": "use [REDACTED:credential:deadbeef] now"}})
+    assert res["block"] and "Unresolvable" in res["blockReason"]
+    # clean params pass through untouched
+    assert st.on_before_tool_call({"params": {"cmd": "plain"}}) is None
+    assert st.on_before_tool_call({"params": None}) is None
+
+
+def test_hooks_pii_channel_allowlist():
+    st = _state({"enabled": True, "allowlist": {"piiAllowedChannels": ["internal"]}})
+    msg = "mail bob@corp.io the key sk-abcdefghij0123456789XY"
+    # allowlisted channel: PII passes, credential still redacted
+    out = st.on_message_sending({"agentId": "a", "channel": "internal", "content": msg})
+    assert "bob@corp.io" in out["content"]
+    assert "sk-abcdefghij0123456789XY" not in out["content"]
+    # other channels redact both
+    out2 = st.on_message_sending({"agentId": "a", "channel": "public", "content": msg})
+    assert "bob@corp.io" not in out2["content"]
+    assert "sk-abcdefghij0123456789XY" not in out2["content"]
+
+
+def test_hooks_sync_write_gate_skips_pii_only():
+    st = _state()
+    # PII-only content passes the synchronous gate (RFC-007 §5.4)
+    assert st.on_before_message_write({"agentId": "a",
+                                       "content": "reach me at alice@x.io"}) is None
+    # credentials and financial data are still redacted
+    out = st.on_before_message_write({"agentId": "a",
+                                      "content": "card 4111 1111 1111 1111"})
+    assert "4111" not in out["content"]
+    out2 = st.on_before_message_write({"agentId": "a",
+                                       "content": "Bearer " + "t" * 24})
+    assert "t" * 24 not in out2["content"]
+    # empty / absent content
+    assert st.on_before_message_write({"agentId": "a", "content": ""}) is None
+    assert st.on_before_message_write({"agentId": "a"}) is None
+
+
+def test_hooks_layer1_nested_and_none_results():
+    st = _state()
+    assert st.on_tool_result_persist({"toolName": "t", "result": None}) is None
+    out = st.on_tool_result_persist({"toolName": "t",
+                                     "result": {"deep": ["ok", {"k": "glpat-" + "x" * 22}]}})
+    import json as _json
+    assert "glpat-" not in _json.dumps(out["result"])
+
+
+def test_hooks_registration_registers_four():
+    bus = HookBus()
+    api = PluginApi(id="r", plugin_config={}, logger=NullLogger(), config={}, bus=bus)
+    register_redaction_hooks(api, {"enabled": True})
+    for hook in ("tool_result_persist", "before_tool_call", "message_sending",
+                 "before_message_write"):
+        assert bus.handlers(hook), hook
