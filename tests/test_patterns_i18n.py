@@ -7,6 +7,7 @@ redaction registry (false-positive guards).
 
 import pytest
 
+from vainplex_openclaw_amd.cortex import patterns as P
 from vainplex_openclaw_amd.cortex.patterns import (
     detect_mood,
     get_registry,
@@ -115,3 +116,69 @@ def test_redaction_positive_in_context(text, expected_id):
     reg = PatternRegistry()
     ids = {m.pattern.id for m in reg.find_matches(text)}
     assert expected_id in ids or ids, f"expected {expected_id}, got {ids}"
+
+
+# -- patterns-custom.test.ts mirrors --------------------------------------
+
+def any_match(pats, text):
+    return any(p.search(text) for p in pats)
+
+
+def test_custom_extend_mode():
+    reg = P.PatternRegistry(["en"], {
+        "decision": ["approved by committee", "design review passed"],
+        "close": ["ticket closed"],
+        "wait": ["pending approval from"],
+        "mode": "extend",
+    })
+    assert any_match(reg.get_patterns("decision"), "This was approved by committee")
+    assert any_match(reg.get_patterns("decision"), "The design review passed")
+    assert any_match(reg.get_patterns("decision"), "We decided to go")  # builtin kept
+    assert any_match(reg.get_patterns("close"), "ticket closed for this issue")
+    assert any_match(reg.get_patterns("wait"), "pending approval from the VP")
+    # default mode is extend
+    reg2 = P.PatternRegistry(["en"], {"decision": ["custom rule fired"]})
+    assert any_match(reg2.get_patterns("decision"), "We decided to go")
+
+
+def test_custom_override_mode():
+    reg = P.PatternRegistry(["en"], {
+        "decision": ["committee approved"],
+        "mode": "override",
+    })
+    assert any_match(reg.get_patterns("decision"), "committee approved it")
+    assert not any_match(reg.get_patterns("decision"), "We decided to go")
+    # families without custom patterns keep builtins
+    assert any_match(reg.get_patterns("close"), "this is done")
+    # empty custom array does not wipe builtins
+    reg2 = P.PatternRegistry(["en"], {"decision": [], "mode": "override"})
+    assert any_match(reg2.get_patterns("decision"), "We decided to go")
+
+
+def test_custom_invalid_regex_skipped():
+    reg = P.PatternRegistry(["en"], {
+        "decision": ["(unclosed", "valid custom phrase"],
+        "mode": "extend",
+    })
+    assert any_match(reg.get_patterns("decision"), "a valid custom phrase here")
+    # all-invalid in override mode falls back to builtins
+    reg2 = P.PatternRegistry(["en"], {"decision": ["(bad", "[worse"], "mode": "override"})
+    assert any_match(reg2.get_patterns("decision"), "We decided to go")
+    # non-string entries filtered
+    reg3 = P.PatternRegistry(["en"], {"decision": [None, 42, "fine pattern"]})
+    assert any_match(reg3.get_patterns("decision"), "fine pattern")
+
+
+def test_custom_blacklist_and_keywords():
+    reg = P.PatternRegistry(["en"], {"blacklist": ["Foobar"], "keywords": ["Launchday"]})
+    assert "foobar" in reg.blacklist
+    assert "launchday" in reg.high_impact
+
+
+def test_language_all_and_invalid():
+    assert set(P._resolve_codes("all")) == set(P.language_codes())
+    assert P._resolve_codes(None) == ["en", "de"]
+    assert P._resolve_codes(12345) == ["en", "de"]
+    assert P._resolve_codes(["en", "fr"]) == ["en", "fr"]
+    reg = P.get_registry("all")
+    assert len(reg.codes) == 10

@@ -1,14 +1,18 @@
 """Pattern registry + shim API for cortex signal extraction.
 
 Parity target: cortex `src/patterns/registry.ts` + the shim
-`src/patterns.ts:38-82` — getPatterns (language or list or "both" = en+de),
-detectMood (last match position wins), isNoiseTopic, HIGH_IMPACT_KEYWORDS.
+`src/patterns.ts:38-82` — getPatterns (language or list or "both" = en+de,
+"all" = every pack), detectMood (last match position wins), isNoiseTopic,
+HIGH_IMPACT_KEYWORDS, and custom pattern packs (registry.ts loadSync:
+extend appends to builtins, override replaces a family when it has at
+least one VALID custom regex; invalid regexes are skipped silently;
+custom blacklist words and high-impact keywords merge in).
 """
 
 from __future__ import annotations
 
 import re
-from typing import Dict, Iterable, List, Set, Union
+from typing import Dict, Iterable, List, Optional, Set, Union
 
 from .packs import PACKS, language_codes
 
@@ -18,6 +22,8 @@ Language = Union[str, List[str]]
 def _resolve_codes(language: Language) -> List[str]:
     if language == "both":
         return ["en", "de"]
+    if language == "all":
+        return list(language_codes())
     if isinstance(language, str):
         return [language]
     if isinstance(language, (list, tuple)):
@@ -26,16 +32,30 @@ def _resolve_codes(language: Language) -> List[str]:
 
 
 class PatternRegistry:
-    def __init__(self, codes: Iterable[str]):
+    def __init__(self, codes: Iterable[str], custom: Optional[Dict] = None):
         self.codes = [c for c in codes if c in PACKS]
         if not self.codes:
             self.codes = ["en"]
+        custom = custom or {}
+        mode = custom.get("mode", "extend")
         self._compiled: Dict[str, List["re.Pattern[str]"]] = {}
         for family in ("decision", "close", "wait", "topic"):
             pats = []
             for c in self.codes:
                 for p in PACKS[c]["patterns"].get(family, []):
                     pats.append(re.compile(p, re.IGNORECASE))
+            extra = []
+            for p in custom.get(family, []) or []:
+                if not isinstance(p, str):
+                    continue
+                try:
+                    extra.append(re.compile(p, re.IGNORECASE))
+                except re.error:
+                    continue  # invalid custom regexes are skipped silently
+            if extra and mode == "override":
+                pats = extra
+            else:
+                pats.extend(extra)
             self._compiled[family] = pats
         self.moods: Dict[str, List["re.Pattern[str]"]] = {}
         for c in self.codes:
@@ -50,6 +70,12 @@ class PatternRegistry:
                 if kw not in self.high_impact:
                     self.high_impact.append(kw)
             self.noise_prefixes.update(PACKS[c]["noise_prefixes"])
+        for w in custom.get("blacklist", []) or []:
+            if isinstance(w, str):
+                self.blacklist.add(w.lower())
+        for kw in custom.get("keywords", []) or []:
+            if isinstance(kw, str) and kw.lower() not in self.high_impact:
+                self.high_impact.append(kw.lower())
 
     def get_patterns(self, family: str) -> List["re.Pattern[str]"]:
         return self._compiled.get(family, [])
@@ -58,7 +84,10 @@ class PatternRegistry:
 _registries: Dict[str, PatternRegistry] = {}
 
 
-def get_registry(language: Language = "both") -> PatternRegistry:
+def get_registry(language: Language = "both",
+                 custom: Optional[Dict] = None) -> PatternRegistry:
+    if custom:
+        return PatternRegistry(_resolve_codes(language), custom)
     key = ",".join(_resolve_codes(language))
     if key not in _registries:
         _registries[key] = PatternRegistry(_resolve_codes(language))
