@@ -246,3 +246,13 @@ def test_extract_signals_families():
                         "waiting for the security review. let's talk about budgets.")
     assert s["decisions"] and s["closures"] and s["waits"] and s["topics"]
     assert extract_signals("nothing interesting here")["decisions"] == []
+
+
+def test_custom_patterns_via_workspace_config(workspace):
+    try:
+        ws = CortexWorkspace(workspace, {"customPatterns": {
+            "decision": ["committee greenlit"], "mode": "extend"}})
+        ws.process_message("the committee greenlit the rollout", "user")
+        assert ws.decisions.decisions, "custom decision pattern should fire"
+    finally:
+        P.set_custom_patterns(None)

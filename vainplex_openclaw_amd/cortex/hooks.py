@@ -28,6 +28,10 @@ class CortexWorkspace:
 
     def __init__(self, workspace: str, config: Dict[str, Any], clock=time.time):
         language = config.get("language", "both")
+        if config.get("customPatterns"):
+            from . import patterns as _P
+
+            _P.set_custom_patterns(config["customPatterns"])
         self.workspace = workspace
         self.threads = ThreadTracker(
             workspace,

@@ -82,10 +82,21 @@ class PatternRegistry:
 
 
 _registries: Dict[str, PatternRegistry] = {}
+_custom_global: Optional[Dict] = None
+
+
+def set_custom_patterns(custom: Optional[Dict]) -> None:
+    """Install plugin-level custom patterns (registry.ts loadSync is a
+    module singleton in the reference too); clears the registry cache."""
+    global _custom_global
+    _custom_global = custom or None
+    _registries.clear()
 
 
 def get_registry(language: Language = "both",
                  custom: Optional[Dict] = None) -> PatternRegistry:
+    if custom is None:
+        custom = _custom_global
     if custom:
         return PatternRegistry(_resolve_codes(language), custom)
     key = ",".join(_resolve_codes(language))
@@ -135,6 +146,7 @@ def high_impact_keywords(language: Language = "both") -> List[str]:
 
 
 __all__ = [
+    "set_custom_patterns",
     "PACKS",
     "language_codes",
     "PatternRegistry",
