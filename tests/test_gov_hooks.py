@@ -155,14 +155,17 @@ def test_erc8004_encoding_and_provider():
     calls = []
 
     def rpc(method, params):
-        calls.append(method)
-        return "0x" + hex(72)[2:].rjust(64, "0")
+        calls.append(params[0]["data"])
+        if len(calls) == 1:  # ownerOf -> a non-zero owner
+            return "0x" + "ab" * 12 + "cd" * 20
+        # profile: score 72, feedbackCount 3, lastUpdated 99
+        return "0x" + hex(72)[2:].rjust(64, "0") + hex(3)[2:].rjust(64, "0")             + hex(99)[2:].rjust(64, "0")
 
     prov = ERC8004Provider(ERC8004Client(rpc))
     rep = prov.lookup_reputation("0x" + "ab" * 20)
-    assert rep["score"] == 72 and rep["tier"] == "trusted"
+    assert rep["registered"] and rep["score"] == 72 and rep["tier"] == "high"
     prov.lookup_reputation("0x" + "ab" * 20)
-    assert len(calls) == 1  # cached
+    assert len(calls) == 2  # ownerOf + getProfile, then cached
 
 
 def test_agentproof_queue_and_breaker():
@@ -359,3 +362,79 @@ def test_2fa_timeout_denies_all():
     expired = ap.expire_stale()
     assert len(expired) == 2 and all(r["status"] == "expired" for r in expired)
     assert ap.pending_requests() == []
+
+
+def test_erc8004_reference_behaviors():
+    """erc8004-client.test.ts mirrors: ABI encode/decode, profile
+    decoding, reputation tiers, LRU, fail-open."""
+    from vainplex_openclaw_amd.governance.security.erc8004 import (
+        LRUCache, ZERO_ADDRESS, classify_reputation, decode_address,
+        decode_profile, encode_uint256_arg,
+    )
+
+    # uint256 encoding always 64 chars
+    assert encode_uint256_arg(0) == "0" * 64
+    assert encode_uint256_arg(1).endswith("1") and len(encode_uint256_arg(1)) == 64
+    assert int(encode_uint256_arg(16700), 16) == 16700
+    # address decode: padded, zero, short, no-prefix
+    assert decode_address("0x" + "0" * 24 + "ab" * 20) == "0x" + "ab" * 20
+    assert decode_address("0x" + "0" * 64) == ZERO_ADDRESS
+    assert decode_address("0x1234") == ZERO_ADDRESS
+    assert decode_address("0" * 24 + "cd" * 20) == "0x" + "cd" * 20
+    # profile decode: full, short, empty, zeros
+    full = "0x" + format(55, "x").rjust(64, "0") + format(7, "x").rjust(64, "0") \
+        + format(1234, "x").rjust(64, "0")
+    assert decode_profile(full) == {"score": 55, "feedbackCount": 7, "lastUpdated": 1234}
+    assert decode_profile("0xabc") == {"score": 0, "feedbackCount": 0, "lastUpdated": 0}
+    assert decode_profile("") == {"score": 0, "feedbackCount": 0, "lastUpdated": 0}
+    assert decode_profile("0x" + "0" * 192)["score"] == 0
+    # reputation tiers
+    assert classify_reputation(False, 5, 90) == "unregistered"
+    assert classify_reputation(True, 0, 90) == "none"
+    assert classify_reputation(True, 3, 70) == "high"
+    assert classify_reputation(True, 3, 69) == "medium"
+    assert classify_reputation(True, 3, 29) == "low"
+    assert classify_reputation(True, 1, 0) == "low"
+    # LRU: miss, store, evict oldest, TTL, update-without-evict, clear
+    t = [0.0]
+    c = LRUCache(capacity=2, ttl_s=10, clock=lambda: t[0])
+    assert c.get("x") is None
+    c.put("a", 1); t[0] += 1; c.put("b", 2)
+    assert c.get("a") == 1 and c.get("b") == 2
+    c.put("a", 11)  # update, no eviction
+    assert len(c) == 2 and c.get("a") == 11
+    t[0] += 1; c.put("c", 3)  # evicts oldest (b at t=1? a updated at t=0... oldest entry)
+    assert len(c) == 2 and c.get("c") == 3
+    t[0] += 20
+    assert c.get("c") is None and not c.has("a")
+    c.clear(); assert len(c) == 0
+
+
+def test_erc8004_unregistered_and_fail_open():
+    def rpc_zero(method, params):
+        return "0x" + "0" * 64
+
+    cli = ERC8004Client(rpc_zero)
+    rep = cli.lookup_reputation("0x" + "11" * 20)
+    assert rep["registered"] is False and rep["tier"] == "unregistered"
+    assert rep["score"] == 0
+
+    def rpc_err(method, params):
+        raise ConnectionError("rpc down")
+
+    assert ERC8004Client(rpc_err).lookup_reputation("0x" + "22" * 20) is None
+
+    # score clamped to 0-100
+    calls = []
+
+    def rpc_big(method, params):
+        calls.append(1)
+        if len(calls) == 1:
+            return "0x" + "ab" * 32
+        return "0x" + format(5000, "x").rjust(64, "0") + format(2, "x").rjust(64, "0") \
+            + "0" * 64
+
+    rep2 = ERC8004Client(rpc_big).lookup_reputation("0x" + "33" * 20)
+    assert rep2["score"] == 100 and rep2["tier"] == "high"
+    # Phase 2 stub
+    assert ERC8004Client(rpc_big).submit_feedback("0x" + "33" * 20, 50) is None

@@ -48,7 +48,7 @@ def decode_uint256(hex_data: str) -> int:
 
 
 def classify_tier(score: int) -> str:
-    """Reputation score -> tier (erc8004-client.ts classification)."""
+    """Reputation score -> governance trust tier (logging helper)."""
     if score >= 80:
         return "elevated"
     if score >= 60:
@@ -60,27 +60,136 @@ def classify_tier(score: int) -> str:
     return "untrusted"
 
 
+ZERO_ADDRESS = "0x" + "0" * 40
+
+
+def encode_uint256_arg(value: int) -> str:
+    """uint256 -> 64 hex chars (erc8004-client.ts encodeUint256)."""
+    return format(max(0, int(value)), "x").rjust(64, "0")
+
+
+def decode_address(hex_data: str) -> str:
+    """Left-padded 32-byte slot -> 0x-address; short input -> zero addr."""
+    data = str(hex_data).replace("0x", "")
+    if len(data) < 64:
+        return ZERO_ADDRESS
+    return "0x" + data[24:64]
+
+
+def decode_profile(hex_data: str) -> Dict[str, int]:
+    """3-slot AgentProfile (score, feedbackCount, lastUpdated); short or
+    empty responses decode to all-zero defaults."""
+    data = str(hex_data).replace("0x", "")
+    if len(data) < 192:
+        return {"score": 0, "feedbackCount": 0, "lastUpdated": 0}
+    return {
+        "score": int(data[0:64], 16),
+        "feedbackCount": int(data[64:128], 16),
+        "lastUpdated": int(data[128:192], 16),
+    }
+
+
+def classify_reputation(registered: bool, feedback_count: int, score: int) -> str:
+    """ERC-8004 reputation tier (erc8004-client.ts classifyTier):
+    unregistered / none (no feedback) / high >= 70 / medium 30-69 / low."""
+    if not registered:
+        return "unregistered"
+    if feedback_count <= 0:
+        return "none"
+    if score >= 70:
+        return "high"
+    if score >= 30:
+        return "medium"
+    return "low"
+
+
+class LRUCache:
+    """Capacity + TTL cache (erc8004-client.ts LRUCache)."""
+
+    def __init__(self, capacity: int = 100, ttl_s: float = 300.0, clock=time.time):
+        self.capacity = capacity
+        self.ttl_s = ttl_s
+        self.clock = clock
+        self._data: Dict[str, Dict[str, Any]] = {}
+
+    def get(self, key: str) -> Optional[Any]:
+        e = self._data.get(key)
+        if e is None:
+            return None
+        if self.clock() - e["at"] >= self.ttl_s:
+            del self._data[key]
+            return None
+        return e["value"]
+
+    def has(self, key: str) -> bool:
+        return self.get(key) is not None
+
+    def put(self, key: str, value: Any) -> None:
+        if key not in self._data and len(self._data) >= self.capacity:
+            oldest = min(self._data, key=lambda k: self._data[k]["at"])
+            del self._data[oldest]
+        self._data[key] = {"at": self.clock(), "value": value}
+
+    def clear(self) -> None:
+        self._data.clear()
+
+    def __len__(self) -> int:
+        return len(self._data)
+
+
 class ERC8004Client:
     def __init__(
         self,
         rpc_call: Optional[Callable[[str, list], Any]] = None,
         registry: str = IDENTITY_REGISTRY,
+        cache: Optional[LRUCache] = None,
     ):
         self.rpc_call = rpc_call
         self.registry = registry
+        self.cache = cache or LRUCache()
+
+    def get_cache(self) -> LRUCache:
+        return self.cache
+
+    def _call(self, data: str) -> Optional[str]:
+        result = self.rpc_call("eth_call", [{"to": self.registry, "data": data}, "latest"])
+        return None if result is None else str(result)
 
     def lookup_reputation(self, agent_address: str) -> Optional[Dict[str, Any]]:
+        """Two-step lookup: ownerOf (registered?) then the profile.
+        Fail-open: any transport/RPC error returns None."""
         if self.rpc_call is None:
             return None
-        data = encode_call("getReputation(address)", agent_address)
+        cached = self.cache.get(agent_address)
+        if cached is not None:
+            return cached
         try:
-            result = self.rpc_call(
-                "eth_call", [{"to": self.registry, "data": data}, "latest"]
-            )
+            owner_raw = self._call(encode_call("ownerOf(address)", agent_address))
+            owner = decode_address(owner_raw or "")
+            if owner == ZERO_ADDRESS:
+                rep = {
+                    "address": agent_address, "registered": False,
+                    "score": 0, "feedbackCount": 0,
+                    "tier": classify_reputation(False, 0, 0),
+                }
+                self.cache.put(agent_address, rep)
+                return rep
+            prof_raw = self._call(encode_call("getProfile(address)", agent_address))
+            prof = decode_profile(prof_raw or "")
+            score = min(100, max(0, prof["score"]))
+            rep = {
+                "address": agent_address, "registered": True,
+                "score": score, "feedbackCount": prof["feedbackCount"],
+                "tier": classify_reputation(True, prof["feedbackCount"], score),
+            }
+            self.cache.put(agent_address, rep)
+            return rep
         except Exception:
             return None
-        score = decode_uint256(str(result))
-        return {"address": agent_address, "score": score, "tier": classify_tier(score)}
+
+    def submit_feedback(self, agent_address: str, score: int) -> None:
+        """Phase 2 stub (erc8004-client.ts submitFeedback returns null)."""
+        return None
 
 
 class ERC8004Provider:
