@@ -438,3 +438,57 @@ def test_erc8004_unregistered_and_fail_open():
     assert rep2["score"] == 100 and rep2["tier"] == "high"
     # Phase 2 stub
     assert ERC8004Client(rpc_big).submit_feedback("0x" + "33" * 20, 50) is None
+
+
+def test_agentproof_profile_and_batch(tmp_path):
+    """agentproof-rest.test.ts mirrors: key file loading/caching/trim,
+    /trust/{id}, tier classification, clamping, batch, failure nulls."""
+    keyf = tmp_path / "key.txt"
+    keyf.write_text("  sekrit-key \n")
+    gets, posts = [], []
+
+    def http_get(url, headers):
+        gets.append((url, dict(headers)))
+        return {"agentId": "main", "reputationScore": 150, "feedbackCount": 2}
+
+    def http_post(url, headers, body):
+        posts.append((url, dict(headers), body))
+        return {"results": [{"agentId": "a"}, {"agentId": "b"}]}
+
+    cli = AgentProofRestClient(base_url="https://ap.example/",
+                               api_key_file=str(keyf),
+                               http_get=http_get, http_post=http_post)
+    prof = cli.get_agent_profile("main")
+    assert gets[0][0] == "https://ap.example/trust/main"  # trailing slash stripped
+    assert gets[0][1] == {"X-API-Key": "sekrit-key"}      # trimmed key header
+    assert prof["score"] == 100 and prof["tier"] == "high"  # clamped
+    # key cached: rewrite the file, header unchanged
+    keyf.write_text("other")
+    cli.get_agent_profile("main")
+    assert gets[1][1] == {"X-API-Key": "sekrit-key"}
+    # batch
+    assert cli.batch_lookup([]) == []
+    res = cli.batch_lookup(["a", "missing", "b"])
+    assert posts[0][0] == "https://ap.example/trust/batch"
+    assert posts[0][2] == {"agentIds": ["a", "missing", "b"]}
+    assert res[0]["agentId"] == "a" and res[1] is None and res[2]["agentId"] == "b"
+
+
+def test_agentproof_failure_nulls(tmp_path):
+    def boom(*a):
+        raise OSError("net down")
+
+    cli = AgentProofRestClient(http_get=boom, http_post=boom)
+    assert cli.get_agent_profile("x") is None
+    assert cli.batch_lookup(["x", "y"]) == [None, None]
+    # malformed profile (no agentId) and missing results array
+    cli2 = AgentProofRestClient(http_get=lambda u, h: {"reputationScore": 50},
+                                http_post=lambda u, h, b: {"nope": 1})
+    assert cli2.get_agent_profile("x") is None
+    assert cli2.batch_lookup(["x"]) == [None]
+    # no key file: no auth header
+    calls = []
+    cli3 = AgentProofRestClient(http_get=lambda u, h: calls.append(dict(h)) or
+                                {"agentId": "m", "feedbackCount": 0})
+    p = cli3.get_agent_profile("m")
+    assert calls[0] == {} and p["tier"] == "none"

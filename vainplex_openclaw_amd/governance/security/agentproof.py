@@ -67,12 +67,24 @@ class AgentProofRestClient:
         self._thread: Optional[threading.Thread] = None
         self.dropped = 0
         self.flushed = 0
+        self._key_cached: Optional[str] = None
 
     def _bearer(self) -> Optional[str]:
+        # API key is read once and cached (agentproof-rest.ts:35-80);
+        # missing or empty files mean unauthenticated requests
+        if self._key_cached is not None:
+            return self._key_cached or None
         if not self.api_key_file or not os.path.isfile(self.api_key_file):
+            self._key_cached = ""
             return None
         with open(self.api_key_file, "r", encoding="utf-8") as fh:
-            return fh.read().strip() or None
+            self._key_cached = fh.read().strip()
+        return self._key_cached or None
+
+    def _headers(self) -> Dict[str, str]:
+        key = self._bearer()
+        # the REST API authenticates via X-API-Key (agentproof-rest.ts)
+        return {"X-API-Key": key} if key else {}
 
     # -- signal queue ------------------------------------------------------
     def enqueue_signal(self, agent_id: str, signal: str, detail: Optional[Dict[str, Any]] = None) -> bool:
@@ -95,8 +107,7 @@ class AgentProofRestClient:
             # no transport / breaker open: drop silently (fire-and-forget)
             self.dropped += len(batch)
             return 0
-        key = self._bearer()
-        headers = {"Authorization": f"Bearer {key}"} if key else {}
+        headers = self._headers()
         try:
             self.http_post(f"{self.base_url}/v1/signals", headers, {"signals": batch})
             self.breaker.record_success()
@@ -113,15 +124,55 @@ class AgentProofRestClient:
     def lookup_reputation(self, agent_id: str) -> Optional[Dict[str, Any]]:
         if self.http_get is None or self.breaker.is_open:
             return None
-        key = self._bearer()
-        headers = {"Authorization": f"Bearer {key}"} if key else {}
         try:
-            out = self.http_get(f"{self.base_url}/v1/reputation/{agent_id}", headers)
+            out = self.http_get(f"{self.base_url}/v1/reputation/{agent_id}", self._headers())
             self.breaker.record_success()
             return out
         except Exception:
             self.breaker.record_failure()
             return None
+
+    def get_agent_profile(self, agent_id: str) -> Optional[Dict[str, Any]]:
+        """GET /trust/{agentId} (agentproof-rest.ts getAgentProfile):
+        None on HTTP/network error or malformed response (no agentId);
+        score clamped to 0-100; tier classified from the REST data."""
+        from .erc8004 import classify_reputation
+
+        if self.http_get is None:
+            return None
+        try:
+            out = self.http_get(f"{self.base_url}/trust/{agent_id}", self._headers())
+        except Exception:
+            return None
+        if not isinstance(out, dict) or not out.get("agentId"):
+            return None
+        score = min(100, max(0, int(out.get("reputationScore") or 0)))
+        feedback = int(out.get("feedbackCount") or 0)
+        return {
+            "agentId": out["agentId"],
+            "registered": bool(out.get("registered", True)),
+            "score": score,
+            "feedbackCount": feedback,
+            "tier": classify_reputation(bool(out.get("registered", True)), feedback, score),
+        }
+
+    def batch_lookup(self, agent_ids: List[str]) -> List[Optional[Dict[str, Any]]]:
+        """POST /trust/batch (agentproof-rest.ts batchLookup): empty in ->
+        empty out; any failure or missing results array -> all None."""
+        if not agent_ids:
+            return []
+        if self.http_post is None:
+            return [None] * len(agent_ids)
+        try:
+            out = self.http_post(f"{self.base_url}/trust/batch", self._headers(),
+                                 {"agentIds": list(agent_ids)})
+        except Exception:
+            return [None] * len(agent_ids)
+        results = out.get("results") if isinstance(out, dict) else None
+        if not isinstance(results, list):
+            return [None] * len(agent_ids)
+        by_id = {r.get("agentId"): r for r in results if isinstance(r, dict)}
+        return [by_id.get(a) for a in agent_ids]
 
     # -- background flusher ------------------------------------------------
     def start(self) -> None:
