@@ -298,3 +298,32 @@ def test_chain_deterministic_ids_and_empty():
     a = reconstruct_chains(list(events))
     b = reconstruct_chains(list(events))
     assert a[0].id == b[0].id and len(a[0].id) == 16
+
+
+def test_report_aggregates_and_incremental_totals(workspace):
+    """report.test.ts mirrors: stats, signalStats, topAgents, timeRange,
+    incremental totals across runs."""
+    events = [ev("msg.in", content="x")]
+    for _ in range(3):
+        events.extend(tool_fail_pair())
+    src = MockTraceSource(events)
+    an = TraceAnalyzer(workspace, src)
+    rep = an.run()
+    assert rep["stats"]["events"] == len(events)
+    assert rep["stats"]["chains"] == rep["chains"]
+    assert rep["stats"]["findings"] == len(rep["findings"])
+    assert rep["stats"]["findingsClassified"] <= rep["stats"]["findings"]
+    assert sum(rep["signalStats"].values()) == len(rep["findings"])
+    for sig, agents in rep["topAgents"].items():
+        assert sig in rep["signalStats"]
+        assert agents and agents[0]["count"] >= agents[-1]["count"]
+    assert rep["timeRange"] is None or rep["timeRange"]["end"] >= rep["timeRange"]["start"]
+    # totals accumulate
+    t1 = an.state["totalEventsAnalyzed"]
+    an.run()
+    assert an.state["totalEventsAnalyzed"] >= t1
+    assert an.state["runsCompleted"] == 2
+    # empty stream handled
+    empty = TraceAnalyzer(workspace, MockTraceSource([]))
+    rep2 = empty.run()
+    assert rep2["stats"]["findings"] == 0 and rep2["timeRange"] is None

@@ -247,11 +247,45 @@ class TraceAnalyzer:
         classified = [c for c in classified if c.get("confidence", 0) >= self.config.min_confidence]
         outputs = generate_outputs(classified)
         now_ms = self.clock() * 1000
+        # aggregates (report.ts assembleReport): per-signal stats, top
+        # agents per signal, chain time range, classified-finding count
+        signal_stats: Dict[str, int] = {}
+        agents_by_signal: Dict[str, Dict[str, int]] = {}
+        for f in classified:
+            sig = f.get("signalType", "unknown")
+            signal_stats[sig] = signal_stats.get(sig, 0) + 1
+            agent = f.get("agent", "unknown")
+            agents_by_signal.setdefault(sig, {})[agent] = (
+                agents_by_signal.get(sig, {}).get(agent, 0) + 1
+            )
+        top_agents = {
+            sig: [
+                {"agent": a, "count": n}
+                for a, n in sorted(counts.items(), key=lambda kv: -kv[1])[:5]
+            ]
+            for sig, counts in agents_by_signal.items()
+        }
+        time_range = (
+            {"start": min(c.start_ts for c in chains),
+             "end": max(c.end_ts for c in chains)}
+            if chains else None
+        )
         report = {
             "version": 1,
             "generatedAt": int(now_ms),
             "eventsAnalyzed": len(events),
             "chains": len(chains),
+            "stats": {
+                "events": len(events),
+                "chains": len(chains),
+                "findings": len(classified),
+                "findingsClassified": sum(
+                    1 for f in classified if f.get("classification") is not None
+                ),
+            },
+            "signalStats": signal_stats,
+            "topAgents": top_agents,
+            "timeRange": time_range,
             "findings": classified,
             "outputs": outputs,
         }
@@ -260,6 +294,8 @@ class TraceAnalyzer:
             "lastProcessedSeq": max([e.seq for e in events], default=self.state.get("lastProcessedSeq", 0)),
             "lastRunAt": int(now_ms),
             "runsCompleted": int(self.state.get("runsCompleted", 0)) + 1,
+            "totalEventsAnalyzed": int(self.state.get("totalEventsAnalyzed", 0)) + len(events),
+            "totalFindings": int(self.state.get("totalFindings", 0)) + len(classified),
         }
         save_json(self.report_path, report)
         save_json(self.state_path, self.state)
