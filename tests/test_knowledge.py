@@ -380,3 +380,20 @@ def test_http_client_post_loopback():
         assert ei.value.status == 500
     finally:
         srv.shutdown()
+
+
+def test_fact_get_boosts_on_access_and_multi_filter(tmp_path):
+    fs = FactStore(str(tmp_path))
+    fs.load()
+    f = fs.add_fact("svc", "runs-on", "k8s")
+    f["relevance"] = 0.4
+    before_access = f["lastAccessed"]
+    got = fs.get_fact(f["id"])
+    assert got["relevance"] == boost_relevance(0.4)
+    assert got["lastAccessed"] >= before_access
+    assert fs.get_fact("nope") is None
+    # multiple filters AND together; empty query returns everything
+    fs.add_fact("svc", "runs-on", "vm")
+    fs.add_fact("db", "runs-on", "k8s")
+    assert len(fs.query(subject="svc", obj="k8s")) == 1
+    assert len(fs.query()) == 3
