@@ -327,3 +327,23 @@ def test_report_aggregates_and_incremental_totals(workspace):
     empty = TraceAnalyzer(workspace, MockTraceSource([]))
     rep2 = empty.run()
     assert rep2["stats"]["findings"] == 0 and rep2["timeRange"] is None
+
+
+def test_resolve_analyzer_llm_config():
+    from vainplex_openclaw_amd.cortex.trace.analyzer import resolve_analyzer_llm_config
+
+    TOP = {"endpoint": "http://localhost:11434/v1", "model": "mistral:7b",
+           "apiKey": "", "timeoutMs": 15000}
+    assert resolve_analyzer_llm_config(TOP, {"enabled": False})["enabled"] is False
+    assert resolve_analyzer_llm_config(TOP, None)["enabled"] is False
+    r = resolve_analyzer_llm_config(TOP, {"enabled": True})
+    assert r == {"enabled": True, "endpoint": "http://localhost:11434/v1",
+                 "model": "mistral:7b", "apiKey": "", "timeoutMs": 15000}
+    r2 = resolve_analyzer_llm_config(TOP, {"enabled": True, "endpoint": "http://c/v1",
+                                           "model": "gpt-4o", "apiKey": "sk-test",
+                                           "timeoutMs": 30000})
+    assert r2["endpoint"] == "http://c/v1" and r2["model"] == "gpt-4o"
+    assert r2["apiKey"] == "sk-test" and r2["timeoutMs"] == 30000
+    r3 = resolve_analyzer_llm_config(TOP, {"enabled": True, "model": "gpt-4o"})
+    assert r3["model"] == "gpt-4o" and r3["endpoint"] == TOP["endpoint"]
+    assert r3["apiKey"] == ""

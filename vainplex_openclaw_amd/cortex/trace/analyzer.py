@@ -39,6 +39,28 @@ from .signals import Finding, detect_all_signals
 
 # -- sources ----------------------------------------------------------------
 
+def resolve_analyzer_llm_config(top_level, override):
+    """Merge the top-level cortex LLM config with the trace-analyzer
+    override (classifier.ts resolveAnalyzerLlmConfig): the override's
+    `enabled` gates the result; any field it omits falls back to the
+    top-level value."""
+    top = dict(top_level or {})
+    ov = dict(override or {})
+    if not ov.get("enabled", False):
+        return {"enabled": False}
+    merged = {
+        "enabled": True,
+        "endpoint": top.get("endpoint", ""),
+        "model": top.get("model", ""),
+        "apiKey": top.get("apiKey", ""),
+        "timeoutMs": top.get("timeoutMs", 15000),
+    }
+    for key in ("endpoint", "model", "apiKey", "timeoutMs"):
+        if key in ov:
+            merged[key] = ov[key]
+    return merged
+
+
 class MockTraceSource:
     """Deterministic in-memory source (test helper parity:
     openclaw-cortex/test/trace-analyzer/helpers.ts)."""
