@@ -531,3 +531,26 @@ def test_hooks_subagent_spawn_and_failed_tool(workspace):
                         "toolName": "exec", "error": "boom"})
     assert all(e["toolName"] != "exec"
                for e in gh.tool_call_log.get("agent:parent", []))
+
+
+def test_make_call_llm_factory():
+    from vainplex_openclaw_amd.governance.plugin import make_call_llm
+
+    assert make_call_llm({}) is None
+    assert make_call_llm({"enabled": True}) is None  # no endpoint
+    posts = []
+
+    def fake_post(url, body, headers=None, timeout_s=None):
+        posts.append((url, body, headers, timeout_s))
+        return '{"choices": [{"message": {"content": "the answer"}}]}'
+
+    fn = make_call_llm({"enabled": True, "endpoint": "http://llm:11434/v1/",
+                        "model": "mistral:7b", "apiKey": "k", "timeoutMs": 5000},
+                       http_post=fake_post)
+    out = fn("analyze this")
+    assert out == "the answer"
+    url, body, headers, timeout_s = posts[0]
+    assert url == "http://llm:11434/v1/chat/completions"
+    assert body["model"] == "mistral:7b"
+    assert body["messages"][0]["content"] == "analyze this"
+    assert headers == {"Authorization": "Bearer k"} and timeout_s == 5.0
