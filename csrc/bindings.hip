@@ -48,6 +48,7 @@ __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigne
                                         float*, float*, int);
 __global__ void trust_recompute_kernel(float*, float*, const float*, const float*,
                                        const float*, float*, const float*, float*, int);
+__global__ void edit_distance_kernel(const uint8_t*, const int32_t*, int32_t*, int);
 struct AuditRecord64;
 __global__ void audit_pack_kernel(const int8_t*, const float*, const unsigned long long*,
                                   const unsigned long long*, const int32_t*, const float*,
@@ -467,6 +468,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   }, "MXFP4-X threshold scan");
   m.def("topk_scan_threshold_fp4x4", &topk_scan_threshold_fp4x4,
         "full-MXFP4 threshold scan (both operands e2m1 + e8m0 group scales)");
+  m.def("edit_distance", [](torch::Tensor bytes, torch::Tensor offsets) {
+    // batched Levenshtein over string PAIRS: offsets has 2*n+1 entries;
+    // pair i = strings (2i, 2i+1) in the packed byte buffer
+    CHECK_GPU(bytes); CHECK_CONTIG(bytes); CHECK_GPU(offsets); CHECK_CONTIG(offsets);
+    TORCH_CHECK(bytes.dtype() == torch::kUInt8 && offsets.dtype() == torch::kInt32);
+    TORCH_CHECK(offsets.numel() % 2 == 1, "offsets must cover 2*n strings");
+    int n_pairs = (int)(offsets.numel() / 2);
+    auto out = torch::empty({n_pairs}, torch::dtype(torch::kInt32).device(bytes.device()));
+    if (n_pairs > 0)
+      hipLaunchKernelGGL(edit_distance_kernel, dim3(n_pairs), dim3(64), 0, cur_stream(),
+                         bytes.data_ptr<uint8_t>(), offsets.data_ptr<int32_t>(),
+                         out.data_ptr<int32_t>(), n_pairs);
+    return out;
+  }, "Batched Levenshtein edit distance (wavefront DP)");
   m.def("firewall_verdict", &firewall_verdict, "Fused verdict/risk/trust-delta");
   m.def("trust_recompute", &trust_recompute, "Agent trust score recompute");
   m.def("audit_pack", &audit_pack, "Pack 64-byte audit records");

@@ -22,6 +22,7 @@ SOURCES = [
     "csrc/gemm_nt.hip",
     "csrc/topk_recall.hip",
     "csrc/firewall.hip",
+    "csrc/edit_distance.hip",
 ]
 
 setup(

@@ -393,3 +393,21 @@ def test_topk_recall_threshold_fp4x4():
         inter = len(set(ids_np[q]) & set(ref_ids[q]))
         assert inter >= k - 2, f"fp4x4 q={q}: {inter}/{k}"
         assert abs(vals[q][0] - ref.values[q, 0].item()) < 2e-2
+
+
+def test_edit_distance_vs_cpu():
+    """Batched Levenshtein kernel vs the plain CPU DP."""
+    rng = random.Random(5)
+    pairs = []
+    alphabet = b"abcdefg "
+    for i in range(64):
+        la, lb = rng.randint(0, 120), rng.randint(0, 120)
+        a = bytes(rng.choice(alphabet) for _ in range(la))
+        b = bytes(rng.choice(alphabet) for _ in range(lb))
+        pairs.append((a, b))
+    pairs += [(b"", b""), (b"", b"abc"), (b"kitten", b"sitting"),
+              (b"flaw", b"lawn"), (b"x" * 500, b"x" * 499 + b"y"),
+              (b"same-string", b"same-string")]
+    got = g.edit_distance_batch(pairs).cpu().tolist()
+    for (a, b), d in zip(pairs, got):
+        assert d == g.reference_edit_distance(a, b), (a[:20], b[:20], d)

@@ -274,6 +274,34 @@ def topk_recall_two_stage(
     return top.values, torch.gather(ids8, 1, top.indices)
 
 
+def edit_distance_batch(pairs: Sequence[Tuple[bytes, bytes]], device="cuda") -> torch.Tensor:
+    """Batched Levenshtein over (a, b) byte-string pairs (csrc
+    edit_distance_kernel, one 64-lane wave per pair, anti-diagonal DP).
+    Inputs are capped at 512 bytes each — the doom-loop detector caps
+    commands at 500 chars (SURVEY §2.8 / reference doom-loop.ts:76-90)."""
+    flat: list = []
+    for a, b in pairs:
+        flat.append(a[:512])
+        flat.append(b[:512])
+    if not flat:
+        return torch.empty(0, dtype=torch.int32)
+    bytes_t, offsets = pack_messages(flat, device=device)
+    return ext().edit_distance(bytes_t, offsets)
+
+
+def reference_edit_distance(a: bytes, b: bytes) -> int:
+    """Plain CPU Levenshtein for the numerics tests."""
+    la, lb = len(a), len(b)
+    prev = list(range(lb + 1))
+    for i in range(1, la + 1):
+        cur = [i] + [0] * lb
+        for j in range(1, lb + 1):
+            cur[j] = min(prev[j] + 1, cur[j - 1] + 1,
+                         prev[j - 1] + (a[i - 1] != b[j - 1]))
+        prev = cur
+    return prev[lb]
+
+
 # -- firewall tail ----------------------------------------------------------
 
 def firewall_verdict(
