@@ -161,3 +161,106 @@ def test_cortex_trace_analyzer_wired_to_journal(tmp_path, monkeypatch):
     assert report["eventsAnalyzed"] >= 8
     assert any(f["signalType"] in ("doom_loop", "repeat_fail", "tool_fail")
                for f in report["findings"])
+
+
+# -- governance integration.test.ts mirrors -------------------------------
+
+def test_subagent_policy_cascade_and_trust_cap(workspace):
+    """Parent deny policies cascade to spawned sub-agents; child trust is
+    capped at the parent's (integration.test.ts:297-364)."""
+    from vainplex_openclaw_amd.core.gateway import Gateway
+    from vainplex_openclaw_amd.governance.plugin import GovernancePlugin
+
+    gw = Gateway(config={"agents": ["parent"]})
+    plugin = GovernancePlugin(workspace)
+    gw.load(plugin, plugin_config={
+        "workspace": workspace,
+        "trust": {"enabled": True, "defaults": {"parent": 70, "*": 40}},
+        "policies": [{
+            "id": "parent-no-deploy", "priority": 50,
+            "scope": {"agents": ["parent"]},
+            "rules": [{"id": "r", "conditions": [{"type": "tool", "name": "deploy"}],
+                       "effect": {"action": "deny", "reason": "parent forbids deploys"}}],
+        }],
+    })
+    gw.start()
+    try:
+        # spawn a sub-agent under the parent session
+        gw.emit("after_tool_call", {
+            "agentId": "parent", "sessionKey": "agent:parent",
+            "toolName": "sessions_spawn",
+            "result": {"sessionId": "agent:parent:subagent:w:1"},
+        })
+        ca = plugin.engine.cross_agent
+        assert ca.get_parent("agent:parent:subagent:w:1") is not None
+        # the child's effective policies include the parent's deny
+        eff = ca.resolve_effective_policies(
+            {"sessionKey": "agent:parent:subagent:w:1", "agentId": "w",
+             "hook": "before_tool_call"},
+            plugin.engine.policy_index)
+        assert any(p["id"] == "parent-no-deploy" for p in eff)
+        # child trust ceiling <= parent score
+        cap = ca.compute_trust_ceiling("agent:parent:subagent:w:1")
+        assert cap <= plugin.engine.trust_manager.score("parent")
+    finally:
+        gw.stop()
+
+
+def test_policy_eval_and_frequency_performance(workspace):
+    """10 regex policies evaluate fast; 1000 frequency entries don't
+    degrade (integration.test.ts:389-442; bounds relaxed for CPython)."""
+    import time
+    from vainplex_openclaw_amd.governance.engine import GovernanceEngine
+
+    policies = [{
+        "id": f"p{i}", "priority": i, "scope": {},
+        "rules": [{"id": "r", "conditions": [
+            {"type": "tool", "params": {"command": {"matches": rf"cmd-{i}-\d+"}}}],
+            "effect": {"action": "deny"}}],
+    } for i in range(10)]
+    engine = GovernanceEngine({"policies": policies}, workspace)
+    engine.start()
+    try:
+        ctx = engine.build_context("before_tool_call", "a1",
+                                   tool_name="exec", tool_params={"command": "harmless"})
+        start = time.perf_counter()
+        for _ in range(50):
+            engine.evaluate(ctx)
+        per_eval_ms = (time.perf_counter() - start) / 50 * 1000
+        assert per_eval_ms < 50, per_eval_ms
+        for i in range(1000):
+            engine.frequency.record(f"a{i % 4}", f"agent:a{i % 4}", "exec")
+        start = time.perf_counter()
+        engine.evaluate(ctx)
+        assert (time.perf_counter() - start) < 0.1
+    finally:
+        engine.stop()
+
+
+def test_output_validation_end_to_end_with_audit(workspace):
+    """Output validation through the ENGINE entry point, including the
+    trust-proportional verdict and an audit trail record
+    (integration.test.ts:443-604)."""
+    from vainplex_openclaw_amd.governance.engine import GovernanceEngine
+
+    engine = GovernanceEngine({
+        "trust": {"enabled": True, "defaults": {"low": 20, "high": 80, "*": 40}},
+        "outputValidation": {
+            "factRegistries": [{"facts": [
+                {"subject": "nginx", "predicate": "state", "value": "stopped"}]}],
+        },
+    }, workspace)
+    engine.start()
+    try:
+        engine.trust_manager.get("low")
+        engine.trust_manager.get("high")
+        blocked = engine.validate_output("nginx is running", "low")
+        assert blocked["verdict"] == "block"
+        passed = engine.validate_output("nginx is running", "high")
+        assert passed["verdict"] == "pass"
+        ok = engine.validate_output("nginx is stopped", "low")
+        assert ok["verdict"] == "pass"
+        # unverified ignored by default
+        assert engine.validate_output("mystery-svc is running", "low")["verdict"] == "pass"
+    finally:
+        engine.stop()
