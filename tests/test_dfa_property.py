@@ -43,6 +43,11 @@ SEED_FRAGMENTS = [
     "password: ", "-----BEGIN ", "@example.com", "411 1-11", "ignore previous",
     "system prompt", "http://", "https://evil", " Inc.", "v2.5", "4111",
     "123-45-6789", "DE44", "+4915123", "is running", "definitely",
+    # round-1 regression shapes: alternation where only one branch ends in \b
+    # (pending lookahead must not leak across branches), and bounded loops
+    # followed by a failing continuation
+    "Bearer v2.5sk-", "do anything now", "DAN", "Claude III", " IIIx",
+    "v2.5.1sk", "Claude 3X", "count is 42x", "there is no podX",
 ]
 
 text_strategy = st.lists(
@@ -56,7 +61,7 @@ text_strategy = st.lists(
 
 
 @pytest.mark.parametrize("family", sorted(FAMILIES))
-@settings(max_examples=200, deadline=None)
+@settings(max_examples=2000, deadline=None)
 @given(text=text_strategy)
 def test_family_parity_fuzz(family, text):
     patterns, icase = FAMILIES[family]
@@ -64,6 +69,41 @@ def test_family_parity_fuzz(family, text):
     got = mdfa.scan(text.encode("utf-8", "replace"))
     want = ref_mask(patterns, icase, text)
     assert got == want, f"{family}: {text!r}: got {got:#x} want {want:#x}"
+
+
+# Minimized regressions from round 1's advisor findings (dfa.py alternation
+# pending-lookahead leak): the shared alt end-state let one branch's trailing
+# \b impose a next-byte-not-word requirement on every branch.
+REGRESSION_CASES = [
+    # (family, text) — parity with `re` is the assertion; these inputs
+    # produced wrong masks before the frontier-based _build fix.
+    ("entity", "Bearer v2.5sk-"),        # product_name bit 7 was dropped
+    ("entity", "Bearer v2.5"),
+    ("entity", "Claude IIIx"),           # roman branch \b must still bind
+    ("entity", "Claude III done"),
+    ("claims", "pod count is 42x"),
+    ("injection", "do anything nowadays"),
+    ("redaction", "Bearer abc.def-123x"),
+]
+
+
+@pytest.mark.parametrize("family,text", REGRESSION_CASES)
+def test_family_parity_regressions(family, text):
+    patterns, icase = FAMILIES[family]
+    got = _DFAS[family].scan(text.encode())
+    want = ref_mask(patterns, icase, text)
+    assert got == want, f"{family}: {text!r}: got {got:#x} want {want:#x}"
+
+
+def test_alt_branch_pending_isolation():
+    """'\\bDAN\\b|do anything now': branch 2 has no trailing boundary, so
+    'do anything nowX' must hit (advisor high-severity repro)."""
+    dfa = compile_patterns([(r"\bDAN\b|do anything now", 0)])
+    assert scan_with_eof(dfa, b"do anything nowX") == 1
+    assert scan_with_eof(dfa, b"do anything now") == 1
+    assert scan_with_eof(dfa, b" DAN ") == 1
+    assert scan_with_eof(dfa, b"DANX") == 0
+    assert scan_with_eof(dfa, b"DAN") == 1  # EOF matures the \b
 
 
 @settings(max_examples=150, deadline=None)

@@ -39,6 +39,11 @@ DOT = frozenset(b for b in range(256) if b != ord("\n"))
 
 EPS = None  # epsilon edge label
 
+# Bump on any change to the NFA/DFA construction so on-disk compiled-family
+# caches (ops/pattern_sets.py:_family_hash) are invalidated; round-1's
+# alternation pending-lookahead bug shipped through a stale cache.
+DFA_COMPILER_VERSION = 2
+
 
 class RegexError(ValueError):
     pass
@@ -293,70 +298,83 @@ class NFA:
         self.trans[src].append((label, dst))
 
 
-def _build(nfa: NFA, node, start: int) -> Tuple[int, Optional[FrozenSet[int]]]:
-    """Build node starting at `start`; return (end_state, pending_nla).
-    A trailing (?!X) becomes a pending negative-lookahead attached to the
-    accept."""
+Frontier = List[Tuple[int, Optional[FrozenSet[int]]]]
+
+
+def _join(nfa: NFA, frontier: Frontier) -> int:
+    """Collapse an all-pendingless frontier into one state (error if any
+    end still carries a pending negative lookahead — lookahead must be the
+    last element of its branch)."""
+    if any(p is not None for _e, p in frontier):
+        raise RegexError("lookahead must be last in a branch")
+    if len(frontier) == 1:
+        return frontier[0][0]
+    end = nfa.new_state()
+    for e, _p in frontier:
+        nfa.add(e, ("eps",), end)
+    return end
+
+
+def _build(nfa: NFA, node, start: int) -> Frontier:
+    """Build node starting at `start`; return the accept frontier — a list
+    of (end_state, pending_nla) pairs. A trailing (?!X) becomes a pending
+    negative-lookahead attached to that branch's accept ONLY: alternation
+    branches keep separate ends when their pendings differ, so e.g.
+    `\\bDAN\\b|do anything now` does not leak branch 1's trailing \\b onto
+    branch 2 (round-1 advisor finding, dfa alternation bug)."""
     kind = node[0]
     if kind == "char":
         end = nfa.new_state()
         nfa.add(start, node[1], end)
-        return end, None
+        return [(end, None)]
     if kind == "cat":
         items = list(node[1])
         # trailing \b means "next byte is not word (or EOF)" -> lookahead
         if items and items[-1] == ("bound",):
             items[-1] = ("nla", WORD)
         cur = start
-        pending = None
+        frontier: Frontier = [(start, None)]
         for item in items:
-            if pending is not None:
-                raise RegexError("lookahead must be last in a branch")
-            cur, pending = _build(nfa, item, cur)
-        return cur, pending
+            cur = _join(nfa, frontier)
+            frontier = _build(nfa, item, cur)
+        return frontier
     if kind == "alt":
-        end = nfa.new_state()
-        pendings = set()
+        out: Frontier = []
+        plain: Frontier = []
         for b in node[1]:
             s = nfa.new_state()
             nfa.add(start, ("eps",), s)
-            e, pend = _build(nfa, b, s)
-            if pend is not None:
-                pendings.add(pend)
-            nfa.add(e, ("eps",), end)
-        if pendings:
-            if len(pendings) > 1:
-                raise RegexError("mixed lookaheads unsupported")
-            return end, next(iter(pendings))
-        return end, None
+            for e, pend in _build(nfa, b, s):
+                (plain if pend is None else out).append((e, pend))
+        if plain:
+            # merge pendingless branch ends into one shared state (keeps the
+            # subset construction small); pending ends stay distinct.
+            out.append((_join(nfa, plain), None))
+        return out
     if kind in ("star", "plus", "opt"):
         inner = node[1]
         s = nfa.new_state()
         e = nfa.new_state()
         nfa.add(start, ("eps",), s)
-        ie, pend = _build(nfa, inner, s)
-        if pend is not None:
-            raise RegexError("lookahead inside quantifier unsupported")
+        ie = _join(nfa, _build(nfa, inner, s))
         nfa.add(ie, ("eps",), e)
         if kind in ("star", "plus"):
             nfa.add(ie, ("eps",), s)
         if kind in ("star", "opt"):
             nfa.add(start, ("eps",), e)
-        return e, None
+        return [(e, None)]
     if kind == "rep":
         _, inner, m, n = node
         cur = start
         for _i in range(m):
-            cur, pend = _build(nfa, inner, cur)
-            if pend is not None:
-                raise RegexError("lookahead inside repeat unsupported")
+            cur = _join(nfa, _build(nfa, inner, cur))
         if n is None:
             # {m,} -> m copies + star
-            cur, _ = _build(nfa, ("star", inner), cur)
+            cur = _join(nfa, _build(nfa, ("star", inner), cur))
         else:
             for _i in range(n - m):
-                cur, _ = _build(nfa, ("opt", inner), cur)
-        return cur, None
+                cur = _join(nfa, _build(nfa, ("opt", inner), cur))
+        return [(cur, None)]
     if kind == "bound":
         # \b before a word char: previous byte must NOT be word (or start).
         # \b after a word char: next byte must not be word — approximated by
@@ -364,14 +382,14 @@ def _build(nfa: NFA, node, start: int) -> Tuple[int, Optional[FrozenSet[int]]]:
         # attach a pending lookahead on WORD.
         end = nfa.new_state()
         nfa.add(start, ("guard", "boundary", WORD), end)
-        return end, None
+        return [(end, None)]
     if kind == "nlb":
         end = nfa.new_state()
         nfa.add(start, ("guard", "prev_not_in", node[1]), end)
-        return end, None
+        return [(end, None)]
     if kind == "nla":
         # pending negative lookahead — resolved at accept time
-        return start, node[1]
+        return [(start, node[1])]
     raise RegexError(f"unknown node {kind}")
 
 
@@ -484,10 +502,10 @@ def compile_patterns(
         ast = _Parser(pattern, ignore_case=pid in ignore_case_ids).parse()
         s = nfa.new_state()
         nfa.add(root, ("eps",), s)
-        end, pending_nla = _build(nfa, ast, s)
-        nfa.accepts.setdefault(end, set()).add((pid, pending_nla))
-        if pending_nla is not None:
-            guard_sets.add(pending_nla)
+        for end, pending_nla in _build(nfa, ast, s):
+            nfa.accepts.setdefault(end, set()).add((pid, pending_nla))
+            if pending_nla is not None:
+                guard_sets.add(pending_nla)
     # collect guard charsets used
     for edges in nfa.trans:
         for label, _ in edges:

@@ -23,7 +23,7 @@ import os
 
 import numpy as np
 
-from .dfa import DFA, MultiDFA, compile_multi
+from .dfa import DFA, DFA_COMPILER_VERSION, MultiDFA, compile_multi
 
 # -- redaction (bit order = registry order; category via bit ranges) --------
 REDACTION_PATTERNS: List[Tuple[str, int, str]] = [
@@ -141,7 +141,7 @@ _CACHE_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_dfa_cach
 
 def _family_hash(name: str) -> str:
     patterns, icase = _FAMILIES[name]
-    blob = repr((patterns, sorted(icase))).encode()
+    blob = repr((DFA_COMPILER_VERSION, patterns, sorted(icase))).encode()
     return hashlib.sha256(blob).hexdigest()[:16]
 
 
