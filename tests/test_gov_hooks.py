@@ -408,6 +408,15 @@ def test_erc8004_reference_behaviors():
     t[0] += 20
     assert c.get("c") is None and not c.has("a")
     c.clear(); assert len(c) == 0
+    # recency: get() refreshes lastAccess, so eviction is true LRU —
+    # a hot old entry survives, the cold newer one goes (erc8004-client.ts
+    # updates entry.lastAccess on get)
+    t[0] = 0.0
+    c2 = LRUCache(capacity=2, ttl_s=100, clock=lambda: t[0])
+    c2.put("old", 1); t[0] = 1; c2.put("cold", 2)
+    t[0] = 2; assert c2.get("old") == 1   # refresh "old"
+    t[0] = 3; c2.put("new", 3)             # must evict "cold", not "old"
+    assert c2.get("old") == 1 and c2.get("cold") is None and c2.get("new") == 3
 
 
 def test_erc8004_unregistered_and_fail_open():

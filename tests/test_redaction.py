@@ -207,6 +207,33 @@ def test_hooks_sync_write_gate_skips_pii_only():
     assert st.on_before_message_write({"agentId": "a"}) is None
 
 
+def test_hooks_sync_write_gate_full_scan_when_critical_present():
+    """hooks.ts:405-456 parity: the sync gate runs the FULL engine; when a
+    credential/financial hit is present the returned content has PII
+    redacted too (not just the credential)."""
+    st = _state()
+    out = st.on_before_message_write({
+        "agentId": "a",
+        "content": "key sk-abcdefghij0123456789XY for alice@example.com",
+    })
+    assert out is not None
+    assert "sk-abcdefghij0123456789XY" not in out["content"]
+    # PII rides along once enforcement triggers
+    assert "alice@example.com" not in out["content"]
+
+
+def test_hooks_sync_write_gate_exempt_agent_passes_everything():
+    """Exempt agents bypass the before_message_write gate entirely —
+    credentials included (hooks.ts:420-423)."""
+    st = _state({"enabled": True, "allowlist": {"exemptAgents": ["main"]}})
+    ev = {"agentId": "main", "content": "key sk-abcdefghij0123456789XY"}
+    assert st.on_before_message_write(ev) is None
+    # non-exempt agent on the same state is still gated
+    out = st.on_before_message_write({"agentId": "other",
+                                      "content": "key sk-abcdefghij0123456789XY"})
+    assert out is not None and "sk-abcdefghij0123456789XY" not in out["content"]
+
+
 def test_hooks_layer1_nested_and_none_results():
     st = _state()
     assert st.on_tool_result_persist({"toolName": "t", "result": None}) is None

@@ -104,7 +104,10 @@ def classify_reputation(registered: bool, feedback_count: int, score: int) -> st
 
 
 class LRUCache:
-    """Capacity + TTL cache (erc8004-client.ts LRUCache)."""
+    """Capacity + TTL cache (erc8004-client.ts LRUCache). TTL is measured
+    from insertion time; eviction is by least-recent ACCESS — get()
+    refreshes recency (lastAccess) without extending the TTL, matching the
+    reference's `entry.lastAccess` update on get."""
 
     def __init__(self, capacity: int = 100, ttl_s: float = 300.0, clock=time.time):
         self.capacity = capacity
@@ -119,6 +122,7 @@ class LRUCache:
         if self.clock() - e["at"] >= self.ttl_s:
             del self._data[key]
             return None
+        e["last_access"] = self.clock()
         return e["value"]
 
     def has(self, key: str) -> bool:
@@ -126,9 +130,10 @@ class LRUCache:
 
     def put(self, key: str, value: Any) -> None:
         if key not in self._data and len(self._data) >= self.capacity:
-            oldest = min(self._data, key=lambda k: self._data[k]["at"])
-            del self._data[oldest]
-        self._data[key] = {"at": self.clock(), "value": value}
+            coldest = min(self._data, key=lambda k: self._data[k]["last_access"])
+            del self._data[coldest]
+        now = self.clock()
+        self._data[key] = {"at": now, "last_access": now, "value": value}
 
     def clear(self) -> None:
         self._data.clear()
