@@ -201,3 +201,67 @@ def test_multidfa_pack_roundtrip():
 
     for text in SECRET_SAMPLES + CLEAN_SAMPLES:
         assert packed_scan(text.encode()) == mdfa.scan(text.encode()), text
+
+
+# -- cortex signal family (thread-tracker.ts:42-82, patterns.ts:47-66) ------
+
+def test_cortex_family_signal_bits():
+    from vainplex_openclaw_amd.ops import pattern_sets as ps
+
+    fam = ps.get_family("cortex")
+    D, C, W, T, H = (ps.CORTEX_BIT_DECISION, ps.CORTEX_BIT_CLOSE,
+                     ps.CORTEX_BIT_WAIT, ps.CORTEX_BIT_TOPIC,
+                     ps.CORTEX_BIT_HIGH_IMPACT)
+    M0 = ps.CORTEX_MOOD_BIT0
+    cases = [
+        ("we decided to go with plan B", D, True),
+        ("nothing to see here", D, False),
+        ("it works now", C, True),
+        ("the fix is done.", C, True),
+        ("✅", C, True),
+        ("waiting for the deploy", W, True),
+        ("blocked by upstream", W, True),
+        ("let's talk about the new schema design", T, True),
+        ("regarding the migration plan", T, True),
+        ("we should delete the production db", H, True),
+        ("this is so annoying", M0 + 0, True),       # frustrated
+        ("awesome, great news!", M0 + 1, True),      # excited
+        ("deadline is asap, careful", M0 + 2, True), # tense
+        ("merged and deployed", M0 + 3, True),       # productive
+        ("what if we experiment", M0 + 4, True),     # exploratory
+        # German / Spanish / Russian / Japanese coverage (one per signal)
+        ("wir haben das beschlossen", D, True),
+        ("está resuelto", C, True),
+        ("esperando a luis", W, True),
+        ("решено", D, True),
+        ("決定です", D, True),
+    ]
+    for text, bit, want in cases:
+        m = fam.scan(text.encode())
+        assert bool(m >> bit & 1) == want, f"{text!r} bit {bit}: mask {m:#x}"
+
+
+def test_cortex_family_mood_parity_with_packs():
+    """Each mood bit must agree with the interactive packs' regexes
+    (re.IGNORECASE, Unicode) on representative texts."""
+    import re
+
+    from vainplex_openclaw_amd.cortex.patterns.packs import PACKS
+    from vainplex_openclaw_amd.ops import pattern_sets as ps
+
+    fam = ps.get_family("cortex")
+    texts = [
+        "broken again, damn", "BROKEN AGAIN", "love it", "risky deadline",
+        "shipped it", "maybe an idea", "just text", "kaputt schon wieder",
+        "das ist genial", "vorsicht riskant", "erledigt und fertig",
+        "probieren wir", "", "the plan is good",
+    ]
+    for text in texts:
+        m = fam.scan(text.encode("utf-8"))
+        for mi, mood in enumerate(ps.CORTEX_MOODS):
+            want = any(
+                re.search(pack["moods"][mood], text, re.IGNORECASE)
+                for pack in PACKS.values() if mood in pack.get("moods", {})
+            )
+            got = bool(m >> (ps.CORTEX_MOOD_BIT0 + mi) & 1)
+            assert got == want, f"{text!r} mood {mood}: got {got} want {want}"

@@ -130,3 +130,46 @@ def test_adhoc_literal_compile_fuzz(text):
     hay = ("zz" + text + "yy").encode("utf-8", "replace")
     want = 1 if re.search(pat, hay.decode("utf-8", "replace")) else 0
     assert scan_with_eof(dfa, hay) == want
+
+
+# -- cortex family fuzz -----------------------------------------------------
+# The cortex family compiles with ignore_case + unicode_word (bytes >= 0x80
+# count as \w). Parity reference: re.IGNORECASE with default Unicode
+# semantics. The fuzz alphabet sticks to characters where both agree
+# (ASCII + letters; no non-letter symbols >= U+0080, whose \w-ness
+# diverges by design — WORD_U8 is hit-detection grade).
+
+CORTEX_SEEDS = [
+    "decided", "we decided to", "done", "it works", "fixed!", "✅",
+    "waiting for", "blocked by", "let's talk about the schema",
+    "regarding", "production", "deploy", "asap", "deadline",
+    "awesome", "annoying", "maybe", "experiment", "shipped",
+    "beschlossen", "erledigt", "funktioniert", "warte auf", "dringend",
+    "decidimos", "resuelto", "esperando a", "urgente",
+    "решено", "готово", "ждём",
+]
+
+
+@settings(max_examples=1000, deadline=None)
+@given(text=st.lists(
+    st.one_of(
+        st.sampled_from(CORTEX_SEEDS),
+        st.text(alphabet=st.characters(min_codepoint=32, max_codepoint=126), max_size=10),
+        st.text(alphabet="äöüßéабвгдежз", max_size=4),
+    ),
+    max_size=6,
+).map(" ".join))
+def test_cortex_family_parity_fuzz(text):
+    from vainplex_openclaw_amd.ops import pattern_sets as ps
+
+    patterns, icase = ps._FAMILIES.get("cortex") or (None, None)
+    if patterns is None:
+        ps.get_family("cortex")
+        patterns, icase = ps._FAMILIES["cortex"]
+    mdfa = ps.get_family("cortex")
+    got = mdfa.scan(text.encode("utf-8"))
+    want = 0
+    for pat, bit, _name in patterns:
+        if re.search(pat, text, re.IGNORECASE):
+            want |= 1 << bit
+    assert got == want, f"{text!r}: got {got:#x} want {want:#x}"

@@ -42,7 +42,7 @@ EPS = None  # epsilon edge label
 # Bump on any change to the NFA/DFA construction so on-disk compiled-family
 # caches (ops/pattern_sets.py:_family_hash) are invalidated; round-1's
 # alternation pending-lookahead bug shipped through a stale cache.
-DFA_COMPILER_VERSION = 2
+DFA_COMPILER_VERSION = 3
 
 
 class RegexError(ValueError):
@@ -58,11 +58,17 @@ class RegexError(ValueError):
 # --------------------------------------------------------------------------
 
 
+WORD_U8 = WORD | frozenset(range(128, 256))  # unicode_word: any UTF-8
+# continuation/lead byte counts as a word byte (approximates re.UNICODE \w
+# for hit detection over UTF-8 text)
+
+
 class _Parser:
-    def __init__(self, pattern: str, ignore_case: bool = False):
+    def __init__(self, pattern: str, ignore_case: bool = False, unicode_word: bool = False):
         self.p = pattern
         self.i = 0
         self.ignore_case = ignore_case
+        self.word = WORD_U8 if unicode_word else WORD
 
     def error(self, msg: str):
         raise RegexError(f"{msg} at {self.i} in {self.p!r}")
@@ -195,25 +201,60 @@ class _Parser:
             self.next()
             return node
         if c == "[":
-            return ("char", self._charset(self.parse_class()))
+            ascii_bytes, hi_cps, negated = self.parse_class()
+            if negated:
+                if hi_cps:
+                    self.error("negated class with non-ASCII members unsupported")
+                # fold case on the member set BEFORE negating (re.IGNORECASE
+                # [^a-z] excludes A-Z as well)
+                return ("char", frozenset(ANY - self._charset(frozenset(ascii_bytes))))
+            branches = []
+            if ascii_bytes:
+                branches.append(("char", self._charset(frozenset(ascii_bytes))))
+            for cp in sorted(hi_cps):
+                branches.append(self._literal(chr(cp)))
+            if not branches:
+                self.error("empty class")
+            return branches[0] if len(branches) == 1 else ("alt", branches)
         if c == ".":
             return ("char", DOT)
         if c == "\\":
             return self.parse_escape()
         if c in "^$":
-            # anchors are noise for unanchored hit detection; ^ treated as \b-ish
-            return ("bound",) if c == "^" else ("cat", [])
+            # ^ = start-of-input (re.search without MULTILINE); $ = end of
+            # input, expressed as a pending (?![\s\S]) lookahead that only
+            # matures at EOF
+            return ("bof",) if c == "^" else ("nla", ANY)
         if c in "*+?":
             self.error("dangling quantifier")
-        return ("char", self._charset(frozenset([ord(c)])))
+        return self._literal(c)
+
+    def _literal(self, c: str):
+        """A literal character; non-ASCII expands to its UTF-8 byte
+        sequence (with simple-case-fold variants when ignore_case)."""
+        if ord(c) < 128:
+            return ("char", self._charset(frozenset([ord(c)])))
+        variants = {c}
+        if self.ignore_case:
+            # single-char case pairs only ('ß'.upper() == 'SS' is a
+            # multi-char expansion re.IGNORECASE does not perform)
+            variants |= {v for v in (c.lower(), c.upper()) if len(v) == 1}
+        branches = []
+        for v in sorted(variants):
+            seq = v.encode("utf-8")
+            if len(seq) == 1:
+                branches.append(("char", frozenset(seq)))
+            else:
+                branches.append(("cat", [("char", frozenset([b])) for b in seq]))
+        return branches[0] if len(branches) == 1 else ("alt", branches)
 
     def parse_escape(self):
         c = self.next()
         table = {
             "d": DIGITS,
             "D": ANY - DIGITS,
-            "w": WORD,
-            "W": ANY - WORD,
+            "w": self.word,
+            "W": ANY - self.word,
             "s": SPACE,
             "S": ANY - SPACE,
             "n": frozenset([10]),
@@ -223,15 +264,19 @@ class _Parser:
         if c in table:
             return ("char", table[c])
         if c == "b":
-            return ("bound",)
-        return ("char", self._charset(frozenset([ord(c)])))
+            return ("bound", self.word)
+        return self._literal(c)
 
-    def parse_class(self) -> FrozenSet[int]:
+    def parse_class(self) -> Tuple[Set[int], Set[int], bool]:
+        """Returns (ascii/escape byte set, non-ASCII literal codepoints,
+        negated). Non-ASCII literal members become UTF-8 alternations in
+        parse_atom; escape tables contribute BYTE values directly."""
         negate = False
         if self.peek() == "^":
             self.next()
             negate = True
         chars: Set[int] = set()
+        hi_cps: Set[int] = set()
         first = True
         while True:
             c = self.peek()
@@ -244,7 +289,7 @@ class _Parser:
             self.next()
             if c == "\\":
                 e = self.next()
-                table = {"d": DIGITS, "w": WORD, "s": SPACE, "n": {10}, "t": {9}, "r": {13}}
+                table = {"d": DIGITS, "w": self.word, "s": SPACE, "n": {10}, "t": {9}, "r": {13}}
                 if e in table:
                     chars.update(table[e])
                     continue
@@ -253,10 +298,15 @@ class _Parser:
             if self.peek() == "-" and self.i + 1 < len(self.p) and self.p[self.i + 1] != "]":
                 self.next()
                 hi = ord(self.next())
-                chars.update(range(lo, hi + 1))
+                if hi < lo:
+                    self.error("reversed class range")
+                if hi - lo > 4096:
+                    self.error("class range too wide")
+                for cp in range(lo, hi + 1):
+                    (chars if cp < 128 else hi_cps).add(cp)
             else:
-                chars.add(lo)
-        return frozenset(ANY - chars) if negate else frozenset(chars)
+                (chars if lo < 128 else hi_cps).add(lo)
+        return chars, hi_cps, negate
 
 
 def _single_charset(node) -> Optional[FrozenSet[int]]:
@@ -330,8 +380,8 @@ def _build(nfa: NFA, node, start: int) -> Frontier:
     if kind == "cat":
         items = list(node[1])
         # trailing \b means "next byte is not word (or EOF)" -> lookahead
-        if items and items[-1] == ("bound",):
-            items[-1] = ("nla", WORD)
+        if items and items[-1][0] == "bound":
+            items[-1] = ("nla", items[-1][1])
         cur = start
         frontier: Frontier = [(start, None)]
         for item in items:
@@ -379,9 +429,15 @@ def _build(nfa: NFA, node, start: int) -> Frontier:
         # \b before a word char: previous byte must NOT be word (or start).
         # \b after a word char: next byte must not be word — approximated by
         # the prev-guard form on the following edge; for trailing \b we
-        # attach a pending lookahead on WORD.
+        # attach a pending lookahead on the word set (node[1]: WORD, or
+        # WORD_U8 under unicode_word).
         end = nfa.new_state()
-        nfa.add(start, ("guard", "boundary", WORD), end)
+        nfa.add(start, ("guard", "boundary", node[1]), end)
+        return [(end, None)]
+    if kind == "bof":
+        # start-of-input: passes only when there is no previous byte
+        end = nfa.new_state()
+        nfa.add(start, ("guard", "bof", frozenset()), end)
         return [(end, None)]
     if kind == "nlb":
         end = nfa.new_state()
@@ -464,7 +520,9 @@ def _closure(nfa: NFA, states: FrozenSet[int], prev_is: Dict[str, bool]) -> Froz
                         ok = prev_is["in:" + _cs_key(cs)]
                     elif kind == "boundary":
                         # \b before consuming a word char: prev not word
-                        ok = not prev_is["in:" + _cs_key(WORD)]
+                        ok = not prev_is["in:" + _cs_key(cs)]
+                    elif kind == "bof":
+                        ok = prev_is.get("bof", False)
                 if ok and dst not in out:
                     out.add(dst)
                     stack.append(dst)
@@ -486,20 +544,25 @@ def compile_patterns(
     patterns: Sequence[Tuple[str, int]],
     ignore_case_ids: Optional[Set[int]] = None,
     max_states: int = 60000,
+    unicode_word: bool = False,
 ) -> DFA:
     """Compile [(regex, pattern_id)] into one unanchored multi-pattern DFA.
 
-    pattern_id is the bit set in the accept mask (0..63).
+    pattern_id is the bit set in the accept mask (0..63); several patterns
+    may share a bit (the cortex family ORs 10 languages into one signal
+    bit). unicode_word widens \w/\b to treat any byte >= 0x80 as a word
+    byte (UTF-8 text under re.UNICODE semantics, hit-detection grade).
     """
     ignore_case_ids = ignore_case_ids or set()
     nfa = NFA()
     root = nfa.new_state()
 
-    guard_sets: Set[FrozenSet[int]] = {WORD}
+    guard_sets: Set[FrozenSet[int]] = {WORD_U8 if unicode_word else WORD}
     for pattern, pid in patterns:
         if pid < 0 or pid > 63:
             raise RegexError("pattern_id must fit a u64 bitmask")
-        ast = _Parser(pattern, ignore_case=pid in ignore_case_ids).parse()
+        ast = _Parser(pattern, ignore_case=pid in ignore_case_ids,
+                      unicode_word=unicode_word).parse()
         s = nfa.new_state()
         nfa.add(root, ("eps",), s)
         for end, pending_nla in _build(nfa, ast, s):
@@ -539,7 +602,7 @@ def compile_patterns(
     def prev_ctx(byte: Optional[int]) -> Dict[str, bool]:
         if byte in _ctx_cache:
             return _ctx_cache[byte]
-        ctx = {}
+        ctx = {"bof": byte is None}
         for cs in guard_sets:
             ctx["in:" + _cs_key(cs)] = byte is not None and byte in cs
         _ctx_cache[byte] = ctx
@@ -806,6 +869,7 @@ def compile_multi(
     patterns: Sequence[Tuple[str, int]],
     ignore_case_ids: Optional[Set[int]] = None,
     per_dfa_state_budget: int = 3000,
+    unicode_word: bool = False,
 ) -> MultiDFA:
     """Greedily pack patterns into as few DFAs as fit the state budget."""
     ignore_case_ids = ignore_case_ids or set()
@@ -816,14 +880,17 @@ def compile_multi(
     for pat in patterns:
         trial = group + [pat]
         try:
-            d = compile_patterns(trial, ignore_case_ids=ignore_case_ids, max_states=per_dfa_state_budget)
+            d = compile_patterns(trial, ignore_case_ids=ignore_case_ids,
+                                 max_states=per_dfa_state_budget, unicode_word=unicode_word)
         except RegexError:
             d = None
         if d is None:
             if group_dfa is not None:
                 dfas.append(group_dfa)
             group = [pat]
-            group_dfa = compile_patterns(group, ignore_case_ids=ignore_case_ids, max_states=per_dfa_state_budget * 8)
+            group_dfa = compile_patterns(group, ignore_case_ids=ignore_case_ids,
+                                         max_states=per_dfa_state_budget * 8,
+                                         unicode_word=unicode_word)
         else:
             group = trial
             group_dfa = d

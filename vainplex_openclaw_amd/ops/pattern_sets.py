@@ -127,6 +127,45 @@ ENTITY_PATTERNS: List[Tuple[str, int, str]] = [
 ENTITY_IGNORECASE: Set[int] = set()
 
 
+# -- cortex signal family (thread-tracker.ts:42-82 + patterns.ts:47-66) ----
+# Built from the 10-language packs (cortex/patterns/packs.py): each signal
+# bit ORs every language's patterns for that signal, so one GPU scan
+# covers all enabled languages — matching the reference's extractSignals
+# loop over language packs. Bits:
+CORTEX_BIT_DECISION = 0
+CORTEX_BIT_CLOSE = 1
+CORTEX_BIT_WAIT = 2
+CORTEX_BIT_TOPIC = 3
+CORTEX_BIT_HIGH_IMPACT = 4   # any high-impact keyword (impact inference)
+CORTEX_MOOD_BIT0 = 8         # bits 8..12: frustrated, excited, tense,
+CORTEX_MOODS = ("frustrated", "excited", "tense", "productive", "exploratory")
+
+
+def _build_cortex_patterns() -> List[Tuple[str, int, str]]:
+    import re as _re
+
+    from ..cortex.patterns.packs import PACKS
+
+    out: List[Tuple[str, int, str]] = []
+    sig_bits = {"decision": CORTEX_BIT_DECISION, "close": CORTEX_BIT_CLOSE,
+                "wait": CORTEX_BIT_WAIT, "topic": CORTEX_BIT_TOPIC}
+    for code, pack in PACKS.items():
+        for sig, bit in sig_bits.items():
+            for i, pat in enumerate(pack["patterns"].get(sig, ())):
+                # hit-detection equivalence: a TRAILING bounded repeat
+                # {m,n} matches somewhere iff {m} does (the first m copies
+                # of a longer run) — truncating keeps the DFA small
+                pat = pat.replace("{3,40})", "{3})")
+                out.append((pat, bit, f"{code}-{sig}-{i}"))
+        for kw in pack.get("high_impact", ()):
+            out.append((_re.escape(kw), CORTEX_BIT_HIGH_IMPACT, f"{code}-impact-{kw}"))
+        for mi, mood in enumerate(CORTEX_MOODS):
+            pat = pack.get("moods", {}).get(mood)
+            if pat:
+                out.append((pat, CORTEX_MOOD_BIT0 + mi, f"{code}-mood-{mood}"))
+    return out
+
+
 _cache: Dict[str, MultiDFA] = {}
 
 _FAMILIES = {
@@ -136,12 +175,22 @@ _FAMILIES = {
     "entity": (ENTITY_PATTERNS, ENTITY_IGNORECASE),
 }
 
+# families compiled with unicode_word (\w, \b include bytes >= 0x80)
+_UNICODE_WORD_FAMILIES = {"cortex"}
+
+
+def _ensure_family_registered(name: str) -> None:
+    if name == "cortex" and "cortex" not in _FAMILIES:
+        pats = _build_cortex_patterns()
+        _FAMILIES["cortex"] = (pats, set(p[1] for p in pats))  # all icase
+
 _CACHE_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_dfa_cache")
 
 
 def _family_hash(name: str) -> str:
     patterns, icase = _FAMILIES[name]
-    blob = repr((DFA_COMPILER_VERSION, patterns, sorted(icase))).encode()
+    blob = repr((DFA_COMPILER_VERSION, patterns, sorted(icase),
+                 name in _UNICODE_WORD_FAMILIES)).encode()
     return hashlib.sha256(blob).hexdigest()[:16]
 
 
@@ -149,6 +198,7 @@ def get_family(name: str) -> MultiDFA:
     """Compile (and cache, incl. on disk) one of the named families."""
     if name in _cache:
         return _cache[name]
+    _ensure_family_registered(name)
     patterns, icase = _FAMILIES[name]
     cache_path = os.path.join(_CACHE_DIR, f"{name}-{_family_hash(name)}.npz")
     if os.path.isfile(cache_path):
@@ -165,7 +215,8 @@ def get_family(name: str) -> MultiDFA:
             return mdfa
         except Exception:
             pass
-    mdfa = compile_multi([(p, bit) for p, bit, _ in patterns], ignore_case_ids=icase)
+    mdfa = compile_multi([(p, bit) for p, bit, _ in patterns], ignore_case_ids=icase,
+                         unicode_word=name in _UNICODE_WORD_FAMILIES)
     try:
         os.makedirs(_CACHE_DIR, exist_ok=True)
         payload = {"n_dfas": np.int32(len(mdfa.dfas))}
@@ -182,4 +233,4 @@ def get_family(name: str) -> MultiDFA:
 
 
 def family_names() -> List[str]:
-    return ["redaction", "injection", "claims", "entity"]
+    return ["redaction", "injection", "claims", "entity", "cortex"]
