@@ -42,7 +42,7 @@ EPS = None  # epsilon edge label
 # Bump on any change to the NFA/DFA construction so on-disk compiled-family
 # caches (ops/pattern_sets.py:_family_hash) are invalidated; round-1's
 # alternation pending-lookahead bug shipped through a stale cache.
-DFA_COMPILER_VERSION = 3
+DFA_COMPILER_VERSION = 4
 
 
 class RegexError(ValueError):
@@ -351,6 +351,90 @@ class NFA:
 Frontier = List[Tuple[int, Optional[FrozenSet[int]]]]
 
 
+class _Unsupported(Exception):
+    pass
+
+
+def _restrict_first(node, allowed: FrozenSet[int]):
+    """Restrict the FIRST consumed byte of `node` to `allowed`.
+
+    Returns the rewritten node, or None when the restriction empties the
+    branch. Raises _Unsupported when the first byte cannot be isolated
+    (leading star/opt/guards) — callers fall back to the approximate
+    prev-byte-only \b guard for those."""
+    kind = node[0]
+    if kind == "char":
+        cs = node[1] & allowed
+        return ("char", cs) if cs else None
+    if kind == "cat":
+        items = list(node[1])
+        if not items:
+            raise _Unsupported
+        r = _restrict_first(items[0], allowed)
+        if r is None:
+            return None
+        return ("cat", [r] + items[1:])
+    if kind == "alt":
+        branches = []
+        for b in node[1]:
+            r = _restrict_first(b, allowed)
+            if r is not None:
+                branches.append(r)
+        if not branches:
+            return None
+        return branches[0] if len(branches) == 1 else ("alt", branches)
+    if kind == "plus":
+        r = _restrict_first(node[1], allowed)
+        if r is None:
+            return None
+        return ("cat", [r, ("star", node[1])])
+    if kind == "rep" and node[2] >= 1:
+        _, inner, m, n = node
+        r = _restrict_first(inner, allowed)
+        if r is None:
+            return None
+        return ("cat", [r, ("rep", inner, m - 1, None if n is None else n - 1)])
+    raise _Unsupported
+
+
+def _rewrite_exact_bounds(items: list) -> list:
+    """Rewrite mid-sequence \b into its exact two-sided form:
+    (prev not word AND next word) OR (prev word AND next non-word),
+    by intersecting the continuation's first-byte class. Fixes e.g.
+    '\b[\w.%+-]+@...' wrongly hitting '.@example.com' (leading '.' is
+    non-word, so re requires a word char before it — the old prev-only
+    guard passed at BOF)."""
+    out = []
+    i = 0
+    while i < len(items):
+        it = items[i]
+        if it[0] == "bound" and i + 1 < len(items):
+            ws = it[1]
+            nxt = items[i + 1]
+            try:
+                b_word = _restrict_first(nxt, ws)
+                b_nonw = _restrict_first(nxt, ANY - ws)
+            except _Unsupported:
+                out.append(it)
+                i += 1
+                continue
+            branches = []
+            if b_word is not None:
+                branches.append(("cat", [("nlb", ws), b_word]))
+            if b_nonw is not None:
+                branches.append(("cat", [("plb", ws), b_nonw]))
+            if not branches:
+                out.append(it)
+                i += 1
+                continue
+            out.append(branches[0] if len(branches) == 1 else ("alt", branches))
+            i += 2
+            continue
+        out.append(it)
+        i += 1
+    return out
+
+
 def _join(nfa: NFA, frontier: Frontier) -> int:
     """Collapse an all-pendingless frontier into one state (error if any
     end still carries a pending negative lookahead — lookahead must be the
@@ -382,6 +466,7 @@ def _build(nfa: NFA, node, start: int) -> Frontier:
         # trailing \b means "next byte is not word (or EOF)" -> lookahead
         if items and items[-1][0] == "bound":
             items[-1] = ("nla", items[-1][1])
+        items = _rewrite_exact_bounds(items)
         cur = start
         frontier: Frontier = [(start, None)]
         for item in items:
@@ -442,6 +527,11 @@ def _build(nfa: NFA, node, start: int) -> Frontier:
     if kind == "nlb":
         end = nfa.new_state()
         nfa.add(start, ("guard", "prev_not_in", node[1]), end)
+        return [(end, None)]
+    if kind == "plb":
+        # positive lookbehind guard: previous byte must be in the set
+        end = nfa.new_state()
+        nfa.add(start, ("guard", "prev_in", node[1]), end)
         return [(end, None)]
     if kind == "nla":
         # pending negative lookahead — resolved at accept time
