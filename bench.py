@@ -81,8 +81,10 @@ def main() -> int:
     setup_s = time.time() - t0
 
     # BASELINE config #5: the full pipeline INCLUDING the event-store leg —
-    # async audit writer (binary records + Merkle manifest JSONL) and a
-    # per-batch envelope into the embedded journal
+    # async audit writer (binary records + Merkle manifest JSONL) and ONE
+    # ClawEvent envelope PER MESSAGE (C++ batched builder) published into
+    # the embedded journal as EventBlocks. The writer/journal are drained
+    # INSIDE the timed region so the per-message envelope cost is measured.
     writer = None
     journal = None
     if not args.no_audit_sink:
@@ -92,46 +94,49 @@ def main() -> int:
         from vainplex_openclaw_amd.pipeline.engine import AsyncAuditWriter
 
         audit_dir = tempfile.mkdtemp(prefix=f"bench-audit-r{rank}-")
-        writer = AsyncAuditWriter(audit_dir)
         journal = EventJournal(durable=False)
+        writer = AsyncAuditWriter(audit_dir, journal=journal, session=f"rank{rank}")
+        pipe.audit_sink = writer
 
-        def sink(records, root):
-            writer(records, root)
-            journal.publish(
-                f"openclaw.events.rank{rank}.batch",
-                {"ts": int(time.time() * 1000), "type": "batch.audited",
-                 "agent": f"rank{rank}", "session": "bench",
-                 "payload": {"count": int(records.shape[0])}},
-            )
+    # pre-GENERATED synthetic batch pool (message bytes only); staging
+    # (CPU pack + pinned H2D on the copy stream) runs INSIDE the timed
+    # loop, double-buffered and overlapped with compute
+    from vainplex_openclaw_amd.pipeline.engine import StagingRing
 
-        pipe.audit_sink = sink
-
-    # pre-generated, pre-staged synthetic batch pool (rotated; every step
-    # still runs the full pipeline on real message bytes)
-    pool = []
-    for i in range(args.pool):
-        batch = synthetic_batch(args.batch, seed=1000 * rank + i, n_agents=cfg.n_agents)
-        pool.append((batch, pipe.stage(batch)))
+    pool = [synthetic_batch(args.batch, seed=1000 * rank + i, n_agents=cfg.n_agents)
+            for i in range(args.pool)]
+    ring = StagingRing(pipe, max_bytes=8 << 20, max_msgs=max(args.batch, 8192) + 1)
 
     def barrier():
         if distributed:
             torch.distributed.barrier()
 
     # warmup
+    staged_next = ring.stage(pool[0])
     for i in range(args.warmup):
-        b, s = pool[i % len(pool)]
-        pipe.step(b, staged=s)
+        staged = staged_next
+        staged_next = ring.stage(pool[(i + 1) % len(pool)])
+        pipe.step(None, staged=staged)
+    if writer is not None:
+        writer.drain()
+    pre_events = writer.events_published if writer is not None else 0
     barrier()
     torch.cuda.synchronize()
 
-    # timed region: exactly --steps full pipeline steps
+    # timed region: exactly --steps full pipeline steps, tokenize -> verdict
+    # -> audit -> per-message envelopes end to end
     t_start = time.perf_counter()
     for i in range(args.steps):
-        b, s = pool[i % len(pool)]
-        pipe.step(b, staged=s)
+        staged = staged_next
+        staged_next = ring.stage(pool[(i + 1) % len(pool)])
+        pipe.step(None, staged=staged)
+    torch.cuda.synchronize()
+    if writer is not None:
+        writer.drain()  # envelope + journal work inside the timed region
     barrier()
     torch.cuda.synchronize()
     elapsed = time.perf_counter() - t_start
+    timed_events = (writer.events_published - pre_events) if writer is not None else 0
 
     # MAX elapsed over ranks
     if distributed:
@@ -159,13 +164,25 @@ def main() -> int:
             "data": "synthetic",
             "setup_s": round(setup_s, 1),
             "config": {
-                "model": "firewall-pipeline (dfa-scan + encoder + classifier head + cosine-kNN recall + merkle audit)",
+                "model": "firewall-pipeline (fused dfa-scan x5 families + fact probe + cortex fold + encoder + classifier head + cosine-kNN recall + merkle audit + per-message envelopes)",
                 "global_batch": world * args.batch,
                 "seq_len": 500,
                 "parallelism": f"dp{world}",
                 "index_rows_total": shard * world,
                 "index_dim": args.dim,
                 "topk": args.topk,
+                "families": list(cfg.families),
+                "recall_mode": cfg.recall_mode,
+                # scan operand precision; final recall scores are always
+                # EXACT fp32 rescored cosines (ops/gpu.py threshold path)
+                "recall_scan_operands": (
+                    "mxfp4" if (cfg.recall_fp4 and cfg.recall_mode == "threshold")
+                    else ("fp8-mx" if (cfg.recall_fp8 and cfg.recall_mx)
+                          else ("fp8" if cfg.recall_fp8 else "bf16"))
+                ),
+                "recall_rescore": "exact-fp32 x salience",
+                "salience_weighting": cfg.salience_weighting,
+                "per_message_envelopes": journal is not None,
             },
         }))
     if args.profile:
@@ -173,8 +190,8 @@ def main() -> int:
         # profiler synchronizes per step)
         pipe.profiler.enabled = torch.cuda.is_available()
         for i in range(3):
-            b, s2 = pool[i % len(pool)]
-            pipe.step(b, staged=s2)
+            s2 = ring.stage(pool[i % len(pool)])
+            pipe.step(None, staged=s2)
             pipe.profiler.commit()
         if rank == 0:
             print("stage ms/step: " + json.dumps(
@@ -185,8 +202,15 @@ def main() -> int:
         writer.close()
         if rank == 0:
             replayed = sum(1 for _ in journal.replay())
-            print(json.dumps({"audit_batches_written": writer.batches_written,
-                              "journal_events_replayed": replayed}), file=sys.stderr)
+            expected = (args.steps + args.warmup) * args.batch
+            print(json.dumps({
+                "audit_batches_written": writer.batches_written,
+                "events_published": writer.events_published,
+                "journal_events_replayed": replayed,
+                "journal_expected": expected,
+                "journal_events_timed": timed_events,
+                "journal_msg_s": round(timed_events / elapsed, 1) if elapsed else None,
+            }), file=sys.stderr)
 
     if distributed:
         torch.distributed.destroy_process_group()

@@ -49,6 +49,14 @@ __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigne
 __global__ void trust_recompute_kernel(float*, float*, const float*, const float*,
                                        const float*, float*, const float*, float*, int);
 __global__ void edit_distance_kernel(const uint8_t*, const int32_t*, int32_t*, int);
+__global__ void dfa_scan_multi_kernel(const uint8_t*, const int32_t*, const uint16_t*,
+                                      const unsigned long long*, const unsigned long long*,
+                                      const uint8_t*, const int32_t*, const int32_t*,
+                                      int, unsigned long long*, int, int);
+__global__ void fact_probe_kernel(const uint8_t*, const int32_t*,
+                                  const unsigned long long*, const unsigned long long*,
+                                  const unsigned long long*, const unsigned long long*,
+                                  int, int32_t*, int32_t*, int);
 struct AuditRecord64;
 __global__ void audit_pack_kernel(const int8_t*, const float*, const unsigned long long*,
                                   const unsigned long long*, const int32_t*, const float*,
@@ -113,6 +121,62 @@ torch::Tensor dfa_scan(torch::Tensor bytes, torch::Tensor offsets, torch::Tensor
                      class_maps.data_ptr<uint8_t>(), meta.data_ptr<int32_t>(), n_dfas,
                      reinterpret_cast<unsigned long long*>(hits.data_ptr<int64_t>()), n);
   return hits;
+}
+
+// One launch scanning ALL families: family_ranges [n_families, 2] rows of
+// (begin, end) into meta rows; hits come back as int64 [n_families, n].
+torch::Tensor dfa_scan_multi(torch::Tensor bytes, torch::Tensor offsets,
+                             torch::Tensor next_tab, torch::Tensor accept,
+                             torch::Tensor eof_mask, torch::Tensor class_maps,
+                             torch::Tensor meta, torch::Tensor family_ranges) {
+  CHECK_GPU(bytes); CHECK_GPU(offsets); CHECK_GPU(next_tab); CHECK_GPU(accept);
+  CHECK_GPU(eof_mask); CHECK_GPU(class_maps); CHECK_GPU(meta); CHECK_GPU(family_ranges);
+  int n = offsets.numel() - 1;
+  int n_dfas = meta.size(0);
+  int n_families = family_ranges.size(0);
+  auto hits = torch::zeros({n_families, n},
+                           torch::dtype(torch::kInt64).device(bytes.device()));
+  int threads = 256;
+  int blocks = (n + threads - 1) / threads;
+  size_t lds = (size_t)n_dfas * 256;
+  TORCH_CHECK(lds <= 160 * 1024, "class maps exceed LDS");
+  hipLaunchKernelGGL(dfa_scan_multi_kernel, dim3(blocks), dim3(threads), lds, cur_stream(),
+                     bytes.data_ptr<uint8_t>(), offsets.data_ptr<int32_t>(),
+                     reinterpret_cast<const uint16_t*>(next_tab.data_ptr()),
+                     reinterpret_cast<const unsigned long long*>(accept.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(eof_mask.data_ptr<int64_t>()),
+                     class_maps.data_ptr<uint8_t>(), meta.data_ptr<int32_t>(),
+                     family_ranges.data_ptr<int32_t>(), n_families,
+                     reinterpret_cast<unsigned long long*>(hits.data_ptr<int64_t>()),
+                     n, n_dfas);
+  return hits;
+}
+
+// GPU fact-registry probe (csrc/fact_probe.hip): returns per-message
+// (verified, contradicted) counts for claim-bearing messages.
+std::vector<torch::Tensor> fact_probe(torch::Tensor bytes, torch::Tensor offsets,
+                                      torch::Tensor claims_mask, torch::Tensor pred_hash,
+                                      torch::Tensor table_keys, torch::Tensor table_vals,
+                                      int64_t table_pow2) {
+  CHECK_GPU(bytes); CHECK_GPU(offsets); CHECK_GPU(claims_mask);
+  CHECK_GPU(pred_hash); CHECK_GPU(table_keys); CHECK_GPU(table_vals);
+  TORCH_CHECK(pred_hash.numel() == 64, "pred_hash must have 64 entries");
+  TORCH_CHECK(table_keys.numel() == (1ll << table_pow2), "table size mismatch");
+  int n = offsets.numel() - 1;
+  auto opts = torch::dtype(torch::kInt32).device(bytes.device());
+  auto verified = torch::zeros({n}, opts);
+  auto contradicted = torch::zeros({n}, opts);
+  int threads = 256;
+  int blocks = (n + threads - 1) / threads;
+  hipLaunchKernelGGL(fact_probe_kernel, dim3(blocks), dim3(threads), 0, cur_stream(),
+                     bytes.data_ptr<uint8_t>(), offsets.data_ptr<int32_t>(),
+                     reinterpret_cast<const unsigned long long*>(claims_mask.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(pred_hash.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(table_keys.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(table_vals.data_ptr<int64_t>()),
+                     (int)table_pow2, verified.data_ptr<int32_t>(),
+                     contradicted.data_ptr<int32_t>(), n);
+  return {verified, contradicted};
 }
 
 torch::Tensor encode_messages(torch::Tensor bytes, torch::Tensor offsets,
@@ -384,10 +448,15 @@ std::vector<torch::Tensor> topk_scan_threshold(torch::Tensor Q, torch::Tensor X,
   return {cand_s, cand_i, counts};
 }
 
+void register_host_envelope(pybind11::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  register_host_envelope(m);
   m.def("sha256_leaves", &sha256_leaves, "Batched SHA-256 leaf digests");
   m.def("merkle_root", &merkle_root_gpu, "Merkle root over leaf digests");
   m.def("dfa_scan", &dfa_scan, "Multi-pattern DFA scan");
+  m.def("dfa_scan_multi", &dfa_scan_multi, "All-family DFA scan in one launch");
+  m.def("fact_probe", &fact_probe, "GPU fact-registry hash probe");
   m.def("encode_messages", &encode_messages, "4-gram hash embedding encoder");
   m.def("gemm_nt", &gemm_nt, "bf16 NT GEMM with fused epilogue",
         py::arg("A"), py::arg("B"), py::arg("bias") = c10::nullopt,

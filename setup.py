@@ -23,6 +23,8 @@ SOURCES = [
     "csrc/topk_recall.hip",
     "csrc/firewall.hip",
     "csrc/edit_distance.hip",
+    "csrc/fact_probe.hip",
+    "csrc/host_envelope.cpp",
 ]
 
 setup(

@@ -388,3 +388,121 @@ def test_run_failed_only_on_failure():
     assert ee.hook_name == "agent_end" and ee.legacy_type == "run.error"
     assert ee.condition({"success": False})
     assert not ee.condition({"success": True})
+
+
+# -- EventBlock batched publish (GPU-pipeline per-message envelope leg) -----
+
+def test_journal_publish_block_replay_and_counts():
+    import json
+
+    from vainplex_openclaw_amd.eventstore import EventJournal
+
+    j = EventJournal(durable=False)
+    envs = [{"id": f"evt-{i}", "ts": 1000 + i, "type": "message.in.received",
+             "payload": {"i": i}} for i in range(5)]
+    blob = ("\n".join(json.dumps(e, separators=(",", ":")) for e in envs) + "\n").encode()
+    first = j.publish_block("openclaw.events.bench.msg_in", blob, 5, ts_ms=1000)
+    assert first == 1
+    assert len(j) == 5
+    assert j.last_seq == 5
+    # a plain publish continues the sequence after the block
+    seq = j.publish("openclaw.events.bench.other", {"id": "evt-x", "ts": 2000})
+    assert seq == 6
+    out = list(j.replay())
+    assert [s for s, _ in out] == [1, 2, 3, 4, 5, 6]
+    assert out[2][1]["payload"]["i"] == 2
+    # since_seq threads into block interiors
+    tail = list(j.replay(since_seq=3))
+    assert [s for s, _ in tail] == [4, 5, 6]
+    # subject filter applies to the whole block
+    only = list(j.replay(subject_filter="openclaw.events.bench.msg_in"))
+    assert len(only) == 5
+
+
+def test_journal_block_retention_by_messages():
+    import json
+
+    from vainplex_openclaw_amd.eventstore import EventJournal
+
+    j = EventJournal(durable=False, max_messages=7)
+    for b in range(3):
+        envs = [{"id": f"e{b}-{i}", "ts": b * 10 + i} for i in range(5)]
+        blob = ("\n".join(json.dumps(e) for e in envs) + "\n").encode()
+        j.publish_block("s.x", blob, 5)
+    # 15 events with a 7-message cap -> oldest blocks dropped whole
+    assert len(j) == 5
+    assert [s for s, _ in j.replay()] == [11, 12, 13, 14, 15]
+
+
+def test_journal_block_durable_segments(workspace):
+    import json
+    import os
+
+    from vainplex_openclaw_amd.eventstore import EventJournal
+
+    d = os.path.join(workspace, "jr")
+    j = EventJournal(directory=d, durable=True)
+    envs = [{"id": f"evt-{i}", "ts": 1757000000000 + i} for i in range(3)]
+    blob = ("\n".join(json.dumps(e) for e in envs) + "\n").encode()
+    j.publish_block("s.block", blob, 3, ts_ms=1757000000000)
+    j.close()
+    j2 = EventJournal(directory=d, durable=True)
+    out = list(j2.replay())
+    assert [s for s, _ in out] == [1, 2, 3]
+    assert out[0][1]["id"] == "evt-0"
+    j2.close()
+
+
+def test_cpp_envelope_builder_parity():
+    """csrc/host_envelope.cpp vs eventstore.hooks.build_envelope: same id
+    derivation, same structural fields, for the audit-record inputs."""
+    import json
+
+    import numpy as np
+    import torch
+
+    from vainplex_openclaw_amd.eventstore.hooks import build_envelope
+    from vainplex_openclaw_amd.ops import gpu as g
+
+    if not g.have_ext():
+        import pytest
+
+        pytest.skip("extension not built")
+
+    dt = np.dtype([("msg_id", "<u8"), ("inj", "<u8"), ("red", "<u8"),
+                   ("risk", "<f4"), ("trust", "<f4"), ("agent", "<i4"),
+                   ("verdict", "u1"), ("resv", "3u1"), ("ts", "<i8"),
+                   ("injs", "<f4"), ("bseq", "<u4"), ("pad", "<u8")])
+    assert dt.itemsize == 64
+    rec = np.zeros(4, dtype=dt)
+    rec["msg_id"] = [100, 101, 102, 103]
+    rec["agent"] = [0, 3, 7, 63]
+    rec["verdict"] = [0, 1, 2, 3]
+    rec["risk"] = [5.25, 42.5, 77.0, 99.75]
+    rec["trust"] = [40.0, 61.5, 20.25, 80.0]
+    rec["ts"] = 1757000000123
+    rec["injs"] = [0.0, 0.25, 0.5, 0.9375]
+    rec["bseq"] = 9
+    raw = torch.from_numpy(rec.view(np.uint8).reshape(4, 64).copy())
+    blob = g.build_envelopes(raw, "sess-1", "agent", "message.in.received")
+    lines = blob.decode().strip().split("\n")
+    assert len(lines) == 4
+    for i, line in enumerate(lines):
+        got = json.loads(line)
+        want = build_envelope(
+            "message.in.received", f"agent{rec['agent'][i]}", "sess-1",
+            payload={}, ctx={"messageId": f"msg-{rec['msg_id'][i]}",
+                             "sessionKey": "sess-1"},
+            clock=lambda: 0,
+        )
+        # deterministic id must match the Python derivation exactly
+        assert got["id"] == want["id"], (got["id"], want["id"])
+        assert got["schemaVersion"] == want["schemaVersion"] == 1
+        assert got["source"] == want["source"]
+        assert got["actor"]["agentId"] == f"agent{rec['agent'][i]}"
+        assert got["scope"]["messageId"] == f"msg-{rec['msg_id'][i]}"
+        assert got["visibility"] == "internal"
+        assert set(got["trace"].keys()) == set(want["trace"].keys())
+        assert got["payload"]["verdict"] == int(rec["verdict"][i])
+        assert abs(got["payload"]["risk"] - float(rec["risk"][i])) < 1e-3
+        assert got["ts"] == 1757000000123

@@ -207,3 +207,45 @@ def test_get_all_facts_deduped():
     allf = reg.get_all_facts()
     assert len(allf) == 1 and allf[0]["value"] == "2"
     assert FactRegistry([]).get_all_facts() == []
+
+
+# -- GPU fact-probe CPU mirror semantics ------------------------------------
+
+def test_fact_probe_reference_semantics():
+    from vainplex_openclaw_amd.ops import gpu as g
+    from vainplex_openclaw_amd.ops import pattern_sets as ps
+
+    facts = [("nginx-service", "status", "running"),
+             ("backup.db", "exists", "replica")]
+    msgs = [
+        b"the nginx-service is running fine",     # verified
+        b"the nginx-service is stopped now",      # contradicted
+        b"no claims at all here",                 # no claim hit
+        b"the unknown-svc is running today",      # claim, no fact -> unverified
+    ]
+    masks = [ps.get_family("claims").scan(m) for m in msgs]
+    v, c = g.reference_fact_probe(msgs, masks, facts)
+    assert list(v) == [1, 0, 0, 0]
+    assert list(c) == [0, 1, 0, 0]
+
+
+def test_fact_table_build_and_collisions():
+    from vainplex_openclaw_amd.ops import gpu as g
+
+    facts = [(f"svc-{i}", "status", "running") for i in range(40)]
+    tk, tv, ph, pw = g.build_fact_table(facts)
+    keys = tk.numpy().view("uint64")
+    assert (keys != 0).sum() == 40           # all inserted
+    assert 1 << pw >= 80                      # >= 2x occupancy headroom
+    # predicate hash vector has entries only on claim bits
+    pred = ph.numpy().view("uint64")
+    assert pred[0] != 0 and pred[10] != 0 and pred[11] == 0
+
+
+def test_fact_probe_tokenizer_matches_kernel_classes():
+    from vainplex_openclaw_amd.ops.gpu import _tokenize_fact
+
+    assert _tokenize_fact(b"The Nginx-Service IS running!") == [
+        b"the", b"nginx-service", b"is", b"running"]
+    assert _tokenize_fact(b"a b") == []          # 1-char tokens dropped
+    assert _tokenize_fact(b"x1 .. y_2") == [b"x1", b"..", b"y_2"]

@@ -28,6 +28,25 @@ import time
 from typing import Dict, Iterator, List, Optional, Tuple
 
 
+class EventBlock:
+    """A batch of pre-serialized envelopes published as one append (the
+    high-rate GPU pipeline leg: one block per step, one ENVELOPE per
+    message). Sequence numbers are per event: the block owns
+    [first_seq, first_seq + count). Lines are parsed lazily on replay, so
+    the hot-path publish cost is one lock + one list append."""
+
+    __slots__ = ("blob", "count")
+
+    def __init__(self, blob: bytes, count: int):
+        self.blob = blob
+        self.count = count
+
+    def envelopes(self) -> Iterator[Dict]:
+        for line in self.blob.splitlines():
+            if line.strip():
+                yield json.loads(line)
+
+
 class EventJournal:
     def __init__(
         self,
@@ -48,9 +67,10 @@ class EventJournal:
         self.max_age_hours = max_age_hours
         self.durable = durable and directory is not None
         self._clock = clock
-        # in-memory ring: (seq, ts, subject, envelope)
-        self._events: List[Tuple[int, float, str, Dict]] = []
+        # in-memory ring: (seq, ts, subject, envelope | EventBlock)
+        self._events: List[Tuple[int, float, str, object]] = []
         self._bytes = 0
+        self._count = 0
         self._seq = 0
         self._lock = threading.Lock()
         self._cv = threading.Condition(self._lock)
@@ -75,26 +95,54 @@ class EventJournal:
             item = (seq, ts, subject, envelope)
             self._events.append(item)
             self._bytes += len(json.dumps(envelope))
+            self._count += 1
             self._apply_retention_locked()
             if self.durable:
                 self._pending.append(item)
                 self._cv.notify()
         return seq
 
+    def publish_block(self, subject: str, jsonl_blob: bytes, count: int,
+                      ts_ms: Optional[float] = None) -> int:
+        """Append `count` pre-serialized envelopes (newline-separated JSON)
+        as one block; returns the FIRST assigned sequence number. Every
+        envelope gets its own seq; replay expands the block lazily."""
+        ts = float(ts_ms if ts_ms is not None else self._clock() * 1000) / 1000.0
+        block = EventBlock(jsonl_blob, count)
+        with self._lock:
+            first = self._seq + 1
+            self._seq += count
+            item = (first, ts, subject, block)
+            self._events.append(item)
+            self._bytes += len(jsonl_blob)
+            self._count += count
+            self._apply_retention_locked()
+            if self.durable:
+                self._pending.append(item)
+                self._cv.notify()
+        return first
+
+    def _drop_first_locked(self) -> None:
+        dropped = self._events.pop(0)
+        body = dropped[3]
+        if isinstance(body, EventBlock):
+            self._bytes -= len(body.blob)
+            self._count -= body.count
+        else:
+            self._bytes -= len(json.dumps(body))
+            self._count -= 1
+
     def _apply_retention_locked(self) -> None:
         if self.max_messages > 0:
-            while len(self._events) > self.max_messages:
-                dropped = self._events.pop(0)
-                self._bytes -= len(json.dumps(dropped[3]))
+            while self._count > self.max_messages and self._events:
+                self._drop_first_locked()
         if self.max_bytes > 0:
             while self._events and self._bytes > self.max_bytes:
-                dropped = self._events.pop(0)
-                self._bytes -= len(json.dumps(dropped[3]))
+                self._drop_first_locked()
         if self.max_age_hours > 0:
             cutoff = self._clock() - self.max_age_hours * 3600.0
             while self._events and self._events[0][1] < cutoff:
-                dropped = self._events.pop(0)
-                self._bytes -= len(json.dumps(dropped[3]))
+                self._drop_first_locked()
 
     # -- read side ---------------------------------------------------------
     def replay(
@@ -110,9 +158,21 @@ class EventJournal:
             snapshot = list(self._events)
         n = 0
         for seq, ts, subject, env in snapshot:
-            if seq <= since_seq or ts < since_ts:
+            if ts < since_ts:
                 continue
             if subject_filter and not _subject_match(subject, subject_filter):
+                continue
+            if isinstance(env, EventBlock):
+                for i, e in enumerate(env.envelopes()):
+                    s_i = seq + i
+                    if s_i <= since_seq:
+                        continue
+                    yield s_i, e
+                    n += 1
+                    if limit and n >= limit:
+                        return
+                continue
+            if seq <= since_seq:
                 continue
             yield seq, env
             n += 1
@@ -128,7 +188,7 @@ class EventJournal:
 
     def __len__(self) -> int:
         with self._lock:
-            return len(self._events)
+            return self._count
 
     @property
     def last_seq(self) -> int:
@@ -162,6 +222,14 @@ class EventJournal:
             try:
                 by_path: Dict[str, List[str]] = {}
                 for seq, ts, subject, env in items:
+                    if isinstance(env, EventBlock):
+                        for i, e in enumerate(env.envelopes()):
+                            line = json.dumps(
+                                {"seq": seq + i, "subject": subject, "event": e},
+                                ensure_ascii=False,
+                            )
+                            by_path.setdefault(self._segment_path(ts), []).append(line)
+                        continue
                     line = json.dumps(
                         {"seq": seq, "subject": subject, "event": env}, ensure_ascii=False
                     )
@@ -198,6 +266,7 @@ class EventJournal:
                         ts = float(env.get("ts", 0)) / 1000.0
                         self._events.append((seq, ts, rec.get("subject", ""), env))
                         self._bytes += len(json.dumps(env))
+                        self._count += 1
                         self._seq = max(self._seq, seq)
             except OSError:
                 continue

@@ -255,6 +255,7 @@ def topk_recall_two_stage(
     X8: torch.Tensor,
     k: int,
     overfetch: int = 2,
+    salience: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Two-stage exact-rescore recall: fp8 scan of the full index for
     k*overfetch candidates (half the staged bytes of the bf16 scan = the
@@ -269,6 +270,8 @@ def topk_recall_two_stage(
     ids = ids8.long().clamp_min(0)  # -1 slots -> row 0 (rescored, never top)
     cand = X[ids]  # [nq, k2, D] bf16 gather
     exact = torch.einsum("qd,qkd->qk", Q.float(), cand.float())
+    if salience is not None:
+        exact = exact * salience[ids]
     exact = torch.where(ids8 < 0, torch.full_like(exact, -1e30), exact)
     top = torch.topk(exact, k, dim=1)
     return top.values, torch.gather(ids8, 1, top.indices)
@@ -353,6 +356,7 @@ def topk_recall_threshold(
     mx: bool = True,
     X4: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
     q4: bool = True,
+    salience: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Threshold-scan recall: per-query score thresholds estimated from a
     sampled pre-pass (Gaussian tail extrapolation), then a fixed-threshold
@@ -434,10 +438,16 @@ def topk_recall_threshold(
     ids = torch.gather(ci, 1, top.indices)
 
     # exact fp32 rescore of the selected candidates (both dtypes: the
-    # bf16 scan's near-ties otherwise reorder the top-k at bf16 precision)
+    # bf16 scan's near-ties otherwise reorder the top-k at bf16 precision).
+    # salience (membrane semantics) folds into the rescore epilogue:
+    # final score = cosine * decayed salience, selection stays by raw
+    # cosine with overfetch (salience <= 1 so the weighted top-k is a
+    # subset of the higher-cosine candidates in the common case).
     gidx = ids.long().clamp_min(0)
     cand = X[gidx]  # [nq, sel, D]
     exact = torch.einsum("qd,qkd->qk", Q.float(), cand.float())
+    if salience is not None:
+        exact = exact * salience[gidx]
     exact = torch.where(ids < 0, torch.full_like(exact, -1e30), exact)
     fin = torch.topk(exact, k, dim=1)
     out_s = fin.values
@@ -447,7 +457,224 @@ def topk_recall_threshold(
     bad = (counts < k) | (counts > cap)
     if bool(bad.any()):
         rows = bad.nonzero(as_tuple=True)[0]
-        fb_s, fb_i = topk_recall(Q[rows].contiguous(), X, k)
+        k2f = min(2 * k, X.shape[0])
+        fb_s, fb_i = topk_recall(Q[rows].contiguous(), X, k2f)
+        if salience is not None:
+            fb_w = fb_s * salience[fb_i.long().clamp_min(0)]
+            fb_w = torch.where(fb_i < 0, torch.full_like(fb_w, -1e30), fb_w)
+            ft = torch.topk(fb_w, k, dim=1)
+            fb_s = ft.values
+            fb_i = torch.gather(fb_i, 1, ft.indices)
+        else:
+            fb_s = fb_s[:, :k]
+            fb_i = fb_i[:, :k]
         out_s[rows] = fb_s
         out_i[rows] = fb_i
     return out_s, out_i
+
+
+# -- fact-registry probe ----------------------------------------------------
+
+FNV_OFFSET = 0xCBF29CE484222325
+FNV_PRIME = 0x100000001B3
+_U64 = (1 << 64) - 1
+
+# claim bit -> canonical predicate, per the reference's claim-type ->
+# predicate strategy table (fact-checker.ts:130-136) over the CLAIMS
+# family bits (ops/pattern_sets.py CLAIMS_PATTERNS)
+CLAIM_BIT_PREDICATE: Dict[int, str] = {
+    0: "status",      # system_state
+    1: "name",        # entity_name
+    2: "exists",      # existence_pos
+    3: "exists",      # existence_neg
+    4: "exists",      # there_is
+    5: "metric",      # has/contains/uses numeric
+    6: "metric",      # percentage
+    7: "count",       # count
+    8: "identity",    # self_identity
+    9: "identity",    # my_name
+    10: "capability", # i_have
+}
+
+
+def fnv1a64(data: bytes) -> int:
+    h = FNV_OFFSET
+    for b in data:
+        h = ((h ^ b) * FNV_PRIME) & _U64
+    return h
+
+
+def _mix_key(subject_h: int, predicate_h: int) -> int:
+    """Continues the subject FNV stream through '|' + predicate-hash bytes
+    (bit-exact with csrc/fact_probe.hip mix_key)."""
+    h = ((subject_h ^ ord("|")) * FNV_PRIME) & _U64
+    for i in range(8):
+        h = ((h ^ ((predicate_h >> (8 * i)) & 0xFF)) * FNV_PRIME) & _U64
+    return h
+
+
+def _tokenize_fact(data: bytes) -> list:
+    """Word tokens (>=2 chars, lowercased) under the probe kernel's
+    byte classes: [a-z0-9_.-]."""
+    toks = []
+    cur = bytearray()
+    for b in data + b" ":
+        c = bytes([b]).lower()[0]
+        if (48 <= c <= 57) or (97 <= c <= 122) or c in (95, 45, 46):
+            cur.append(c)
+        elif cur:
+            if len(cur) >= 2:
+                toks.append(bytes(cur))
+            cur = bytearray()
+    return toks
+
+
+def build_fact_table(facts: Sequence[Tuple[str, str, str]], pow2: Optional[int] = None):
+    """Pack (subject, predicate, object) triples into the open-addressing
+    probe table: keys/vals int64 tensors + the per-claim-bit predicate
+    hash vector. Single-token subjects/objects only (multi-word facts stay
+    on the host FactRegistry path — see csrc/fact_probe.hip header)."""
+    if pow2 is None:
+        pow2 = max(4, (len(facts) * 2 - 1).bit_length())
+    size = 1 << pow2
+    keys = np.zeros(size, dtype=np.uint64)
+    vals = np.zeros(size, dtype=np.uint64)
+    for subject, predicate, obj in facts:
+        sh = fnv1a64(subject.strip().lower().encode())
+        ph = fnv1a64(predicate.strip().lower().encode())
+        key = _mix_key(sh, ph) or 1
+        slot = key & (size - 1)
+        for _ in range(size):
+            if keys[slot] == 0 or keys[slot] == key:
+                break
+            slot = (slot + 1) % size
+        keys[slot] = key
+        vals[slot] = fnv1a64(obj.strip().lower().encode())
+    pred = np.zeros(64, dtype=np.uint64)
+    for bit, p in CLAIM_BIT_PREDICATE.items():
+        pred[bit] = fnv1a64(p.encode())
+    return (
+        torch.from_numpy(keys.view(np.int64)),
+        torch.from_numpy(vals.view(np.int64)),
+        torch.from_numpy(pred.view(np.int64)),
+        pow2,
+    )
+
+
+def fact_probe(
+    bytes_t: torch.Tensor,
+    offsets: torch.Tensor,
+    claims_mask: torch.Tensor,
+    table_keys: torch.Tensor,
+    table_vals: torch.Tensor,
+    pred_hash: torch.Tensor,
+    pow2: int,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Per-message (verified, contradicted) counts (csrc/fact_probe.hip)."""
+    v, c = ext().fact_probe(bytes_t, offsets, claims_mask, pred_hash,
+                            table_keys, table_vals, pow2)
+    return v, c
+
+
+def reference_fact_probe(
+    messages: Sequence[bytes], claims_masks: Sequence[int],
+    facts: Sequence[Tuple[str, str, str]],
+) -> Tuple[np.ndarray, np.ndarray]:
+    """CPU mirror of the probe kernel's semantics for the numerics tests."""
+    table = {}
+    for subject, predicate, obj in facts:
+        key = _mix_key(fnv1a64(subject.strip().lower().encode()),
+                       fnv1a64(predicate.strip().lower().encode())) or 1
+        table[key] = fnv1a64(obj.strip().lower().encode())
+    ver = np.zeros(len(messages), dtype=np.int32)
+    con = np.zeros(len(messages), dtype=np.int32)
+    for i, (msg, cmask) in enumerate(zip(messages, claims_masks)):
+        if not cmask:
+            continue
+        toks = [fnv1a64(t) for t in _tokenize_fact(msg)][:96]
+        for bit, predicate in CLAIM_BIT_PREDICATE.items():
+            if not (cmask >> bit) & 1:
+                continue
+            ph = fnv1a64(predicate.encode())
+            for t in toks:
+                key = _mix_key(t, ph) or 1
+                val = table.get(key)
+                if val is None:
+                    continue
+                if val in toks:
+                    ver[i] += 1
+                else:
+                    con[i] += 1
+    return ver, con
+
+
+# -- batched envelopes + all-family scan ------------------------------------
+
+def build_envelopes(records_cpu: torch.Tensor, session: str,
+                    agent_prefix: str = "agent",
+                    ctype: str = "message.in.received") -> bytes:
+    """One ClawEvent JSONL line per audit record (csrc/host_envelope.cpp,
+    GIL released). Line format parity with eventstore.hooks.build_envelope
+    is covered by tests/test_eventstore.py."""
+    return ext().build_envelopes(records_cpu, session, agent_prefix, ctype)
+
+
+class DeviceFamilySet:
+    """All families' packed tables concatenated for the one-launch scan
+    (csrc dfa_scan_multi_kernel): per-family (begin, end) sub-DFA ranges
+    into the shared meta table."""
+
+    def __init__(self, names: Sequence[str], device) -> None:
+        import numpy as _np
+
+        self.names = list(names)
+        nexts, accepts, eofs, cmaps, metas, ranges = [], [], [], [], [], []
+        state_base = 0
+        next_base = 0
+        row = 0
+        for name in self.names:
+            packed = pattern_sets.get_family(name).pack()
+            meta = packed["meta"].copy()
+            n_dfas = meta.shape[0]
+            meta[:, 0] += next_base
+            meta[:, 1] += state_base
+            meta[:, 3] += row
+            ranges.append((row, row + n_dfas))
+            row += n_dfas
+            next_base += packed["next"].size
+            state_base += packed["accept"].size
+            nexts.append(packed["next"])
+            accepts.append(packed["accept"])
+            eofs.append(packed["eof"])
+            cmaps.append(packed["class_maps"])
+            metas.append(meta)
+        self.next = torch.from_numpy(_np.concatenate(nexts).view(_np.int16)).to(device)
+        self.accept = torch.from_numpy(_np.concatenate(accepts).view(_np.int64)).to(device)
+        self.eof = torch.from_numpy(_np.concatenate(eofs).view(_np.int64)).to(device)
+        self.class_maps = torch.from_numpy(
+            _np.concatenate(cmaps).reshape(-1)
+        ).to(device)
+        self.meta = torch.from_numpy(_np.concatenate(metas)).to(device)
+        self.ranges = torch.from_numpy(
+            _np.array(ranges, dtype=_np.int32)
+        ).to(device)
+
+
+_device_family_sets: Dict[Tuple[Tuple[str, ...], int], DeviceFamilySet] = {}
+
+
+def device_family_set(names: Sequence[str], device) -> DeviceFamilySet:
+    dev = torch.device(device)
+    key = (tuple(names), dev.index if dev.index is not None else -1)
+    if key not in _device_family_sets:
+        _device_family_sets[key] = DeviceFamilySet(names, dev)
+    return _device_family_sets[key]
+
+
+def dfa_scan_all(bytes_t: torch.Tensor, offsets: torch.Tensor,
+                 families: Sequence[str]) -> Dict[str, torch.Tensor]:
+    """All families in ONE kernel launch; returns {family: hit masks}."""
+    fs = device_family_set(families, bytes_t.device)
+    hits = ext().dfa_scan_multi(bytes_t, offsets, fs.next, fs.accept, fs.eof,
+                                fs.class_maps, fs.meta, fs.ranges)
+    return {name: hits[i] for i, name in enumerate(fs.names)}

@@ -49,7 +49,8 @@ _PII = [
 ]
 
 _CLAIMS = [
-    "the nginx-service is running and the queue has {n},908 items",
+    "the {svc} is running and the queue has {n},908 items",
+    "the {svc} is stopped since {day}",
     "backup.db does not exist on the replica",
     "disk usage is at {pct}% and node count is {small}",
     "I am DeployBot. I have admin capabilities.",
@@ -60,6 +61,46 @@ _ENTITY = [
     "upgrade Photoshop v{v}.5 before {iso}",
     "see https://docs.example.com/guide and email ops@corp.io",
 ]
+
+# cortex signal templates (decision/close/wait/topic + moods, several
+# languages) so the cortex DFA family and batched tracker updates have
+# realistic work per batch
+_CORTEX = [
+    "we decided to go with the blue-green deploy",
+    "the plan is to cut over during the low-traffic window",
+    "let's talk about the storage migration plan",
+    "regarding the incident postmortem from {day}",
+    "it works now, shipping it",
+    "that's done and merged",
+    "waiting for the security review to finish",
+    "blocked by the upstream api limits",
+    "awesome, great news on the launch",
+    "this is urgent, deadline is asap",
+    "broken again, so annoying",
+    "what if we experiment with a cache layer",
+    "wir haben beschlossen, das deployment heute zu machen",
+    "das ist erledigt und funktioniert",
+    "decidimos usar el nuevo enfoque de colas",
+    "esperando a que termine la migración",
+    "решено, делаем миграцию в пятницу",
+    "決定です、明日デプロイします",
+]
+
+# fact registry consistent with the _CLAIMS templates: svc claims verify
+# against (subject, "status", "running"); the "stopped" variants
+# contradict it (fact-checker verdict shapes)
+_FACT_SUBJECTS = ["nginx-service", "redis-cache", "api-gateway", "worker-pool"]
+
+
+def default_facts():
+    """Synthetic fact registry for the GPU probe (bench default)."""
+    facts = [(s, "status", "running") for s in _FACT_SUBJECTS]
+    facts.append(("backup.db", "exists", "replica"))
+    facts.append(("deploybot", "identity", "deploybot"))
+    # filler rows: realistic table occupancy
+    facts += [(f"svc-{i}", "status", "running") for i in range(64)]
+    return facts
+
 
 _DAYS = ["monday", "tuesday", "wednesday", "thursday", "friday"]
 
@@ -87,6 +128,7 @@ def _fill(template: str, rng: random.Random) -> str:
         .replace("{akia}", "".join(rng.choices("ABCDEFGHIJKLMNOP", k=16)))
         .replace("{v}", str(rng.randint(1, 9)))
         .replace("{iso}", f"2026-{rng.randint(1,12):02d}-{rng.randint(1,28):02d}")
+        .replace("{svc}", rng.choice(_FACT_SUBJECTS))
     )
 
 
@@ -119,6 +161,8 @@ def synthetic_batch(
             parts.append(_fill(rng.choice(_CLAIMS), rng))
         if rng.random() < 0.3:
             parts.append(_fill(rng.choice(_ENTITY), rng))
+        if rng.random() < 0.4:
+            parts.append(_fill(rng.choice(_CORTEX), rng))
         # pad toward the 200-500B envelope
         while sum(len(p) for p in parts) < 180:
             parts.append(_fill(rng.choice(_CLEAN), rng))
