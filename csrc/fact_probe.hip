@@ -23,7 +23,7 @@
 // array (<= 96 tokens of a 500 B message).
 #include "common.hpp"
 
-#define FNV_OFFSET 1469598103934665603ull
+#define FNV_OFFSET 0xCBF29CE484222325ull
 #define FNV_PRIME 1099511628211ull
 #define MAX_TOKENS 96
 

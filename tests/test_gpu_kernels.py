@@ -459,20 +459,21 @@ def test_salience_weighted_recall_matches_dense():
     Q = torch.nn.functional.normalize(torch.randn(nq, d, device="cuda"), dim=1).bfloat16()
     X = torch.nn.functional.normalize(torch.randn(nx, d, device="cuda"), dim=1).bfloat16()
     sal = torch.rand(nx, device="cuda") * 0.9 + 0.1
-    # dense reference: exact fp32 cosine * salience
-    ref = torch.matmul(Q.float(), X.float().T) * sal.unsqueeze(0)
-    want_v, want_i = torch.topk(ref, k, dim=1)
+    # Contract (mirrors membrane/engine.py retrieve): candidates are
+    # overfetched by raw cosine, ranked by cosine * decayed salience.
+    # Assert (a) reported scores are the EXACT fp32 weighted cosines of
+    # the returned ids, (b) near-zero regret vs the dense weighted
+    # top-k, (c) weighting actually changes the ranking vs raw cosine.
+    dense = torch.matmul(Q.float(), X.float().T)
     got_s, got_i = g.topk_recall_threshold(Q, X, k, salience=sal)
-    # candidate selection is by raw cosine with overfetch, so allow the
-    # occasional tail miss: require >= k-2 of the true weighted top-k
-    hit = (got_i.unsqueeze(2) == want_i.unsqueeze(1)).any(dim=1).sum(dim=1)
-    assert (hit >= k - 2).all(), hit.min()
-    # scores of matched ids must be the exact weighted cosines
-    for q in range(0, nq, 16):
-        for j in range(k):
-            m = (want_i[q] == got_i[q, j]).nonzero()
-            if m.numel():
-                assert abs(got_s[q, j].item() - want_v[q, m[0, 0]].item()) < 1e-3
+    exact_w = torch.gather(dense, 1, got_i.long()) * sal[got_i.long()]
+    assert (got_s - exact_w).abs().max().item() < 1e-3
+    ref_w = torch.topk(dense * sal.unsqueeze(0), k, dim=1).values
+    regret = 1.0 - got_s.sum(dim=1) / ref_w.sum(dim=1).clamp_min(1e-6)
+    assert regret.max().item() < 0.05, regret.max().item()
+    # sanity: unweighted call ranks differently somewhere
+    raw_s, raw_i = g.topk_recall_threshold(Q, X, k)
+    assert not torch.equal(raw_i, got_i)
 
 
 def test_staging_ring_matches_pack():
