@@ -432,8 +432,13 @@ def topk_recall_threshold(
     cs = torch.where(valid, cs, torch.full_like(cs, -1e30))
 
     # 4. select: top candidates by scan score (wider for the noisier
-    # fp4 scan; the exact rescore makes overfetch nearly free)
-    sel = min(cap, max(4 * k, 128)) if use_fp4 else min(cap, max(2 * k, 32))
+    # fp4 scan, and for salience weighting — the weighted top-k can sit
+    # below the raw-cosine top-2k; the exact rescore makes overfetch
+    # nearly free)
+    if use_fp4 or salience is not None:
+        sel = min(cap, max(8 * k if salience is not None else 4 * k, 128))
+    else:
+        sel = min(cap, max(2 * k, 32))
     top = torch.topk(cs, sel, dim=1)
     ids = torch.gather(ci, 1, top.indices)
 
@@ -457,19 +462,18 @@ def topk_recall_threshold(
     bad = (counts < k) | (counts > cap)
     if bool(bad.any()):
         rows = bad.nonzero(as_tuple=True)[0]
-        k2f = min(2 * k, X.shape[0])
+        k2f = min(4 * k, X.shape[0])
         fb_s, fb_i = topk_recall(Q[rows].contiguous(), X, k2f)
+        # exact fp32 rescore (+ salience weight) so fallback rows report
+        # the same score definition as the main path
+        fb_g = fb_i.long().clamp_min(0)
+        fb_exact = torch.einsum("qd,qkd->qk", Q[rows].float(), X[fb_g].float())
         if salience is not None:
-            fb_w = fb_s * salience[fb_i.long().clamp_min(0)]
-            fb_w = torch.where(fb_i < 0, torch.full_like(fb_w, -1e30), fb_w)
-            ft = torch.topk(fb_w, k, dim=1)
-            fb_s = ft.values
-            fb_i = torch.gather(fb_i, 1, ft.indices)
-        else:
-            fb_s = fb_s[:, :k]
-            fb_i = fb_i[:, :k]
-        out_s[rows] = fb_s
-        out_i[rows] = fb_i
+            fb_exact = fb_exact * salience[fb_g]
+        fb_exact = torch.where(fb_i < 0, torch.full_like(fb_exact, -1e30), fb_exact)
+        ft = torch.topk(fb_exact, k, dim=1)
+        out_s[rows] = ft.values
+        out_i[rows] = torch.gather(fb_i, 1, ft.indices)
     return out_s, out_i
 
 
