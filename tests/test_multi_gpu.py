@@ -68,7 +68,7 @@ def _entry(fn, rank, world, port, err_q, *args):
         torch.cuda.set_device(rank % max(torch.cuda.device_count(), 1))
         torch.distributed.init_process_group(
             "nccl", rank=rank, world_size=world,
-            timeout=datetime.timedelta(seconds=120),
+            timeout=datetime.timedelta(seconds=40),
         )
         fn(rank, world, *args)
         torch.distributed.barrier()
@@ -247,7 +247,7 @@ def test_rccl_world2_single_gpu_smoke():
     devices in a communicator — if it does, record that as a skip, not a
     failure (the xfail-style probe VERDICT round 1 asked for)."""
     try:
-        _spawn(2, _w_allgather_dtypes, timeout=150)
+        _spawn(2, _w_allgather_dtypes, timeout=45)
     except AssertionError as exc:
         msg = str(exc)
         if any(t in msg for t in ("Duplicate GPU", "invalid usage", "NCCL",
