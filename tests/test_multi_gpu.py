@@ -68,7 +68,7 @@ def _entry(fn, rank, world, port, err_q, *args):
         torch.cuda.set_device(rank % max(torch.cuda.device_count(), 1))
         torch.distributed.init_process_group(
             "nccl", rank=rank, world_size=world,
-            timeout=datetime.timedelta(seconds=40),
+            timeout=datetime.timedelta(seconds=120),
         )
         fn(rank, world, *args)
         torch.distributed.barrier()
