@@ -22,9 +22,11 @@ class EventStorePlugin:
     description = "Publish agent events for audit, replay, and multi-agent sharing"
     version = "0.1.0"
 
-    def __init__(self, journal: Optional[EventJournal] = None, journal_dir: Optional[str] = None):
+    def __init__(self, journal: Optional[EventJournal] = None, journal_dir: Optional[str] = None,
+                 nats_transport: Any = None):
         self.journal = journal
         self.journal_dir = journal_dir
+        self.nats_transport = nats_transport  # injectable (tests); None -> TCP
         self.publisher: Optional[EventPublisher] = None
         self.config: Dict[str, Any] = {}
 
@@ -34,6 +36,20 @@ class EventStorePlugin:
         if not cfg["enabled"]:
             api.logger.info("[nats-eventstore] Disabled via config")
             return
+        if self.journal is None and (cfg.get("useNats") or self.nats_transport is not None):
+            # live NATS backend over the wire-protocol client; connect +
+            # ensure-stream can fail without egress -> fall back to the
+            # embedded journal (fire-and-forget philosophy)
+            from .nats_client import JetStreamClient, NatsPublishAdapter
+
+            try:
+                client = JetStreamClient(cfg, logger=api.logger,
+                                         transport=self.nats_transport)
+                client.connect()
+                self.journal = NatsPublishAdapter(client)
+            except Exception as exc:
+                api.logger.warn("[nats-eventstore] NATS connect failed (%s); "
+                                "using embedded journal", exc)
         if self.journal is None:
             ret = cfg["retention"]
             self.journal = EventJournal(
