@@ -516,3 +516,133 @@ def test_pipeline_step_cortex_and_validation_outputs():
     v = out["validation"]
     assert torch.equal(v == 2, contra & (tr < 40.0))
     assert torch.equal(v == 1, contra & (tr >= 40.0) & (tr < 60.0))
+
+
+def _verdict_mirror(inj_hit, cred_hit, inj_score, trust, tool_risk, freq, hour,
+                    inj_threshold=0.9):
+    """Python mirror of csrc/firewall.hip firewall_verdict_kernel."""
+    f_tool = tool_risk * 0.01 * 30.0
+    f_time = 15.0 if (hour < 8 or hour >= 23) else 0.0
+    f_trust = (100.0 - trust) * 0.01 * 20.0
+    f_freq = min(freq / 20.0, 1.0) * 15.0
+    injected = inj_hit or inj_score > inj_threshold
+    f_scope = 20.0 if injected else 0.0
+    r = min(f_tool + f_time + f_trust + f_freq + f_scope, 100.0)
+    if cred_hit or (injected and trust < 60.0):
+        v = 3
+    elif injected or (r > 75.0 and trust < 80.0):
+        v = 2
+    elif r > 50.0:
+        v = 1
+    else:
+        v = 0
+    return v, r
+
+
+def test_firewall_verdict_full_matrix():
+    """Exact kernel parity over the cross product the round-1 verdict
+    called out as untested: trust tiers x night-mode hours x frequency
+    edges x injection/credential/classifier combinations."""
+    import itertools
+
+    trusts = [0.0, 39.5, 40.0, 59.5, 60.0, 79.5, 80.0, 100.0]
+    hours = [3, 12, 23]
+    freqs = [0, 10, 20, 40]
+    inj_hits = [0, 1]
+    creds = [0, 1]
+    inj_scores = [0.0, 0.89, 0.91]
+    risks = [0.0, 30.0, 95.0]
+    combos = list(itertools.product(trusts, hours, freqs, inj_hits, creds,
+                                    inj_scores, risks))
+    # one kernel launch per hour value (hour is a scalar kernel arg)
+    for hour in hours:
+        rows = [c for c in combos if c[1] == hour]
+        B_ = len(rows)
+        A_ = len(trusts)
+        t_of = {t: i for i, t in enumerate(trusts)}
+        inj = torch.tensor([c[3] for c in rows], dtype=torch.int64, device="cuda")
+        red = torch.tensor([c[4] for c in rows], dtype=torch.int64, device="cuda")
+        logits = torch.zeros(B_, 8, device="cuda")
+        logits[:, 0] = torch.tensor([c[5] for c in rows], device="cuda")
+        agent_idx = torch.tensor([t_of[c[0]] for c in rows], dtype=torch.int32,
+                                 device="cuda")
+        trust_vec = torch.tensor(trusts, device="cuda")
+        tool_risk = torch.tensor([c[6] for c in rows], device="cuda")
+        freq = torch.tensor([c[2] for c in rows], dtype=torch.int32, device="cuda")
+        v, r, sd, vd = g.firewall_verdict(inj, red, logits, agent_idx, trust_vec,
+                                          tool_risk, freq, hour=hour, n_agents=A_)
+        v_np, r_np = v.cpu().numpy(), r.cpu().numpy()
+        for j, c in enumerate(rows):
+            wv, wr = _verdict_mirror(c[3], c[4], c[5], c[0], c[6], c[2], hour)
+            assert v_np[j] == wv, (c, v_np[j], wv)
+            assert abs(r_np[j] - wr) < 1e-3, (c, r_np[j], wr)
+        # trust-delta bookkeeping: every deny is a violation, the rest
+        # successes (engine.ts trust learning on deny)
+        n_deny = int((v == 3).sum())
+        assert abs(vd.sum().item() - n_deny) < 1e-3
+        assert abs(sd.sum().item() - (B_ - n_deny)) < 1e-3
+
+
+def test_firewall_verdict_agrees_with_host_engine_invariants():
+    """Cross-check the fused kernel against the HOST GovernanceEngine on
+    the safety-critical scenarios both implement (the kernel is the
+    batched distillation of the policy engine, not a line-for-line port;
+    agreement is asserted on invariants, not raw verdict codes):
+      - credential material in params  -> both deny
+      - clean call, high trust, day    -> host allows, kernel allow/audit
+      - severity monotone in risk factors on the kernel side."""
+    import tempfile
+
+    from vainplex_openclaw_amd.governance.engine import GovernanceEngine
+
+    eng = GovernanceEngine({"builtinPolicies": {"credentialGuard": True}},
+                           tempfile.mkdtemp(prefix="gpu-verdict-"))
+    eng.start()
+
+    # credential scenario: host credential-guard policy denies credential
+    # FILE access (builtin-policies.ts:60-100); the kernel's analog is
+    # the credential-bit deny
+    ctx = eng.build_context("before_tool_call", "agentX", tool_name="exec",
+                            tool_params={"command": "cat secrets/.env"})
+    host = eng.evaluate(ctx)
+    assert host["action"] == "deny"
+    red = torch.tensor([1], dtype=torch.int64, device="cuda")  # cred bit
+    inj = torch.zeros(1, dtype=torch.int64, device="cuda")
+    logits = torch.zeros(1, 8, device="cuda")
+    aidx = torch.zeros(1, dtype=torch.int32, device="cuda")
+    trust = torch.full((1,), 40.0, device="cuda")
+    v, _, _, _ = g.firewall_verdict(inj, red, logits, aidx, trust,
+                                    torch.zeros(1, device="cuda"),
+                                    torch.zeros(1, dtype=torch.int32, device="cuda"),
+                                    hour=12, n_agents=1)
+    assert int(v[0]) == 3  # deny
+
+    # clean low-risk call: host allows; kernel must not deny/2fa
+    ctx2 = eng.build_context("before_tool_call", "agentX", tool_name="read",
+                             tool_params={"path": "notes.txt"})
+    host2 = eng.evaluate(ctx2)
+    assert host2["action"] in ("allow", "audit")
+    v2, _, _, _ = g.firewall_verdict(
+        torch.zeros(1, dtype=torch.int64, device="cuda"),
+        torch.zeros(1, dtype=torch.int64, device="cuda"),
+        logits, aidx, torch.full((1,), 80.0, device="cuda"),
+        torch.full((1,), 10.0, device="cuda"),
+        torch.zeros(1, dtype=torch.int32, device="cuda"),
+        hour=12, n_agents=1)
+    assert int(v2[0]) in (0, 1)
+
+    # monotonicity: verdict severity never decreases as factors worsen
+    sev = []
+    for trust_v, hour, freq_v, risk_v in [
+        (90.0, 12, 0, 5.0), (60.0, 12, 10, 30.0), (40.0, 23, 20, 60.0),
+        (10.0, 3, 40, 95.0),
+    ]:
+        vv, _, _, _ = g.firewall_verdict(
+            torch.zeros(1, dtype=torch.int64, device="cuda"),
+            torch.zeros(1, dtype=torch.int64, device="cuda"),
+            logits, aidx, torch.full((1,), trust_v, device="cuda"),
+            torch.full((1,), risk_v, device="cuda"),
+            torch.full((1,), freq_v, dtype=torch.int32, device="cuda"),
+            hour=hour, n_agents=1)
+        sev.append(int(vv[0]))
+    assert sev == sorted(sev), sev
