@@ -182,3 +182,216 @@ def test_language_all_and_invalid():
     assert P._resolve_codes(["en", "fr"]) == ["en", "fr"]
     reg = P.get_registry("all")
     assert len(reg.codes) == 10
+
+
+# ===========================================================================
+# Full per-language signal tables (patterns.test.ts parity depth): one
+# positive sample per family per language, asserted against THAT
+# language's registry, plus cross-language negatives.
+# ===========================================================================
+
+CLOSE_SAMPLES = {
+    "en": "ok that is done now",
+    "de": "das ist erledigt und gut",
+    "es": "está resuelto por fin",
+    "fr": "c'est fait enfin",
+    "it": "è fatto finalmente",
+    "pt": "está feito afinal",
+    "ru": "всё готово наконец",
+    "ja": "タスクは完了です",
+    "ko": "작업 완료 했습니다",
+    "zh": "任务完成了",
+}
+
+WAIT_SAMPLES = {
+    "en": "waiting for the review to land",
+    "de": "wir warten auf die freigabe",
+    "es": "esperando a la revisión",
+    "fr": "en attente de la validation",
+    "it": "in attesa di conferma",
+    "pt": "aguardando a aprovação",
+    "ru": "ждём подтверждения",
+    "ja": "レビューを待っています",
+    "ko": "승인 대기 중 입니다",
+    "zh": "等待审核结果",
+}
+
+TOPIC_SAMPLES = {
+    "en": "let's talk about the billing system",
+    "de": "zurück zu dem deployment plan",
+    "es": "hablemos de la nueva arquitectura",
+    "fr": "parlons de la migration batch",
+    "it": "parliamo di architettura nuova",
+    "pt": "agora sobre o plano de backup",
+    "ru": "давайте обсудим новую схему",
+    "ja": "データベースについて 設計を考えます",
+    "ko": "배포 에 대해 이야기해 봅시다 계획을",
+    "zh": "关于 数据库迁移的事",
+}
+
+MOOD_SAMPLES = {
+    ("en", "frustrated"): "this is broken again",
+    ("en", "excited"): "awesome work team",
+    ("en", "tense"): "careful, deadline is near",
+    ("en", "productive"): "merged and shipped",
+    ("en", "exploratory"): "what if we cache it",
+    ("de", "frustrated"): "alles kaputt, mist",
+    ("de", "excited"): "das ist genial",
+    ("de", "tense"): "vorsicht, das ist riskant",
+    ("de", "productive"): "behoben und fertig",
+    ("de", "exploratory"): "vielleicht ein experiment",
+    ("es", "frustrated"): "está roto otra vez",
+    ("es", "excited"): "quedó genial",
+    ("es", "tense"): "es urgente, cuidado",
+    ("es", "productive"): "arreglado y desplegado",
+    ("es", "exploratory"): "quizás un experimento",
+    ("fr", "frustrated"): "c'est cassé encore, zut",
+    ("fr", "excited"): "c'est génial",
+    ("fr", "tense"): "attention c'est risqué",
+    ("fr", "productive"): "déployé et réglé",
+    ("fr", "exploratory"): "et si on essayait une idée",
+    ("it", "frustrated"): "è rotto di nuovo, accidenti",
+    ("it", "excited"): "è fantastico",
+    ("it", "tense"): "attenzione, è rischioso",
+    ("it", "productive"): "risolto e deployato",
+    ("it", "exploratory"): "forse un esperimento",
+    ("pt", "frustrated"): "quebrado de novo, droga",
+    ("pt", "excited"): "ficou incrível",
+    ("pt", "tense"): "cuidado, é arriscado",
+    ("pt", "productive"): "resolvido e implantado",
+    ("pt", "exploratory"): "talvez um experimento",
+    ("ru", "frustrated"): "опять сломалось, блин",
+    ("ru", "excited"): "получилось отлично",
+    ("ru", "tense"): "осторожно, это срочно",
+    ("ru", "productive"): "сделано и работает",
+    ("ru", "exploratory"): "может быть эксперимент",
+    ("ja", "frustrated"): "また壊れた、最悪",
+    ("ja", "excited"): "すごい、最高です",
+    ("ja", "tense"): "注意、締め切りが近い",
+    ("ja", "productive"): "修正してデプロイ済み",
+    ("ja", "exploratory"): "もし試してみたら",
+    ("ko", "frustrated"): "또 고장났어, 짜증",
+    ("ko", "excited"): "대박 멋지다",
+    ("ko", "tense"): "조심해, 마감이 긴급해",
+    ("ko", "productive"): "수정 후 배포됨",
+    ("ko", "exploratory"): "아이디어 하나 해보자",
+    ("zh", "frustrated"): "又坏了，真烦人",
+    ("zh", "excited"): "太棒了",
+    ("zh", "tense"): "小心，这个紧急",
+    ("zh", "productive"): "修好了，已部署",
+    ("zh", "exploratory"): "也许可以试试",
+}
+
+NEGATIVE_TEXT = "the quarterly numbers look flat across regions"
+
+
+def _fires(reg, family, text):
+    return any(rx.search(text) for rx in reg.get_patterns(family))
+
+
+@pytest.mark.parametrize("code", language_codes())
+def test_decision_fires_own_language(code):
+    reg = get_registry(code)
+    assert _fires(reg, "decision", DECISION_SAMPLES[code]), code
+
+
+@pytest.mark.parametrize("code", language_codes())
+def test_close_fires_own_language(code):
+    reg = get_registry(code)
+    assert _fires(reg, "close", CLOSE_SAMPLES[code]), code
+
+
+@pytest.mark.parametrize("code", language_codes())
+def test_wait_fires_own_language(code):
+    reg = get_registry(code)
+    assert _fires(reg, "wait", WAIT_SAMPLES[code]), code
+
+
+@pytest.mark.parametrize("code", language_codes())
+def test_topic_fires_own_language(code):
+    reg = get_registry(code)
+    assert _fires(reg, "topic", TOPIC_SAMPLES[code]), code
+
+
+@pytest.mark.parametrize("code", language_codes())
+def test_no_family_fires_on_neutral_text(code):
+    reg = get_registry(code)
+    for family in ("decision", "close", "wait", "topic"):
+        assert not _fires(reg, family, NEGATIVE_TEXT), (code, family)
+
+
+@pytest.mark.parametrize("code,mood", sorted(MOOD_SAMPLES))
+def test_mood_detected_per_language(code, mood):
+    got = detect_mood(MOOD_SAMPLES[(code, mood)], code)
+    assert got == mood, (code, mood, got)
+
+
+@pytest.mark.parametrize("code", language_codes())
+def test_mood_neutral_on_neutral_text(code):
+    assert detect_mood(NEGATIVE_TEXT, code) == "neutral"
+    assert detect_mood("", code) == "neutral"
+
+
+def test_close_checkmark_fires_in_every_language():
+    for code in language_codes():
+        assert _fires(get_registry(code), "close", "✅"), code
+
+
+def test_both_is_en_plus_de():
+    reg = get_registry("both")
+    assert _fires(reg, "decision", DECISION_SAMPLES["en"])
+    assert _fires(reg, "decision", DECISION_SAMPLES["de"])
+    assert not _fires(reg, "decision", DECISION_SAMPLES["ru"])
+
+
+def test_all_covers_every_language():
+    reg = get_registry("all")
+    for code in language_codes():
+        assert _fires(reg, "decision", DECISION_SAMPLES[code]), code
+
+
+def test_language_list_selection():
+    reg = get_registry(["es", "ru"])
+    assert _fires(reg, "decision", DECISION_SAMPLES["es"])
+    assert _fires(reg, "decision", DECISION_SAMPLES["ru"])
+    assert not _fires(reg, "wait", WAIT_SAMPLES["ja"])
+
+
+@pytest.mark.parametrize("code", ["en", "de", "es", "fr", "it", "pt"])
+def test_case_insensitive_decision(code):
+    reg = get_registry(code)
+    assert _fires(reg, "decision", DECISION_SAMPLES[code].upper()), code
+
+
+def test_mood_last_match_wins_ordering():
+    # 'broken again' (frustrated) appears AFTER 'awesome' (excited)
+    assert detect_mood("awesome start but broken again", "en") == "frustrated"
+    assert detect_mood("broken again but awesome now", "en") == "excited"
+    # same rule across languages
+    assert detect_mood("genial aber alles kaputt", "de") == "frustrated"
+
+
+def test_undecided_partial_word_no_decision_fire():
+    # patterns.test.ts: 'undecided' must not fire the decision family
+    reg = get_registry("en")
+    assert not _fires(reg, "decision", "the team is still undecided about it")
+
+
+NOISE_TOPICS = [
+    ("it", True), ("that", True), ("the", True), ("abc", True),
+    ("i something", True), ("billing system", False),
+    ("storage migration", False), ("x" * 61, True),
+    ("two\nlines", True), ("today", True), ("deploy plan", False),
+]
+
+
+@pytest.mark.parametrize("topic,noisy", NOISE_TOPICS)
+def test_noise_topic_table(topic, noisy):
+    assert is_noise_topic(topic, "en") == noisy, topic
+
+
+@pytest.mark.parametrize("code", language_codes())
+def test_high_impact_keywords_nonempty_and_lower(code):
+    kws = high_impact_keywords(code)
+    assert len(kws) >= 8
+    assert all(k == k.lower() for k in kws)

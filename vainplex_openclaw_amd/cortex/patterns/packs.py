@@ -21,7 +21,7 @@ PACKS: Dict[str, dict] = {
     "en": {
         "name": "English",
         "patterns": {
-            "decision": [r"(?:decided|decision|agreed|let'?s do|the plan is|approach:|we(?:'ll| will) go with)"],
+            "decision": [r"\b(?:decided|decision|agreed|let'?s do|the plan is|approach:|we(?:'ll| will) go with)"],
             "close": [
                 r"(?:^|\s)(?:is |it's |that's |all )?(?:done|fixed|solved|closed|resolved)(?:\s|[.!]|$)",
                 r"(?:^|\s)(?:it |that )works(?:\s|[.!]|$)",
@@ -51,13 +51,13 @@ PACKS: Dict[str, dict] = {
     "de": {
         "name": "Deutsch",
         "patterns": {
-            "decision": [r"(?:entschieden|entscheidung|beschlossen|einigen wir uns|der plan ist|machen wir so|vorgehen:)"],
+            "decision": [r"\b(?:entschieden|entscheidung|beschlossen|einigen wir uns|der plan ist|machen wir so|vorgehen:)"],
             "close": [
                 r"(?:^|\s)(?:ist |das ist |alles )?(?:erledigt|fertig|gelöst|behoben|abgeschlossen)(?:\s|[.!]|$)",
                 r"(?:^|\s)(?:es |das )funktioniert(?:\s|[.!]|$)",
                 r"✅",
             ],
-            "wait": [r"(?:warte auf|wartet auf|blockiert durch|braucht zuerst|erst wenn)"],
+            "wait": [r"(?:warte[tn]? auf|blockiert durch|braucht zuerst|erst wenn)"],
             "topic": [r"(?:zurück zu|jetzt zu|bezüglich|lass uns über|thema)\s+(?:de[mnr]\s+|die\s+|das\s+)?(\w[\w\s-]{3,40})"],
         },
         "topic_blacklist": ["es", "das", "dies", "der", "die", "was", "dort", "nichts", "etwas", "alles", "ich", "du", "heute", "morgen", "gestern"],
@@ -74,7 +74,7 @@ PACKS: Dict[str, dict] = {
     "es": {
         "name": "Español",
         "patterns": {
-            "decision": [r"(?:decidido|decidimos|decisión|acordado|acordamos|el plan es|vamos a hacer|enfoque:)"],
+            "decision": [r"\b(?:decidido|decidimos|decisión|acordado|acordamos|el plan es|vamos a hacer|enfoque:)"],
             "close": [r"(?:^|\s)(?:está |todo )?(?:hecho|listo|resuelto|arreglado|cerrado|terminado)(?:\s|[.!]|$)", r"(?:^|\s)funciona(?:\s|[.!]|$)", r"✅"],
             "wait": [r"(?:esperando a|esperando por|bloqueado por|necesita primero|en espera)"],
             "topic": [r"(?:volviendo a|ahora sobre|respecto a|hablemos de|miremos)\s+(?:el\s+|la\s+|los\s+)?(\w[\w\s-]{3,40})"],
@@ -93,7 +93,7 @@ PACKS: Dict[str, dict] = {
     "fr": {
         "name": "Français",
         "patterns": {
-            "decision": [r"(?:décidé|décision|convenu|d'accord pour|le plan est|on va faire|approche\s*:)"],
+            "decision": [r"\b(?:décidé|décision|convenu|d'accord pour|le plan est|on va faire|approche\s*:)"],
             "close": [r"(?:^|\s)(?:c'est |tout est )?(?:fait|réglé|résolu|corrigé|terminé|fermé)(?:\s|[.!]|$)", r"(?:^|\s)ça (?:marche|fonctionne)(?:\s|[.!]|$)", r"✅"],
             "wait": [r"(?:en attente de|attend|bloqué par|besoin d'abord|en pause jusqu)"],
             "topic": [r"(?:revenons à|maintenant sur|concernant|parlons de|regardons)\s+(?:le\s+|la\s+|les\s+|l')?(\w[\w\s-]{3,40})"],
@@ -112,7 +112,7 @@ PACKS: Dict[str, dict] = {
     "it": {
         "name": "Italiano",
         "patterns": {
-            "decision": [r"(?:deciso|decisione|concordato|il piano è|facciamo così|approccio:)"],
+            "decision": [r"\b(?:deciso|decisione|concordato|il piano è|facciamo così|approccio:)"],
             "close": [r"(?:^|\s)(?:è |tutto )?(?:fatto|risolto|sistemato|chiuso|completato)(?:\s|[.!]|$)", r"(?:^|\s)funziona(?:\s|[.!]|$)", r"✅"],
             "wait": [r"(?:in attesa di|aspettando|bloccato da|serve prima|in sospeso)"],
             "topic": [r"(?:torniamo a|ora su|riguardo a|parliamo di|guardiamo)\s+(?:il\s+|la\s+|lo\s+|l')?(\w[\w\s-]{3,40})"],
@@ -131,7 +131,7 @@ PACKS: Dict[str, dict] = {
     "pt": {
         "name": "Português",
         "patterns": {
-            "decision": [r"(?:decidido|decidimos|decisão|combinado|acordamos|o plano é|vamos fazer|abordagem:)"],
+            "decision": [r"\b(?:decidido|decidimos|decisão|combinado|acordamos|o plano é|vamos fazer|abordagem:)"],
             "close": [r"(?:^|\s)(?:está |tudo )?(?:feito|pronto|resolvido|corrigido|fechado|concluído)(?:\s|[.!]|$)", r"(?:^|\s)funciona(?:\s|[.!]|$)", r"✅"],
             "wait": [r"(?:esperando por|aguardando|bloqueado por|precisa primeiro|em espera)"],
             "topic": [r"(?:voltando a|agora sobre|sobre o|falemos de|vejamos)\s+(?:o\s+|a\s+|os\s+)?(\w[\w\s-]{3,40})"],
@@ -150,7 +150,7 @@ PACKS: Dict[str, dict] = {
     "ru": {
         "name": "Русский",
         "patterns": {
-            "decision": [r"(?:решено|решили|решение|договорились|план такой|будем делать|подход:)"],
+            "decision": [r"\b(?:решено|решили|решение|договорились|план такой|будем делать|подход:)"],
             "close": [r"(?:^|\s)(?:всё |все )?(?:готово|сделано|решено|исправлено|закрыто|завершено)(?:\s|[.!]|$)", r"(?:^|\s)работает(?:\s|[.!]|$)", r"✅"],
             "wait": [r"(?:ждём|ожидаем|заблокировано|сначала нужно|в ожидании)"],
             "topic": [r"(?:вернёмся к|теперь о|насчёт|давайте обсудим|посмотрим на)\s+(\w[\w\s-]{3,40})"],
