@@ -250,3 +250,71 @@ def test_hooks_registration_registers_four():
     for hook in ("tool_result_persist", "before_tool_call", "message_sending",
                  "before_message_write"):
         assert bus.handlers(hook), hook
+
+
+# ===========================================================================
+# allowlist.test.ts depth: the credential invariant, channel/tool/agent
+# exemption precedence, category filtering
+# ===========================================================================
+
+def test_allowlist_credential_invariant_every_field_set():
+    al = normalize_allowlist({
+        "piiAllowedChannels": ["x"], "financialAllowedChannels": ["x"],
+        "exemptTools": ["t"], "exemptAgents": ["a"],
+    })
+    ctx = {"channel": "x", "toolName": "t", "agentId": "a"}
+    assert not evaluate_allowlist("credential", ctx, al)["allowed"]
+    assert not evaluate_allowlist("credential", {}, al)["allowed"]
+
+
+def test_allowlist_financial_channels():
+    al = normalize_allowlist({"financialAllowedChannels": ["finance"]})
+    assert evaluate_allowlist("financial", {"channel": "finance"}, al)["allowed"]
+    assert not evaluate_allowlist("financial", {"channel": "public"}, al)["allowed"]
+    assert not evaluate_allowlist("financial", {}, al)["allowed"]
+
+
+def test_allowlist_pii_requires_channel():
+    al = normalize_allowlist({"piiAllowedChannels": ["internal"]})
+    assert not evaluate_allowlist("pii", {}, al)["allowed"]
+    empty = normalize_allowlist({})
+    assert not evaluate_allowlist("pii", {"channel": "internal"}, empty)["allowed"]
+
+
+def test_allowlist_tool_exemption_not_for_credentials():
+    from vainplex_openclaw_amd.governance.redaction.allowlist import is_tool_exempt
+
+    al = normalize_allowlist({"exemptTools": ["read", "fetch"]})
+    assert is_tool_exempt("read", al) and not is_tool_exempt("exec", al)
+    assert evaluate_allowlist("pii", {"toolName": "read"}, al)["allowed"]
+    assert not evaluate_allowlist("credential", {"toolName": "read"}, al)["allowed"]
+
+
+def test_allowlist_agent_exemption_not_for_credentials():
+    from vainplex_openclaw_amd.governance.redaction.allowlist import is_agent_exempt
+
+    al = normalize_allowlist({"exemptAgents": ["main"]})
+    assert is_agent_exempt("main", al) and not is_agent_exempt("forge", al)
+    assert evaluate_allowlist("pii", {"agentId": "main"}, al)["allowed"]
+    assert not evaluate_allowlist("credential", {"agentId": "main"}, al)["allowed"]
+
+
+def test_redactable_categories_matrix():
+    al = normalize_allowlist({"piiAllowedChannels": ["internal"],
+                              "financialAllowedChannels": ["finance"]})
+    cats = ["credential", "pii", "financial", "custom"]
+    # nothing allowlisted for this context
+    assert get_redactable_categories(cats, {"channel": "public"}, al) == cats
+    # pii drops on its channel, credential always stays
+    got = get_redactable_categories(cats, {"channel": "internal"}, al)
+    assert "pii" not in got and "credential" in got
+    got2 = get_redactable_categories(cats, {"channel": "finance"}, al)
+    assert "financial" not in got2 and "credential" in got2
+    # custom category passes through untouched
+    assert "custom" in got and "custom" in got2
+
+
+def test_redactable_categories_empty_allowlist_keeps_all():
+    al = normalize_allowlist(None)
+    cats = ["credential", "pii", "financial"]
+    assert get_redactable_categories(cats, {"channel": "anything"}, al) == cats

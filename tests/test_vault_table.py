@@ -5,6 +5,8 @@ hash8->hash12 collision extension, resolve-all, lifecycle, perf."""
 import hashlib
 import time
 
+import pytest
+
 from vainplex_openclaw_amd.governance.redaction.vault import (
     PLACEHOLDER_RX,
     RedactionVault,
@@ -120,3 +122,120 @@ def test_perf_1000_entries():
     out = v.resolve(text)
     assert "REDACTED" not in out
     assert (time.perf_counter() - start) < 2.0
+
+
+# ===========================================================================
+# vault.test.ts depth: hash properties, TTL/eviction edges, resolve-all
+# behavior, perf budgets, placeholder grammar
+# ===========================================================================
+
+import re
+
+from vainplex_openclaw_amd.governance.redaction.vault import (
+    PLACEHOLDER_RX,
+    RedactionVault,
+    format_placeholder,
+)
+
+
+def test_placeholder_uses_sha256_prefix():
+    import hashlib
+
+    v = RedactionVault()
+    ph = v.store("known-value", "pii")
+    want8 = hashlib.sha256("known-value".encode()).hexdigest()[:8]
+    assert ph == f"[REDACTED:pii:{want8}]"
+
+
+def test_hash_deterministic_and_not_trivial():
+    v = RedactionVault()
+    a = v.store("value-one", "custom")
+    v2 = RedactionVault()
+    assert v2.store("value-one", "custom") == a  # deterministic across vaults
+    # different values, different hashes (not a length/первый-char hash)
+    b = v.store("value-two", "custom")
+    assert a != b
+
+
+def test_same_value_reuses_placeholder_across_categories():
+    # vault.ts store() keys on the value hash alone: re-storing the same
+    # value under another category returns the EXISTING placeholder
+    # (first category wins) — one secret, one placeholder.
+    v = RedactionVault()
+    a = v.store("shared-secret", "credential")
+    b = v.store("shared-secret", "pii")
+    assert a == b and "credential" in a
+    assert v.lookup(a) == "shared-secret"
+
+
+def test_lookup_unknown_and_malformed():
+    v = RedactionVault()
+    assert v.lookup("[REDACTED:credential:deadbeef]") is None
+    assert v.lookup("not-a-placeholder") is None
+    assert v.lookup("[REDACTED:credential:xyz]") is None  # non-hex
+
+
+def test_restore_after_expiry_creates_fresh_entry():
+    t = [0.0]
+    v = RedactionVault(expiry_seconds=10, clock=lambda: t[0])
+    ph = v.store("secret-val", "pii")
+    t[0] = 11
+    assert v.lookup(ph) is None
+    ph2 = v.store("secret-val", "pii")
+    assert ph2 == ph            # same hash, fresh entry
+    assert v.lookup(ph2) == "secret-val"
+
+
+def test_resolve_all_mixed_and_plain():
+    v = RedactionVault()
+    a = v.store("alpha-secret", "credential")
+    b = v.store("beta-secret", "pii")
+    text = f"use {a} then {b} and [REDACTED:pii:00000000] stays"
+    out = v.resolve(text)
+    assert "alpha-secret" in out and "beta-secret" in out
+    assert "[REDACTED:pii:00000000]" in out  # unresolvable left intact
+    assert v.resolve("no placeholders here") == "no placeholders here"
+
+
+def test_clear_removes_everything():
+    v = RedactionVault()
+    ph = v.store("to-clear", "custom")
+    v.clear()
+    assert v.size == 0
+    assert v.lookup(ph) is None
+
+
+def test_perf_1000_store_resolve():
+    import time as _time
+
+    v = RedactionVault()
+    t0 = _time.perf_counter()
+    phs = [v.store(f"secret-{i:04d}", "credential") for i in range(1000)]
+    for ph in phs[:200]:
+        assert v.lookup(ph)
+    assert (_time.perf_counter() - t0) * 1000 < 50
+    text = " ".join(phs[:100])
+    t0 = _time.perf_counter()
+    out = v.resolve(text)
+    assert (_time.perf_counter() - t0) * 1000 < 5
+    assert "secret-0000" in out
+
+
+@pytest.mark.parametrize("text,matches", [
+    ("[REDACTED:credential:abcd1234]", True),
+    ("[REDACTED:pii:abcdef123456]", True),      # hash12
+    ("[REDACTED:financial:00ff00ff]", True),
+    ("[REDACTED:custom:12345678]", True),
+    ("[REDACTED:unknown:abcd1234]", False),     # unknown category
+    ("[REDACTED:credential:abc]", False),       # too short
+    ("[REDACTED:credential:abcd1234567890]", False),  # too long
+    ("[redacted:credential:abcd1234]", False),  # case-sensitive
+    ("REDACTED:credential:abcd1234", False),    # missing brackets
+])
+def test_placeholder_grammar(text, matches):
+    assert bool(PLACEHOLDER_RX.fullmatch(text)) == matches
+
+
+def test_format_placeholder_roundtrip():
+    ph = format_placeholder("financial", "aabbccdd")
+    assert PLACEHOLDER_RX.fullmatch(ph)
