@@ -182,3 +182,124 @@ def test_legacy_array_format(workspace):
         json.dump([thread("legacy-thread")], fh)
     out = gen(workspace).generate()
     assert "legacy-thread" in out
+
+
+# ===========================================================================
+# boot-context.test.ts depth: execution-mode table, per-tier staleness
+# messages, sort stability, thread caps, footer stats, overwrite
+# ===========================================================================
+
+from vainplex_openclaw_amd.cortex.boot_context import execution_mode
+
+
+@pytest.mark.parametrize("hour,mode", [
+    (2, "night-watch"), (5, "night-watch"), (7, "night-watch"),
+    (8, "business-hours"), (13, "business-hours"), (17, "business-hours"),
+    (18, "evening"), (22, "evening"), (23, "night-watch"),
+])
+def test_execution_mode_by_hour(hour, mode):
+    # boot-context.ts:18-24 tiers: 8-18 business, 18-23 evening, else night
+    assert execution_mode(hour) == mode
+
+
+def test_open_threads_priority_then_recency():
+    data = {"threads": [
+        thread("old-critical", "critical", age_s=7200),
+        thread("new-critical", "critical", age_s=60),
+        thread("new-high", "high", age_s=10),
+        thread("new-low", "low", age_s=5),
+    ]}
+    got = [t["title"] for t in get_open_threads(data, 10)]
+    assert got == ["new-critical", "old-critical", "new-high", "new-low"]
+
+
+def test_open_threads_excludes_closed_and_respects_limit():
+    data = {"threads": [
+        thread("a", status="closed"),
+        thread("b"), thread("c"), thread("d"),
+    ]}
+    got = get_open_threads(data, 2)
+    assert len(got) == 2 and all(t["status"] == "open" for t in got)
+
+
+def test_open_threads_missing_file_shape():
+    assert get_open_threads({}, 5) == []
+    assert get_open_threads({"threads": []}, 5) == []
+
+
+@pytest.mark.parametrize("age_h,expect", [
+    (0.5, ""),                 # fresh -> no warning
+    (3.0, "h old"),            # > 2h -> aging warning
+    (9.0, "STALE"),            # > 8h -> STALE DATA
+])
+def test_integrity_tier_messages(age_h, expect):
+    data = {"integrity": {"last_event_timestamp": iso(-age_h * 3600)}}
+    w = integrity_warning(data, NOW)
+    if expect == "":
+        assert w == ""
+    else:
+        assert expect in w, w
+
+
+def test_integrity_no_data_warns():
+    assert integrity_warning({}, NOW) != ""
+
+
+def test_generate_includes_footer_stats(workspace):
+    write_threads(workspace, [thread("t1"), thread("t2")])
+    out = gen(workspace).generate()
+    assert out.startswith("#")          # markdown header
+    assert "2" in out                   # thread count surfaces in footer
+
+
+def test_generate_thread_cap(workspace):
+    write_threads(workspace, [thread(f"t{i}") for i in range(20)])
+    out = gen(workspace, max_threads=3).generate()
+    listed = [ln for ln in out.splitlines() if ln.strip().startswith(("-", "*"))
+              and " t" in ln]
+    assert len([ln for ln in listed if any(f"t{i}" in ln for i in range(20))]) <= 3
+
+
+def test_write_overwrites_existing(workspace):
+    write_threads(workspace, [thread("first-run")])
+    g = gen(workspace)
+    assert g.write()
+    path = os.path.join(workspace, "BOOTSTRAP.md")
+    first = open(path).read()
+    assert "first-run" in first
+    write_threads(workspace, [thread("second-run")])
+    assert gen(workspace).write()
+    second = open(path).read()
+    assert "second-run" in second and "first-run" not in second
+
+
+def test_mood_emoji_for_frustrated(workspace):
+    write_threads(workspace, [thread("t")], mood="frustrated")
+    out = gen(workspace).generate()
+    assert "frustrated" in out.lower()
+
+
+def test_narrative_included_only_when_fresh(workspace):
+    d = write_threads(workspace, [thread("t")])
+    with open(os.path.join(d, "narrative.md"), "w") as fh:
+        fh.write("NARRATIVE-CONTENT-HERE")
+    # fresh mtime -> included
+    out = gen(workspace).generate()
+    assert "NARRATIVE-CONTENT-HERE" in out
+    # stale (>36h) -> excluded
+    old = NOW - 37 * 3600
+    os.utime(os.path.join(d, "narrative.md"), (old, old))
+    out2 = gen(workspace).generate()
+    assert "NARRATIVE-CONTENT-HERE" not in out2
+
+
+def test_hot_snapshot_included_only_within_1h(workspace):
+    d = write_threads(workspace, [thread("t")])
+    snap = os.path.join(d, "hot-snapshot.md")
+    with open(snap, "w") as fh:
+        fh.write("SNAPSHOT-LINES")
+    out = gen(workspace).generate()
+    assert "SNAPSHOT-LINES" in out
+    old = NOW - 2 * 3600
+    os.utime(snap, (old, old))
+    assert "SNAPSHOT-LINES" not in gen(workspace).generate()
