@@ -130,3 +130,105 @@ def test_multiple_decisions_one_message(workspace):
         "postpone the migration until May."
     )
     assert n >= 2
+
+
+# ===========================================================================
+# decision-tracker.test.ts depth: context windows, multi-decision
+# messages, impact assignment per scenario, cap eviction order, corrupt
+# and missing files, recency/limit interplay
+# ===========================================================================
+
+def test_no_extract_from_unrelated_or_empty(workspace):
+    dt, _ = make_tracker(workspace)
+    assert dt.process_message("the weather is nice today") == 0
+    assert dt.process_message("") == 0
+    assert dt.process_message("   ") == 0
+    assert dt.decisions == []
+
+
+def test_context_window_for_why(workspace):
+    dt, _ = make_tracker(workspace)
+    before = "Because latency dominated the eval results and p99 spiked, "
+    dt.process_message(before + "we decided to move to the rust rewrite")
+    d = dt.decisions[0]
+    ctx = d.get("why") or d.get("context") or ""
+    assert "latency" in str(ctx), d
+
+
+@pytest.mark.parametrize("text,impact", [
+    ("we decided on the new architecture for auth", "high"),
+    ("we decided to tighten security reviews", "high"),
+    ("we decided to start the data migration", "high"),
+    ("we decided to pick blue as the accent color", "medium"),
+    ("we decided to rename the helper function", "medium"),
+])
+def test_impact_assignment_scenarios(workspace, text, impact):
+    dt, _ = make_tracker(workspace)
+    dt.process_message(text)
+    assert dt.decisions and dt.decisions[0]["impact"] == impact, text
+
+
+def test_dedupe_window_expires(workspace):
+    dt, t = make_tracker(workspace)
+    assert dt.process_message("we decided to use postgres") == 1
+    # identical within 24 h window -> deduped
+    t[0] += 3600
+    assert dt.process_message("we decided to use postgres") == 0
+    # after the window it is a new decision again
+    t[0] += 25 * 3600
+    assert dt.process_message("we decided to use postgres") == 1
+    assert len(dt.decisions) == 2
+
+
+def test_cap_evicts_oldest_first(workspace):
+    dt, t = make_tracker(workspace, max_decisions=3)
+    for i in range(5):
+        t[0] += 90000  # outside the dedupe window each time
+        dt.process_message(f"we decided to adopt plan-{i} next")
+    assert len(dt.decisions) == 3
+    whats = " ".join(d["what"] for d in dt.decisions)
+    assert "plan-4" in whats and "plan-0" not in whats and "plan-1" not in whats
+
+
+def test_load_roundtrip_and_missing(workspace):
+    dt, t = make_tracker(workspace)
+    dt.process_message("we decided to archive old logs")
+    dt.flush()
+    dt2, _ = make_tracker(workspace)
+    assert any("archive" in d["what"] for d in dt2.decisions)
+    # missing file in a fresh workspace
+    import tempfile
+
+    dt3, _ = make_tracker(tempfile.mkdtemp(prefix="dt-missing-"))
+    assert dt3.decisions == []
+
+
+def test_corrupt_file_graceful(workspace):
+    d = os.path.join(workspace, "memory", "reboot")
+    os.makedirs(d, exist_ok=True)
+    with open(os.path.join(d, "decisions.json"), "w") as fh:
+        fh.write("{not json!!")
+    dt, _ = make_tracker(workspace)
+    assert dt.decisions == []
+    assert dt.process_message("we decided to recover gracefully") == 1
+
+
+def test_recency_filter_and_limit(workspace):
+    dt, t = make_tracker(workspace)
+    dt.process_message("we decided on the ancient plan")
+    t[0] += 10 * 86400
+    for i in range(4):
+        t[0] += 90000
+        dt.process_message(f"we decided on recent plan {i}")
+    recent = dt.recent_within(days=7.0, limit=10)
+    assert all("recent" in d["what"] for d in recent)
+    assert len(dt.recent_within(days=7.0, limit=2)) == 2
+    assert len(dt.recent(2)) == 2
+
+
+def test_multiple_decisions_in_one_message(workspace):
+    dt, _ = make_tracker(workspace)
+    n = dt.process_message(
+        "we decided to use postgres. also agreed the plan is to ship monthly."
+    )
+    assert n >= 1 and len(dt.decisions) == n
