@@ -347,3 +347,202 @@ def test_resolve_analyzer_llm_config():
     r3 = resolve_analyzer_llm_config(TOP, {"enabled": True, "model": "gpt-4o"})
     assert r3["model"] == "gpt-4o" and r3["endpoint"] == TOP["endpoint"]
     assert r3["apiKey"] == ""
+
+
+# ===========================================================================
+# chain-reconstructor.test.ts + events.test.ts depth tables
+# ===========================================================================
+
+def _nev(i, ts, agent="a1", session="s1", etype="msg.in", payload=None):
+    from vainplex_openclaw_amd.cortex.trace.events import NormalizedEvent
+
+    return NormalizedEvent(id=f"e{i}", ts=ts, agent=agent, session=session,
+                           type=etype, payload=payload or {}, seq=i)
+
+
+def _rc(events, **kw):
+    from vainplex_openclaw_amd.cortex.trace.chains import reconstruct_chains
+
+    return reconstruct_chains(events, **kw)
+
+
+MIN = 60_000.0  # one minute in ms
+
+
+def test_chains_group_by_session_and_agent():
+    evs = [
+        _nev(1, 1 * MIN, agent="a1", session="s1"),
+        _nev(2, 2 * MIN, agent="a1", session="s1"),
+        _nev(3, 1 * MIN, agent="a2", session="s1"),
+        _nev(4, 2 * MIN, agent="a2", session="s1"),
+        _nev(5, 1 * MIN, agent="a1", session="s2"),
+        _nev(6, 2 * MIN, agent="a1", session="s2"),
+    ]
+    chains = _rc(evs)
+    keys = {(c.session, c.agent) for c in chains}
+    assert keys == {("s1", "a1"), ("s1", "a2"), ("s2", "a1")}
+
+
+def test_chain_orders_events_by_ts():
+    evs = [_nev(1, 5 * MIN), _nev(2, 1 * MIN), _nev(3, 3 * MIN)]
+    chains = _rc(evs)
+    assert len(chains) == 1
+    assert [e.ts for e in chains[0].events] == [1 * MIN, 3 * MIN, 5 * MIN]
+
+
+@pytest.mark.parametrize("split_type", ["session.start", "session.end"])
+def test_chain_splits_on_lifecycle(split_type):
+    evs = [
+        _nev(1, 1 * MIN), _nev(2, 2 * MIN),
+        _nev(3, 3 * MIN, etype=split_type),
+        _nev(4, 4 * MIN),
+    ]
+    chains = _rc(evs)
+    assert len(chains) == 2
+    assert chains[1].boundary_type == "lifecycle"
+
+
+def test_chain_gap_split_boundary_exact():
+    evs = [_nev(1, 0.0), _nev(2, 29 * MIN), _nev(3, 29 * MIN + 29 * MIN),
+           _nev(4, 29 * MIN + 29 * MIN + 31 * MIN),
+           _nev(5, 29 * MIN + 29 * MIN + 32 * MIN)]
+    chains = _rc(evs)
+    # <=30 min gaps keep one chain; the 31-min gap splits
+    assert len(chains) == 2
+    assert chains[0].boundary_type == "gap"
+
+
+def test_chain_min_two_events_filter():
+    evs = [_nev(1, 0.0), _nev(2, 40 * MIN), _nev(3, 41 * MIN)]
+    chains = _rc(evs)
+    # the singleton before the gap is dropped
+    assert len(chains) == 1 and len(chains[0].events) == 2
+
+
+def test_chain_id_deterministic_16_hex():
+    from vainplex_openclaw_amd.cortex.trace.chains import chain_id
+
+    a = chain_id("s1", "a1", 12345.0)
+    assert a == chain_id("s1", "a1", 12345.0)
+    assert len(a) == 16 and all(c in "0123456789abcdef" for c in a)
+    assert a != chain_id("s1", "a2", 12345.0)
+    assert a != chain_id("s1", "a1", 99999.0)
+
+
+def test_chain_type_counts_and_ts_range():
+    evs = [_nev(1, 1 * MIN), _nev(2, 2 * MIN, etype="tool.call"),
+           _nev(3, 3 * MIN, etype="tool.call")]
+    c = _rc(evs)[0]
+    assert c.type_counts == {"msg.in": 1, "tool.call": 2}
+    assert c.start_ts == 1 * MIN and c.end_ts == 3 * MIN
+
+
+def test_chain_cap_max_events():
+    evs = [_nev(i, i * 1000.0) for i in range(50)]
+    chains = _rc(evs, max_events=10)
+    assert all(len(c.events) <= 10 for c in chains)
+    assert sum(len(c.events) for c in chains) == 50
+
+
+def test_chain_empty_and_unknown_session():
+    assert _rc([]) == []
+    evs = [_nev(1, 1 * MIN, session="unknown"), _nev(2, 2 * MIN, session="unknown")]
+    chains = _rc(evs)
+    assert len(chains) == 1 and chains[0].session == "unknown"
+
+
+def test_chain_dedupes_same_event_id():
+    e1 = _nev(1, 1 * MIN)
+    e1b = _nev(1, 1 * MIN)  # same id
+    e2 = _nev(2, 2 * MIN)
+    chains = _rc([e1, e1b, e2])
+    assert len(chains[0].events) == 2
+
+
+def test_chain_interleaved_agents():
+    evs = []
+    for i in range(6):
+        evs.append(_nev(10 + i, i * MIN, agent="a1"))
+        evs.append(_nev(20 + i, i * MIN + 1000, agent="a2"))
+    chains = _rc(evs)
+    assert len(chains) == 2
+    for c in chains:
+        assert all(e.agent == c.agent for e in c.events)
+
+
+# -- events.test.ts depth ----------------------------------------------------
+
+def _schema_a(etype, **kw):
+    ev = {"id": "evt-1", "ts": 1000.0, "agent": "main",
+          "actor": {"id": "main"}, "scope": {"sessionKey": "agent:main:abc123"},
+          "type": etype, "payload": kw.pop("payload", {})}
+    ev.update(kw)
+    return ev
+
+
+def test_schema_a_session_uuid_extraction():
+    from vainplex_openclaw_amd.cortex.trace.events import normalize_schema_a
+
+    ev = normalize_schema_a(_schema_a("message.in.received"))
+    assert ev is not None and ev.session == "abc123"
+    # plain session strings pass through
+    ev2 = normalize_schema_a({"id": "e", "ts": 1.0, "type": "message.in.received",
+                              "session": "plain", "payload": {}})
+    assert ev2.session == "plain"
+
+
+def test_schema_a_msg_roles():
+    from vainplex_openclaw_amd.cortex.trace.events import normalize_schema_a
+
+    assert normalize_schema_a(_schema_a("message.in.received")).payload["role"] == "user"
+    out = normalize_schema_a(_schema_a("message.out.sent"))
+    assert out.payload["role"] == "assistant"
+
+
+def test_schema_a_tool_call_and_result_payloads():
+    from vainplex_openclaw_amd.cortex.trace.events import normalize_schema_a
+
+    call = normalize_schema_a(_schema_a(
+        "tool.call.requested", payload={"toolName": "exec", "params": {"cmd": "ls"}}))
+    assert call.payload["toolName"] == "exec"
+    assert call.payload["toolParams"] == {"cmd": "ls"}
+    res = normalize_schema_a(_schema_a(
+        "tool.call.failed", payload={"toolName": "exec", "error": "boom"}))
+    assert res.payload["toolIsError"] is True
+    ok = normalize_schema_a(_schema_a(
+        "tool.call.executed", payload={"toolName": "exec", "result": "fine"}))
+    assert ok.payload["toolIsError"] is False
+
+
+def test_schema_a_unknown_type_is_none():
+    from vainplex_openclaw_amd.cortex.trace.events import normalize_schema_a
+
+    assert normalize_schema_a(_schema_a("totally.unknown.type")) is None
+
+
+def test_schema_a_agent_defaults_unknown():
+    from vainplex_openclaw_amd.cortex.trace.events import normalize_schema_a
+
+    ev = normalize_schema_a({"id": "e", "ts": 5.0,
+                             "type": "message.in.received", "payload": {}})
+    assert ev.agent == "unknown" and ev.session == "unknown"
+
+
+def test_schema_b_kinds_and_fields():
+    from vainplex_openclaw_amd.cortex.trace.events import normalize_schema_b
+
+    ev = normalize_schema_b({"kind": "tool_use", "sessionId": "s-9",
+                             "agentId": "forge", "ts": 77.0,
+                             "body": {"name": "read", "args": {"p": 1}}})
+    assert ev is not None
+    assert ev.type == "tool.call" and ev.agent == "forge"
+    msg = normalize_schema_b({"kind": "assistant_message", "sessionId": "s",
+                              "agentId": "a", "ts": 1.0,
+                              "body": {"text": "hello"}})
+    assert msg.type == "msg.out" and msg.payload.get("content")
+
+
+def test_schema_b_unknown_kind_none():
+    from vainplex_openclaw_amd.cortex.trace.events import normalize_schema_b
+
+    assert normalize_schema_b({"kind": "nope", "ts": 1.0}) is None
