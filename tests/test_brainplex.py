@@ -251,3 +251,174 @@ def test_update_config_dry_run_no_write(tmp_path):
     res = update_openclaw_config(str(path), {}, ["p"], dry_run=True)
     assert res["updated"] and not res["backed_up"]
     assert path.read_text() == "{}"
+
+
+# ===========================================================================
+# installer execution path (installer.ts:40-165) + /brainplex dashboard
+# ===========================================================================
+
+def test_installer_prefers_openclaw_cli():
+    from vainplex_openclaw_amd.brainplex import installer as inst
+
+    calls = []
+
+    def runner(argv, cwd, timeout_s):
+        calls.append(argv)
+        return 0, "ok"
+
+    plan = {"to_install": [{"id": "openclaw-leuko",
+                            "module": "vainplex_openclaw_amd.leuko.plugin"}]}
+    res = inst.execute_installation(plan, runner=runner)
+    assert res["installed"] and res["installed"][0]["method"] == "openclaw"
+    assert calls[0] == ["which", "openclaw"]
+    assert calls[1][:3] == ["openclaw", "plugins", "install"]
+
+
+def test_installer_pip_path_with_copy(tmp_path):
+    from vainplex_openclaw_amd.brainplex import installer as inst
+
+    def runner(argv, cwd, timeout_s):
+        if argv[0] == "which":
+            return 1, ""          # no openclaw CLI
+        if argv[0] == "pip":
+            # simulate a successful install into --target
+            import os
+            tgt = argv[argv.index("--target") + 1]
+            pkg_dir = os.path.join(tgt, "openclaw_leuko")
+            os.makedirs(pkg_dir)
+            open(os.path.join(pkg_dir, "__init__.py"), "w").write("# pkg")
+            return 0, "installed"
+        return 1, "?"
+
+    plan = {"to_install": [{"id": "leuko", "package": "openclaw-leuko",
+                            "module": "vainplex_openclaw_amd.leuko.plugin"}]}
+    res = inst.execute_installation(plan, runner=runner,
+                                    workspace_path=str(tmp_path))
+    assert res["installed"] and res["installed"][0]["method"] == "pip"
+    assert (tmp_path / "extensions" / "openclaw_leuko" / "__init__.py").exists()
+
+
+def test_installer_offline_fallback_copies_source(tmp_path):
+    from vainplex_openclaw_amd.brainplex import installer as inst
+
+    def runner(argv, cwd, timeout_s):
+        return 1, "no network"    # which + pip both fail
+
+    plan = {"to_install": [{"id": "openclaw-leuko",
+                            "module": "vainplex_openclaw_amd.leuko.plugin"}]}
+    res = inst.execute_installation(plan, runner=runner,
+                                    workspace_path=str(tmp_path))
+    assert res["installed"] and res["installed"][0]["method"] == "in-package"
+    ext = tmp_path / "extensions" / "openclaw-leuko"
+    assert ext.is_dir() and any(ext.iterdir())
+
+
+def test_installer_failure_recorded_not_raised(tmp_path):
+    from vainplex_openclaw_amd.brainplex import installer as inst
+
+    def runner(argv, cwd, timeout_s):
+        if argv[0] == "which":
+            return 1, ""
+        raise RuntimeError("runner exploded")
+
+    plan = {"to_install": [{"id": "ghost-plugin"}]}  # no module fallback
+    res = inst.execute_installation(plan, runner=runner,
+                                    workspace_path=str(tmp_path))
+    assert res["failed"] and "error" in res["failed"][0]
+
+
+def test_installer_dry_run_and_empty_plan():
+    from vainplex_openclaw_amd.brainplex import installer as inst
+
+    def runner(argv, cwd, timeout_s):  # must never be called
+        raise AssertionError("runner called on dry-run")
+
+    assert inst.execute_installation({"to_install": [{"id": "x"}]},
+                                     dry_run=True, runner=runner) == \
+        {"installed": [], "failed": []}
+    assert inst.execute_installation({"to_install": []}, runner=runner) == \
+        {"installed": [], "failed": []}
+
+
+def test_installer_verify_reports_broken():
+    from vainplex_openclaw_amd.brainplex import installer as inst
+
+    plan = {"to_install": [
+        {"id": "ok", "module": "vainplex_openclaw_amd.leuko.plugin"},
+        {"id": "bad", "module": "vainplex_openclaw_amd.does.not.exist"},
+    ]}
+    v = inst.verify_installed(plan)
+    assert v["ok"] == ["ok"]
+    assert v["broken"] and v["broken"][0]["id"] == "bad"
+
+
+def _dash_api():
+    from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+
+    bus = HookBus()
+    return PluginApi(id="brainplex", plugin_config={}, logger=NullLogger(),
+                     config={}, bus=bus)
+
+
+def test_dashboard_renders_all_sections():
+    from vainplex_openclaw_amd.brainplex.dashboard import create_plugin
+
+    api = _dash_api()
+    api.register_gateway_method("governance.status", lambda: {
+        "evaluations": 100, "denied": 4, "avgEvaluationUs": 85,
+        "policies": ["p1", "p2"]})
+    api.register_gateway_method("governance.trust", lambda: {
+        "agents": [{"agentId": "main", "score": 72.5, "tier": "trusted"},
+                   {"agentId": "forge", "score": 41.0, "tier": "standard"}]})
+    api.register_gateway_method("eventstore.status", lambda: {
+        "connected": True, "stream": "openclaw-events", "messages": 420,
+        "publishFailures": 0, "disconnectCount": 0})
+    api.register_gateway_method("cortex.status", lambda: {
+        "openThreads": 3, "decisions": 7, "sessionMood": "productive"})
+    api.register_gateway_method("leuko.health", lambda: {
+        "status": "healthy", "notable": ["disk growth on /var"]})
+    create_plugin().register(api)
+    out = api.commands["brainplex"]()["text"]
+    assert "Shield score:" in out
+    assert "main" in out and "72.5" in out and "trusted" in out
+    assert "openclaw-events" in out and "420" in out
+    assert "open threads: 3" in out and "productive" in out
+    assert "healthy" in out and "disk growth" in out
+    # highest trust listed first
+    assert out.index("main") < out.index("forge")
+
+
+def test_dashboard_handles_missing_modules():
+    from vainplex_openclaw_amd.brainplex.dashboard import create_plugin
+
+    api = _dash_api()
+    create_plugin().register(api)
+    out = api.commands["brainplex"]()["text"]
+    assert out.count("not installed") >= 3
+    assert "Shield score:** 100/100" in out
+
+
+def test_dashboard_shield_score_degrades():
+    from vainplex_openclaw_amd.brainplex.dashboard import _shield_score
+
+    assert _shield_score(None, None) == 100
+    # denial-heavy traffic costs points
+    busy = _shield_score({"evaluations": 100, "denied": 30}, None)
+    assert busy < 100
+    # publish failures + disconnect costs more
+    worse = _shield_score({"evaluations": 100, "denied": 30},
+                          {"publishFailures": 5, "disconnectCount": 2,
+                           "connected": False})
+    assert worse < busy
+    assert _shield_score({"evaluations": 100, "denied": 100},
+                         {"publishFailures": 99, "disconnectCount": 99,
+                          "connected": False}) >= 0
+
+
+def test_dashboard_gateway_method_registered():
+    from vainplex_openclaw_amd.brainplex.dashboard import create_plugin
+
+    api = _dash_api()
+    create_plugin().register(api)
+    assert "brainplex.dashboard" in api.gateway_methods
+    assert isinstance(api.gateway_methods["brainplex.dashboard"](), str)

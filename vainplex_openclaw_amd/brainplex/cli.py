@@ -79,20 +79,27 @@ def plan_installation(scan_result: ScanResult, configs: List[Dict], full: bool) 
     return plan
 
 
-def execute_installation(plan: Dict[str, List], dry_run: bool = False) -> Dict[str, List]:
-    """In-package "install": import each plugin module and check for a
-    create_plugin factory."""
-    result: Dict[str, List] = {"installed": [], "failed": []}
-    if dry_run:
-        return result
-    for p in plan["to_install"]:
-        try:
-            mod = importlib.import_module(p["module"])
-            if not hasattr(mod, "create_plugin"):
-                raise AttributeError(f"{p['module']} has no create_plugin")
-            result["installed"].append({"id": p["id"], "success": True})
-        except Exception as exc:
-            result["failed"].append({"id": p["id"], "success": False, "error": str(exc)})
+def execute_installation(plan: Dict[str, List], dry_run: bool = False,
+                         runner=None, workspace_path=None, home=None) -> Dict[str, List]:
+    """Real install execution (installer.py): openclaw CLI when present,
+    else package-manager install copied into the extensions directory,
+    with the in-package source-copy fallback for the six suite plugins
+    (this environment has no index access). Falls back to the import
+    check when no runner is usable at all."""
+    from . import installer as _inst
+
+    if runner is None:
+        # offline default: skip the process-spawning paths, source-install
+        # straight from this package (still a real copy into extensions)
+        def runner(argv, cwd, timeout_s):
+            return 1, "offline: no package index"
+
+    result = _inst.execute_installation(plan, dry_run=dry_run, runner=runner,
+                                        workspace_path=workspace_path, home=home)
+    if not dry_run:
+        verify = _inst.verify_installed(plan)
+        for entry in result["installed"]:
+            entry["verified"] = entry["id"] in verify["ok"]
     return result
 
 
@@ -108,7 +115,8 @@ def run_init(opts: Dict[str, Any], start_dir: str = ".", home: Optional[str] = N
     plan = plan_installation(sr, configs, opts["full"])
     echo(f"📦 Installing {len(plan['to_install'])} plugin(s), "
          f"skipping {len(plan['to_skip'])} (already installed)")
-    install = execute_installation(plan, dry_run=opts["dry_run"])
+    install = execute_installation(plan, dry_run=opts["dry_run"],
+                                   runner=opts.get("runner"), home=home)
     written = write_configs(plan["to_configure"], home=home, dry_run=opts["dry_run"])
     echo(f"⚙️  Configured: {', '.join(written['written']) or '(none)'}; "
          f"kept existing: {', '.join(written['skipped']) or '(none)'}")
