@@ -39,6 +39,9 @@ __global__ void topk_scan_mx4_kernel(const uint8_t*, const uint8_t*, const uint8
 __global__ void topk_scan_fp4_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
                                      const uint8_t*, int, int, int, int, int,
                                      float*, int32_t*, const float*, int32_t*, int);
+__global__ void topk_scan_fp4_v2_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
+                                        const uint8_t*, int, int, int, int, int,
+                                        float*, int32_t*, const float*, int32_t*, int);
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
                                   float*, int32_t*);
 __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigned long long*,
@@ -384,7 +387,14 @@ std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
   auto cand_i = torch::full({(long long)nq, cap}, -1, i32opts);
   auto counts = torch::zeros({(long long)nq}, i32opts);
   dim3 grid((unsigned)(n_qblocks * n_swaths));
-  hipLaunchKernelGGL(topk_scan_fp4_kernel, grid, dim3(512), 0, cur_stream(),
+  static const bool use_v2 = [] {
+    const char* e = getenv("VAINPLEX_FP4_V2");
+    return e != nullptr && e[0] == '1';  // v1 default: at production
+    // survivor rates it matches/beats the pipelined v2 (A/B in
+    // docs/NOTES-NEXT.md "fp4 scan v2/v3 exploration")
+  }();
+  auto kern = use_v2 ? topk_scan_fp4_v2_kernel : topk_scan_fp4_kernel;
+  hipLaunchKernelGGL(kern, grid, dim3(512), 0, cur_stream(),
                      Q4.data_ptr<uint8_t>(), QS.data_ptr<uint8_t>(),
                      X4.data_ptr<uint8_t>(), XS.data_ptr<uint8_t>(),
                      nq, (int)nx, D, 1, (int)n_swaths,

@@ -1262,3 +1262,233 @@ topk_scan_fp4_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__
         }
   }
 }
+
+
+// ===========================================================================
+// fp4x4 threshold scan v2: software-pipelined fragment reads. v1 reads
+// all 24 LDS fragments, drains lgkmcnt(0) twice, then bursts 32 MFMAs —
+// with only 2 waves/SIMD the drain serializes LDS latency against the
+// MFMA pipe (measured 2.15 PF/s vs the instruction's 8.9 PF/s ceiling,
+// tools/probe_mfma_rate). v2 issues the X fragments + the first two Q
+// fragments, then per m-group overlaps the NEXT Q-fragment read with
+// the current 4-MFMA burst, waiting lgkmcnt(2) instead of 0; ordering
+// is pinned by "+v" passthroughs of the registers each wait guarantees.
+// Zero-padding of the unused high halves of the fp4 operands is dropped
+// (cbsz/blgp=4 consume 4 dwords; the probe's decode map covers only
+// those) — that removes ~96 v_mov per pair from the hot loop.
+// ===========================================================================
+
+// Transposed, 40 B-padded scale staging for the v2 scan: LDS byte
+// [row*40 + kgrp*8 + p] = global scale byte [row][4p + kgrp]. The
+// per-pair consumers then read ONE conflict-free ds_read_b64 per
+// fragment per x-tile (16 lanes x 40 B stride covers 16 distinct even
+// bank residues) instead of 12 bank-conflicted ds_read_u8 per pair —
+// PMC showed 6.0e9 LDS conflict cycles (~5 replays/read) on the u8 path.
+#define SCALE_PITCH 40
+DEVINL void stage_scale_rows_t(const uint8_t* __restrict__ src, int sb,
+                               long long row0, long long row_max,
+                               uint8_t* lds_base, int tile_rows) {
+  int nd = sb >> 2;  // dwords per row actually present (np)
+  for (int task = threadIdx.x; task < tile_rows * 4; task += TK_THREADS) {
+    int row = task >> 2, kg = task & 3;
+    long long gr = row0 + row;
+    if (gr >= row_max) gr = row_max - 1;
+    const uint8_t* rp = src + gr * (long long)sb;
+    uint32_t lo = 0, hi = 0;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      if (q < nd)
+        lo |= (uint32_t)rp[q * 4 + kg] << (8 * q);
+      if (q + 4 < nd)
+        hi |= (uint32_t)rp[(q + 4) * 4 + kg] << (8 * q);
+    }
+    uint8_t* dst = lds_base + row * SCALE_PITCH + kg * 8;
+    *(uint32_t*)dst = lo;
+    *(uint32_t*)(dst + 4) = hi;
+  }
+}
+
+DEVINL uint32_t scale_byte(uint32_t lo, uint32_t hi, int p) {
+  uint32_t d = (p & 4) ? hi : lo;
+  return (d >> (8 * (p & 3))) & 0xffu;
+}
+
+DEVINL v8i_mx fp4_frag(bf16x8 raw) {
+  int4 l = __builtin_bit_cast(int4, raw);
+  v8i_mx f;
+  f[0] = l.x; f[1] = l.y; f[2] = l.z; f[3] = l.w;
+  // f[4..7] intentionally uninitialized: fp4 operands (cbsz/blgp=4)
+  // consume only the low 4 dwords of the 8-dword tuple
+  return f;
+}
+
+extern "C" __global__ void __launch_bounds__(TK_THREADS)
+topk_scan_fp4_v2_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__ QS,
+                        const uint8_t* __restrict__ X4, const uint8_t* __restrict__ XS,
+                        int nq, int nx, int D, int k, int n_swaths,
+                        float* __restrict__ cand_scores,
+                        int32_t* __restrict__ cand_ids,
+                        const float* __restrict__ theta,
+                        int32_t* __restrict__ tc_n, int cap) {
+  __shared__ bf16 lds_q[4 * BM * BK];
+  __shared__ bf16 lds_x[4 * BN * BK];
+  __shared__ uint8_t lds_qs[BM * SCALE_PITCH];
+  __shared__ uint8_t lds_xs2[BN * SCALE_PITCH];
+  __shared__ float row_min[BM];
+
+  int S = n_swaths;
+  int qb = blockIdx.x / S;
+  int swath = blockIdx.x % S;
+  long long row0 = (long long)qb * BM;
+
+  long long per = ((long long)nx + S - 1) / S;
+  per = ((per + BN - 1) / BN) * BN;
+  long long x_begin = (long long)swath * per;
+  long long x_end = min((long long)nx, x_begin + per);
+
+  for (int i = threadIdx.x; i < BM; i += blockDim.x)
+    row_min[i] = (theta != nullptr && row0 + i < nq) ? theta[row0 + i] : -1e30f;
+  __syncthreads();
+
+  int wid = wave_id();
+  int wm = wid >> 2, wn = wid & 3;
+  int lane = lane_id();
+  int lrow = lane & 15;
+  int kgrp = lane >> 4;
+  int np = D / (2 * BK_F8);
+  int sb = D / 32;
+  long long p4_ld = D / 4;
+
+  for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
+    f32x4 acc[8][4] = {};
+    stage_scale_rows_t(QS, sb, row0, (long long)nq, lds_qs, BM);
+    stage_scale_rows_t(XS, sb, x0, (long long)nx, lds_xs2, BN);
+    __syncthreads();  // transposed ds_writes visible before any read
+    // per-x-tile scale registers: one b64 per fragment, conflict-free
+    uint32_t xs_lo[4], xs_hi[4], qs_lo[8], qs_hi[8];
+    {
+      uint32_t xrow_base = (uint32_t)(wn * 64 + lrow);
+      uint32_t qrow_base = (uint32_t)(wm * 128 + lrow);
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        const uint8_t* a = lds_xs2 + (xrow_base + 16u * n) * SCALE_PITCH + kgrp * 8;
+        xs_lo[n] = *(const uint32_t*)a;
+        xs_hi[n] = *(const uint32_t*)(a + 4);
+      }
+#pragma unroll
+      for (int m = 0; m < 8; ++m) {
+        const uint8_t* a = lds_qs + (qrow_base + 16u * m) * SCALE_PITCH + kgrp * 8;
+        qs_lo[m] = *(const uint32_t*)a;
+        qs_hi[m] = *(const uint32_t*)(a + 4);
+      }
+    }
+    for (int pp = 0; pp < 3 && pp < np; ++pp) {
+      stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, pp * 32, QP4(pp), BM);
+      stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, pp * 32, XP4(pp), BN);
+    }
+    for (int p = 0; p < np; ++p) {
+      if (p + 2 < np)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else if (p + 1 < np)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (2 * p + 6 < 2 * np) {
+        stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, (p + 3) * 32,
+                   QP4((p + 3) & 3), BM);
+        stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, (p + 3) * 32,
+                   XP4((p + 3) & 3), BN);
+      }
+      uint32_t xrow_base = (uint32_t)(wn * 64 + lrow);
+      uint32_t qrow_base = (uint32_t)(wm * 128 + lrow);
+      uint32_t xaddr = (uint32_t)(size_t)XP4(p & 3)
+                       + lds_off_bytes(xrow_base, (uint32_t)kgrp);
+      uint32_t qaddr = (uint32_t)(size_t)QP4(p & 3)
+                       + lds_off_bytes(qrow_base, (uint32_t)kgrp);
+      bf16x8 xf[4], qf[8];
+      // per-pair scale bytes extracted from the hoisted b64 registers
+      uint32_t xsb[4], qsb[8];
+#pragma unroll
+      for (int n = 0; n < 4; ++n) xsb[n] = scale_byte(xs_lo[n], xs_hi[n], p);
+#pragma unroll
+      for (int m = 0; m < 8; ++m) qsb[m] = scale_byte(qs_lo[m], qs_hi[m], p);
+      // prologue: X fragments + Q fragments 0,1; lgkmcnt(1) leaves
+      // exactly qf1 in flight
+      asm volatile(
+          "ds_read_b128 %0, %6\n\t"
+          "ds_read_b128 %1, %6 offset:1024\n\t"
+          "ds_read_b128 %2, %6 offset:2048\n\t"
+          "ds_read_b128 %3, %6 offset:3072\n\t"
+          "ds_read_b128 %4, %7\n\t"
+          "ds_read_b128 %5, %7 offset:1024\n\t"
+          "s_waitcnt lgkmcnt(1)"
+          : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
+            "=&v"(qf[0]), "=&v"(qf[1])
+          : "v"(xaddr), "v"(qaddr));
+      v8i_mx xv[4];
+#pragma unroll
+      for (int n = 0; n < 4; ++n) xv[n] = fp4_frag(xf[n]);
+
+#define FP4V2_MFMA_GROUP(mm)                                                  \
+  {                                                                           \
+    v8i_mx qv = fp4_frag(qf[mm]);                                             \
+    _Pragma("unroll") for (int n = 0; n < 4; ++n)                             \
+        acc[mm][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(        \
+            qv, xv[n], acc[mm][n], 4, 4, 0, (int)qsb[mm], 0, (int)xsb[n]);    \
+  }
+// issue fragment mm+2 while group mm computes; the lgkmcnt(1) then
+// guarantees fragment mm+1, whose registers pass through to pin order
+#define FP4V2_READ_NEXT(mm)                                                   \
+  asm volatile(                                                               \
+      "ds_read_b128 %0, %2\n\t"                                               \
+      "s_waitcnt lgkmcnt(1)"                                                  \
+      : "=&v"(qf[(mm) + 2]), "+v"(qf[(mm) + 1])                               \
+      : "v"(qaddr + ((mm) + 2) * 1024u));
+
+      FP4V2_MFMA_GROUP(0)
+      FP4V2_READ_NEXT(0)
+      FP4V2_MFMA_GROUP(1)
+      FP4V2_READ_NEXT(1)
+      FP4V2_MFMA_GROUP(2)
+      FP4V2_READ_NEXT(2)
+      FP4V2_MFMA_GROUP(3)
+      FP4V2_READ_NEXT(3)
+      FP4V2_MFMA_GROUP(4)
+      FP4V2_READ_NEXT(4)
+      FP4V2_MFMA_GROUP(5)
+      FP4V2_READ_NEXT(5)
+      FP4V2_MFMA_GROUP(6)
+      asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(qf[7]));
+      FP4V2_MFMA_GROUP(7)
+#undef FP4V2_MFMA_GROUP
+#undef FP4V2_READ_NEXT
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    if (k < 0) {
+      if (acc[0][0][0] > 1e29f) cand_scores[0] = acc[0][0][0];
+      continue;
+    }
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+          float v = acc[m][n][r];
+          if (!(v > row_min[row])) continue;
+          long long grow = row0 + row;
+          if (grow >= nq) continue;
+          long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+          if (col >= x_end) continue;
+          int pos = atomicAdd(&tc_n[grow], 1);
+          if (pos < cap) {
+            cand_scores[grow * cap + pos] = v;
+            cand_ids[grow * cap + pos] = int32_t(col);
+          }
+        }
+  }
+}
