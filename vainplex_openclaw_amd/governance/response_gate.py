@@ -90,7 +90,12 @@ class ResponseGate:
         return {"passed": True}
 
     def _render_fallback(self, agent_id: str, failed: List[str], reasons: List[str]) -> Optional[str]:
-        template = self.config.get("fallbackMessage") or self.config.get("fallbackTemplate")
+        # fallbackMessage is STATIC and takes precedence; fallbackTemplate
+        # renders {reasons}/{validators}/{agent} (response-gate.ts:153-169)
+        static = self.config.get("fallbackMessage")
+        if static:
+            return static
+        template = self.config.get("fallbackTemplate")
         if not template:
             return None
         return (
