@@ -397,3 +397,101 @@ def test_fact_get_boosts_on_access_and_multi_filter(tmp_path):
     fs.add_fact("db", "runs-on", "k8s")
     assert len(fs.query(subject="svc", obj="k8s")) == 1
     assert len(fs.query()) == 3
+
+
+# ===========================================================================
+# fact-store.test.ts depth: load-gate, access boost, query matrix, decay
+# floor, embedding lifecycle, prune order
+# ===========================================================================
+
+def _store(workspace, **kw):
+    from vainplex_openclaw_amd.knowledge.fact_store import FactStore
+
+    t = [1_700_000_000.0]
+    fs = FactStore(workspace, clock=lambda: t[0], **kw)
+    fs.load()
+    return fs, t
+
+
+def test_fact_store_add_before_load_raises(workspace):
+    from vainplex_openclaw_amd.knowledge.fact_store import FactStore
+
+    fs = FactStore(workspace)
+    with pytest.raises(RuntimeError):
+        fs.add_fact("a", "b", "c")
+
+
+def test_fact_store_get_boosts_and_touches(workspace):
+    # relevance starts at 1.0 and boost moves 50% closer to 1.0, so the
+    # boost is only visible after decay (fact-store.ts boostRelevance)
+    fs, t = _store(workspace)
+    f = fs.add_fact("svc", "status", "running")
+    fs.decay_facts(0.5)
+    assert f["relevance"] == 0.5
+    t[0] += 100
+    got = fs.get_fact(f["id"])
+    assert got["relevance"] == 0.75  # 0.5 + (1-0.5)*0.5
+    assert got["lastAccessed"] != f["createdAt"]
+    assert fs.get_fact("nope") is None
+
+
+def test_fact_store_query_matrix(workspace):
+    fs, _ = _store(workspace)
+    fs.add_fact("nginx", "status", "running")
+    fs.add_fact("nginx", "port", "443")
+    fs.add_fact("redis", "status", "running")
+    assert len(fs.query(subject="nginx")) == 2
+    assert len(fs.query(predicate="status")) == 2
+    assert len(fs.query(obj="running")) == 2
+    assert len(fs.query(subject="nginx", predicate="status")) == 1
+    assert len(fs.query()) == 3
+    assert fs.query(subject="ghost") == []
+
+
+def test_fact_store_query_sorted_by_relevance(workspace):
+    fs, _ = _store(workspace)
+    fs.add_fact("a", "p", "1")
+    fs.add_fact("b", "p", "2")
+    fs.decay_facts(0.5)          # both at 0.5
+    fs.add_fact("b", "p", "2")   # dedupe boost lifts b to 0.75
+    out = fs.query(predicate="p")
+    assert out[0]["subject"] == "b" and out[1]["subject"] == "a"
+    assert out[0]["relevance"] > out[1]["relevance"]
+
+
+def test_fact_store_decay_floor_and_counts(workspace):
+    fs, _ = _store(workspace)
+    fs.add_fact("a", "p", "1")
+    assert fs.decay_facts(0.5) == 1
+    for _ in range(10):
+        fs.decay_facts(0.5)
+    assert all(f["relevance"] >= 0.1 for f in fs.query())
+    empty, _t = _store(workspace + "/sub2")
+    assert empty.decay_facts(0.5) == 0
+
+
+def test_fact_store_embedding_lifecycle(workspace):
+    fs, _ = _store(workspace)
+    a = fs.add_fact("a", "p", "1")
+    b = fs.add_fact("b", "p", "2")
+    assert {f["id"] for f in fs.unembedded_facts()} == {a["id"], b["id"]}
+    fs.mark_embedded([a["id"], "ghost-id"])
+    left = fs.unembedded_facts()
+    assert [f["id"] for f in left] == [b["id"]]
+    assert fs.facts[a["id"]]["embedded"]
+    fs.mark_embedded([b["id"]])
+    assert fs.unembedded_facts() == []
+
+
+def test_fact_store_prune_least_relevant_first(workspace):
+    fs, t = _store(workspace, max_facts=3)
+    keep = fs.add_fact("hot", "p", "v")
+    cold = fs.add_fact("cold0", "p", "v0")
+    fs.decay_facts(0.5)
+    fs.add_fact("hot", "p", "v")     # boost the keeper back to 0.75
+    for i in range(1, 3):
+        t[0] += 1
+        fs.add_fact(f"cold{i}", "p", f"v{i}")
+    assert len(fs.facts) == 3
+    assert keep["id"] in fs.facts     # boosted fact survived
+    assert cold["id"] not in fs.facts  # least relevant evicted
