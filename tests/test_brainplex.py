@@ -422,3 +422,91 @@ def test_dashboard_gateway_method_registered():
     create_plugin().register(api)
     assert "brainplex.dashboard" in api.gateway_methods
     assert isinstance(api.gateway_methods["brainplex.dashboard"](), str)
+
+
+# ===========================================================================
+# scanner.test.ts depth: JSON5 tolerance, walk-up, agent extraction
+# shapes, plugin detection sources
+# ===========================================================================
+
+def test_scanner_json5_tolerance():
+    from vainplex_openclaw_amd.brainplex.scanner import parse_config
+
+    assert parse_config('{"a": 1}') == {"a": 1}
+    assert parse_config('{\n// comment\n"a": 1}') == {"a": 1}
+    assert parse_config('{/* multi\nline */ "a": 1}') == {"a": 1}
+    assert parse_config('{"a": 1, "b": [1, 2,], }') == {"a": 1, "b": [1, 2]}
+    import pytest as _pt
+
+    with _pt.raises(Exception):
+        parse_config("not even close {{{")
+
+
+def test_scanner_find_config_variants(tmp_path):
+    from vainplex_openclaw_amd.brainplex.scanner import find_config
+
+    # direct openclaw.json
+    d1 = tmp_path / "w1"; d1.mkdir()
+    (d1 / "openclaw.json").write_text("{}")
+    assert find_config(str(d1), home=str(tmp_path / "nohome")) == str(d1 / "openclaw.json")
+    # .openclaw/openclaw.json
+    d2 = tmp_path / "w2"; (d2 / ".openclaw").mkdir(parents=True)
+    (d2 / ".openclaw" / "openclaw.json").write_text("{}")
+    assert find_config(str(d2), home=str(tmp_path / "nohome")).endswith(
+        ".openclaw/openclaw.json")
+    # walk-up from a nested dir
+    nested = d1 / "a" / "b"; nested.mkdir(parents=True)
+    assert find_config(str(nested), home=str(tmp_path / "nohome")) == str(d1 / "openclaw.json")
+    # not found anywhere
+    d3 = tmp_path / "w3"; d3.mkdir()
+    assert find_config(str(d3), home=str(tmp_path / "nohome")) is None
+
+
+@pytest.mark.parametrize("config,want", [
+    ({"agents": {"definitions": [{"id": "a1"}, {"name": "a2"}]}}, ["a1", "a2"]),
+    ({"agents": {"alpha": {}, "beta": {}}}, ["alpha", "beta"]),
+    ({"agents": ["x", "y"]}, ["x", "y"]),
+    ({"agents": {"list": [{"id": "main"}, {"id": "forge"}]}}, ["main", "forge"]),
+    ({"agents": {"list": [{"id": "byid", "name": "byname"}]}}, ["byid"]),
+    ({}, []),
+    ({"agents": {"definitions": [], "defaults": {}}}, []),
+])
+def test_scanner_agent_extraction_shapes(config, want):
+    from vainplex_openclaw_amd.brainplex.scanner import extract_agents
+
+    assert extract_agents(config) == want
+
+
+def test_scanner_plugin_detection_sources(tmp_path):
+    from vainplex_openclaw_amd.brainplex.scanner import detect_installed_plugins
+
+    cfg = {"plugins": {
+        "entries": {"openclaw-governance": {"enabled": True}},
+        "allow": ["openclaw-cortex"],
+        "installs": ["openclaw-leuko"],
+    }}
+    got = detect_installed_plugins(cfg, str(tmp_path))
+    assert {"openclaw-governance", "openclaw-cortex", "openclaw-leuko"} <= got
+    assert detect_installed_plugins({}, str(tmp_path)) == set()
+
+
+def test_scanner_configured_plugins(tmp_path):
+    from vainplex_openclaw_amd.brainplex.scanner import detect_configured_plugins
+
+    pdir = tmp_path / ".openclaw" / "plugins"
+    (pdir / "with-config").mkdir(parents=True)
+    (pdir / "with-config" / "config.json").write_text("{}")
+    (pdir / "without-config").mkdir()
+    got = detect_configured_plugins(home=str(tmp_path))
+    assert got == {"with-config"}
+    assert detect_configured_plugins(home=str(tmp_path / "ghost")) == set()
+
+
+def test_scanner_extensions_dir_detection(tmp_path):
+    from vainplex_openclaw_amd.brainplex.scanner import detect_installed_plugins
+
+    ext = tmp_path / "extensions" / "openclaw-membrane"
+    ext.mkdir(parents=True)
+    (tmp_path / "extensions" / "not-a-dir.txt").write_text("x")
+    got = detect_installed_plugins({}, str(tmp_path))
+    assert got == {"openclaw-membrane"}

@@ -72,12 +72,31 @@ def extract_agents(config: Dict[str, Any]) -> List[str]:
 
 
 def detect_installed_plugins(config: Dict[str, Any], config_dir: str) -> Set[str]:
-    """A plugin counts as installed when it has an entries record."""
+    """Installed = plugins.entries keys + plugins.allow strings +
+    plugins.installs keys (dict) or items (list) + directories under the
+    extensions path (scanner.ts detectInstalledPlugins)."""
+    out: Set[str] = set()
     plugins = config.get("plugins")
-    if not isinstance(plugins, dict):
-        return set()
-    entries = plugins.get("entries")
-    return set(entries.keys()) if isinstance(entries, dict) else set()
+    if isinstance(plugins, dict):
+        entries = plugins.get("entries")
+        if isinstance(entries, dict):
+            out.update(entries.keys())
+        allow = plugins.get("allow")
+        if isinstance(allow, list):
+            out.update(x for x in allow if isinstance(x, str))
+        installs = plugins.get("installs")
+        if isinstance(installs, dict):
+            out.update(installs.keys())
+        elif isinstance(installs, list):
+            out.update(x for x in installs if isinstance(x, str))
+    ext = os.path.join(config_dir, "extensions")
+    if os.path.isdir(ext):
+        try:
+            out.update(d for d in os.listdir(ext)
+                       if os.path.isdir(os.path.join(ext, d)))
+        except OSError:
+            pass
+    return out
 
 
 def detect_configured_plugins(home: Optional[str] = None) -> Set[str]:
