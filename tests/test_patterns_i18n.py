@@ -395,3 +395,95 @@ def test_high_impact_keywords_nonempty_and_lower(code):
     kws = high_impact_keywords(code)
     assert len(kws) >= 8
     assert all(k == k.lower() for k in kws)
+
+
+# ===========================================================================
+# patterns-custom.test.ts depth: extend/override per family, invalid
+# regex skip, custom blacklist/keywords, language resolution table
+# ===========================================================================
+
+from vainplex_openclaw_amd.cortex.patterns import _resolve_codes
+
+
+@pytest.mark.parametrize("family,phrase", [
+    ("decision", "XDECIDEX now"),
+    ("close", "XCLOSEX done-ish"),
+    ("wait", "XWAITX pending"),
+    ("topic", "XTOPICX something"),
+])
+def test_custom_extend_appends_per_family(family, phrase):
+    reg = get_registry("en", custom={family: [phrase.split()[0]]})
+    assert any(rx.search(phrase) for rx in reg.get_patterns(family))
+    # builtins still present in extend mode
+    assert any(rx.search(DECISION_SAMPLES["en"])
+               for rx in reg.get_patterns("decision"))
+
+
+def test_custom_override_replaces_only_that_family():
+    reg = get_registry("en", custom={"mode": "override",
+                                     "decision": ["ONLYTHIS"]})
+    pats = reg.get_patterns("decision")
+    assert len(pats) == 1 and pats[0].search("ONLYTHIS here")
+    assert not any(rx.search(DECISION_SAMPLES["en"]) for rx in pats)
+    # close family untouched by a decision-only override
+    assert any(rx.search(CLOSE_SAMPLES["en"]) for rx in reg.get_patterns("close"))
+
+
+def test_custom_override_empty_array_keeps_builtins():
+    reg = get_registry("en", custom={"mode": "override", "decision": []})
+    assert any(rx.search(DECISION_SAMPLES["en"])
+               for rx in reg.get_patterns("decision"))
+
+
+def test_custom_invalid_regex_skipped_silently():
+    reg = get_registry("en", custom={"decision": ["(unclosed", "VALIDPAT"]})
+    pats = reg.get_patterns("decision")
+    assert any(rx.search("VALIDPAT") for rx in pats)
+    # all-invalid: builtins remain, nothing raises
+    reg2 = get_registry("en", custom={"mode": "override",
+                                      "decision": ["(bad", "[worse"]})
+    assert any(rx.search(DECISION_SAMPLES["en"])
+               for rx in reg2.get_patterns("decision"))
+
+
+def test_custom_blacklist_and_keywords_merge():
+    reg = get_registry("en", custom={"blacklist": ["Widget"],
+                                     "keywords": ["Rollout"]})
+    assert "widget" in reg.blacklist
+    assert "rollout" in reg.high_impact
+    assert "architecture" in reg.high_impact  # builtins kept
+
+
+def test_custom_non_string_values_filtered():
+    reg = get_registry("en", custom={"decision": [42, None, "OKPAT"],
+                                     "blacklist": [7, "ok"],
+                                     "keywords": [None, "kw"]})
+    assert any(rx.search("OKPAT") for rx in reg.get_patterns("decision"))
+    assert "ok" in reg.blacklist and 7 not in reg.blacklist
+    assert "kw" in reg.high_impact
+
+
+@pytest.mark.parametrize("arg,want", [
+    ("en", ["en"]),
+    ("de", ["de"]),
+    ("both", ["en", "de"]),
+    (["en", "fr"], ["en", "fr"]),
+    ("ja", ["ja"]),
+    (42, ["en", "de"]),
+    (None, ["en", "de"]),
+])
+def test_resolve_codes_table(arg, want):
+    assert _resolve_codes(arg) == want
+
+
+def test_resolve_all_is_every_language():
+    assert _resolve_codes("all") == list(language_codes())
+
+
+def test_all_languages_merge_blacklist_and_keywords():
+    reg = get_registry("all")
+    # one word from each of several languages' blacklists
+    for w in ("it", "das", "eso", "ça", "это"):
+        assert w in reg.blacklist, w
+    for kw in ("architecture", "sicherheit", "seguridad", "миграция", "安全"):
+        assert kw in reg.high_impact, kw
