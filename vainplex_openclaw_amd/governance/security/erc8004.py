@@ -170,7 +170,9 @@ class ERC8004Client:
             return cached
         try:
             owner_raw = self._call(encode_call("ownerOf(address)", agent_address))
-            owner = decode_address(owner_raw or "")
+            if owner_raw is None:
+                return None  # transport failure: fail-open, not "unregistered"
+            owner = decode_address(owner_raw)
             if owner == ZERO_ADDRESS:
                 rep = {
                     "address": agent_address, "registered": False,
