@@ -546,3 +546,69 @@ def test_schema_b_unknown_kind_none():
     from vainplex_openclaw_amd.cortex.trace.events import normalize_schema_b
 
     assert normalize_schema_b({"kind": "nope", "ts": 1.0}) is None
+
+
+# -- multilang-detectors.test.ts depth ---------------------------------------
+
+COMPLETION_WORDS = {
+    "fr": "c'est terminé", "ja": "タスクは完了です", "ko": "작업 완료 했어요",
+    "es": "está completado", "ru": "всё готово", "de": "das ist erledigt",
+    "pt": "está concluído", "zh": "任务完成", "it": "tutto completato",
+    "en": "the task is done",
+}
+
+
+@pytest.mark.parametrize("lang", sorted(COMPLETION_WORDS))
+def test_hallucination_completion_claim_after_tool_error(lang):
+    events = [
+        ev("msg.out", content=COMPLETION_WORDS[lang]),
+        ev("tool.result", toolError="command failed with exit 1", toolIsError=True),
+    ]
+    chains = reconstruct_chains(events)
+    kinds = {f.signal_type for f in detect_all_signals(chains, ["hallucination"])}
+    assert "hallucination" in kinds, lang
+
+
+def test_hallucination_completion_claim_without_error_is_clean():
+    events = [
+        ev("msg.out", content="c'est terminé"),
+        ev("tool.result", toolResult="ok"),
+    ]
+    chains = reconstruct_chains(events)
+    assert not detect_all_signals(chains, ["hallucination"])
+
+
+STATE_CLAIMS = [
+    ("there are 5 errors in the log", True),
+    ("es gibt 5 fehler im log", True),
+    ("il y a 3 erreurs", True),
+    ("hay 5 errores en total", True),
+    ("ci sono 2 errori", True),
+    ("есть 4 ошибки", True),
+    ("je crois qu'il y a 3 erreurs", False),      # opinion exclusion
+    ("ich glaube es gibt 5 fehler", False),
+    ("creo que hay 5 errores", False),
+    ("наверное есть 4 ошибки", False),
+    ("the weather is nice", False),
+]
+
+
+@pytest.mark.parametrize("text,want", STATE_CLAIMS)
+def test_unverified_state_claims_with_opinion_exclusions(text, want):
+    events = [
+        ev("msg.in", content="status?"),
+        ev("msg.out", content=text),
+    ]
+    chains = reconstruct_chains(events)
+    kinds = {f.signal_type for f in detect_all_signals(chains, ["unverified_claim"])}
+    assert ("unverified_claim" in kinds) == want, text
+
+
+def test_state_claim_backed_by_tool_result_passes():
+    events = [
+        ev("tool.call", toolName="grep"),
+        ev("tool.result", toolResult="5 matches"),
+        ev("msg.out", content="there are 5 errors in the log"),
+    ]
+    chains = reconstruct_chains(events)
+    assert not detect_all_signals(chains, ["unverified_claim"])

@@ -272,6 +272,54 @@ _ABS_CLAIM_RX = re.compile(
     r"\b(?:definitely|certainly|guaranteed|always|never fails|100%|no doubt)\b", re.I
 )
 
+# localized completion-claim words (multilang hallucination detection:
+# signal-lang packs in the reference's multilang-detectors suite)
+_COMPLETION_RX = re.compile(
+    "|".join([
+        r"\b(?:done|completed|finished|fixed)\b",
+        r"\b(?:fertig|erledigt|abgeschlossen)\b",
+        r"\b(?:terminé|fini|corrigé)\b",
+        r"\b(?:completado|terminado|arreglado)\b",
+        r"\b(?:completato|finito|risolto)\b",
+        r"\b(?:concluído|pronto|consertado)\b",
+        r"готово|завершено|исправлено",
+        "完了|完成した|修正済み",
+        "완료|끝났|수정됨",
+        "完成|搞定|修好",
+    ]),
+    re.I,
+)
+
+# localized system-state claim shapes + opinion exclusions
+# ("es gibt 5 fehler" counts; "je crois qu'il y a..." does not)
+_STATE_CLAIM_RX = re.compile(
+    "|".join([
+        r"\bthere (?:are|is)\s+\d+\s+error",
+        r"\bes gibt\s+\d+\s+fehler",
+        r"\bil y a\s+\d+\s+erreurs?",
+        r"\bhay\s+\d+\s+errores?",
+        r"\bci sono\s+\d+\s+errori",
+        r"\bhá\s+\d+\s+erros?",
+        r"есть\s+\d+\s+ошиб",
+    ]),
+    re.I,
+)
+_OPINION_RX = re.compile(
+    "|".join([
+        r"\b(?:i think|i believe|maybe|probably|perhaps)\b",
+        r"\b(?:ich glaube|ich denke|vielleicht|wahrscheinlich)\b",
+        r"\b(?:je crois|je pense|peut-être)\b",
+        r"\b(?:creo que|quizás|tal vez)\b",
+        r"\b(?:credo che|forse)\b",
+        r"\b(?:acho que|talvez)\b",
+        r"думаю|наверное|возможно",
+        "たぶん|と思う",
+        "아마|같아요",
+        "可能|大概|我觉得",
+    ]),
+    re.I,
+)
+
 
 def detect_hallucination(chain: ConversationChain) -> List[Finding]:
     """Agent asserts a fact that the next tool result contradicts
@@ -283,9 +331,12 @@ def detect_hallucination(chain: ConversationChain) -> List[Finding]:
             continue
         content = str(ev.payload.get("content") or "")
         claims = re.findall(r"([\w.-]{2,40})\s+(?:is|are)\s+(?:running|done|deployed|fixed|available)", content, re.I)
-        if not claims:
+        completion = _COMPLETION_RX.search(content) is not None
+        if not claims and not completion:
             continue
-        # next tool result errors referencing the same subject -> contradiction
+        # next tool result errors -> contradiction (subject-matched for
+        # the English state claims; any error for localized completion
+        # claims, the multilang-detectors contract)
         for nxt in evs[i + 1 : i + 5]:
             if nxt.type == "tool.result" and (nxt.payload.get("toolError") or nxt.payload.get("toolIsError")):
                 err = str(nxt.payload.get("toolError") or "")
@@ -296,6 +347,12 @@ def detect_hallucination(chain: ConversationChain) -> List[Finding]:
                                         {"subject": subj, "predicate": "state",
                                          "value": "error", "claim": content[:150],
                                          "error": err[:150]}, 0.65))
+                elif completion:
+                    findings.append(_mk(chain, "hallucination", "medium",
+                                        "Completion claim followed by tool error",
+                                        {"claim": content[:150],
+                                         "error": err[:150]}, 0.55))
+                break
     return findings
 
 
@@ -307,7 +364,9 @@ def detect_unverified_claim(chain: ConversationChain) -> List[Finding]:
         if ev.type != "msg.out":
             continue
         content = str(ev.payload.get("content") or "")
-        if _ABS_CLAIM_RX.search(content):
+        state_claim = (_STATE_CLAIM_RX.search(content) is not None
+                       and _OPINION_RX.search(content) is None)
+        if _ABS_CLAIM_RX.search(content) or state_claim:
             # only a SUCCESSFUL tool result counts as verification
             tool_before = any(
                 e.type == "tool.result"
