@@ -42,6 +42,7 @@ __global__ void topk_scan_fp4_kernel(const uint8_t*, const uint8_t*, const uint8
 __global__ void topk_scan_fp4_v2_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
                                         const uint8_t*, int, int, int, int, int,
                                         float*, int32_t*, const float*, int32_t*, int);
+
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
                                   float*, int32_t*);
 __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigned long long*,
@@ -374,6 +375,10 @@ std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
   TORCH_CHECK(QS.size(0) == nq && QS.size(1) == D / 32);
   TORCH_CHECK(X4.size(1) == D / 2 && XS.size(0) == nx && XS.size(1) == D / 32);
   TORCH_CHECK(theta.numel() == nq && cap >= 32 && cap <= 4096);
+  static const int variant = [] {
+    const char* e = getenv("VAINPLEX_FP4_V2");
+    return (e != nullptr && e[0] == '1') ? 1 : 0;  // pipelined v2 opt-in
+  }();
   int n_qblocks = (nq + 255) / 256;
   if (n_swaths <= 0) {
     int want = 256 / (n_qblocks > 0 ? n_qblocks : 1);
@@ -387,13 +392,7 @@ std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
   auto cand_i = torch::full({(long long)nq, cap}, -1, i32opts);
   auto counts = torch::zeros({(long long)nq}, i32opts);
   dim3 grid((unsigned)(n_qblocks * n_swaths));
-  static const bool use_v2 = [] {
-    const char* e = getenv("VAINPLEX_FP4_V2");
-    return e != nullptr && e[0] == '1';  // v1 default: at production
-    // survivor rates it matches/beats the pipelined v2 (A/B in
-    // docs/NOTES-NEXT.md "fp4 scan v2/v3 exploration")
-  }();
-  auto kern = use_v2 ? topk_scan_fp4_v2_kernel : topk_scan_fp4_kernel;
+  auto kern = variant == 1 ? topk_scan_fp4_v2_kernel : topk_scan_fp4_kernel;
   hipLaunchKernelGGL(kern, grid, dim3(512), 0, cur_stream(),
                      Q4.data_ptr<uint8_t>(), QS.data_ptr<uint8_t>(),
                      X4.data_ptr<uint8_t>(), XS.data_ptr<uint8_t>(),
@@ -417,6 +416,10 @@ std::vector<torch::Tensor> topk_scan_threshold(torch::Tensor Q, torch::Tensor X,
   TORCH_CHECK(theta.numel() == nq);
   TORCH_CHECK(X.size(1) == D && D % 64 == 0);
   TORCH_CHECK(cap >= 32 && cap <= 4096);
+  static const int variant = [] {
+    const char* e = getenv("VAINPLEX_FP4_V2");
+    return (e != nullptr && e[0] == '1') ? 1 : 0;  // pipelined v2 opt-in
+  }();
   int n_qblocks = (nq + 255) / 256;
   if (n_swaths <= 0) {
     int want = 256 / (n_qblocks > 0 ? n_qblocks : 1);
