@@ -92,30 +92,44 @@ def resolve_audit(raw: Any) -> Dict[str, Any]:
     }
 
 
+def _str_list(v: Any, default: list) -> list:
+    if not isinstance(v, list):
+        return list(default)
+    return [x for x in v if isinstance(x, str)]
+
+
 def resolve_llm_validator(raw: Any) -> Dict[str, Any]:
     r = _rec(raw)
     return {
         "enabled": _boolv(r.get("enabled"), False),
-        "externalChannels": r.get("externalChannels")
-        if isinstance(r.get("externalChannels"), list)
-        else ["twitter", "linkedin", "email"],
-        "externalCommands": r.get("externalCommands")
-        if isinstance(r.get("externalCommands"), list) else ["bird tweet"],
+        "externalChannels": _str_list(r.get("externalChannels"),
+                                      ["twitter", "linkedin", "email"]),
+        "externalCommands": _str_list(r.get("externalCommands"), ["bird tweet"]),
         "cacheTtlSeconds": _num(r.get("cacheTtlSeconds"), 300),
         "model": _str(r.get("model"), "mistral:7b"),
         "endpoint": _str(r.get("endpoint"), "http://localhost:11434/api/generate"),
     }
 
 
+def _policy3(v: Any) -> str:
+    return v if v in ("ignore", "flag", "block") else "ignore"
+
+
 def resolve_output_validation(raw: Any) -> Dict[str, Any]:
     r = _rec(raw)
-    detectors = r.get("detectors")
+    # accept both the reference's `enabledDetectors`/`claimDetectors`
+    # spellings and the legacy `detectors`
+    detectors = (r.get("enabledDetectors") if isinstance(r.get("enabledDetectors"), list)
+                 else r.get("claimDetectors") if isinstance(r.get("claimDetectors"), list)
+                 else r.get("detectors"))
     return {
         "enabled": _boolv(r.get("enabled"), True),
         "detectors": [d for d in detectors if d in ALL_DETECTOR_IDS]
         if isinstance(detectors, list) else list(ALL_DETECTOR_IDS),
         "facts": r.get("facts") if isinstance(r.get("facts"), list) else [],
         "factsFile": r.get("factsFile") if isinstance(r.get("factsFile"), str) else None,
+        "unverifiedClaimPolicy": _policy3(r.get("unverifiedClaimPolicy")),
+        "selfReferentialPolicy": _policy3(r.get("selfReferentialPolicy")),
         "llmValidator": resolve_llm_validator(r.get("llmValidator")),
     }
 
