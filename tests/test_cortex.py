@@ -256,3 +256,76 @@ def test_custom_patterns_via_workspace_config(workspace):
         assert ws.decisions.decisions, "custom decision pattern should fire"
     finally:
         P.set_custom_patterns(None)
+
+
+# -- commitment-tracker.test.ts depth ---------------------------------------
+
+def _ct(workspace, **cfg):
+    from vainplex_openclaw_amd.cortex.commitment_tracker import (
+        CommitmentTracker,
+        CommitmentTrackerConfig,
+    )
+
+    t = [1_700_000_000.0]
+    return CommitmentTracker(workspace, CommitmentTrackerConfig(**cfg),
+                             clock=lambda: t[0]), t
+
+
+def test_commitment_detection_and_what_capture(workspace):
+    ct, _ = _ct(workspace)
+    n = ct.process_message("I'll send the report by friday", "agent")
+    assert n == 1
+    c = ct.open_commitments()[0]
+    assert "report" in c["action"]
+    assert c["status"] == "open" and c["by"] == "agent"
+    assert c["language"] == "en" and c["completed"] is None
+
+
+def test_commitment_dedupe_same_message(workspace):
+    ct, _ = _ct(workspace)
+    ct.process_message("I'll update the docs. I'll update the docs.", "agent")
+    assert len(ct.open_commitments()) == 1
+
+
+def test_commitment_non_commitment_messages(workspace):
+    ct, _ = _ct(workspace)
+    assert ct.process_message("the weather is nice", "agent") == 0
+    assert ct.process_message("", "agent") == 0
+    assert ct.open_commitments() == []
+
+
+def test_commitment_overdue_marking(workspace):
+    ct, t = _ct(workspace)
+    ct.process_message("I'll fix the flaky test tomorrow", "agent")
+    assert ct.overdue() == []                 # fresh: not overdue
+    t[0] += 8 * 86400                          # past the 7-day window
+    od = ct.overdue()
+    assert len(od) == 1 and od[0]["status"] in ("open", "overdue")
+    # recent commitment not flagged
+    ct.process_message("I'll write the changelog", "agent")
+    assert len(ct.overdue()) == 1
+
+
+def test_commitment_done_not_overdue(workspace):
+    ct, t = _ct(workspace)
+    ct.process_message("I'll deploy the fix", "agent")
+    cid = ct.open_commitments()[0]["id"]
+    assert ct.complete(cid)
+    assert not ct.complete("ghost")
+    t[0] += 30 * 86400
+    assert ct.overdue() == []
+
+
+def test_commitment_flush_roundtrip(workspace):
+    import json
+    import os
+
+    ct, _ = _ct(workspace)
+    ct.process_message("I'll benchmark the kernel", "agent")
+    ct.flush()
+    path = os.path.join(workspace, "memory", "reboot", "commitments.json")
+    data = json.load(open(path))
+    blob = json.dumps(data)
+    assert "benchmark" in blob
+    ct2, _ = _ct(workspace)
+    assert any("benchmark" in c["action"] for c in ct2.open_commitments())
