@@ -204,3 +204,88 @@ def test_index_gpu_matches_cpu_reference():
         got_c = [r for r, _ in cpu.search("a", q, k=5)]
         got_g = [r for r, _ in gpu.search("a", q, k=5)]
         assert len(set(got_c) & set(got_g)) >= 3  # bf16 vs fp32 tie tolerance
+
+
+# ===========================================================================
+# Membrane depth: sensitivity ceilings, retrieve limits, decay pass,
+# configurator-key parity, stats counters
+# ===========================================================================
+
+def _eng(tmp_path, **cfg):
+    from vainplex_openclaw_amd.membrane.engine import MembraneEngine
+
+    t = [1_700_000_000.0]
+    e = MembraneEngine(str(tmp_path), config=cfg or None, clock=lambda: t[0])
+    return e, t
+
+
+def test_engine_sensitivity_ceiling(tmp_path):
+    e, _ = _eng(tmp_path, retrieve_max_sensitivity="medium")
+    e.remember("a", "public fact about storage", sensitivity="low")
+    e.remember("a", "medium fact about storage", sensitivity="medium")
+    e.remember("a", "secret fact about storage", sensitivity="high")
+    out = e.retrieve("a", "fact about storage", limit=10, min_salience=0.0)
+    texts = [r["record"]["text"] for r in out]
+    assert any("public" in t for t in texts)
+    assert any("medium" in t for t in texts)
+    assert not any("secret" in t for t in texts)
+    # per-call override widens the ceiling
+    out2 = e.retrieve("a", "fact about storage", limit=10, min_salience=0.0,
+                      max_sensitivity="high")
+    assert any("secret" in r["record"]["text"] for r in out2)
+
+
+def test_engine_retrieve_limit_and_score_order(tmp_path):
+    e, _ = _eng(tmp_path)
+    for i in range(6):
+        e.remember("a", f"note number {i} about the deploy pipeline")
+    out = e.retrieve("a", "deploy pipeline", limit=3, min_salience=0.0)
+    assert len(out) == 3
+    scores = [r["score"] for r in out]
+    assert scores == sorted(scores, reverse=True)
+    for r in out:
+        assert abs(r["score"] - r["cosine"] * r["salience"]) < 1e-9
+
+
+def test_engine_decay_pass_prunes(tmp_path):
+    e, t = _eng(tmp_path)
+    e.remember("a", "ephemeral memory one")
+    e.flush_buffer()
+    t[0] += 400 * 24 * 3600       # far past many half-lives
+    n = e.decay_pass("a")
+    assert n >= 1
+    assert e.store.count("a") == 0
+
+
+def test_engine_stats_counters(tmp_path):
+    e, _ = _eng(tmp_path)
+    e.remember("a", "stat memory about things")
+    e.retrieve("a", "memory about things", limit=1, min_salience=0.0)
+    st = dict(e.stats)
+    assert st.get("remembered", st.get("records", 1)) >= 1
+    assert st["retrievals"] == 1 and st["retrieved"] >= 0
+
+
+def test_engine_agent_isolation_in_retrieve(tmp_path):
+    e, _ = _eng(tmp_path)
+    e.remember("alice", "alice private plan for launch")
+    e.remember("bob", "bob different note entirely")
+    out = e.retrieve("bob", "private plan for launch", limit=5, min_salience=0.0)
+    assert all("alice" not in r["record"]["text"] for r in out)
+
+
+def test_configurator_membrane_keys_match_defaults():
+    """brainplex configurator's Membrane keys (configurator.ts:137-148)
+    line up with the engine's accepted config."""
+    from vainplex_openclaw_amd.brainplex.configurator import generate_configs
+    from vainplex_openclaw_amd.membrane.engine import MembraneEngine
+
+    cfgs = generate_configs(["main"], "UTC", full=True)
+    mem = next(c for c in cfgs if "membrane" in c["pluginId"])
+    keys = set(mem["config"])
+    assert {"buffer_size", "default_sensitivity", "retrieve_limit",
+            "retrieve_min_salience", "retrieve_max_sensitivity"} <= keys
+    # the engine accepts exactly these keys without error
+    import tempfile
+
+    MembraneEngine(tempfile.mkdtemp(), config=mem["config"])
