@@ -260,3 +260,115 @@ def test_resolve_config_merges_collectors(tmp_path):
     assert cfg["collectors"]["goals"]["enabled"] is True
     assert cfg["collectors"]["systemd_timers"]["enabled"] is True
     assert cfg["intervalMinutes"] == 5
+
+
+# ===========================================================================
+# Anomaly-detector depth: growth slope, declining metric, failure
+# correlation window/threshold, bootstrap integrity, metric history
+# ===========================================================================
+
+def _det(tmp_path):
+    from vainplex_openclaw_amd.leuko.anomaly import AnomalyDetector, MetricHistory
+
+    t = [1_700_000_000.0]
+    h = MetricHistory(str(tmp_path / "hist.json"), clock=lambda: t[0])
+    return AnomalyDetector(h, clock=lambda: t[0]), h, t
+
+
+def test_metric_history_roundtrip_and_cap(tmp_path):
+    from vainplex_openclaw_amd.leuko.anomaly import MetricHistory
+
+    t = [0.0]
+    h = MetricHistory(str(tmp_path / "m.json"), max_points=5, clock=lambda: t[0])
+    for i in range(8):
+        t[0] += 60
+        h.record("x", float(i))
+    assert len(h.series("x")) == 5           # capped
+    h2 = MetricHistory(str(tmp_path / "m.json"), max_points=5, clock=lambda: t[0])
+    assert len(h2.series("x")) == 5          # persisted (cap applied on load)
+    h3 = MetricHistory(str(tmp_path / "m.json"), clock=lambda: t[0])
+    assert len(h3.series("x")) == 8          # full JSONL kept on disk
+
+
+def test_slope_per_hour_math(tmp_path):
+    from vainplex_openclaw_amd.leuko.anomaly import MetricHistory
+
+    t = [0.0]
+    h = MetricHistory(str(tmp_path / "m.json"), clock=lambda: t[0])
+    for i in range(10):
+        h.record("lin", 100.0 * i, ts=i * 3600.0)  # +100/hour
+    slope = h.slope_per_hour("lin")
+    assert slope is not None and abs(slope - 100.0) < 1e-6
+    assert h.slope_per_hour("missing") is None
+
+
+def test_declining_metric_thresholds(tmp_path):
+    det, h, t = _det(tmp_path)
+    # decline of 50/h, threshold 10/h -> warn
+    for i in range(10):
+        t[0] = i * 3600.0
+        it = det.check_declining_metric("throughput", 1000.0 - 50.0 * i, 10.0)
+    assert it is not None and "declining" in it["title"]
+    # rising metric never warns
+    for i in range(10):
+        t[0] = 100_000 + i * 3600.0
+        it2 = det.check_declining_metric("rising", 10.0 * i, 10.0)
+    assert it2 is None
+
+
+def test_failure_correlation_window_and_threshold(tmp_path):
+    from vainplex_openclaw_amd.eventstore import EventJournal
+
+    det, _h, t = _det(tmp_path)
+    j = EventJournal(durable=False, clock=lambda: t[0])
+    t[0] = 1_700_000_000.0
+    for i, agent in enumerate(("a", "b", "a")):
+        j.publish(f"s.{agent}", {"ts": t[0] * 1000, "canonicalType": "tool.call.failed",
+                                 "agent": agent})
+    it = det.correlate_failures(j, window_s=600, threshold=3)
+    assert it is not None and it["severity"] == "critical"
+    assert "2 agent(s)" in it["title"]
+    # below threshold -> None
+    j2 = EventJournal(durable=False, clock=lambda: t[0])
+    j2.publish("s.a", {"ts": t[0] * 1000, "canonicalType": "tool.call.failed", "agent": "a"})
+    assert det.correlate_failures(j2, window_s=600, threshold=3) is None
+    # outside the window -> None
+    j3 = EventJournal(durable=False, clock=lambda: t[0])
+    for i in range(3):
+        j3.publish("s.a", {"ts": (t[0] - 1200) * 1000,
+                           "canonicalType": "tool.call.failed", "agent": "a"})
+    assert det.correlate_failures(j3, window_s=600, threshold=3) is None
+    assert det.correlate_failures(None) is None
+
+
+def test_bootstrap_integrity_matrix(tmp_path):
+    import os
+
+    det, _h, _t = _det(tmp_path)
+    ws = tmp_path / "ws"
+    (ws / "governance").mkdir(parents=True)
+    (ws / "memory" / "reboot").mkdir(parents=True)
+    # missing files = first boot = fine
+    assert det.check_bootstrap_integrity(str(ws)) == []
+    # valid file = fine
+    (ws / "governance" / "trust.json").write_text('{"agents": {}}')
+    assert det.check_bootstrap_integrity(str(ws)) == []
+    # corrupt file = critical
+    (ws / "memory" / "reboot" / "threads.json").write_text("{broken")
+    items = det.check_bootstrap_integrity(str(ws))
+    assert len(items) == 1 and items[0]["severity"] == "critical"
+    assert "threads.json" in items[0]["title"]
+
+
+def test_directory_growth_warn(tmp_path):
+    det, h, t = _det(tmp_path)
+    d = tmp_path / "grow"
+    d.mkdir()
+    assert det.check_directory_growth(str(d), warn_mb_per_hour=100.0) is None
+    (d / "blob1.bin").write_bytes(b"x" * 2_000_000)   # +2 MB
+    t[0] += 36.0                                       # 0.01 h
+    assert det.check_directory_growth(str(d), warn_mb_per_hour=100.0) is None  # 2 pts: no slope yet
+    (d / "blob2.bin").write_bytes(b"x" * 2_000_000)   # +2 MB more
+    t[0] += 36.0                                       # ~200 MB/h over 3 points
+    it = det.check_directory_growth(str(d), warn_mb_per_hour=100.0)
+    assert it is not None and "growing" in it["title"]
