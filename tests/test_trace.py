@@ -612,3 +612,113 @@ def test_state_claim_backed_by_tool_result_passes():
     ]
     chains = reconstruct_chains(events)
     assert not detect_all_signals(chains, ["unverified_claim"])
+
+
+# -- trace-to-facts-bridge.test.ts file-level depth -------------------------
+
+def _report(findings):
+    return {"findings": findings}
+
+
+def _finding(subject="svc", predicate="status", value="running",
+             conf=0.8, sig="hallucination"):
+    return {"id": "f1", "signalType": sig, "confidence": conf,
+            "evidence": {"subject": subject, "predicate": predicate,
+                         "value": value}}
+
+
+def test_bridge_file_formats(tmp_path):
+    import json
+
+    from vainplex_openclaw_amd.governance.trace_to_facts import extract_facts_from_file
+
+    p = tmp_path / "r.json"
+    p.write_text(json.dumps(_report([_finding()])))
+    assert extract_facts_from_file(str(p))[0]["subject"] == "svc"
+    # bare array
+    p2 = tmp_path / "arr.json"
+    p2.write_text(json.dumps([_finding(subject="x")]))
+    assert extract_facts_from_file(str(p2))[0]["subject"] == "x"
+    # missing / invalid / wrong shape
+    assert extract_facts_from_file(str(tmp_path / "ghost.json")) == []
+    bad = tmp_path / "bad.json"
+    bad.write_text("{nope")
+    assert extract_facts_from_file(str(bad)) == []
+    odd = tmp_path / "odd.json"
+    odd.write_text('{"not_findings": 1}')
+    assert extract_facts_from_file(str(odd)) == []
+
+
+def test_bridge_confidence_and_signal_filter(tmp_path):
+    import json
+
+    from vainplex_openclaw_amd.governance.trace_to_facts import extract_facts_from_file
+
+    p = tmp_path / "r.json"
+    p.write_text(json.dumps(_report([
+        _finding(conf=0.3),                         # below threshold
+        _finding(sig="doom_loop"),                  # wrong signal type
+        _finding(subject=None),                     # no factCorrection payload
+        _finding(subject="keeper"),
+    ])))
+    facts = extract_facts_from_file(str(p))
+    assert [f["subject"] for f in facts] == ["keeper"]
+
+
+def test_bridge_merge_dedupe_later_wins():
+    from vainplex_openclaw_amd.governance.trace_to_facts import merge_facts
+
+    existing = [{"subject": "svc", "predicate": "status", "value": "old"}]
+    new = [{"subject": "SVC", "predicate": "Status", "value": "new"},
+           {"subject": "svc", "predicate": "port", "value": "443"}]
+    out = merge_facts(existing, new)
+    assert len(out) == 2                          # case-insensitive dedupe
+    by_pred = {f["predicate"].lower(): f for f in out}
+    assert by_pred["status"]["value"] == "new"    # later wins
+    assert by_pred["port"]["value"] == "443"
+
+
+def test_bridge_run_write_and_merge(tmp_path):
+    import json
+
+    from vainplex_openclaw_amd.governance.trace_to_facts import run_bridge
+
+    r1 = tmp_path / "r1.json"
+    r1.write_text(json.dumps(_report([_finding(subject="a", value="1")])))
+    r2 = tmp_path / "r2.json"
+    r2.write_text(json.dumps(_report([_finding(subject="b", value="2")])))
+    out = tmp_path / "deep" / "facts.json"        # dir created on demand
+    n = run_bridge([str(r1), str(r2)], str(out))
+    assert n == 2 and out.is_file()
+    data = json.loads(out.read_text())
+    assert {f["subject"] for f in data["facts"]} == {"a", "b"}
+    # second run merges + dedupes (same subject|predicate overrides)
+    r3 = tmp_path / "r3.json"
+    r3.write_text(json.dumps(_report([_finding(subject="a", value="9")])))
+    n2 = run_bridge([str(r3)], str(out))
+    assert n2 == 2
+    data2 = json.loads(out.read_text())
+    va = next(f["value"] for f in data2["facts"] if f["subject"] == "a")
+    assert va == "9"
+    # no new facts -> existing count returned
+    empty = tmp_path / "empty.json"
+    empty.write_text(json.dumps(_report([])))
+    assert run_bridge([str(empty)], str(out)) == 2
+
+
+def test_bridge_timer_idempotent():
+    import time as _time
+
+    from vainplex_openclaw_amd.governance.trace_to_facts import BridgeTimer
+
+    hits = []
+    t = BridgeTimer(0.02, lambda: hits.append(1))
+    t.start()
+    t.start()                                     # idempotent
+    _time.sleep(0.08)
+    t.stop()
+    t.stop()                                      # safe twice
+    n = len(hits)
+    assert n >= 1
+    _time.sleep(0.05)
+    assert len(hits) == n                         # actually stopped
