@@ -563,3 +563,113 @@ def test_make_call_llm_factory():
     assert body["model"] == "mistral:7b"
     assert body["messages"][0]["content"] == "analyze this"
     assert headers == {"Authorization": "Bearer k"} and timeout_s == 5.0
+
+
+# ===========================================================================
+# approval-2fa.test.ts depth: batching windows, per-session batches,
+# cooldown status, notification contents, has_pending edge cases
+# ===========================================================================
+
+def _fa(**kw):
+    from vainplex_openclaw_amd.governance.approval_2fa import Approval2FA, generate_secret
+
+    t = [1_700_000_000.0]
+    notes = []
+    fa = Approval2FA(secret=generate_secret(), notify=notes.append,
+                     clock=lambda: t[0], **kw)
+    return fa, t, notes
+
+
+def _code(fa, t):
+    from vainplex_openclaw_amd.governance.approval_2fa import totp_at
+
+    return totp_at(fa.secret, t[0])
+
+
+def test_2fa_same_session_batches_within_debounce():
+    fa, t, notes = _fa()
+    r1 = fa.request("s1", "a", "cmd one")
+    t[0] += 1.0
+    r2 = fa.request("s1", "a", "cmd two")
+    assert len(notes) == 1                       # one batch, one notification
+    batch = notes[0]
+    assert len(batch["requests"]) == 2
+    out = fa.try_resolve_any(_code(fa, t))
+    assert {r["status"] for r in out} == {"approved"}
+    assert r1["status"] == "approved" and r2["status"] == "approved"
+
+
+def test_2fa_different_sessions_separate_batches():
+    fa, t, notes = _fa()
+    fa.request("s1", "a", "one")
+    fa.request("s2", "b", "two")
+    assert len(notes) == 2
+    assert notes[0]["sessionKey"] == "s1" and notes[1]["sessionKey"] == "s2"
+
+
+def test_2fa_notification_includes_all_commands():
+    fa, t, notes = _fa()
+    fa.request("s1", "a", "deploy the service", details={"command": "x" * 50})
+    t[0] += 0.5
+    fa.request("s1", "a", "restart the db")
+    reasons = [r["reason"] for r in notes[0]["requests"]]
+    assert "deploy the service" in reasons and "restart the db" in reasons
+
+
+def test_2fa_wrong_code_attempts_then_cooldown():
+    fa, t, _ = _fa(max_attempts=3)
+    fa.request("s1", "a", "cmd")
+    for _i in range(3):
+        assert fa.try_resolve_any("000000") == []
+    assert fa.in_attempt_cooldown()
+    # during cooldown even the RIGHT code is refused
+    assert fa.try_resolve_any(_code(fa, t)) == []
+    # cooldown expires
+    t[0] += fa.cooldown_s + 1
+    fa.request("s1b", "a", "cmd2")
+    assert fa.try_resolve_any(_code(fa, t)) != []
+
+
+def test_2fa_no_pending_and_has_pending():
+    fa, t, _ = _fa()
+    assert fa.try_resolve_any(_code(fa, t)) == []   # nothing pending
+    assert not fa.has_pending_batch()
+    t[0] += 31  # the empty resolve consumed this TOTP counter (replay set)
+    fa.request("s1", "a", "cmd")
+    assert fa.has_pending_batch() and fa.has_pending_batch("s1")
+    assert not fa.has_pending_batch("other-session")
+    fa.try_resolve_any(_code(fa, t))
+    assert not fa.has_pending_batch()
+
+
+def test_2fa_session_auto_approval_window():
+    fa, t, _ = _fa()
+    fa.request("s1", "a", "first")
+    fa.try_resolve_any(_code(fa, t))
+    # within 10 min: auto-approved without a new batch
+    t[0] += 300
+    r = fa.request("s1", "a", "second")
+    assert r["status"] == "approved" and r.get("auto")
+    # other sessions do not inherit the approval
+    r2 = fa.request("s-other", "a", "third")
+    assert r2["status"] == "pending"
+
+
+def test_2fa_timeout_expires_batches():
+    fa, t, _ = _fa()
+    r = fa.request("s1", "a", "cmd")
+    t[0] += fa.timeout_s + 1
+    expired = fa.expire_stale()
+    assert expired and r["status"] in ("expired", "denied")
+    assert not fa.has_pending_batch()
+
+
+def test_2fa_replay_protection_across_batches():
+    fa, t, _ = _fa()
+    fa.request("s1", "a", "one")
+    code = _code(fa, t)
+    assert fa.try_resolve_any(code)
+    # bypass the auto-approval window to force a real second batch
+    fa._session_approved_until.clear()
+    fa.request("s1", "a", "two")
+    assert fa.try_resolve_any(code) == []      # same counter refused
