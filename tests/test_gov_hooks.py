@@ -673,3 +673,129 @@ def test_2fa_replay_protection_across_batches():
     fa._session_approved_until.clear()
     fa.request("s1", "a", "two")
     assert fa.try_resolve_any(code) == []      # same counter refused
+
+
+# ===========================================================================
+# agentproof-rest.test.ts depth: API-key lifecycle, URL shapes, error
+# nulls, ring overflow, breaker timing, requeue-on-failure
+# ===========================================================================
+
+def _ap(tmp_path=None, key=None, **kw):
+    from vainplex_openclaw_amd.governance.security.agentproof import AgentProofRestClient
+
+    t = [1_700_000_000.0]
+    keyfile = None
+    if tmp_path is not None and key is not None:
+        keyfile = str(tmp_path / "key.txt")
+        (tmp_path / "key.txt").write_text(key)
+    cli = AgentProofRestClient(api_key_file=keyfile, clock=lambda: t[0], **kw)
+    return cli, t
+
+
+def test_agentproof_key_read_cache_and_trim(tmp_path):
+    reads = []
+    cli, _ = _ap(tmp_path, key="  secret-key \n")
+    real_open = open
+
+    assert cli._headers() == {"X-API-Key": "secret-key"}   # trimmed
+    # cached: mutate the file, header unchanged
+    (tmp_path / "key.txt").write_text("changed")
+    assert cli._headers() == {"X-API-Key": "secret-key"}
+
+
+def test_agentproof_missing_and_empty_key_unauthenticated(tmp_path):
+    cli, _ = _ap()                                          # no file
+    assert cli._headers() == {}
+    cli2, _ = _ap(tmp_path, key="   \n")                    # empty file
+    assert cli2._headers() == {}
+
+
+def test_agentproof_base_url_trailing_slash():
+    from vainplex_openclaw_amd.governance.security.agentproof import AgentProofRestClient
+
+    urls = []
+
+    def get(url, headers):
+        urls.append(url)
+        return {"agentId": "a", "score": 50}
+
+    cli = AgentProofRestClient(base_url="https://ap.example/", http_get=get)
+    cli.lookup_reputation("a")
+    assert urls[0].startswith("https://ap.example/v1/")
+    assert "//v1" not in urls[0].replace("https://", "")
+
+
+def test_agentproof_error_and_network_nulls():
+    from vainplex_openclaw_amd.governance.security.agentproof import AgentProofRestClient
+
+    def boom(url, headers):
+        raise OSError("network down")
+
+    cli = AgentProofRestClient(http_get=boom)
+    assert cli.lookup_reputation("a") is None
+    assert AgentProofRestClient().lookup_reputation("a") is None  # no transport
+
+
+def test_agentproof_ring_overflow_drops_oldest():
+    from vainplex_openclaw_amd.governance.security.agentproof import (
+        QUEUE_CAPACITY,
+        AgentProofRestClient,
+    )
+
+    cli = AgentProofRestClient()
+    for i in range(QUEUE_CAPACITY + 5):
+        cli.enqueue_signal(f"a{i}", "success")
+    assert cli.queue_depth == QUEUE_CAPACITY
+    assert cli.dropped == 5
+    with cli._lock:
+        first = cli._queue[0]["agentId"]
+    assert first == "a5"                      # oldest dropped
+
+
+def test_agentproof_breaker_opens_and_resets():
+    from vainplex_openclaw_amd.governance.security.agentproof import (
+        BREAKER_RESET_S,
+        BREAKER_THRESHOLD,
+        AgentProofRestClient,
+    )
+
+    calls = []
+    t = [1_700_000_000.0]
+
+    def flaky_get(url, headers):
+        calls.append(url)
+        raise OSError("500")
+
+    cli = AgentProofRestClient(http_get=flaky_get, clock=lambda: t[0])
+    for _ in range(BREAKER_THRESHOLD):
+        cli.lookup_reputation("a")
+    assert cli.breaker.is_open
+    n = len(calls)
+    cli.lookup_reputation("a")                # breaker short-circuits
+    assert len(calls) == n
+    t[0] += BREAKER_RESET_S + 1               # circuit half-opens
+    cli.lookup_reputation("a")
+    assert len(calls) == n + 1
+
+
+def test_agentproof_flush_requeues_on_failure():
+    from vainplex_openclaw_amd.governance.security.agentproof import AgentProofRestClient
+
+    posts = []
+
+    def failing_post(url, headers, body):
+        posts.append(body)
+        raise OSError("temporarily down")
+
+    cli = AgentProofRestClient(http_post=failing_post)
+    cli.enqueue_signal("a", "success")
+    cli.enqueue_signal("b", "policyBlock")
+    assert cli.flush() == 0
+    assert cli.queue_depth == 2             # requeued for retry
+    # transport recovers
+    ok = []
+    cli.http_post = lambda u, h, b: ok.append(b)
+    # breaker may be counting failures but not open after one miss
+    assert cli.flush() == 2
+    assert cli.queue_depth == 0
+    assert {s["agentId"] for s in ok[0]["signals"]} == {"a", "b"}
