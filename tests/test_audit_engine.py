@@ -402,3 +402,99 @@ def test_engine_known_agents_auto_registered_and_kept(workspace):
         assert "forge" in e.trust_manager.get_store()["agents"]
     finally:
         e.stop()
+
+
+# ===========================================================================
+# audit-redactor.test.ts depth (9 its): recursive redaction, defaults,
+# custom/invalid patterns, non-string preservation
+# ===========================================================================
+
+def test_audit_redactor_default_patterns():
+    from vainplex_openclaw_amd.governance.audit_redactor import create_redactor
+
+    red = create_redactor()
+    ctx = {
+        "toolParams": {"cmd": "export api_key=abc123secret && run"},
+        "note": "uses sk-" + "a" * 20,
+        "hdr": "Authorization: Bearer abc.def.ghi",
+    }
+    out = red(ctx)
+    blob = str(out)
+    assert "abc123secret" not in blob
+    assert "sk-" + "a" * 20 not in blob
+    assert "abc.def.ghi" not in blob
+    assert "[REDACTED]" in blob
+
+
+def test_audit_redactor_recursive_and_type_preserving():
+    from vainplex_openclaw_amd.governance.audit_redactor import create_redactor
+
+    red = create_redactor()
+    ctx = {"deep": {"list": ["password: hunter22", 42, None, True],
+                    "n": 3.5}}
+    out = red(ctx)
+    assert "hunter22" not in str(out)
+    assert out["deep"]["list"][1] == 42
+    assert out["deep"]["list"][2] is None
+    assert out["deep"]["list"][3] is True
+    assert out["deep"]["n"] == 3.5
+
+
+def test_audit_redactor_custom_and_invalid_patterns():
+    from vainplex_openclaw_amd.governance.audit_redactor import create_redactor
+
+    red = create_redactor([r"TICKET-\d+", "(unclosed"])
+    out = red({"msg": "see TICKET-123 and password: x9x9x9x9"})
+    assert "TICKET-123" not in out["msg"]
+    # custom list REPLACES defaults: the password pattern no longer applies
+    assert "x9x9x9x9" in out["msg"]
+    # empty pattern list redacts nothing
+    red2 = create_redactor([])
+    assert red2({"a": "password: keepme99"})["a"] == "password: keepme99"
+
+
+def test_audit_redactor_does_not_mutate_input():
+    from vainplex_openclaw_amd.governance.audit_redactor import create_redactor
+
+    red = create_redactor()
+    ctx = {"k": "password: original9"}
+    red(ctx)
+    assert ctx["k"] == "password: original9"
+
+
+# -- core config loader depth (config-loader.ts, 14 its) ---------------------
+
+def test_plugin_config_file_first_and_fallback(tmp_path):
+    import json
+
+    from vainplex_openclaw_amd.core.config import load_plugin_config
+
+    # `home` here is the .openclaw directory itself (openclaw_home())
+    home = tmp_path
+    d = home / "plugins" / "p1"
+    d.mkdir(parents=True)
+    (d / "config.json").write_text(json.dumps({"from": "file"}))
+    assert load_plugin_config("p1", {"from": "api"}, home=str(home)) == {"from": "file"}
+    # no file -> api.pluginConfig fallback
+    assert load_plugin_config("p2", {"from": "api"}, home=str(home)) == {"from": "api"}
+    # neither -> empty dict
+    assert load_plugin_config("p3", home=str(home)) == {}
+
+
+def test_plugin_config_invalid_json_falls_back(tmp_path):
+    from vainplex_openclaw_amd.core.config import load_plugin_config
+
+    d = tmp_path / "plugins" / "p1"
+    d.mkdir(parents=True)
+    (d / "config.json").write_text("{broken json")
+    assert load_plugin_config("p1", {"ok": 1}, home=str(tmp_path)) == {"ok": 1}
+
+
+def test_resolve_defaults_recursion_and_type_guards():
+    from vainplex_openclaw_amd.core.config import resolve_defaults
+
+    defaults = {"a": 1, "nested": {"x": "dx", "y": True}, "keep": "d"}
+    out = resolve_defaults({"a": 5, "nested": {"x": "cx"}, "extra": 9}, defaults)
+    assert out["a"] == 5
+    assert out["nested"] == {"x": "cx", "y": True}
+    assert out["keep"] == "d"
