@@ -92,8 +92,13 @@ def load_plugin_config(
     fallback: Optional[Dict[str, Any]] = None,
     home: Optional[str] = None,
 ) -> Dict[str, Any]:
-    """External file first, api.pluginConfig fallback (index.ts:73-76)."""
-    data = load_json_file(plugin_config_path(plugin_id, home))
+    """External file first, api.pluginConfig fallback (index.ts:73-76).
+    A corrupt external file falls back instead of breaking plugin
+    registration (config-loader.ts handles parse errors gracefully)."""
+    try:
+        data = load_json_file(plugin_config_path(plugin_id, home))
+    except Exception:
+        data = None
     if data is not None:
         return data
     return dict(fallback or {})
