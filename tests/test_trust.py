@@ -280,3 +280,96 @@ def test_age_days_refreshed_on_load(workspace):
     tm = TrustManager(TrustConfig(), workspace, clock=lambda: t[0])
     tm.load()
     assert tm.get_store()["agents"]["aged"]["signals"]["ageDays"] == 10
+
+
+# ===========================================================================
+# cross-agent.test.ts depth (13 its)
+# ===========================================================================
+
+def _ca(workspace):
+    from vainplex_openclaw_amd.governance.cross_agent import CrossAgentManager
+    from vainplex_openclaw_amd.governance.trust import TrustConfig, TrustManager
+
+    tm = TrustManager(TrustConfig.from_dict(None), workspace)
+    tm.load()
+    return CrossAgentManager(tm), tm
+
+
+PARENT = "agent:main"
+CHILD = "agent:main:subagent:forge:abc"
+
+
+def test_cross_agent_relationship_lifecycle(workspace):
+    ca, _ = _ca(workspace)
+    ca.register_relationship(PARENT, CHILD)
+    p = ca.get_parent(CHILD)
+    assert p and p["parentAgentId"] == "main"
+    assert [c["childAgentId"] for c in ca.get_children(PARENT)] == ["forge"]
+    ca.remove_relationship(CHILD)
+    assert ca.get_children(PARENT) == []
+
+
+def test_cross_agent_root_has_no_parent(workspace):
+    ca, _ = _ca(workspace)
+    assert ca.get_parent(PARENT) is None
+
+
+def test_cross_agent_enrich_sub_vs_root(workspace):
+    ca, _ = _ca(workspace)
+    ca.register_relationship(PARENT, CHILD)
+    sub = ca.enrich_context({"sessionKey": CHILD, "agentId": "forge"})
+    assert sub.get("crossAgent", {}).get("parentAgentId") == "main"
+    root = ca.enrich_context({"sessionKey": PARENT, "agentId": "main"})
+    assert "crossAgent" not in root or not root["crossAgent"].get("parentAgentId")
+
+
+def test_cross_agent_trust_ceiling_caps_child_only(workspace):
+    ca, tm = _ca(workspace)
+    tm.set_score("main", 55.0)
+    tm.set_score("forge", 90.0)
+    ca.register_relationship(PARENT, CHILD)
+    assert ca.compute_trust_ceiling(CHILD) <= 55.0
+    # root agents are uncapped
+    assert ca.compute_trust_ceiling(PARENT) >= 90.0 or \
+        ca.compute_trust_ceiling(PARENT) == tm.get("main")["score"]
+
+
+def test_cross_agent_policy_cascade(workspace):
+    from vainplex_openclaw_amd.governance.policies import build_policy_index
+
+    ca, _ = _ca(workspace)
+    ca.register_relationship(PARENT, CHILD)
+    idx = build_policy_index({"policies": [
+        {"id": "parent-only", "scope": {"agents": ["main"], "hooks": ["before_tool_call"]},
+         "rules": [{"id": "r", "conditions": [], "effect": {"action": "deny"}}]},
+        {"id": "global", "scope": {"hooks": ["before_tool_call"]},
+         "rules": [{"id": "r", "conditions": [], "effect": {"action": "audit"}}]},
+        {"id": "other-agent", "scope": {"agents": ["zeta"], "hooks": ["before_tool_call"]},
+         "rules": [{"id": "r", "conditions": [], "effect": {"action": "deny"}}]},
+    ]})
+    # sub-agent inherits parent-scoped + global, not other agents'
+    sub_pols = {p["id"] for p in ca.resolve_effective_policies(
+        {"sessionKey": CHILD, "agentId": "forge", "hook": "before_tool_call"}, idx)}
+    assert "parent-only" in sub_pols and "global" in sub_pols
+    assert "other-agent" not in sub_pols
+    # root gets only own + global
+    root_pols = {p["id"] for p in ca.resolve_effective_policies(
+        {"sessionKey": "agent:zeta", "agentId": "zeta", "hook": "before_tool_call"}, idx)}
+    assert root_pols == {"other-agent", "global"}
+
+
+def test_cross_agent_implicit_subagent_detection(workspace):
+    ca, _ = _ca(workspace)
+    # no explicit registration: the session key alone identifies the parent
+    ctx = ca.enrich_context({"sessionKey": CHILD, "agentId": "forge"})
+    assert ctx.get("crossAgent", {}).get("parentAgentId") == "main"
+
+
+def test_cross_agent_graph_summary(workspace):
+    ca, _ = _ca(workspace)
+    ca.register_relationship(PARENT, CHILD)
+    ca.register_relationship(PARENT, "agent:main:subagent:scout:xyz")
+    g = ca.graph_summary()
+    assert len(g["relationships"]) == 2
+    kids = {r["childAgentId"] for r in g["relationships"]}
+    assert kids == {"forge", "scout"}
