@@ -43,6 +43,7 @@ __global__ void topk_scan_fp4_v2_kernel(const uint8_t*, const uint8_t*, const ui
                                         const uint8_t*, int, int, int, int, int,
                                         float*, int32_t*, const float*, int32_t*, int);
 
+
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
                                   float*, int32_t*);
 __global__ void firewall_verdict_kernel(const unsigned long long*, const unsigned long long*,
