@@ -292,3 +292,107 @@ def test_controls_propagate_into_matches():
                "rules": [{"id": "r", "conditions": [], "effect": {"action": "audit"}}]}
     res2 = ev.evaluate(ctx_base(), [without], {"level": "low", "score": 0})
     assert res2["matches"][0]["controls"] == []
+
+
+# ===========================================================================
+# policy-loader.test.ts depth: file loading, disabled filter, merge,
+# hook/agent indexing, regex cache robustness
+# ===========================================================================
+
+def _pol(pid, hooks=None, agents=None, enabled=True, action="deny", pattern=None):
+    cond = []
+    if pattern is not None:
+        cond = [{"type": "tool", "params": {"cmd": {"matches": pattern}}}]
+    p = {"id": pid, "enabled": enabled,
+         "rules": [{"id": "r", "conditions": cond, "effect": {"action": action}}]}
+    scope = {}
+    if hooks is not None:
+        scope["hooks"] = hooks
+    if agents is not None:
+        scope["agents"] = agents
+    if scope:
+        p["scope"] = scope
+    return p
+
+
+def test_loader_policy_dir_files(tmp_path):
+    import json
+
+    from vainplex_openclaw_amd.governance.policies import load_policies
+
+    d = tmp_path / "policies"
+    d.mkdir()
+    (d / "one.json").write_text(json.dumps(_pol("from-file")))
+    (d / "many.json").write_text(json.dumps([_pol("list-a"), _pol("list-b")]))
+    (d / "junk.json").write_text("{broken")
+    (d / "noid.json").write_text(json.dumps({"rules": []}))
+    (d / "notes.txt").write_text("ignored")
+    try:
+        pols = load_policies({"policies": [_pol("inline")]}, policy_dir=str(d))
+    except Exception as exc:
+        import pytest as _pt
+
+        _pt.fail(f"loader must tolerate junk files: {exc}")
+    ids = {p["id"] for p in pols}
+    assert {"inline", "from-file", "list-a", "list-b"} <= ids
+    assert "noid" not in ids
+
+
+def test_loader_disabled_policies_filtered_at_index(tmp_path):
+    from vainplex_openclaw_amd.governance.policies import build_policy_index
+
+    idx = build_policy_index({"policies": [
+        _pol("on", hooks=["before_tool_call"]),
+        _pol("off", hooks=["before_tool_call"], enabled=False),
+    ]})
+    hooked = {p["id"] for p in idx.for_hook("before_tool_call")}
+    assert "on" in hooked and "off" not in hooked
+
+
+def test_index_by_hook_and_global(tmp_path):
+    from vainplex_openclaw_amd.governance.policies import build_policy_index
+
+    idx = build_policy_index({"policies": [
+        _pol("tool-only", hooks=["before_tool_call"]),
+        _pol("msg-only", hooks=["message_sending"]),
+        _pol("everywhere"),                       # no hooks = all hooks
+    ]})
+    t = {p["id"] for p in idx.for_hook("before_tool_call")}
+    m = {p["id"] for p in idx.for_hook("message_sending")}
+    assert t == {"tool-only", "everywhere"}
+    assert m == {"msg-only", "everywhere"}
+
+
+def test_index_regex_cache_and_invalid_regex(tmp_path):
+    from vainplex_openclaw_amd.governance.policies import build_policy_index
+
+    idx = build_policy_index({"policies": [
+        _pol("good", hooks=["before_tool_call"], pattern=r"rm\s+-rf"),
+        _pol("bad", hooks=["before_tool_call"], pattern="(unclosed"),
+    ]})
+    assert idx.regex_cache.get(r"rm\s+-rf") is not None
+    assert idx.regex_cache.get("(unclosed") is None  # cached as invalid, no crash
+
+
+# -- audit-trail query depth -------------------------------------------------
+
+def test_audit_query_filters(workspace):
+    from vainplex_openclaw_amd.governance.audit import AuditTrail
+
+    t = [1_700_000_000.0]
+    trail = AuditTrail({}, workspace, clock=lambda: t[0])
+    trail.load()
+    for i, (agent, verdict) in enumerate([("a1", "allow"), ("a1", "deny"),
+                                          ("a2", "allow"), ("a2", "2fa")]):
+        t[0] += 10
+        trail.record(verdict, "r", {"hook": "h", "agentId": agent}, {}, {}, [], 5)
+    trail.flush()
+    assert len(trail.query({"agentId": "a1"})) == 2
+    assert len(trail.query({"verdict": "allow"})) == 2
+    assert len(trail.query({"agentId": "a2", "verdict": "2fa"})) == 1
+    assert len(trail.query({"limit": 3})) == 3
+    # time-range filter
+    mid = 1_700_000_000.0 + 25
+    after = trail.query({"after": mid * 1000})
+    assert all(r["timestamp"] >= mid * 1000 for r in after)
+    assert len(trail.query({})) == 4

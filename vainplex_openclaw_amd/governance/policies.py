@@ -233,19 +233,24 @@ def load_policies(
     """User policies (inline config list + *.json files in policy_dir) plus
     builtins (policy-loader.ts)."""
     config = config or {}
+
+    def _usable(p: Any) -> bool:
+        # explicit enabled:false is filtered at load (policy-loader.ts:80-82)
+        return isinstance(p, dict) and bool(p.get("id")) and p.get("enabled") is not False
+
     policies: List[Dict[str, Any]] = []
     for p in config.get("policies", []):
-        if isinstance(p, dict) and p.get("id"):
+        if _usable(p):
             policies.append(p)
     if policy_dir and os.path.isdir(policy_dir):
         for name in sorted(os.listdir(policy_dir)):
             if not name.endswith(".json"):
                 continue
             data = read_json(os.path.join(policy_dir, name))
-            if isinstance(data, dict) and data.get("id"):
+            if isinstance(data, dict) and _usable(data):
                 policies.append(data)
             elif isinstance(data, list):
-                policies.extend(d for d in data if isinstance(d, dict) and d.get("id"))
+                policies.extend(d for d in data if _usable(d))
     policies.extend(get_builtin_policies(config.get("builtinPolicies")))
     return policies
 
