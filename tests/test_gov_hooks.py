@@ -799,3 +799,85 @@ def test_agentproof_flush_requeues_on_failure():
     assert cli.flush() == 2
     assert cli.queue_depth == 0
     assert {s["agentId"] for s in ok[0]["signals"]} == {"a", "b"}
+
+
+# ===========================================================================
+# Matrix-poller depth: code extraction across messages, pagination
+# token, non-message events, failure tolerance, end-to-end 2FA resolve
+# ===========================================================================
+
+def _poller(responses):
+    from vainplex_openclaw_amd.governance.approval_2fa import Approval2FA, generate_secret
+    from vainplex_openclaw_amd.governance.matrix_poller import MatrixPoller
+
+    t = [1_700_000_000.0]
+    fa = Approval2FA(secret=generate_secret(), clock=lambda: t[0])
+    calls = []
+
+    def http_get(url, headers):
+        calls.append((url, headers))
+        if isinstance(responses[0], Exception):
+            raise responses.pop(0)
+        return responses.pop(0) if len(responses) > 1 else responses[0]
+
+    return MatrixPoller(fa, homeserver="https://hs.example/", room_id="!r:x",
+                        access_token="tok", http_get=http_get), fa, t, calls
+
+
+def _msg(body):
+    return {"type": "m.room.message", "content": {"body": body}}
+
+
+def test_matrix_poll_resolves_pending_batch():
+    from vainplex_openclaw_amd.governance.approval_2fa import totp_at
+
+    poller, fa, t, calls = _poller([{"chunk": [], "end": "t1"}])
+    r = fa.request("s1", "a", "deploy")
+    code = totp_at(fa.secret, t[0])
+    poller.http_get = lambda u, h: {"chunk": [_msg(f"approve: {code}")], "end": "t2"}
+    found = poller.poll_once()
+    assert found == [code]
+    assert r["status"] == "approved"
+    assert poller.codes_seen == 1
+
+
+def test_matrix_poll_extracts_multiple_and_ignores_noise():
+    poller, _fa, _t, _ = _poller([{
+        "chunk": [
+            _msg("codes 123456 and 654321 here"),
+            {"type": "m.reaction", "content": {"body": "999999"}},  # not a message
+            _msg("no digits"),
+            _msg("1234567 is seven digits"),   # 7-digit run: no 6-digit match
+        ],
+        "end": "tkn",
+    }])
+    codes = poller.poll_once()
+    assert "123456" in codes and "654321" in codes
+    assert "999999" not in codes
+    assert len([c for c in codes if c == "234567"]) == 0
+
+
+def test_matrix_poll_pagination_and_auth_header():
+    poller, _fa, _t, calls = _poller([
+        {"chunk": [], "end": "page-1"},
+        {"chunk": [], "end": "page-2"},
+    ])
+    poller.poll_once()
+    poller.poll_once()
+    assert "from=page-1" in calls[1][0]
+    assert calls[0][1]["Authorization"] == "Bearer tok"
+    assert "/_matrix/client/v3/rooms/!r:x/messages" in calls[0][0]
+
+
+def test_matrix_poll_failures_tolerated():
+    poller, _fa, _t, _ = _poller([OSError("matrix down"), {"chunk": []}])
+    assert poller.poll_once() == []        # error swallowed
+    assert poller.poll_once() == []        # next poll works
+    # no transport configured -> no-op
+    from vainplex_openclaw_amd.governance.approval_2fa import Approval2FA
+    from vainplex_openclaw_amd.governance.matrix_poller import MatrixPoller
+
+    silent = MatrixPoller(Approval2FA(), http_get=None)
+    assert silent.poll_once() == []
+    silent.start()                          # refuses without transport
+    assert silent._thread is None
