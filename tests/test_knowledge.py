@@ -495,3 +495,94 @@ def test_fact_store_prune_least_relevant_first(workspace):
     assert len(fs.facts) == 3
     assert keep["id"] in fs.facts     # boosted fact survived
     assert cold["id"] not in fs.facts  # least relevant evicted
+
+
+# ===========================================================================
+# KE llm-enhancer depth: prompt shape, Ollama envelope parsing,
+# transforms (type/importance normalization), batch trigger
+# ===========================================================================
+
+def test_enhancer_prompt_shape():
+    from vainplex_openclaw_amd.knowledge.llm_enhancer import construct_prompt
+
+    p = construct_prompt(["Alice met Bob.", "Initech shipped v2."])
+    assert "Alice met Bob." in p and "Initech shipped v2." in p
+    assert '"entities"' in p and '"facts"' in p
+    assert "JSON" in p
+
+
+def test_enhancer_parse_envelope_and_bare():
+    import json as _json
+
+    from vainplex_openclaw_amd.knowledge.llm_enhancer import parse_llm_response
+
+    inner = {"entities": [{"type": "person", "value": "Alice"}], "facts": []}
+    # Ollama envelope
+    out = parse_llm_response(_json.dumps({"response": _json.dumps(inner)}))
+    assert out["entities"][0]["value"] == "Alice"
+    # bare object
+    out2 = parse_llm_response(_json.dumps(inner))
+    assert out2["entities"] and out2["facts"] == []
+    # non-list fields degrade to []
+    out3 = parse_llm_response(_json.dumps({"entities": "no", "facts": 4}))
+    assert out3 == {"entities": [], "facts": []}
+    import pytest as _pt
+
+    with _pt.raises(Exception):
+        parse_llm_response("[1,2]")
+
+
+def test_enhancer_transform_entities_normalization():
+    from vainplex_openclaw_amd.knowledge.llm_enhancer import transform_entities
+
+    raw = [
+        {"type": "Person", "value": "  Alice Johnson "},
+        {"type": "org", "value": "Initech", "importance": 3.5},   # clamped
+        {"type": "concept", "value": "caching", "importance": -1},
+        {"value": "no-type"}, "junk", {"type": "person", "value": 42},
+    ]
+    out = transform_entities(raw, clock=lambda: 1_700_000_000.0)
+    assert len(out) == 3
+    alice = out[0]
+    assert alice["id"] == "person:alice-johnson"
+    assert alice["type"] == "person" and alice["value"] == "Alice Johnson"
+    assert out[1]["importance"] == 1.0 and out[2]["importance"] == 0.0
+    assert all(e["source"] == ["llm"] and e["count"] == 1 for e in out)
+
+
+def test_enhancer_transform_facts_normalization():
+    from vainplex_openclaw_amd.knowledge.llm_enhancer import transform_facts
+
+    raw = [
+        {"subject": " Alice ", "predicate": "Works At", "object": " Initech "},
+        {"subject": "x", "predicate": "p"},          # incomplete
+        {"subject": 1, "predicate": "p", "object": "o"},
+        "junk",
+    ]
+    out = transform_facts(raw)
+    assert out == [{"subject": "Alice", "predicate": "works-at",
+                    "object": "Initech", "source": "extracted-llm"}]
+
+
+def test_enhancer_batch_triggers_on_size():
+    import json as _json
+
+    from vainplex_openclaw_amd.knowledge.llm_enhancer import LlmEnhancer
+
+    calls = []
+
+    def llm(prompt):
+        calls.append(prompt)
+        return _json.dumps({"entities": [{"type": "person", "value": "A"}],
+                            "facts": []})
+
+    results = []
+    enh = LlmEnhancer(call_llm=llm, batch_size=2, clock=lambda: 0.0)
+    enh.set_result_handler(results.append)
+    assert enh.add_to_batch("m1", "first text") is None
+    out = enh.add_to_batch("m2", "second text")
+    assert out is not None and len(calls) == 1
+    assert "first text" in calls[0] and "second text" in calls[0]
+    # disabled enhancer is inert
+    assert not LlmEnhancer(None).enabled
+    assert LlmEnhancer(None).add_to_batch("x", "y") is None
