@@ -109,3 +109,36 @@ def test_service_on_real_pipeline_gpu():
         assert inj["hits"]["injection"] != 0
     finally:
         svc.close()
+
+
+@pytest.mark.gpu
+def test_service_sync_check_latency_budget_gpu():
+    """The synchronous check() path (before_message_write contract) must
+    close a single-message micro-batch within the deadline budget: with
+    max_wait_ms=5 and a small warm pipeline, p95 end-to-end latency for
+    lone messages stays under 150 ms (deadline + step + slack) and the
+    deadline trigger actually fires (latency >> instant, >= the wait)."""
+    import time as _time
+
+    from vainplex_openclaw_amd.pipeline.engine import FirewallPipeline, PipelineConfig
+
+    cfg = PipelineConfig(batch=256, index_size=65536, topk=8,
+                         recall_mode="direct", recall_fp8=False, recall_fp4=False)
+    pipe = FirewallPipeline(cfg, device="cuda:0")
+    svc = FirewallService(pipe, max_batch=256, max_wait_ms=5.0)
+    try:
+        svc.check(b"warmup message one")          # warm kernels/caches
+        lats = []
+        for i in range(12):
+            t0 = _time.perf_counter()
+            out = svc.check(f"lone message {i} about the deploy".encode())
+            lats.append((_time.perf_counter() - t0) * 1000)
+            assert out["verdict"] in ("allow", "audit", "2fa", "deny")
+        lats.sort()
+        p95 = lats[int(len(lats) * 0.95) - 1]
+        assert p95 < 150.0, lats
+        # the micro-batch deadline is respected (not closing instantly
+        # on partial batches, not hanging either)
+        assert lats[0] >= 2.0, lats
+    finally:
+        svc.close()
