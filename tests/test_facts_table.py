@@ -215,7 +215,7 @@ def test_fact_probe_reference_semantics():
     from vainplex_openclaw_amd.ops import gpu as g
     from vainplex_openclaw_amd.ops import pattern_sets as ps
 
-    facts = [("nginx-service", "status", "running"),
+    facts = [("nginx-service", "state", "running"),
              ("backup.db", "exists", "replica")]
     msgs = [
         b"the nginx-service is running fine",     # verified
@@ -232,14 +232,15 @@ def test_fact_probe_reference_semantics():
 def test_fact_table_build_and_collisions():
     from vainplex_openclaw_amd.ops import gpu as g
 
-    facts = [(f"svc-{i}", "status", "running") for i in range(40)]
+    facts = [(f"svc-{i}", "state", "running") for i in range(40)]
     tk, tv, ph, pw = g.build_fact_table(facts)
     keys = tk.numpy().view("uint64")
     assert (keys != 0).sum() == 40           # all inserted
     assert 1 << pw >= 80                      # >= 2x occupancy headroom
     # predicate hash vector has entries only on claim bits
     pred = ph.numpy().view("uint64")
-    assert pred[0] != 0 and pred[10] != 0 and pred[11] == 0
+    assert pred[0] != 0 and pred[7] != 0
+    assert pred[1] == 0 and pred[8] == 0  # subject-only strategies skipped
 
 
 def test_fact_probe_tokenizer_matches_kernel_classes():

@@ -264,3 +264,120 @@ def test_output_validation_end_to_end_with_audit(workspace):
         assert engine.validate_output("mystery-svc is running", "low")["verdict"] == "pass"
     finally:
         engine.stop()
+
+
+# ===========================================================================
+# integration.test.ts depth: deny-wins, tier gates, night mode flow,
+# fail-open engine errors, validation trust thresholds end to end
+# ===========================================================================
+
+def _engine(workspace, config=None):
+    from vainplex_openclaw_amd.governance.engine import GovernanceEngine
+
+    eng = GovernanceEngine(config or {}, workspace)
+    eng.start()
+    return eng
+
+
+def test_deny_wins_across_policies(workspace):
+    eng = _engine(workspace, {"policies": [
+        {"id": "p-allow", "scope": {"hooks": ["before_tool_call"]}, "priority": 10,
+         "rules": [{"id": "r", "conditions": [], "effect": {"action": "audit"}}]},
+        {"id": "p-deny", "scope": {"hooks": ["before_tool_call"]}, "priority": 1,
+         "rules": [{"id": "r", "conditions": [{"type": "tool", "name": "exec"}],
+                    "effect": {"action": "deny"}}]},
+    ]})
+    out = eng.evaluate(eng.build_context("before_tool_call", "a1", tool_name="exec"))
+    assert out["action"] == "deny"
+    out2 = eng.evaluate(eng.build_context("before_tool_call", "a1", tool_name="read"))
+    # audit aggregates to allow-with-logging (policy-evaluator.ts:44-78)
+    assert out2["action"] == "allow" and "audit logging" in out2["reason"]
+
+
+def test_trust_tier_gate_on_rule(workspace):
+    eng = _engine(workspace, {"policies": [
+        {"id": "p", "scope": {"hooks": ["before_tool_call"]},
+         "rules": [{"id": "r", "minTrust": "standard",
+                    "conditions": [{"type": "tool", "name": "deploy"}],
+                    "effect": {"action": "allow"}},
+                   {"id": "r2",
+                    "conditions": [{"type": "tool", "name": "deploy"}],
+                    "effect": {"action": "deny",
+                               "reason": "untrusted deploy"}}]},
+    ]})
+    # minTrust is a TIER gate on the SESSION tier
+    # (policy-evaluator.ts:134-139): agent 40 seeds session 28 ->
+    # "restricted" < "standard" -> the allow rule is skipped, deny hits
+    out = eng.evaluate(eng.build_context("before_tool_call", "lowT", tool_name="deploy"))
+    assert out["action"] == "deny"
+    eng.trust_manager.set_score("highT", 90.0)  # session 63 -> "trusted"
+    out2 = eng.evaluate(eng.build_context("before_tool_call", "highT", tool_name="deploy"))
+    assert out2["action"] == "allow"
+
+
+def test_night_mode_flow_and_violation_skip(workspace):
+    t = [0.0]
+
+    def clock():
+        return t[0]
+
+    from vainplex_openclaw_amd.governance.engine import GovernanceEngine
+
+    eng = GovernanceEngine({"builtinPolicies": {"nightMode": True}}, workspace,
+                           clock=clock)
+    eng.start()
+    # 03:00 UTC: destructive exec denied by builtin-night-mode
+    t[0] = 1_700_000_000.0
+    import time as _time
+
+    # find a timestamp at 03:00 UTC
+    base = 1_700_000_000.0
+    while _time.gmtime(base).tm_hour != 3:
+        base += 3600
+    t[0] = base
+    ctx = eng.build_context("before_tool_call", "nite", tool_name="exec",
+                            tool_params={"command": "rm -rf /var/data"})
+    out = eng.evaluate(ctx)
+    assert out["action"] == "deny"
+    assert any(m.get("policyId") == "builtin-night-mode" for m in out["matchedPolicies"])
+    # night-mode denials do NOT count trust violations (engine.ts:248-263)
+    assert eng.trust_manager.get("nite")["signals"].get("violations", 0) == 0
+    # same call at 14:00 passes the time window
+    while _time.gmtime(base).tm_hour != 14:
+        base += 3600
+    t[0] = base
+    out2 = eng.evaluate(eng.build_context("before_tool_call", "nite", tool_name="exec",
+                                          tool_params={"command": "rm -rf /var/data"}))
+    assert out2["action"] != "deny" or not any(
+        m.get("policyId") == "builtin-night-mode" for m in out2["matchedPolicies"])
+
+
+def test_engine_error_fail_open_and_closed(workspace):
+    eng = _engine(workspace, {"failMode": "open"})
+    eng.policy_index = None          # force an internal failure
+    out = eng.evaluate({"hook": "before_tool_call", "agentId": "x", "toolName": "t"})
+    assert out["action"] == "allow" and "error" in str(out.get("reason", "")).lower()
+
+    eng2 = _engine(workspace + "/c", {"failMode": "closed"})
+    eng2.policy_index = None
+    out2 = eng2.evaluate({"hook": "before_tool_call", "agentId": "x", "toolName": "t"})
+    assert out2["action"] == "deny"
+
+
+def test_validation_trust_thresholds_end_to_end(workspace):
+    cfg = {"outputValidation": {
+        "enabled": True,
+        "factRegistries": [{"facts": [
+            {"subject": "nginx", "predicate": "state", "value": "running"}]}],
+    }}
+    eng = _engine(workspace, cfg)
+    text = "nginx is stopped."
+    eng.trust_manager.set_score("low", 20.0)
+    v_low = eng.validate_output(text, "low")
+    assert v_low["verdict"] == "block"
+    eng.trust_manager.set_score("mid", 50.0)
+    assert eng.validate_output(text, "mid")["verdict"] == "flag"
+    eng.trust_manager.set_score("high", 85.0)
+    assert eng.validate_output(text, "high")["verdict"] == "pass"
+    # matching claims pass at any trust
+    assert eng.validate_output("nginx is running.", "low")["verdict"] == "pass"

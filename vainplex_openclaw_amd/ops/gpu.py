@@ -484,20 +484,20 @@ FNV_PRIME = 0x100000001B3
 _U64 = (1 << 64) - 1
 
 # claim bit -> canonical predicate, per the reference's claim-type ->
-# predicate strategy table (fact-checker.ts:130-136) over the CLAIMS
-# family bits (ops/pattern_sets.py CLAIMS_PATTERNS)
+# predicate strategy table (fact-checker.ts:127-136) over the CLAIMS
+# family bits (ops/pattern_sets.py CLAIMS_PATTERNS). Types whose
+# strategy is NOT a single predicate (entity_name matches by subject
+# alone; self_referential matches the "self" subject) have no entry —
+# the GPU probe skips those bits and the host FactRegistry path covers
+# them.
 CLAIM_BIT_PREDICATE: Dict[int, str] = {
-    0: "status",      # system_state
-    1: "name",        # entity_name
+    0: "state",       # system_state -> "state"
     2: "exists",      # existence_pos
     3: "exists",      # existence_neg
     4: "exists",      # there_is
     5: "metric",      # has/contains/uses numeric
-    6: "metric",      # percentage
+    6: "percentage",  # percentage
     7: "count",       # count
-    8: "identity",    # self_identity
-    9: "identity",    # my_name
-    10: "capability", # i_have
 }
 
 

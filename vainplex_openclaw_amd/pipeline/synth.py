@@ -93,12 +93,13 @@ _FACT_SUBJECTS = ["nginx-service", "redis-cache", "api-gateway", "worker-pool"]
 
 
 def default_facts():
-    """Synthetic fact registry for the GPU probe (bench default)."""
-    facts = [(s, "status", "running") for s in _FACT_SUBJECTS]
+    """Synthetic fact registry for the GPU probe (bench default);
+    predicates follow the reference's claim-type strategy table
+    (system_state -> "state", fact-checker.ts:127-136)."""
+    facts = [(s, "state", "running") for s in _FACT_SUBJECTS]
     facts.append(("backup.db", "exists", "replica"))
-    facts.append(("deploybot", "identity", "deploybot"))
     # filler rows: realistic table occupancy
-    facts += [(f"svc-{i}", "status", "running") for i in range(64)]
+    facts += [(f"svc-{i}", "state", "running") for i in range(64)]
     return facts
 
 
