@@ -9,6 +9,7 @@ agentId), "session" (same sessionKey), "tool" (same agentId+toolName),
 
 from __future__ import annotations
 
+import threading
 import time
 from typing import List, Optional, Tuple
 
@@ -20,11 +21,16 @@ class FrequencyTracker:
         self._buf: List[Optional[Tuple[float, str, str, str]]] = [None] * capacity
         self._head = 0
         self._size = 0
+        # the reference is single-threaded; here the service micro-batcher
+        # and hook threads may record concurrently — a lost update means a
+        # rate limit silently under-counts, so writes take a lock
+        self._lock = threading.Lock()
 
     def record(self, agent_id: str, session_key: str = "", tool_name: str = "") -> None:
-        self._buf[self._head] = (self.clock(), agent_id, session_key, tool_name)
-        self._head = (self._head + 1) % self.capacity
-        self._size = min(self._size + 1, self.capacity)
+        with self._lock:
+            self._buf[self._head] = (self.clock(), agent_id, session_key, tool_name)
+            self._head = (self._head + 1) % self.capacity
+            self._size = min(self._size + 1, self.capacity)
 
     def count(
         self,
