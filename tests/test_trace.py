@@ -722,3 +722,80 @@ def test_bridge_timer_idempotent():
     assert n >= 1
     _time.sleep(0.05)
     assert len(hits) == n                         # actually stopped
+
+
+# -- redactor.test.ts depth --------------------------------------------------
+
+def _red_chain(payload, etype="msg.in"):
+    from vainplex_openclaw_amd.cortex.trace.analyzer import ChainRedactor
+    from vainplex_openclaw_amd.cortex.trace.chains import ConversationChain
+    from vainplex_openclaw_amd.cortex.trace.events import NormalizedEvent
+
+    ev = NormalizedEvent(id="e", ts=1.0, agent="a", session="s",
+                         type=etype, payload=payload)
+    chain = ConversationChain(id="c", session="s", agent="a", events=[ev])
+    return ChainRedactor().redact_chain(chain)[0]["payload"], ev
+
+
+REDACT_CASES = [
+    ("sk-" + "a1B2" * 6, True),                       # OpenAI-style
+    ("pk_live_" + "x" * 20, True),                    # Stripe publishable
+    ("sk_test_" + "y" * 20, True),                    # Stripe secret
+    ("Bearer " + "t" * 24, True),
+    ("https://bob:s3cretpw@db.example/x", True),      # url-embedded password
+    ("-----BEGIN RSA PRIVATE KEY-----", True),
+    ("PASSWORD=hunter22x", True),
+    ("ghp_" + "f" * 36, True),
+    ("eyJhbGciOiJIUzI1NiJ9.eyJzdWIiOiIxMjMifQ.sig-part-here", True),  # JWT
+    ("perfectly ordinary sentence with no secrets", False),
+    ("", False),
+]
+
+
+@pytest.mark.parametrize("text,redacts", REDACT_CASES)
+def test_trace_redactor_content_table(text, redacts):
+    out, ev = _red_chain({"content": f"before {text} after"})
+    changed = out["content"] != f"before {text} after"
+    assert changed == redacts, (text, out["content"])
+
+
+def test_trace_redactor_tool_fields_and_nesting():
+    key = "sk-abcdefghij0123456789XY"
+    out, ev = _red_chain({
+        "toolError": f"auth failed for {key}",
+        "toolResult": {"nested": {"token": key}, "list": [key, "ok"]},
+        "toolParams": {"cmd": f"use {key}"},
+    }, etype="tool.result")
+    import json as _json
+
+    blob = _json.dumps(out)
+    assert key not in blob
+    assert out["toolResult"]["list"][1] == "ok"
+    # original event untouched
+    assert key in ev.payload["toolError"]
+    assert key in ev.payload["toolParams"]["cmd"]
+
+
+def test_trace_redactor_multiple_types_one_text():
+    out, _ = _red_chain({"content":
+        "key sk-abcdefghij0123456789XY then ghp_" + "f" * 36 +
+        " and Bearer " + "b" * 24})
+    c = out["content"]
+    assert "sk-abcdefghij" not in c and "ghp_" + "f" * 36 not in c
+    assert "Bearer " + "b" * 24 not in c
+
+
+def test_trace_redactor_custom_patterns_and_invalid():
+    from vainplex_openclaw_amd.cortex.trace.analyzer import ChainRedactor
+    from vainplex_openclaw_amd.cortex.trace.chains import ConversationChain
+    from vainplex_openclaw_amd.cortex.trace.events import NormalizedEvent
+
+    red = ChainRedactor(custom_patterns=[
+        {"name": "ticket", "regex": r"TICKET-\d{4}"},
+        {"name": "bad", "regex": "(unclosed"},        # skipped silently
+    ])
+    ev = NormalizedEvent(id="e", ts=1.0, agent="a", session="s", type="msg.in",
+                         payload={"content": "see TICKET-9999 now"})
+    chain = ConversationChain(id="c", session="s", agent="a", events=[ev])
+    out = red.redact_chain(chain)[0]["payload"]["content"]
+    assert "TICKET-9999" not in out

@@ -96,19 +96,44 @@ def create_nats_source(url: Optional[str] = None):
 
 # -- redactor ---------------------------------------------------------------
 
+# trace-specific credential shapes beyond the 17 registry builtins
+# (redactor.ts: Stripe keys, raw JWTs, url-embedded passwords, env
+# assignments) — chains quote raw tool output, so the surface is wider
+_TRACE_EXTRA_PATTERNS = [
+    {"name": "stripe-key", "category": "custom",
+     "regex": r"\b[ps]k_(?:live|test)_[A-Za-z0-9]{16,}"},
+    {"name": "jwt", "category": "custom",
+     "regex": r"\beyJ[A-Za-z0-9_-]{8,}\.[A-Za-z0-9_-]{8,}\.[A-Za-z0-9_-]{8,}"},
+    {"name": "url-password", "category": "custom",
+     "regex": r"://[^/\s:]+:[^@\s]{4,}@"},
+    {"name": "env-secret", "category": "custom",
+     "regex": r"\b(?:PASSWORD|SECRET|TOKEN|API_KEY)=\S{4,}"},
+]
+
+
 class ChainRedactor:
-    def __init__(self) -> None:
-        self.engine = RedactionEngine(PatternRegistry(), RedactionVault())
+    def __init__(self, custom_patterns: Optional[List[Dict[str, Any]]] = None) -> None:
+        patterns = list(_TRACE_EXTRA_PATTERNS)
+        for p in custom_patterns or []:
+            if isinstance(p, dict) and isinstance(p.get("regex"), str):
+                patterns.append({"name": p.get("name", "custom"),
+                                 "category": "custom", "regex": p["regex"]})
+        # PatternRegistry drops invalid regexes silently (redactor.ts:141)
+        self.engine = RedactionEngine(PatternRegistry(None, patterns), RedactionVault())
 
     def redact_chain(self, chain: ConversationChain) -> List[Dict[str, Any]]:
+        """Returns redacted copies; the chain itself is never mutated."""
         out = []
         for ev in chain.events:
             payload = dict(ev.payload)
             for key in ("content", "toolError"):
                 if isinstance(payload.get(key), str):
                     payload[key] = self.engine.scan_string(payload[key])["output"]
-            if isinstance(payload.get("toolParams"), dict):
-                payload["toolParams"] = self.engine.scan(payload["toolParams"])["output"]
+            for key in ("toolParams", "toolResult"):
+                if isinstance(payload.get(key), (dict, list)):
+                    payload[key] = self.engine.scan(payload[key])["output"]
+                elif isinstance(payload.get(key), str):
+                    payload[key] = self.engine.scan_string(payload[key])["output"]
             out.append({"type": ev.type, "ts": ev.ts, "payload": payload})
         return out
 
