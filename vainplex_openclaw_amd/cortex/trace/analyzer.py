@@ -271,12 +271,14 @@ class TraceAnalyzer:
         config: Optional[AnalyzerConfig] = None,
         call_llm: Optional[Callable[[str], str]] = None,
         clock=time.time,
+        signal_registry=None,
     ):
         self.workspace = workspace
         self.source = source
         self.config = config or AnalyzerConfig()
         self.classifier = FindingClassifier(call_llm)
         self.clock = clock
+        self.signal_registry = signal_registry
         self.state_path = os.path.join(workspace, "memory", "reboot", "trace-analyzer-state.json")
         self.report_path = os.path.join(workspace, "memory", "reboot", "trace-analysis-report.json")
         self.state: Dict[str, Any] = load_json(self.state_path) or {}
@@ -289,7 +291,8 @@ class TraceAnalyzer:
         events = self.source.fetch(since_ts=since)
         chains = reconstruct_chains(events)
         chain_map = {c.id: c for c in chains}
-        findings = detect_all_signals(chains, self.config.detectors)
+        findings = detect_all_signals(chains, self.config.detectors,
+                                      registry=self.signal_registry)
         classified = self.classifier.classify(findings, chain_map)
         classified = [c for c in classified if c.get("confidence", 0) >= self.config.min_confidence]
         outputs = generate_outputs(classified)

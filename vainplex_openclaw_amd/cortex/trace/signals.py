@@ -176,15 +176,19 @@ def _similar_attempts(a: Dict[str, Any], b: Dict[str, Any]) -> bool:
 
 # -- detectors --------------------------------------------------------------
 
-def detect_correction(chain: ConversationChain) -> List[Finding]:
-    """User messages that correct the agent right after an agent message."""
+def detect_correction(chain: ConversationChain, registry=None) -> List[Finding]:
+    """User messages that correct the agent right after an agent message.
+    A SignalPatternRegistry (signals_lang) extends the builtin phrase
+    tables with its merged correction indicators — runtime-registered
+    language packs feed detection (signals/lang/index.ts)."""
     findings = []
     phrases = _phrases("correction")
+    extra = registry.get_patterns()["correction"]["indicators"] if registry else []
     for i, ev in enumerate(chain.events):
         if ev.type != "msg.in":
             continue
         content = str(ev.payload.get("content") or "").lower()
-        if any(p in content for p in phrases):
+        if any(p in content for p in phrases) or any(rx.search(content) for rx in extra):
             prior_agent = any(e.type == "msg.out" for e in chain.events[:i])
             if prior_agent:
                 findings.append(_mk(chain, "correction", "medium",
@@ -239,13 +243,18 @@ def detect_doom_loop(chain: ConversationChain) -> List[Finding]:
     return findings
 
 
-def detect_dissatisfied(chain: ConversationChain) -> List[Finding]:
+def detect_dissatisfied(chain: ConversationChain, registry=None) -> List[Finding]:
     phrases = _phrases("dissatisfied")
+    pats = registry.get_patterns()["dissatisfaction"] if registry else None
     for ev in chain.events:
         if ev.type != "msg.in":
             continue
         content = str(ev.payload.get("content") or "").lower()
-        if any(p in content for p in phrases):
+        hit = any(p in content for p in phrases) or (
+            pats and any(rx.search(content) for rx in pats["indicators"]))
+        if hit and pats and any(rx.search(content) for rx in pats["satisfactionOverrides"]):
+            continue  # "thanks, forget it" reads as satisfied (lang packs)
+        if hit:
             return [_mk(chain, "dissatisfied", "high", "User expressed dissatisfaction",
                         {"message": content[:200]}, 0.7)]
     return []
@@ -391,14 +400,24 @@ DETECTORS: Dict[str, Callable[[ConversationChain], List[Finding]]] = {
 }
 
 
+# detectors that consult the signal-language registry's merged patterns
+_REGISTRY_AWARE = {"correction", "dissatisfied"}
+
+
 def detect_all_signals(
-    chains: List[ConversationChain], enabled: Optional[List[str]] = None
+    chains: List[ConversationChain],
+    enabled: Optional[List[str]] = None,
+    registry=None,
 ) -> List[Finding]:
     names = enabled if enabled is not None else list(DETECTORS.keys())
     findings: List[Finding] = []
     for chain in chains:
         for name in names:
             fn = DETECTORS.get(name)
-            if fn:
+            if fn is None:
+                continue
+            if name in _REGISTRY_AWARE:
+                findings.extend(fn(chain, registry=registry))
+            else:
                 findings.extend(fn(chain))
     return findings
