@@ -146,8 +146,12 @@ TRIAGE_PROMPT = (
 )
 ANALYSIS_PROMPT = (
     "You analyze an agent failure. Reply ONLY JSON: "
-    '{"rootCause": "...", "actionText": "...", "confidence": 0.0}\n'
+    '{"rootCause": "...", "actionType": "soul_rule"|"governance_policy"|'
+    '"cortex_pattern"|"manual_review", "actionText": "...", "confidence": 0.0}\n'
 )
+
+# classifier.ts: unknown actionType values default to manual_review
+ACTION_TYPES = ("soul_rule", "governance_policy", "cortex_pattern", "manual_review")
 
 
 class FindingClassifier:
@@ -157,9 +161,11 @@ class FindingClassifier:
         self,
         call_llm: Optional[Callable[[str], str]] = None,
         triage_enabled: bool = True,
+        model: str = "",
     ):
         self.call_llm = call_llm
         self.triage_enabled = triage_enabled
+        self.model = model
         self.redactor = ChainRedactor()
 
     @staticmethod
@@ -191,11 +197,26 @@ class FindingClassifier:
                         continue
                     if triage.get("severity") in ("low", "medium", "high"):
                         d["severity"] = triage["severity"]
-                analysis = self._parse(self.call_llm(ANALYSIS_PROMPT + context)) or {}
+                analysis = self._parse(self.call_llm(ANALYSIS_PROMPT + context))
+                if analysis is None:
+                    # invalid JSON / HTTP error / timeout: classification
+                    # stays null (classifier.ts), default action text kept
+                    d["actionText"] = default_action_text(f)
+                    out.append(d)
+                    continue
                 d["rootCause"] = analysis.get("rootCause", "")
                 d["actionText"] = analysis.get("actionText") or default_action_text(f)
                 if isinstance(analysis.get("confidence"), (int, float)):
                     d["confidence"] = float(analysis["confidence"])
+                atype = analysis.get("actionType")
+                d["classification"] = {
+                    "rootCause": d["rootCause"],
+                    "actionType": atype if atype in ACTION_TYPES else "manual_review",
+                    "actionText": d["actionText"],
+                    "confidence": float(analysis["confidence"])
+                    if isinstance(analysis.get("confidence"), (int, float)) else 0.5,
+                    "model": self.model,
+                }
             except Exception:
                 d["actionText"] = default_action_text(f)
             out.append(d)
@@ -253,6 +274,85 @@ def generate_outputs(classified: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
     return outputs
 
 
+# classification-driven generator (output-generator.ts R-021/22/23):
+# consumes the nested LLM classification; unclassified and manual_review
+# findings produce no output.
+
+_HOOKS_BY_SIGNAL = {
+    "doom_loop": ["before_tool_call"],
+    "tool_fail": ["before_tool_call"],
+    "hallucination": ["message_sending"],
+    "unverified_claim": ["message_sending"],
+}
+
+
+def _classified(findings: List[Dict[str, Any]], atype: str) -> List[Dict[str, Any]]:
+    return [f for f in findings
+            if (f.get("classification") or {}).get("actionType") == atype]
+
+
+def generate_classified_outputs(findings: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+    outputs: List[Dict[str, Any]] = []
+
+    # soul rules: grouped by normalized actionText (first 80 chars),
+    # content carries the observation count + finding-id refs
+    groups: Dict[str, List[Dict[str, Any]]] = {}
+    for f in _classified(findings, "soul_rule"):
+        key = " ".join(f["classification"]["actionText"].lower().split())[:80]
+        groups.setdefault(key, []).append(f)
+    for items in groups.values():
+        action = items[0]["classification"]["actionText"]
+        ids = [i["id"] for i in items]
+        id_ref = ", ".join(i[:8] for i in ids[:3])
+        outputs.append({
+            "id": f"out-{uuid.uuid4().hex[:12]}",
+            "type": "soul_rule",
+            "content": f"{action} [{len(items)}× beobachtet in Traces, Findings: {id_ref}]",
+            "sourceFindings": ids,
+            "observationCount": len(items),
+            "confidence": sum(i["classification"]["confidence"] for i in items) / len(items),
+        })
+
+    # governance policies: one audit-effect policy JSON per finding
+    for f in _classified(findings, "governance_policy"):
+        c = f["classification"]
+        sig = str(f.get("signalType", "unknown"))
+        policy = {
+            "id": f"trace-gen-{sig.replace('_', '-')}-{f['id'][:8]}",
+            "name": f"Auto: {str(f.get('summary', ''))[:60]}",
+            "version": "1.0.0",
+            "description": f"Auto-generated from trace finding {f['id']}. "
+                           f"Root cause: {c.get('rootCause', '')}",
+            "scope": {"hooks": _HOOKS_BY_SIGNAL.get(sig, ["message_sent"])},
+            "rules": [{
+                "id": f"rule-{f['id'][:8]}",
+                "description": c["actionText"],
+                "conditions": {"signal": sig, "severity": f.get("severity")},
+                "effect": {"action": "audit", "reason": c["actionText"]},
+            }],
+        }
+        outputs.append({
+            "id": f"out-{uuid.uuid4().hex[:12]}",
+            "type": "governance_policy",
+            "content": json.dumps(policy, indent=2),
+            "sourceFindings": [f["id"]],
+            "observationCount": 1,
+            "confidence": c["confidence"],
+        })
+
+    # cortex patterns: the actionText IS the regex
+    for f in _classified(findings, "cortex_pattern"):
+        outputs.append({
+            "id": f"out-{uuid.uuid4().hex[:12]}",
+            "type": "cortex_pattern",
+            "content": f["classification"]["actionText"],
+            "sourceFindings": [f["id"]],
+            "observationCount": 1,
+            "confidence": f["classification"]["confidence"],
+        })
+    return outputs
+
+
 # -- report + orchestrator --------------------------------------------------
 
 @dataclass
@@ -296,6 +396,7 @@ class TraceAnalyzer:
         classified = self.classifier.classify(findings, chain_map)
         classified = [c for c in classified if c.get("confidence", 0) >= self.config.min_confidence]
         outputs = generate_outputs(classified)
+        classified_outputs = generate_classified_outputs(classified)
         now_ms = self.clock() * 1000
         # aggregates (report.ts assembleReport): per-signal stats, top
         # agents per signal, chain time range, classified-finding count
@@ -338,6 +439,7 @@ class TraceAnalyzer:
             "timeRange": time_range,
             "findings": classified,
             "outputs": outputs,
+            "classifiedOutputs": classified_outputs,
         }
         self.state = {
             "lastProcessedTs": max([e.ts for e in events], default=last_ts),
