@@ -361,6 +361,9 @@ class AnalyzerConfig:
     detectors: Optional[List[str]] = None
     incremental_context_window_ms: float = 30 * 60 * 1000
     min_confidence: float = 0.0
+    chain_gap_minutes: float = 30.0
+    redact_patterns: List[str] = field(default_factory=list)
+    max_findings: int = 1000
 
 
 class TraceAnalyzer:
@@ -379,9 +382,34 @@ class TraceAnalyzer:
         self.classifier = FindingClassifier(call_llm)
         self.clock = clock
         self.signal_registry = signal_registry
+        if self.config.redact_patterns:
+            self.classifier.redactor = ChainRedactor(
+                [{"name": f"cfg-{i}", "regex": p}
+                 for i, p in enumerate(self.config.redact_patterns)])
         self.state_path = os.path.join(workspace, "memory", "reboot", "trace-analyzer-state.json")
         self.report_path = os.path.join(workspace, "memory", "reboot", "trace-analysis-report.json")
         self.state: Dict[str, Any] = load_json(self.state_path) or {}
+
+    @classmethod
+    def from_config(cls, workspace: str, source, resolved: Dict[str, Any],
+                    call_llm=None, clock=time.time, signal_registry=None):
+        """Build from a resolve_trace_analyzer_config() dict: per-signal
+        enables -> detectors, chainGapMinutes, redactPatterns, output
+        caps (config.ts -> analyzer.ts wiring)."""
+        from .config import enabled_detectors
+
+        cfg = AnalyzerConfig(
+            enabled=bool(resolved.get("enabled")),
+            detectors=enabled_detectors(resolved),
+            chain_gap_minutes=float(resolved["chainGapMinutes"]),
+            redact_patterns=list(resolved.get("redactPatterns", [])),
+            max_findings=int(resolved["output"]["maxFindings"]),
+        )
+        ta = cls(workspace, source, cfg, call_llm=call_llm, clock=clock,
+                 signal_registry=signal_registry)
+        if resolved["output"].get("reportPath"):
+            ta.report_path = resolved["output"]["reportPath"]
+        return ta
 
     def run(self) -> Dict[str, Any]:
         """Full pipeline: fetch -> chains -> detect -> classify -> outputs
@@ -389,12 +417,13 @@ class TraceAnalyzer:
         last_ts = float(self.state.get("lastProcessedTs", 0))
         since = max(0.0, last_ts - self.config.incremental_context_window_ms)
         events = self.source.fetch(since_ts=since)
-        chains = reconstruct_chains(events)
+        chains = reconstruct_chains(events, gap_minutes=self.config.chain_gap_minutes)
         chain_map = {c.id: c for c in chains}
         findings = detect_all_signals(chains, self.config.detectors,
                                       registry=self.signal_registry)
         classified = self.classifier.classify(findings, chain_map)
         classified = [c for c in classified if c.get("confidence", 0) >= self.config.min_confidence]
+        classified = classified[: self.config.max_findings]
         outputs = generate_outputs(classified)
         classified_outputs = generate_classified_outputs(classified)
         now_ms = self.clock() * 1000
