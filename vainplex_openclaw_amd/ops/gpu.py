@@ -477,6 +477,56 @@ def topk_recall_threshold(
     return out_s, out_i
 
 
+def topk_recall_threshold_banded(
+    Q: torch.Tensor,
+    X: torch.Tensor,
+    k: int,
+    salience: torch.Tensor,
+    hot_idx: torch.Tensor,
+    hot_X: Optional[torch.Tensor] = None,
+    is_hot: Optional[torch.Tensor] = None,
+    **kw,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Salience-banded weighted recall for SKEWED salience distributions.
+
+    The bounded-overfetch weighted path (topk_recall_threshold with
+    salience) selects by raw cosine; under heavy skew (e.g. Zipf with a
+    100x range) the dense weighted optimum can sit at cosine rank ~10k —
+    beyond any bounded overfetch (measured ~86% dense-regret at 128
+    candidates; the reference's interactive limit*4 overfetch has the
+    same bound, membrane retrieve path). This variant recovers those
+    optima: the HOT band (top-salience rows, `hot_idx`) is scored
+    exactly — dense weighted bf16 matmul over a small gathered copy — so
+    no cosine-rank bound applies there, while the COLD band keeps the
+    fp4 threshold scan. Bands merge by exact weighted score; cold
+    candidates that are hot rows are masked out (the hot band already
+    scored them exactly), so no duplicates survive.
+
+    `hot_X` (gathered rows) and `is_hot` (membership mask) are optional
+    caches the caller refreshes when salience order drifts.
+    """
+    nh = int(hot_idx.numel())
+    if nh == 0:
+        return topk_recall_threshold(Q, X, k, salience=salience, **kw)
+    if hot_X is None:
+        hot_X = X[hot_idx.long()]
+    hot = torch.matmul(Q, hot_X.T).float() * salience[hot_idx.long()]
+    hs, hsel = torch.topk(hot, min(k, nh), dim=1)
+    hi = hot_idx[hsel.reshape(-1)].reshape(hsel.shape).to(torch.int32)
+
+    cs, ci = topk_recall_threshold(Q, X, k, salience=salience, **kw)
+    if is_hot is None:
+        is_hot = torch.zeros(X.shape[0], dtype=torch.bool, device=X.device)
+        is_hot[hot_idx.long()] = True
+    dup = is_hot[ci.long().clamp_min(0)]
+    cs = torch.where(dup, torch.full_like(cs, -1e30), cs)
+
+    all_s = torch.cat([hs, cs], dim=1)
+    all_i = torch.cat([hi, ci], dim=1)
+    fin = torch.topk(all_s, k, dim=1)
+    return fin.values, torch.gather(all_i, 1, fin.indices)
+
+
 # -- fact-registry probe ----------------------------------------------------
 
 FNV_OFFSET = 0xCBF29CE484222325

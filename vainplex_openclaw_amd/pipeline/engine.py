@@ -81,6 +81,12 @@ class PipelineConfig:
     # (membrane/engine.py retrieve semantics); selection overfetches by
     # raw cosine, exact rescore applies the salience weight
     salience_weighting: bool = True
+    # "threshold_banded" recall: exact weighted scoring of the
+    # top-salience band (hot_frac of rows) + fp4 threshold scan for the
+    # rest — recovers weighted optima that sit beyond any bounded cosine
+    # overfetch under skewed salience (ops.gpu.topk_recall_threshold_banded)
+    hot_frac: float = 0.002
+    band_refresh_steps: int = 64
 
 
 class StageProfiler:
@@ -154,7 +160,8 @@ class FirewallPipeline:
             # e4m3 copy serves the two_stage mode and the fp8/MX scans.
             self.index8 = None
             self.index4 = None
-            if cfg.recall_fp4 and cfg.recall_mode == "threshold" and cfg.dim % 128 == 0:
+            if cfg.recall_fp4 and cfg.recall_mode in ("threshold", "threshold_banded") \
+                    and cfg.dim % 128 == 0:
                 x4 = torch.empty(cfg.index_size, cfg.dim // 2, dtype=torch.uint8,
                                  device=self.device)
                 xs = torch.empty(cfg.index_size, cfg.dim // 32, dtype=torch.uint8,
@@ -165,7 +172,8 @@ class FirewallPipeline:
                     x4[i : i + n] = c4
                     xs[i : i + n] = cs
                 self.index4 = (x4, xs)
-            elif cfg.recall_fp8 and cfg.recall_mode in ("two_stage", "threshold"):
+            elif cfg.recall_fp8 and cfg.recall_mode in ("two_stage", "threshold",
+                                                        "threshold_banded"):
                 self.index8 = torch.empty(
                     cfg.index_size, cfg.dim, dtype=torch.uint8, device=self.device
                 )
@@ -175,6 +183,14 @@ class FirewallPipeline:
 
             # salience state: recall strength + decay (Membrane semantics)
             self.salience = torch.ones(cfg.index_size, device=self.device)
+
+            # hot-band caches for threshold_banded recall; refreshed every
+            # band_refresh_steps as reinforcement reshapes the salience
+            self.hot_idx = None
+            self.hot_X = None
+            self.is_hot = None
+            if cfg.recall_mode == "threshold_banded":
+                self._refresh_hot_band()
 
             # trust state vectors (mirrors governance TrustManager fields)
             A = cfg.n_agents
@@ -218,6 +234,18 @@ class FirewallPipeline:
             }
 
         self.audit_sink: Optional[Any] = None  # callable(records_cpu, root_hex)
+        self._steps_done = 0
+
+    def _refresh_hot_band(self) -> None:
+        """Re-pick the top-salience rows and regather their bf16 copy
+        (threshold_banded mode). Cheap relative to a step: topk over the
+        salience vector + a nh-row gather."""
+        cfg = self.cfg
+        nh = max(1, int(cfg.index_size * cfg.hot_frac))
+        self.hot_idx = torch.topk(self.salience, min(nh, cfg.index_size)).indices
+        self.hot_X = self.index[self.hot_idx]
+        self.is_hot = torch.zeros(cfg.index_size, dtype=torch.bool, device=self.device)
+        self.is_hot[self.hot_idx] = True
 
     # -- input staging -----------------------------------------------------
     def stage(self, batch: SynthBatch) -> Dict[str, torch.Tensor]:
@@ -300,7 +328,15 @@ class FirewallPipeline:
         sal = self.salience if cfg.salience_weighting else None
 
         def local_recall(queries):
-            if cfg.recall_mode == "threshold":
+            if cfg.recall_mode == "threshold_banded" and sal is not None:
+                if self._steps_done and self._steps_done % cfg.band_refresh_steps == 0:
+                    self._refresh_hot_band()
+                return g.topk_recall_threshold_banded(
+                    queries, self.index, cfg.topk, salience=sal,
+                    hot_idx=self.hot_idx, hot_X=self.hot_X, is_hot=self.is_hot,
+                    X8=self.index8, mx=cfg.recall_mx, X4=self.index4,
+                )
+            if cfg.recall_mode in ("threshold", "threshold_banded"):
                 return g.topk_recall_threshold(
                     queries, self.index, cfg.topk, X8=self.index8,
                     mx=cfg.recall_mx, X4=self.index4, salience=sal
@@ -391,6 +427,7 @@ class FirewallPipeline:
 
         prof.mark("end")
         self.batch_seq += 1
+        self._steps_done += 1
         if self.audit_sink is not None:
             self.audit_sink(records, root)
 
