@@ -47,7 +47,7 @@ def main() -> int:
     ap.add_argument("--topk", type=int, default=16)
     ap.add_argument("--pool", type=int, default=4, help="pre-generated batch pool size")
     ap.add_argument("--recall-mode", default="threshold",
-                    choices=["threshold", "two_stage", "direct"])
+                    choices=["threshold", "threshold_banded", "two_stage", "direct"])
     ap.add_argument("--no-fp8", action="store_true")
     ap.add_argument("--no-mx", action="store_true")
     ap.add_argument("--no-fp4", action="store_true")
@@ -176,7 +176,8 @@ def main() -> int:
                 # scan operand precision; final recall scores are always
                 # EXACT fp32 rescored cosines (ops/gpu.py threshold path)
                 "recall_scan_operands": (
-                    "mxfp4" if (cfg.recall_fp4 and cfg.recall_mode == "threshold")
+                    "mxfp4" if (cfg.recall_fp4 and cfg.recall_mode
+                                in ("threshold", "threshold_banded"))
                     else ("fp8-mx" if (cfg.recall_fp8 and cfg.recall_mx)
                           else ("fp8" if cfg.recall_fp8 else "bf16"))
                 ),
