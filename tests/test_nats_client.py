@@ -72,6 +72,7 @@ class FakeNatsServer:
         self.nack_publishes = nack_publishes
         self.mute_acks = mute_acks
         self.messages = []           # (subject, payload)
+        self.stored = {}             # seq -> (subject, payload): stream storage
         self.subs = {}               # sid -> subject pattern
         self.created_configs = []
         self.connect_opts = None
@@ -146,9 +147,22 @@ class FakeNatsServer:
                 name = op[len("STREAM.INFO."):]
                 if name in self.streams:
                     resp = {"type": "io.nats.jetstream.api.v1.stream_info_response",
-                            "config": self.streams[name]}
+                            "config": self.streams[name],
+                            "state": {"first_seq": min(self.stored) if self.stored else 0,
+                                      "last_seq": self._seq,
+                                      "messages": len(self.stored)}}
                 else:
                     resp = {"error": {"code": 404, "description": "stream not found"}}
+            elif op.startswith("STREAM.MSG.GET."):
+                import base64 as _b64
+                q = json.loads(payload.decode())
+                seq = int(q.get("seq", 0))
+                if seq in self.stored:
+                    subj, pl = self.stored[seq]
+                    resp = {"message": {"seq": seq, "subject": subj,
+                                        "data": _b64.b64encode(pl).decode()}}
+                else:
+                    resp = {"error": {"code": 404, "description": "no message found"}}
             elif op.startswith("STREAM.CREATE."):
                 cfg = json.loads(payload.decode())
                 self.streams[cfg["name"]] = cfg
@@ -162,9 +176,10 @@ class FakeNatsServer:
             return
         # regular JetStream publish
         self.messages.append((subject, payload))
+        self._seq += 1
+        self.stored[self._seq] = (subject, payload)
         self._deliver(subject, payload, reply)
         if reply and not self.mute_acks:
-            self._seq += 1
             if self.nack_publishes:
                 ack = {"error": {"code": 503, "description": "no responders"}}
             else:

@@ -87,11 +87,15 @@ class JournalTraceSource:
         return out
 
 
-def create_nats_source(url: Optional[str] = None):
-    """NATS JetStream source — returns None gracefully when no client is
-    available (nats-trace-source.ts:1-12). There is no network in this
-    environment; the journal source is the production path."""
-    return None
+def create_nats_source(url: Optional[str] = None, transport=None, logger=None):
+    """NATS JetStream source — delegates to trace.nats_source (real
+    wire-protocol client over eventstore/nats_client); returns None
+    gracefully when the connection fails (nats-trace-source.ts:103-115),
+    in which case the journal source is the production path."""
+    from .nats_source import create_nats_trace_source
+
+    cfg = {"url": url} if url else {}
+    return create_nats_trace_source(cfg, logger=logger, transport=transport)
 
 
 # -- redactor ---------------------------------------------------------------

@@ -136,3 +136,14 @@ def normalize_schema_b(ev: Dict[str, Any], seq: int = 0) -> Optional[NormalizedE
         payload=payload,
         seq=seq,
     )
+
+
+def normalize_event(ev, seq: int = 0) -> Optional[NormalizedEvent]:
+    """Schema-sniffing normalizer (events.ts normalizeEvent): Schema B
+    carries a `kind` field; everything else is tried as a Schema A
+    ClawEvent envelope."""
+    if not isinstance(ev, dict):
+        return None
+    if "kind" in ev:
+        return normalize_schema_b(ev, seq)
+    return normalize_schema_a(ev, seq)

@@ -307,6 +307,36 @@ class JetStreamClient:
         if self.logger:
             self.logger.info(f'[nats-eventstore] Created stream "{self.stream}"')
 
+    def stream_info(self) -> Optional[Dict]:
+        """$JS.API.STREAM.INFO — returns the info response (with its
+        `state` block: first_seq/last_seq/messages) or None on error.
+        Used by the trace-analyzer NATS source (nats-trace-source.ts)."""
+        try:
+            info = self._js_request(f"STREAM.INFO.{self.stream}", b"")
+        except NatsError:
+            return None
+        return None if (info or {}).get("error") else info
+
+    def get_message(self, seq: int) -> Optional[Dict]:
+        """Stored message by stream sequence ($JS.API.STREAM.MSG.GET):
+        {"seq", "subject", "data": bytes} or None (missing seq / error)."""
+        try:
+            resp = self._js_request(f"STREAM.MSG.GET.{self.stream}",
+                                    json.dumps({"seq": int(seq)}).encode())
+        except NatsError:
+            return None
+        msg = (resp or {}).get("message")
+        if not isinstance(msg, dict):
+            return None
+        import base64
+
+        try:
+            data = base64.b64decode(msg.get("data") or "")
+        except Exception:
+            return None
+        return {"seq": int(msg.get("seq", seq)),
+                "subject": msg.get("subject", ""), "data": data}
+
     def _js_request(self, op: str, payload: bytes) -> Optional[Dict]:
         try:
             raw = self.nc.request(f"$JS.API.{op}", payload,
