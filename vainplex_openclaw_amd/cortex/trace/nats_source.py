@@ -63,15 +63,23 @@ class NatsTraceSource:
         return float(ts) if isinstance(ts, (int, float)) else None
 
     def _find_start_sequence(self, first: int, last: int, target_ms: float) -> int:
-        """First seq with ts >= target_ms (unreadable seqs advance lo)."""
+        """First seq with ts >= target_ms. An unreadable probe (retained-
+        out seq or junk payload) is resolved by scanning a few seqs
+        forward for a readable timestamp before deciding the direction —
+        slightly more robust than the reference's null-means-before-start
+        rule, which skips valid events past an in-range junk record."""
         lo, hi = first, last
         while lo < hi:
             mid = (lo + hi) // 2
-            ts = self._ts_at(mid)
+            ts = None
+            for probe in range(mid, min(mid + 8, hi) + 1):
+                ts = self._ts_at(probe)
+                if ts is not None:
+                    break
             if ts is None or ts < target_ms:
                 lo = mid + 1
             else:
-                hi = mid
+                hi = mid   # a readable seq at/after mid qualifies; mid may too
         return lo
 
     # -- fetch ----------------------------------------------------------------
