@@ -201,42 +201,63 @@ class CortexPlugin:
         self._register_trace_analyzer(api, config, workspace)
 
     def _register_trace_analyzer(self, api: PluginApi, config, workspace: str) -> None:
-        """Conditional trace-analyzer registration + interval timer
+        """Conditional trace-analyzer registration: analyze + status
+        commands, signal-registry preload, optional schedule timer
         (reference cortex hooks.ts:242-253, trace-analyzer/hooks.ts)."""
         ta_cfg = config.get("traceAnalyzer") or {}
         if not ta_cfg.get("enabled", self.journal is not None):
             return
         from .trace.analyzer import AnalyzerConfig, JournalTraceSource, TraceAnalyzer, create_nats_source
+        from .trace.signals_lang import default_registry
 
         source = None
         if self.journal is not None:
             source = JournalTraceSource(self.journal)
         else:
-            source = create_nats_source(ta_cfg.get("natsUrl"))
+            nats_cfg = ta_cfg.get("nats") or {}
+            source = create_nats_source(
+                nats_cfg.get("url") or ta_cfg.get("natsUrl"), logger=api.logger)
         if source is None:
             api.logger.info("[cortex] trace analyzer: no event source available")
             return
-        self.analyzer = TraceAnalyzer(
-            workspace, source,
-            AnalyzerConfig(
-                detectors=ta_cfg.get("detectors"),
-                min_confidence=ta_cfg.get("minConfidence", 0.0),
-            ),
-            call_llm=self.call_llm,
-        )
+        registry = default_registry()
+        api.logger.info("[trace-analyzer] Loaded signal patterns for: "
+                        + ", ".join(registry.loaded_languages()))
+        if "signals" in ta_cfg:   # resolved trace-analyzer config shape
+            from .trace.config import resolve_trace_analyzer_config
+
+            self.analyzer = TraceAnalyzer.from_config(
+                workspace, source, resolve_trace_analyzer_config(ta_cfg),
+                call_llm=self.call_llm, signal_registry=registry)
+        else:                     # ad-hoc keys (detectors/minConfidence)
+            self.analyzer = TraceAnalyzer(
+                workspace, source,
+                AnalyzerConfig(
+                    detectors=ta_cfg.get("detectors"),
+                    min_confidence=ta_cfg.get("minConfidence", 0.0),
+                ),
+                call_llm=self.call_llm,
+                signal_registry=registry,
+            )
         api.register_command("cortexanalyze", lambda *a, **kw: self.analyzer.run())
         api.register_gateway_method("cortex.analyze", lambda *a, **kw: self.analyzer.run())
+        api.register_command("cortextracestatus", lambda *a, **kw: self.trace_status())
+        api.register_gateway_method("cortex.trace.status", lambda *a, **kw: self.trace_status())
 
+        schedule_cfg = ta_cfg.get("schedule") or {}
         interval_min = float(ta_cfg.get("intervalMinutes", 0))
+        if schedule_cfg.get("enabled"):
+            interval_min = float(schedule_cfg.get("intervalHours", 24)) * 60.0
         if interval_min > 0:
             import threading
 
             def schedule():
                 def fire():
                     try:
+                        api.logger.info("[trace-analyzer] Running scheduled analysis...")
                         self.analyzer.run()
-                    except Exception:
-                        pass
+                    except Exception as exc:
+                        api.logger.warn(f"[trace-analyzer] Scheduled analysis failed: {exc}")
                     schedule()
 
                 self._analyzer_timer = threading.Timer(interval_min * 60.0, fire)
@@ -244,9 +265,36 @@ class CortexPlugin:
                 self._analyzer_timer.start()
 
             schedule()
+            api.logger.info(
+                f"[trace-analyzer] Scheduled analysis every {interval_min / 60.0:g}h")
             api.on("gateway_stop",
-                   lambda ev: self._analyzer_timer and self._analyzer_timer.cancel(),
+                   lambda ev: self.cleanup_trace_analyzer(),
                    priority=999)
+
+    def trace_status(self) -> Dict[str, Any]:
+        """/trace-status: analyzer state summary (trace-analyzer/hooks.ts
+        handleTraceStatus)."""
+        if self.analyzer is None:
+            return {"enabled": False}
+        st = self.analyzer.state or {}
+        return {
+            "enabled": True,
+            "scheduled": getattr(self, "_analyzer_timer", None) is not None,
+            "runsCompleted": int(st.get("runsCompleted", 0)),
+            "lastRunAt": st.get("lastRunAt"),
+            "totalEventsAnalyzed": int(st.get("totalEventsAnalyzed", 0)),
+            "totalFindings": int(st.get("totalFindings", 0)),
+            "reportPath": self.analyzer.report_path,
+        }
+
+    def cleanup_trace_analyzer(self) -> None:
+        """Cancel the schedule timer and drop the analyzer; safe to call
+        repeatedly (trace-analyzer/hooks.ts cleanupTraceAnalyzerHooks)."""
+        timer = getattr(self, "_analyzer_timer", None)
+        if timer is not None:
+            timer.cancel()
+            self._analyzer_timer = None
+        self.analyzer = None
 
 
 def create_plugin(workspace: Optional[str] = None, journal=None, call_llm=None) -> CortexPlugin:
