@@ -187,3 +187,80 @@ def test_trace_analyzer_runs_on_nats_source(tmp_path):
     assert report["eventsAnalyzed"] == 6
     assert any(f["signalType"] == "doom_loop" for f in report["findings"])
     src.close()
+
+
+# -- events depth: detect_schema + conversation.* shape + nested errors -------
+# (mirrors test/trace-analyzer/events.test.ts detection/payload sections)
+
+from vainplex_openclaw_amd.cortex.trace.events import (
+    detect_schema,
+    normalize_schema_a,
+)
+
+
+def test_detect_schema_matrix():
+    assert detect_schema({"type": "conversation.message.in", "timestamp": 5}) == "B"
+    assert detect_schema({"type": "msg.in", "ts": 5.0}) == "A"
+    assert detect_schema({"type": "custom.thing", "meta": {"source": "session-sync"},
+                          "timestamp": 5}) == "B"
+    assert detect_schema({"type": "custom.thing", "timestamp": 5}) == "B"
+    assert detect_schema({"type": "msg.in"}) == "A"            # known type fallback
+    assert detect_schema({"type": "totally.unknown"}) is None
+    assert detect_schema({"canonicalType": "message.in.received", "ts": 1}) == "A"
+    assert detect_schema({"kind": "user_message"}) == "B"
+    assert detect_schema("junk") is None
+
+
+def test_conversation_msg_text_preview():
+    ev = normalize_event({"type": "conversation.message.in", "timestamp": 1000.0,
+                          "agent": "dev", "session": "agent:main:u42",
+                          "payload": {"text_preview": [{"text": "hello there"}],
+                                      "sessionId": "u42"}}, seq=3)
+    assert ev.type == "msg.in" and ev.payload["content"] == "hello there"
+    assert ev.payload["role"] == "user" and ev.session == "u42" and ev.seq == 3
+    out = normalize_event({"type": "conversation.message.out", "timestamp": 1001.0,
+                           "payload": {"text_preview": []}})
+    assert out.payload["content"] == "" and out.payload["role"] == "assistant"
+    assert out.agent == "unknown" and out.session == "unknown"
+
+
+def test_conversation_tool_call_and_result():
+    tc = normalize_event({"type": "conversation.tool_call", "timestamp": 2000.0,
+                          "payload": {"data": {"name": "exec", "args": {"c": "ls"}}}})
+    assert tc.type == "tool.call"
+    assert tc.payload["toolName"] == "exec" and tc.payload["toolParams"] == {"c": "ls"}
+    err = normalize_event({"type": "conversation.tool_result", "timestamp": 2001.0,
+                           "payload": {"data": {"name": "exec", "result": "Permission denied",
+                                                "isError": True}}})
+    assert err.payload["toolIsError"] is True
+    assert err.payload["toolError"] == "Permission denied"
+    ok = normalize_event({"type": "conversation.tool_result", "timestamp": 2002.0,
+                          "payload": {"data": {"name": "exec", "result": "file.txt"}}})
+    assert ok.payload["toolIsError"] is False and ok.payload["toolError"] is None
+
+
+def test_conversation_requires_timestamp():
+    assert normalize_event({"type": "conversation.message.in",
+                            "payload": {"text_preview": [{"text": "x"}]}}) is None
+    assert normalize_event({"type": "conversation.message.in", "timestamp": 0,
+                            "payload": {}}) is None
+
+
+@pytest.mark.parametrize("data,want_err,want_is", [
+    ({"error": "boom"}, "boom", True),
+    ({"result": {"details": {"error": "nested fail"}}}, "nested fail", True),
+    ({"result": {"details": {"status": "error"}}}, "status: error", True),
+    ({"result": {"details": {"exitCode": 2}}}, "exit code 2", True),
+    ({"result": {"isError": True, "content": [{"text": "stderr text"}]}}, "stderr text", True),
+    ({"result": {"isError": True}}, "unknown error", True),
+    ({"result": {"details": {"exitCode": 0}}}, None, False),
+    ({"result": "plain ok"}, None, False),
+])
+def test_schema_a_nested_error_extraction(data, want_err, want_is):
+    ev = normalize_schema_a({"canonicalType": "tool.call.executed", "ts": 9.0,
+                             "actor": {"id": "main"},
+                             "scope": {"sessionKey": "agent:main:s"},
+                             "data": {"toolName": "exec", **data}})
+    assert ev.payload["toolIsError"] is want_is
+    if want_err is not None:
+        assert ev.payload["toolError"] == want_err

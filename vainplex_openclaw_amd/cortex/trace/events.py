@@ -87,8 +87,9 @@ def normalize_schema_a(ev: Dict[str, Any], seq: int = 0) -> Optional[NormalizedE
     elif etype == "tool.result":
         payload["toolName"] = data.get("toolName") or data.get("tool")
         payload["toolResult"] = data.get("result")
-        payload["toolError"] = data.get("error")
-        payload["toolIsError"] = bool(data.get("error")) or bool(data.get("isError"))
+        err = _extract_error_from_result(data)
+        payload["toolError"] = err.get("error") if err["isError"] else data.get("error")
+        payload["toolIsError"] = err["isError"] or bool(data.get("isError"))
     ts = ev.get("ts") or ev.get("timestamp") or 0
     return NormalizedEvent(
         id=str(ev.get("id", "")),
@@ -138,12 +139,129 @@ def normalize_schema_b(ev: Dict[str, Any], seq: int = 0) -> Optional[NormalizedE
     )
 
 
-def normalize_event(ev, seq: int = 0) -> Optional[NormalizedEvent]:
-    """Schema-sniffing normalizer (events.ts normalizeEvent): Schema B
-    carries a `kind` field; everything else is tried as a Schema A
-    ClawEvent envelope."""
+# session-sync "conversation.*" types (events.ts:90-93)
+CONVERSATION_MAP = {
+    "conversation.message.in": "msg.in",
+    "conversation.message.out": "msg.out",
+    "conversation.tool_call": "tool.call",
+    "conversation.tool_result": "tool.result",
+}
+
+
+def detect_schema(ev: Dict[str, Any]) -> Optional[str]:
+    """Schema sniffing (events.ts detectSchema): "B" = session-sync
+    (conversation.* type, meta.source == "session-sync", or a `timestamp`
+    field instead of `ts`), "A" = eventstore envelope, None = unknown.
+    This build's kind-based session-sync shape also reads as "B"."""
     if not isinstance(ev, dict):
         return None
     if "kind" in ev:
-        return normalize_schema_b(ev, seq)
+        return "B"
+    etype = ev.get("type")
+    if not isinstance(etype, str):
+        if ev.get("canonicalType"):
+            return "A"
+        return None
+    if etype.startswith("conversation."):
+        return "B"
+    meta = ev.get("meta")
+    if isinstance(meta, dict) and meta.get("source") == "session-sync":
+        return "B"
+    known = etype in SCHEMA_A_MAP
+    if isinstance(ev.get("ts"), (int, float)) and known:
+        return "A"
+    if isinstance(ev.get("timestamp"), (int, float)):
+        return "B"
+    return "A" if known else None
+
+
+def _extract_error_from_result(payload: Dict[str, Any]) -> Dict[str, Any]:
+    """Nested tool-result error extraction (events.ts:218-247):
+    top-level error > result.details.{error,status,exitCode} >
+    result.isError with content[0].text."""
+    top = payload.get("error")
+    if isinstance(top, str) and top:
+        return {"error": top, "isError": True}
+    result = payload.get("result")
+    if isinstance(result, dict):
+        details = result.get("details")
+        if isinstance(details, dict):
+            derr = details.get("error")
+            if isinstance(derr, str) and derr:
+                return {"error": derr, "isError": True}
+            if details.get("status") == "error":
+                return {"error": "status: error", "isError": True}
+            code = details.get("exitCode")
+            if isinstance(code, (int, float)) and code > 0:
+                return {"error": f"exit code {int(code)}", "isError": True}
+        if result.get("isError") is True:
+            text = None
+            content = result.get("content")
+            if isinstance(content, list) and content and isinstance(content[0], dict):
+                t = content[0].get("text")
+                if isinstance(t, str):
+                    text = t[:500]
+            if text is None and isinstance(result.get("result"), str):
+                text = result["result"][:500]
+            return {"error": text or "unknown error", "isError": True}
+    return {"isError": False}
+
+
+def normalize_schema_b_conv(ev: Dict[str, Any], seq: int = 0) -> Optional[NormalizedEvent]:
+    """session-sync conversation.* shape (events.ts Schema B paths):
+    content from payload.text_preview[0].text, tool call/result under
+    payload.data {name, args, result, isError}."""
+    etype = CONVERSATION_MAP.get(str(ev.get("type", "")))
+    if etype is None:
+        return None
+    ts = ev.get("ts") if isinstance(ev.get("ts"), (int, float)) else ev.get("timestamp")
+    if not isinstance(ts, (int, float)) or ts == 0:
+        return None
+    raw_payload = ev.get("payload") if isinstance(ev.get("payload"), dict) else {}
+    payload: Dict[str, Any] = {}
+    if etype in ("msg.in", "msg.out"):
+        payload["role"] = "user" if etype == "msg.in" else "assistant"
+        tp = raw_payload.get("text_preview")
+        if isinstance(tp, list) and tp and isinstance(tp[0], dict) \
+                and isinstance(tp[0].get("text"), str):
+            payload["content"] = tp[0]["text"]
+        else:
+            payload["content"] = ""
+        if isinstance(raw_payload.get("sessionId"), str):
+            payload["sessionId"] = raw_payload["sessionId"]
+    elif etype == "tool.call":
+        data = raw_payload.get("data") if isinstance(raw_payload.get("data"), dict) else {}
+        payload["toolName"] = data.get("name") if isinstance(data.get("name"), str) else None
+        payload["toolParams"] = data.get("args") if isinstance(data.get("args"), dict) else {}
+    elif etype == "tool.result":
+        data = raw_payload.get("data") if isinstance(raw_payload.get("data"), dict) else {}
+        payload["toolName"] = data.get("name") if isinstance(data.get("name"), str) else None
+        payload["toolResult"] = data.get("result")
+        is_err = data.get("isError") is True
+        payload["toolIsError"] = is_err
+        payload["toolError"] = data.get("result") \
+            if is_err and isinstance(data.get("result"), str) else None
+    session_raw = ev.get("session") if isinstance(ev.get("session"), str) else "unknown"
+    return NormalizedEvent(
+        id=str(ev.get("id", "")),
+        ts=float(ts),
+        agent=str(ev.get("agent") or "unknown"),
+        session=_norm_session(session_raw) if session_raw != "unknown" else "unknown",
+        type=etype,
+        payload=payload,
+        seq=seq,
+    )
+
+
+def normalize_event(ev, seq: int = 0) -> Optional[NormalizedEvent]:
+    """Schema-sniffing normalizer (events.ts normalizeEvent): routes by
+    detect_schema — the kind-based and conversation.* session-sync
+    shapes to the Schema B normalizers, everything else to Schema A."""
+    schema = detect_schema(ev)
+    if schema is None:
+        return None
+    if schema == "B":
+        if "kind" in ev:
+            return normalize_schema_b(ev, seq)
+        return normalize_schema_b_conv(ev, seq)
     return normalize_schema_a(ev, seq)
