@@ -42,6 +42,9 @@ __global__ void topk_scan_fp4_kernel(const uint8_t*, const uint8_t*, const uint8
 __global__ void topk_scan_fp4_v2_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
                                         const uint8_t*, int, int, int, int, int,
                                         float*, int32_t*, const float*, int32_t*, int);
+__global__ void topk_scan_fp4_v3_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
+                                        const uint8_t*, int, int, int, int, int,
+                                        float*, int32_t*, const float*, int32_t*, int);
 
 
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
@@ -377,6 +380,8 @@ std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
   TORCH_CHECK(X4.size(1) == D / 2 && XS.size(0) == nx && XS.size(1) == D / 32);
   TORCH_CHECK(theta.numel() == nq && cap >= 32 && cap <= 4096);
   static const int variant = [] {
+    const char* e3 = getenv("VAINPLEX_FP4_V3");
+    if (e3 != nullptr && e3[0] == '1') return 2;   // 2-pair intervals opt-in
     const char* e = getenv("VAINPLEX_FP4_V2");
     return (e != nullptr && e[0] == '1') ? 1 : 0;  // pipelined v2 opt-in
   }();
@@ -393,7 +398,8 @@ std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
   auto cand_i = torch::full({(long long)nq, cap}, -1, i32opts);
   auto counts = torch::zeros({(long long)nq}, i32opts);
   dim3 grid((unsigned)(n_qblocks * n_swaths));
-  auto kern = variant == 1 ? topk_scan_fp4_v2_kernel : topk_scan_fp4_kernel;
+  auto kern = variant == 2 ? topk_scan_fp4_v3_kernel
+            : variant == 1 ? topk_scan_fp4_v2_kernel : topk_scan_fp4_kernel;
   hipLaunchKernelGGL(kern, grid, dim3(512), 0, cur_stream(),
                      Q4.data_ptr<uint8_t>(), QS.data_ptr<uint8_t>(),
                      X4.data_ptr<uint8_t>(), XS.data_ptr<uint8_t>(),

@@ -1492,3 +1492,175 @@ topk_scan_fp4_v2_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restric
         }
   }
 }
+
+// ===========================================================================
+// fp4x4 threshold scan v3: 2-pair barrier intervals. v1 pays one full
+// s_barrier + vmcnt gate per K-pair (np=8 at D=1024); the measured gap
+// to the MFMA floor is per-pair overhead amortized over a 32-MFMA burst
+// (NOTES-NEXT). v3 keeps v1's fragment reads and MFMA body but consumes
+// TWO consecutive pairs per barrier interval: one vmcnt wait + one
+// barrier per 64 MFMAs. Buffer discipline: interval p consumes buffers
+// p&3,(p+1)&3 and stages pairs p+2,p+3 into (p+2)&3,(p+3)&3 — disjoint,
+// and the interval-entry barrier keeps the previous interval's readers
+// ahead of this interval's overwrites (same invariant as v1, half the
+// barriers). No extra VGPR: fragments for the two pairs are read
+// sequentially, not held concurrently.
+// ===========================================================================
+
+extern "C" __global__ void __launch_bounds__(TK_THREADS)
+topk_scan_fp4_v3_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__ QS,
+                        const uint8_t* __restrict__ X4, const uint8_t* __restrict__ XS,
+                        int nq, int nx, int D, int k, int n_swaths,
+                        float* __restrict__ cand_scores,
+                        int32_t* __restrict__ cand_ids,
+                        const float* __restrict__ theta,
+                        int32_t* __restrict__ tc_n, int cap) {
+  __shared__ bf16 lds_q[4 * BM * BK];
+  __shared__ bf16 lds_x[4 * BN * BK];
+  __shared__ uint8_t lds_qs[BM * 32];
+  __shared__ uint8_t lds_xs2[BN * 32];
+  __shared__ float row_min[BM];
+#define QP4A(buf) (lds_q + (buf) * BM * BK)
+#define XP4A(buf) (lds_x + (buf) * BN * BK)
+
+  int S = n_swaths;
+  int qb = blockIdx.x / S;
+  int swath = blockIdx.x % S;
+  long long row0 = (long long)qb * BM;
+
+  long long per = ((long long)nx + S - 1) / S;
+  per = ((per + BN - 1) / BN) * BN;
+  long long x_begin = (long long)swath * per;
+  long long x_end = min((long long)nx, x_begin + per);
+
+  for (int i = threadIdx.x; i < BM; i += blockDim.x)
+    row_min[i] = (theta != nullptr && row0 + i < nq) ? theta[row0 + i] : -1e30f;
+  __syncthreads();
+
+  int wid = wave_id();
+  int wm = wid >> 2, wn = wid & 3;
+  int lane = lane_id();
+  int lrow = lane & 15;
+  int kgrp = lane >> 4;
+  int np = D / (2 * BK_F8);
+  int sb = D / 32;
+  size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
+  long long p4_ld = D / 4;
+
+  for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
+    f32x4 acc[8][4] = {};
+    stage_scale_rows(QS, sb, row0, (long long)nq, lds_qs, BM);
+    stage_scale_rows(XS, sb, x0, (long long)nx, lds_xs2, BN);
+    // prologue: stage the first interval's two pairs
+    for (int pp = 0; pp < 2 && pp < np; ++pp) {
+      stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, pp * 32, QP4A(pp), BM);
+      stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, pp * 32, XP4A(pp), BN);
+    }
+    for (int p = 0; p < np; p += 2) {
+      int pe = min(p + 2, np);
+      // all outstanding glds are exactly this interval's pairs
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      // stage the NEXT interval (pairs p+2, p+3) into the buffers the
+      // PREVIOUS interval consumed; overlaps this interval's compute
+      for (int s = p + 2; s < min(np, p + 4); ++s) {
+        stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, s * 32,
+                   QP4A(s & 3), BM);
+        stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, s * 32,
+                   XP4A(s & 3), BN);
+      }
+      for (int pc = p; pc < pe; ++pc) {
+        uint32_t xrow_base = (uint32_t)(wn * 64 + lrow);
+        uint32_t qrow_base = (uint32_t)(wm * 128 + lrow);
+        uint32_t xaddr = (uint32_t)(size_t)XP4A(pc & 3)
+                         + lds_off_bytes(xrow_base, (uint32_t)kgrp);
+        uint32_t qaddr = (uint32_t)(size_t)QP4A(pc & 3)
+                         + lds_off_bytes(qrow_base, (uint32_t)kgrp);
+        uint32_t sst = 16u * (uint32_t)sb;
+        uint32_t xs_a = (uint32_t)(size_t)lds_xs2 + xrow_base * (uint32_t)sb
+                        + (uint32_t)(pc * 4 + kgrp);
+        uint32_t qs_a = (uint32_t)(size_t)lds_qs + qrow_base * (uint32_t)sb
+                        + (uint32_t)(pc * 4 + kgrp);
+        bf16x8 xf[4], qf[8];
+        uint32_t xs_v[4], qs_v[8];
+        asm volatile(
+            "ds_read_b128 %0, %8\n\t"
+            "ds_read_b128 %1, %8 offset:1024\n\t"
+            "ds_read_b128 %2, %8 offset:2048\n\t"
+            "ds_read_b128 %3, %8 offset:3072\n\t"
+            "ds_read_u8 %4, %9\n\t"
+            "ds_read_u8 %5, %10\n\t"
+            "ds_read_u8 %6, %11\n\t"
+            "ds_read_u8 %7, %12\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3]),
+              "=&v"(xs_v[0]), "=&v"(xs_v[1]), "=&v"(xs_v[2]), "=&v"(xs_v[3])
+            : "v"(xaddr), "v"(xs_a), "v"(xs_a + sst), "v"(xs_a + 2 * sst),
+              "v"(xs_a + 3 * sst));
+        asm volatile(
+            "ds_read_b128 %0, %16\n\t"
+            "ds_read_b128 %1, %16 offset:1024\n\t"
+            "ds_read_b128 %2, %16 offset:2048\n\t"
+            "ds_read_b128 %3, %16 offset:3072\n\t"
+            "ds_read_b128 %4, %16 offset:4096\n\t"
+            "ds_read_b128 %5, %16 offset:5120\n\t"
+            "ds_read_b128 %6, %16 offset:6144\n\t"
+            "ds_read_b128 %7, %16 offset:7168\n\t"
+            "ds_read_u8 %8, %17\n\t"
+            "ds_read_u8 %9, %18\n\t"
+            "ds_read_u8 %10, %19\n\t"
+            "ds_read_u8 %11, %20\n\t"
+            "ds_read_u8 %12, %21\n\t"
+            "ds_read_u8 %13, %22\n\t"
+            "ds_read_u8 %14, %23\n\t"
+            "ds_read_u8 %15, %24\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
+              "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7]),
+              "=&v"(qs_v[0]), "=&v"(qs_v[1]), "=&v"(qs_v[2]), "=&v"(qs_v[3]),
+              "=&v"(qs_v[4]), "=&v"(qs_v[5]), "=&v"(qs_v[6]), "=&v"(qs_v[7])
+            : "v"(qaddr), "v"(qs_a), "v"(qs_a + sst), "v"(qs_a + 2 * sst),
+              "v"(qs_a + 3 * sst), "v"(qs_a + 4 * sst), "v"(qs_a + 5 * sst),
+              "v"(qs_a + 6 * sst), "v"(qs_a + 7 * sst));
+        v8i_mx xv[4], qv[8];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) xv[n] = fp4_frag(xf[n]);
+#pragma unroll
+        for (int m = 0; m < 8; ++m) qv[m] = fp4_frag(qf[m]);
+#pragma unroll
+        for (int m = 0; m < 8; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                qv[m], xv[n], acc[m][n], 4 /*A fp4*/, 4 /*B fp4*/,
+                0, (int)qs_v[m], 0, (int)xs_v[n]);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    if (k < 0) {
+      if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+          float v = acc[m][n][r];
+          if (!(v > row_min[row])) continue;
+          long long grow = row0 + row;
+          if (grow >= nq) continue;
+          long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+          if (col >= x_end) continue;
+          int pos = atomicAdd(&tc_n[grow], 1);
+          if (pos < cap) {
+            cand_scores[grow * cap + pos] = v;
+            cand_ids[grow * cap + pos] = int32_t(col);
+          }
+        }
+  }
+}
