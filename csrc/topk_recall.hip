@@ -1664,3 +1664,4 @@ topk_scan_fp4_v3_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restric
         }
   }
 }
+
