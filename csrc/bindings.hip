@@ -45,6 +45,9 @@ __global__ void topk_scan_fp4_v2_kernel(const uint8_t*, const uint8_t*, const ui
 __global__ void topk_scan_fp4_v3_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
                                         const uint8_t*, int, int, int, int, int,
                                         float*, int32_t*, const float*, int32_t*, int);
+__global__ void topk_scan_fp4_v5_kernel(const uint8_t*, const uint8_t*, const uint8_t*,
+                                        const uint8_t*, int, int, int, int, int,
+                                        float*, int32_t*, const float*, int32_t*, int);
 
 
 __global__ void topk_merge_kernel(const float*, const int32_t*, int, int, int,
@@ -384,6 +387,7 @@ std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
   static const int variant = [] {
     if (const char* e = getenv("VAINPLEX_FP4_V1"); e && e[0] == '1') return 0;
     if (const char* e = getenv("VAINPLEX_FP4_V2"); e && e[0] == '1') return 1;
+    if (const char* e = getenv("VAINPLEX_FP4_V5"); e && e[0] == '1') return 4;
     return 2;
   }();
   int n_qblocks = (nq + 255) / 256;
@@ -399,7 +403,8 @@ std::vector<torch::Tensor> topk_scan_threshold_fp4x4(
   auto cand_i = torch::full({(long long)nq, cap}, -1, i32opts);
   auto counts = torch::zeros({(long long)nq}, i32opts);
   dim3 grid((unsigned)(n_qblocks * n_swaths));
-  auto kern = variant == 2 ? topk_scan_fp4_v3_kernel
+  auto kern = variant == 4 ? topk_scan_fp4_v5_kernel
+            : variant == 2 ? topk_scan_fp4_v3_kernel
             : variant == 1 ? topk_scan_fp4_v2_kernel : topk_scan_fp4_kernel;
   hipLaunchKernelGGL(kern, grid, dim3(512), 0, cur_stream(),
                      Q4.data_ptr<uint8_t>(), QS.data_ptr<uint8_t>(),

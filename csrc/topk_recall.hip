@@ -1665,3 +1665,177 @@ topk_scan_fp4_v3_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restric
   }
 }
 
+
+// ===========================================================================
+// fp4x4 threshold scan v5: v3's 2-pair intervals + tile-resident scales.
+// Q block-scales are invariant across x tiles — staged transposed
+// (SCALE_PITCH sheets) once per BLOCK, read as 8 conflict-free b64 per
+// lane, then every pair derives its scale byte with VALU shifts. X
+// scales load as 4 b64 per tile after the first interval barrier. This
+// deletes all 12 per-pair ds_read_u8 (the PMC-measured 6.0e9
+// bank-conflict replay cycles) at +24 VGPR of block-lifetime scalars.
+// ===========================================================================
+
+extern "C" __global__ void __launch_bounds__(TK_THREADS)
+topk_scan_fp4_v5_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restrict__ QS,
+                        const uint8_t* __restrict__ X4, const uint8_t* __restrict__ XS,
+                        int nq, int nx, int D, int k, int n_swaths,
+                        float* __restrict__ cand_scores,
+                        int32_t* __restrict__ cand_ids,
+                        const float* __restrict__ theta,
+                        int32_t* __restrict__ tc_n, int cap) {
+  __shared__ bf16 lds_q[4 * BM * BK];
+  __shared__ bf16 lds_x[4 * BN * BK];
+  __shared__ uint8_t lds_qs_t[BM * SCALE_PITCH];
+  __shared__ uint8_t lds_xs_t[BN * SCALE_PITCH];
+  __shared__ float row_min[BM];
+#define QP4C(buf) (lds_q + (buf) * BM * BK)
+#define XP4C(buf) (lds_x + (buf) * BN * BK)
+
+  int S = n_swaths;
+  int qb = blockIdx.x / S;
+  int swath = blockIdx.x % S;
+  long long row0 = (long long)qb * BM;
+
+  long long per = ((long long)nx + S - 1) / S;
+  per = ((per + BN - 1) / BN) * BN;
+  long long x_begin = (long long)swath * per;
+  long long x_end = min((long long)nx, x_begin + per);
+
+  for (int i = threadIdx.x; i < BM; i += blockDim.x)
+    row_min[i] = (theta != nullptr && row0 + i < nq) ? theta[row0 + i] : -1e30f;
+  __syncthreads();
+
+  int wid = wave_id();
+  int wm = wid >> 2, wn = wid & 3;
+  int lane = lane_id();
+  int lrow = lane & 15;
+  int kgrp = lane >> 4;
+  int np = D / (2 * BK_F8);
+  int sb = D / 32;
+  size_t cbase = (((size_t)qb * S) + swath) * (size_t)BM * k;
+  long long p4_ld = D / 4;
+
+  // Q block scales are invariant across x tiles: stage the transposed
+  // sheet once, read each lane's 8 m-row b64 slices once, and derive
+  // per-pair scale bytes with VALU shifts — zero per-pair LDS traffic
+  // (v1 paid 12 bank-conflicted ds_read_u8 per pair, PMC 6.0e9 replay
+  // cycles).
+  stage_scale_rows_t(QS, sb, row0, (long long)nq, lds_qs_t, BM);
+  __syncthreads();
+  uint32_t qslo[8], qshi[8];
+  {
+    uint32_t qrow_base0 = (uint32_t)(wm * 128 + (lane_id() & 15));
+#pragma unroll
+    for (int m = 0; m < 8; ++m) {
+      const uint8_t* a = lds_qs_t + (qrow_base0 + m * 16) * SCALE_PITCH
+                         + (lane_id() >> 4) * 8;
+      qslo[m] = *(const uint32_t*)a;
+      qshi[m] = *(const uint32_t*)(a + 4);
+    }
+  }
+
+  for (long long x0 = x_begin; x0 < x_end; x0 += BN) {
+    f32x4 acc[8][4] = {};
+    uint32_t xslo[4], xshi[4];
+    stage_scale_rows_t(XS, sb, x0, (long long)nx, lds_xs_t, BN);
+    // prologue: stage the first interval's two pairs
+    for (int pp = 0; pp < 2 && pp < np; ++pp) {
+      stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, pp * 32, QP4C(pp), BM);
+      stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, pp * 32, XP4C(pp), BN);
+    }
+    for (int p = 0; p < np; p += 2) {
+      int pe = min(p + 2, np);
+      // all outstanding glds are exactly this interval's pairs
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      // stage the NEXT interval (pairs p+2, p+3) into the buffers the
+      // PREVIOUS interval consumed; overlaps this interval's compute
+      for (int s = p + 2; s < min(np, p + 4); ++s) {
+        stage_tile((const bf16*)Q4, p4_ld, row0, (long long)nq, s * 32,
+                   QP4C(s & 3), BM);
+        stage_tile((const bf16*)X4, p4_ld, x0, (long long)nx, s * 32,
+                   XP4C(s & 3), BN);
+      }
+      if (p == 0) {
+        // x scale sheet staged before the interval-0 barrier above
+        uint32_t xrb = (uint32_t)(wn * 64 + lrow);
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const uint8_t* a = lds_xs_t + (xrb + n * 16) * SCALE_PITCH + kgrp * 8;
+          xslo[n] = *(const uint32_t*)a;
+          xshi[n] = *(const uint32_t*)(a + 4);
+        }
+      }
+      for (int pc = p; pc < pe; ++pc) {
+        uint32_t xrow_base = (uint32_t)(wn * 64 + lrow);
+        uint32_t qrow_base = (uint32_t)(wm * 128 + lrow);
+        uint32_t xaddr = (uint32_t)(size_t)XP4C(pc & 3)
+                         + lds_off_bytes(xrow_base, (uint32_t)kgrp);
+        uint32_t qaddr = (uint32_t)(size_t)QP4C(pc & 3)
+                         + lds_off_bytes(qrow_base, (uint32_t)kgrp);
+        bf16x8 xf[4], qf[8];
+        asm volatile(
+            "ds_read_b128 %0, %4\n\t"
+            "ds_read_b128 %1, %4 offset:1024\n\t"
+            "ds_read_b128 %2, %4 offset:2048\n\t"
+            "ds_read_b128 %3, %4 offset:3072"
+            : "=&v"(xf[0]), "=&v"(xf[1]), "=&v"(xf[2]), "=&v"(xf[3])
+            : "v"(xaddr));
+        asm volatile(
+            "ds_read_b128 %0, %8\n\t"
+            "ds_read_b128 %1, %8 offset:1024\n\t"
+            "ds_read_b128 %2, %8 offset:2048\n\t"
+            "ds_read_b128 %3, %8 offset:3072\n\t"
+            "ds_read_b128 %4, %8 offset:4096\n\t"
+            "ds_read_b128 %5, %8 offset:5120\n\t"
+            "ds_read_b128 %6, %8 offset:6144\n\t"
+            "ds_read_b128 %7, %8 offset:7168\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(qf[0]), "=&v"(qf[1]), "=&v"(qf[2]), "=&v"(qf[3]),
+              "=&v"(qf[4]), "=&v"(qf[5]), "=&v"(qf[6]), "=&v"(qf[7])
+            : "v"(qaddr));
+        v8i_mx xv[4], qv[8];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) xv[n] = fp4_frag(xf[n]);
+#pragma unroll
+        for (int m = 0; m < 8; ++m) qv[m] = fp4_frag(qf[m]);
+#pragma unroll
+        for (int m = 0; m < 8; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                qv[m], xv[n], acc[m][n], 4 /*A fp4*/, 4 /*B fp4*/,
+                0, (int)scale_byte(qslo[m], qshi[m], pc), 0,
+                (int)scale_byte(xslo[n], xshi[n], pc));
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    if (k < 0) {
+      if (acc[0][0][0] > 1e29f) cand_scores[cbase] = acc[0][0][0];
+      continue;
+    }
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wm * 128 + m * 16 + (lane >> 4) * 4 + r;
+          float v = acc[m][n][r];
+          if (!(v > row_min[row])) continue;
+          long long grow = row0 + row;
+          if (grow >= nq) continue;
+          long long col = x0 + wn * 64 + n * 16 + (lane & 15);
+          if (col >= x_end) continue;
+          int pos = atomicAdd(&tc_n[grow], 1);
+          if (pos < cap) {
+            cand_scores[grow * cap + pos] = v;
+            cand_ids[grow * cap + pos] = int32_t(col);
+          }
+        }
+  }
+}
+
