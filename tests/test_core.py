@@ -197,3 +197,115 @@ def test_async_audit_writer_cpu(tmp_path):
     assert len(lines) == 2
     m = _json.loads(lines[0])
     assert m["count"] == 2 and m["root"] == "07" * 32
+
+
+# -- shared layered config loader (KE/nats/sitrep config-loader.ts) ----------
+
+def test_layered_legacy_inline_wins(tmp_path):
+    from vainplex_openclaw_amd.core.config import load_layered_config
+
+    defaults = {"enabled": True, "depth": 3, "nested": {"a": 1}}
+    # extra key beyond enabled/configPath -> legacy full inline config
+    cfg = load_layered_config("p1", {"enabled": False, "depth": 9},
+                              defaults, home=str(tmp_path))
+    assert cfg["depth"] == 9 and cfg["enabled"] is False
+    assert cfg["nested"] == {"a": 1}                 # defaults still resolve
+    # and no external file was touched
+    import os
+    assert not os.path.exists(str(tmp_path / "plugins" / "p1" / "config.json"))
+
+
+def test_layered_pointer_loads_config_path(tmp_path):
+    import json
+
+    from vainplex_openclaw_amd.core.config import load_layered_config
+
+    ext = tmp_path / "ext.json"
+    ext.write_text(json.dumps({"depth": 7}))
+    cfg = load_layered_config("p1", {"configPath": str(ext)},
+                              {"enabled": True, "depth": 3}, home=str(tmp_path))
+    assert cfg["depth"] == 7 and cfg["enabled"] is True
+
+
+def test_layered_inline_enabled_overrides_file(tmp_path):
+    import json
+
+    from vainplex_openclaw_amd.core.config import load_layered_config
+
+    ext = tmp_path / "ext.json"
+    ext.write_text(json.dumps({"enabled": True, "depth": 7}))
+    cfg = load_layered_config("p1", {"enabled": False, "configPath": str(ext)},
+                              {"enabled": True, "depth": 3})
+    assert cfg["enabled"] is False and cfg["depth"] == 7
+
+
+def test_layered_malformed_and_non_object_fall_back(tmp_path):
+    from vainplex_openclaw_amd.core.config import load_layered_config
+
+    class Log:
+        def __init__(self):
+            self.warns = []
+
+        def warn(self, m):
+            self.warns.append(m)
+
+    bad = tmp_path / "bad.json"
+    bad.write_text("{corrupt!!")
+    log = Log()
+    cfg = load_layered_config("p1", {"configPath": str(bad)},
+                              {"enabled": True, "depth": 3}, logger=log)
+    assert cfg == {"enabled": True, "depth": 3}
+    arr = tmp_path / "arr.json"
+    arr.write_text("[1, 2]")
+    cfg2 = load_layered_config("p1", {"configPath": str(arr)},
+                               {"enabled": True, "depth": 3}, logger=log)
+    assert cfg2["depth"] == 3
+    assert any("not an object" in w for w in log.warns)
+
+
+def test_layered_bootstraps_default_file(tmp_path):
+    import json
+    import os
+
+    from vainplex_openclaw_amd.core.config import load_layered_config
+
+    defaults = {"enabled": True, "depth": 3}
+    cfg = load_layered_config("kp", {}, defaults, home=str(tmp_path),
+                              bootstrap=True)
+    assert cfg == defaults
+    path = os.path.join(str(tmp_path), "plugins", "kp", "config.json")
+    assert json.load(open(path)) == defaults          # written to disk
+    # second load reads it back
+    cfg2 = load_layered_config("kp", None, defaults, home=str(tmp_path))
+    assert cfg2 == defaults
+
+
+def test_layered_no_bootstrap_returns_defaults(tmp_path):
+    import os
+
+    from vainplex_openclaw_amd.core.config import load_layered_config
+
+    cfg = load_layered_config("kp", None, {"enabled": True, "x": 1},
+                              home=str(tmp_path), bootstrap=False)
+    assert cfg == {"enabled": True, "x": 1}
+    assert not os.path.exists(os.path.join(str(tmp_path), "plugins", "kp"))
+
+
+def test_ke_resolve_config_uses_layered_loader(tmp_path):
+    import json
+    import os
+
+    from vainplex_openclaw_amd.knowledge.config import DEFAULT_CONFIG, resolve_config
+
+    # pointer + external file
+    p = os.path.join(str(tmp_path), "plugins", "openclaw-knowledge-engine")
+    os.makedirs(p)
+    json.dump({"storage": {"maxEntities": 42}}, open(os.path.join(p, "config.json"), "w"))
+    cfg = resolve_config({"enabled": False}, home=str(tmp_path))
+    assert cfg["storage"]["maxEntities"] == 42
+    assert cfg["storage"]["maxFacts"] == DEFAULT_CONFIG["storage"]["maxFacts"]
+    assert cfg["enabled"] is False
+    # legacy full inline
+    cfg2 = resolve_config({"enabled": True, "storage": {"maxEntities": 5}},
+                          home=str(tmp_path / "other"))
+    assert cfg2["storage"]["maxEntities"] == 5

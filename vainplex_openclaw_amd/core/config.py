@@ -133,3 +133,51 @@ def plugin_enabled(openclaw_config: Dict[str, Any], plugin_id: str) -> bool:
     if isinstance(allow, list):
         return plugin_id in allow
     return True
+
+
+def load_layered_config(
+    plugin_id: str,
+    plugin_config: Optional[Dict[str, Any]],
+    defaults: Dict[str, Any],
+    home: Optional[str] = None,
+    logger: Any = None,
+    bootstrap: bool = True,
+) -> Dict[str, Any]:
+    """The shared config-loader pattern (knowledge-engine / nats /
+    sitrep `src/config-loader.ts`): the host's pluginConfig is either a
+    LEGACY full inline config (any key beyond enabled/configPath -> use
+    it directly) or a minimal pointer {enabled?, configPath?}; the real
+    config lives in an external JSON file (configPath override or
+    ~/.openclaw/plugins/<id>/config.json), which is BOOTSTRAPPED from
+    the defaults when missing; a malformed or non-object file falls
+    back to defaults; an inline `enabled` boolean always overrides the
+    file's. Result is defaults-resolved."""
+    raw = plugin_config if isinstance(plugin_config, dict) else {}
+    if raw and any(k not in ("enabled", "configPath") for k in raw):
+        return resolve_defaults(raw, defaults)  # legacy inline config
+    path = raw.get("configPath") if isinstance(raw.get("configPath"), str) \
+        else plugin_config_path(plugin_id, home)
+    data: Optional[Dict[str, Any]] = None
+    try:
+        data = load_json_file(path)
+        if data is None and os.path.isfile(path) and logger is not None:
+            logger.warn(f"[{plugin_id}] Config file is not an object: {path}")
+    except Exception as exc:
+        if logger is not None:
+            logger.warn(f"[{plugin_id}] Failed to read config file {path}: {exc}")
+        data = None
+    if data is None and bootstrap and not os.path.exists(path):
+        try:
+            os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
+            tmp = path + ".tmp"
+            with open(tmp, "w", encoding="utf-8") as fh:
+                json.dump(defaults, fh, indent=2)
+            os.replace(tmp, path)
+            data = dict(defaults)
+        except OSError:
+            data = None
+    if data is None:
+        data = {}
+    if isinstance(raw.get("enabled"), bool):
+        data = {**data, "enabled": raw["enabled"]}
+    return resolve_defaults(data, defaults)

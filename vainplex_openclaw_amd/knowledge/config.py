@@ -39,12 +39,14 @@ DEFAULT_CONFIG: Dict[str, Any] = {
 
 
 def resolve_config(
-    plugin_config: Optional[Dict[str, Any]] = None, home: Optional[str] = None
+    plugin_config: Optional[Dict[str, Any]] = None, home: Optional[str] = None,
+    logger: Any = None, bootstrap: bool = False,
 ) -> Dict[str, Any]:
-    inline = dict(plugin_config or {})
-    raw = load_plugin_config(PLUGIN_ID, fallback=inline, home=home)
-    cfg = resolve_defaults(raw, DEFAULT_CONFIG)
-    # inline `enabled` always wins (config-loader.ts applyInlineOverrides)
-    if isinstance(inline.get("enabled"), bool):
-        cfg["enabled"] = inline["enabled"]
-    return cfg
+    """Full config-loader.ts semantics via the shared layered loader:
+    legacy full inline config wins outright; a minimal {enabled?,
+    configPath?} pointer loads (and optionally bootstraps) the external
+    file; inline `enabled` overrides the file's."""
+    from ..core.config import load_layered_config
+
+    return load_layered_config(PLUGIN_ID, plugin_config, DEFAULT_CONFIG,
+                               home=home, logger=logger, bootstrap=bootstrap)
