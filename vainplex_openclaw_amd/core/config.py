@@ -152,9 +152,25 @@ def load_layered_config(
     the defaults when missing; a malformed or non-object file falls
     back to defaults; an inline `enabled` boolean always overrides the
     file's. Result is defaults-resolved."""
+    data = load_raw_layered(plugin_id, plugin_config, home=home,
+                            logger=logger,
+                            bootstrap_defaults=defaults if bootstrap else None)
+    return resolve_defaults(data, defaults)
+
+
+def load_raw_layered(
+    plugin_id: str,
+    plugin_config: Optional[Dict[str, Any]],
+    home: Optional[str] = None,
+    logger: Any = None,
+    bootstrap_defaults: Optional[Dict[str, Any]] = None,
+) -> Dict[str, Any]:
+    """The layered loading WITHOUT defaults-resolution — for plugins
+    whose resolve_config does its own per-field typed picks (the
+    eventstore / sitrep pattern)."""
     raw = plugin_config if isinstance(plugin_config, dict) else {}
     if raw and any(k not in ("enabled", "configPath") for k in raw):
-        return resolve_defaults(raw, defaults)  # legacy inline config
+        return dict(raw)  # legacy inline config
     path = raw.get("configPath") if isinstance(raw.get("configPath"), str) \
         else plugin_config_path(plugin_id, home)
     data: Optional[Dict[str, Any]] = None
@@ -166,18 +182,18 @@ def load_layered_config(
         if logger is not None:
             logger.warn(f"[{plugin_id}] Failed to read config file {path}: {exc}")
         data = None
-    if data is None and bootstrap and not os.path.exists(path):
+    if data is None and bootstrap_defaults is not None and not os.path.exists(path):
         try:
             os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
             tmp = path + ".tmp"
             with open(tmp, "w", encoding="utf-8") as fh:
-                json.dump(defaults, fh, indent=2)
+                json.dump(bootstrap_defaults, fh, indent=2)
             os.replace(tmp, path)
-            data = dict(defaults)
+            data = dict(bootstrap_defaults)
         except OSError:
             data = None
     if data is None:
         data = {}
     if isinstance(raw.get("enabled"), bool):
         data = {**data, "enabled": raw["enabled"]}
-    return resolve_defaults(data, defaults)
+    return data

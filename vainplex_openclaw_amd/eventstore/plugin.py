@@ -10,7 +10,7 @@ from __future__ import annotations
 from typing import Any, Dict, Optional
 
 from ..core.api import PluginApi
-from ..core.config import load_plugin_config
+from ..core.config import load_raw_layered
 from .config import PLUGIN_ID, resolve_config
 from .hooks import EventPublisher
 from .journal import EventJournal
@@ -31,7 +31,7 @@ class EventStorePlugin:
         self.config: Dict[str, Any] = {}
 
     def register(self, api: PluginApi) -> None:
-        cfg = resolve_config(load_plugin_config(PLUGIN_ID, fallback=api.plugin_config))
+        cfg = resolve_config(load_raw_layered(PLUGIN_ID, api.plugin_config))
         self.config = cfg
         if not cfg["enabled"]:
             api.logger.info("[nats-eventstore] Disabled via config")

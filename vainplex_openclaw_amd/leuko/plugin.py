@@ -14,7 +14,7 @@ import time
 from typing import Any, Dict, Optional
 
 from ..core.api import PluginApi
-from ..core.config import load_plugin_config
+from ..core.config import load_raw_layered
 from .aggregator import generate_sitrep, write_sitrep
 from .anomaly import AnomalyDetector, MetricHistory
 
@@ -120,7 +120,7 @@ class LeukoPlugin:
             self._timer = None
 
     def register(self, api: PluginApi) -> None:
-        self.config = resolve_config(load_plugin_config(PLUGIN_ID, fallback=api.plugin_config))
+        self.config = resolve_config(load_raw_layered(PLUGIN_ID, api.plugin_config))
         if not self.config["enabled"]:
             return
         hist_path = os.path.join(self.workspace, "memory", "leuko", "metrics.jsonl")
