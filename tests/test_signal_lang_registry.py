@@ -227,3 +227,65 @@ def test_registry_extends_correction_detection():
     reg.register_pack(_custom_pack())
     found = detect_correction(chain, registry=reg)
     assert len(found) == 1 and found[0].signal_type == "correction"
+
+
+# -- per-language pack validation (signal-langs.test.ts: minimum pattern
+#    requirements + representative phrases per language, CJK without \b) ------
+
+_REP = {
+    "en": {"correction": "that's wrong", "dissatisfaction": "this is not helpful",
+           "completion": "all set", "systemState": "there are 3 errors"},
+    "de": {"correction": "nicht was ich meinte", "dissatisfaction": "das hilft nicht",
+           "completion": "erledigt", "systemState": "es gibt 5 fehler"},
+    "es": {"correction": "eso está mal", "dissatisfaction": "sigues fallando",
+           "completion": "ya funciona", "systemState": "hay 2 errores"},
+    "fr": {"correction": "pas ce que j'ai demandé", "dissatisfaction": "laisse tomber",
+           "completion": "ça marche maintenant", "systemState": "il y a 4 erreurs"},
+    "it": {"correction": "è sbagliato", "dissatisfaction": "lascia perdere",
+           "completion": "ora funziona", "systemState": "ci sono 2 errori"},
+    "pt": {"correction": "está errado", "dissatisfaction": "você continua falhando",
+           "completion": "funciona agora", "systemState": "há 3 erros"},
+    "ru": {"correction": "это неверно", "dissatisfaction": "это не помогает",
+           "completion": "теперь работает", "systemState": "есть 2 ошибки"},
+    "ja": {"correction": "そうじゃない", "dissatisfaction": "役に立たない",
+           "completion": "できました", "systemState": "エラーが3件"},
+    "ko": {"correction": "그게 아니에요", "dissatisfaction": "도움이 안 돼요",
+           "completion": "작동합니다", "systemState": "오류가 2개"},
+    "zh": {"correction": "不是我要的", "dissatisfaction": "帮不上忙",
+           "completion": "可以用了", "systemState": "有3个错误"},
+}
+
+
+@pytest.mark.parametrize("code", BUILTIN_SIGNAL_LANGUAGES)
+def test_pack_minimum_pattern_requirements(code):
+    pack = builtin_pack(code)
+    assert pack is not None and pack.code == code
+    assert pack.name and pack.name_en
+    assert len(pack.correction["indicators"]) >= 4
+    assert len(pack.correction["shortNegatives"]) >= 1
+    assert len(pack.dissatisfaction["indicators"]) >= 4
+    assert len(pack.dissatisfaction["satisfactionOverrides"]) >= 2
+    assert len(pack.dissatisfaction["resolutionIndicators"]) >= 1
+    assert len(pack.completion["claims"]) >= 4
+    assert len(pack.system_state["claims"]) >= 1
+    assert len(pack.system_state["opinionExclusions"]) >= 2
+    assert len(pack.question["indicators"]) >= 2
+
+
+@pytest.mark.parametrize("code", BUILTIN_SIGNAL_LANGUAGES)
+def test_pack_matches_representative_phrases(code):
+    pack = builtin_pack(code)
+    rep = _REP[code]
+    assert _any(pack.correction["indicators"], rep["correction"]), code
+    assert _any(pack.dissatisfaction["indicators"], rep["dissatisfaction"]), code
+    assert _any(pack.completion["claims"], rep["completion"]), code
+    assert _any(pack.system_state["claims"], rep["systemState"]), code
+
+
+@pytest.mark.parametrize("code", ["ja", "ko", "zh"])
+def test_cjk_packs_have_no_word_boundaries(code):
+    pack = builtin_pack(code)
+    for cat in ("correction", "question", "dissatisfaction", "completion"):
+        for rxs in pack.category(cat).values():
+            for rx in rxs:
+                assert r"\b" not in rx.pattern, (code, rx.pattern)
