@@ -372,3 +372,68 @@ def test_directory_growth_warn(tmp_path):
     t[0] += 36.0                                       # ~200 MB/h over 3 points
     it = det.check_directory_growth(str(d), warn_mb_per_hour=100.0)
     assert it is not None and "growing" in it["title"]
+
+
+# -- resolve_config tables (sitrep test/config.test.ts, 10 its) ---------------
+
+def test_resolve_config_defaults_for_none_and_empty(tmp_path):
+    from vainplex_openclaw_amd.leuko.plugin import default_config, resolve_config
+
+    d = default_config(str(tmp_path))
+    for raw in (None, {}):
+        c = resolve_config(raw, home=str(tmp_path))
+        assert c["enabled"] == d["enabled"]
+        assert set(c["collectors"]) == set(d["collectors"])
+
+
+def test_resolve_config_scalar_overrides(tmp_path):
+    from vainplex_openclaw_amd.leuko.plugin import resolve_config
+
+    c = resolve_config({"enabled": False, "outputPath": "/x/sitrep.md",
+                        "intervalMinutes": 15}, home=str(tmp_path))
+    assert c["enabled"] is False
+    assert c["outputPath"] == "/x/sitrep.md"
+    assert c["intervalMinutes"] == 15
+
+
+def test_resolve_config_merges_collector_configs(tmp_path):
+    from vainplex_openclaw_amd.leuko.plugin import default_config, resolve_config
+
+    c = resolve_config({"collectors": {"goals": {"enabled": True,
+                                                 "staleDays": 9}}},
+                       home=str(tmp_path))
+    assert c["collectors"]["goals"]["enabled"] is True
+    assert c["collectors"]["goals"]["staleDays"] == 9
+    # untouched collectors keep their defaults
+    d = default_config(str(tmp_path))
+    for name, base in d["collectors"].items():
+        if name != "goals":
+            assert c["collectors"][name] == base
+
+
+def test_resolve_config_ignores_non_object_collectors(tmp_path):
+    from vainplex_openclaw_amd.leuko.plugin import default_config, resolve_config
+
+    c = resolve_config({"collectors": {"goals": "yes", "errors": 3}},
+                       home=str(tmp_path))
+    d = default_config(str(tmp_path))
+    assert c["collectors"]["goals"] == d["collectors"]["goals"]
+    assert c["collectors"]["errors"] == d["collectors"]["errors"]
+
+
+def test_resolve_config_scoring_and_custom_collectors(tmp_path):
+    from vainplex_openclaw_amd.leuko.plugin import resolve_config
+
+    custom = [{"name": "disk", "command": "df -h", "threshold": 90}]
+    c = resolve_config({"scoring": {"critical": 40},
+                        "customCollectors": custom}, home=str(tmp_path))
+    assert c["scoring"]["critical"] == 40
+    assert c["customCollectors"] == custom
+
+
+def test_resolve_config_unknown_collector_added(tmp_path):
+    from vainplex_openclaw_amd.leuko.plugin import resolve_config
+
+    c = resolve_config({"collectors": {"brandnew": {"enabled": True}}},
+                       home=str(tmp_path))
+    assert c["collectors"]["brandnew"]["enabled"] is True
