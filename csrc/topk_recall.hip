@@ -1839,3 +1839,4 @@ topk_scan_fp4_v5_kernel(const uint8_t* __restrict__ Q4, const uint8_t* __restric
   }
 }
 
+
