@@ -558,3 +558,48 @@ def test_dashboard_against_real_plugins(tmp_path):
     # cortex + leuko sections populated from the real shapes
     assert "open threads:" in out
     assert "status:" in out
+
+
+# -- configurator trust-score table (configurator.test.ts, 22 its) ------------
+
+@pytest.mark.parametrize("name,score", [
+    ("admin-bot", 70), ("root", 70), ("rootkit-checker", 70),
+    ("main", 60), ("main-agent", 60),
+    ("reviewer", 50), ("cerberus", 50), ("code-review", 50),
+    ("forge", 45), ("builder", 45), ("build-ci", 45),
+    ("random-agent", 40), ("zeta", 40),
+    ("*", 10),
+    ("ADMIN", 70), ("Cerberus", 50),          # case-insensitive
+    ("admin-forge", 70),                       # first-match priority
+    ("main-review", 60),
+])
+def test_trust_score_table(name, score):
+    from vainplex_openclaw_amd.brainplex.configurator import compute_trust_score
+
+    assert compute_trust_score(name) == score
+
+
+def test_trust_defaults_include_all_agents_and_wildcard():
+    from vainplex_openclaw_amd.brainplex.configurator import build_trust_defaults
+
+    d = build_trust_defaults(["main", "forge", "admin"])
+    assert d == {"main": 60, "forge": 45, "admin": 70, "*": 10}
+    assert build_trust_defaults([]) == {"*": 10}
+
+
+def test_governance_config_uses_timezone_and_agents():
+    from vainplex_openclaw_amd.brainplex.configurator import generate_governance_config
+
+    cfg = generate_governance_config(["main", "cerberus"], "Europe/Berlin")
+    assert cfg["timezone"] == "Europe/Berlin"
+    assert cfg["trust"]["defaults"]["cerberus"] == 50
+    assert cfg["trust"]["defaults"]["*"] == 10
+    assert cfg["nightMode"]["start"] == "23:00"
+    assert cfg["trust"]["sessionTrust"]["seedFactor"] == 0.7
+
+
+def test_detect_timezone_never_raises():
+    from vainplex_openclaw_amd.brainplex.configurator import detect_timezone
+
+    tz = detect_timezone()
+    assert isinstance(tz, str) and tz
