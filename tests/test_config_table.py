@@ -118,3 +118,60 @@ def test_builtin_policy_toggles():
 def test_performance_defaults():
     p = resolve_performance(None)
     assert isinstance(p, dict) and p
+
+
+# -- layered loading through the governance plugin (config-loader.test.ts) ---
+
+def test_gov_plugin_legacy_inline_config(tmp_path, monkeypatch):
+    import json
+
+    from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+    from vainplex_openclaw_amd.governance.plugin import GovernancePlugin
+
+    monkeypatch.setenv("OPENCLAW_HOME", str(tmp_path))
+    api = PluginApi(id="openclaw-governance",
+                    plugin_config={"enabled": True, "failMode": "closed"},
+                    logger=NullLogger(), config={}, bus=HookBus())
+    p = GovernancePlugin(workspace=str(tmp_path))
+    p.register(api)
+    assert p.engine.config["failMode"] == "closed"   # legacy inline won
+
+
+def test_gov_plugin_config_path_pointer(tmp_path, monkeypatch):
+    import json
+    import os
+
+    from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+    from vainplex_openclaw_amd.governance.plugin import GovernancePlugin
+
+    monkeypatch.setenv("OPENCLAW_HOME", str(tmp_path))
+    ext = tmp_path / "gov.json"
+    ext.write_text(json.dumps({"enabled": True, "failMode": "closed",
+                               "audit": {"enabled": False}}))
+    api = PluginApi(id="openclaw-governance",
+                    plugin_config={"configPath": str(ext)},
+                    logger=NullLogger(), config={}, bus=HookBus())
+    p = GovernancePlugin(workspace=str(tmp_path))
+    p.register(api)
+    assert p.engine.config["failMode"] == "closed"
+    assert p.engine.config["audit"]["enabled"] is False
+
+
+def test_gov_plugin_external_file_default_location(tmp_path, monkeypatch):
+    import json
+    import os
+
+    from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+    from vainplex_openclaw_amd.governance.plugin import GovernancePlugin
+
+    monkeypatch.setenv("OPENCLAW_HOME", str(tmp_path))
+    d = tmp_path / "plugins" / "openclaw-governance"
+    os.makedirs(d)
+    (d / "config.json").write_text(json.dumps({"failMode": "closed"}))
+    # pointer-only inline: enabled override applies on the file config
+    api = PluginApi(id="openclaw-governance", plugin_config={"enabled": True},
+                    logger=NullLogger(), config={}, bus=HookBus())
+    p = GovernancePlugin(workspace=str(tmp_path))
+    p.register(api)
+    assert p.engine.config["failMode"] == "closed"
+    assert p.engine.config["enabled"] is True

@@ -11,7 +11,7 @@ from __future__ import annotations
 from typing import Any, Dict, List, Optional
 
 from ..core.api import PluginApi
-from ..core.config import load_plugin_config
+from ..core.config import load_raw_layered
 from .approval_2fa import Approval2FA
 from .engine import GovernanceEngine
 from .hooks import register_governance_hooks
@@ -97,7 +97,7 @@ class GovernancePlugin:
     def register(self, api: PluginApi) -> None:
         from .config import resolve_config
 
-        config = resolve_config(load_plugin_config(self.id, fallback=api.plugin_config))
+        config = resolve_config(load_raw_layered(self.id, api.plugin_config))
         workspace = self.workspace or config.get("workspace") or "."
         engine = GovernanceEngine(config, workspace, api.logger)
         engine.set_known_agents(extract_agent_ids(api.config))
