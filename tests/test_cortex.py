@@ -329,3 +329,24 @@ def test_commitment_flush_roundtrip(workspace):
     assert "benchmark" in blob
     ct2, _ = _ct(workspace)
     assert any("benchmark" in c["action"] for c in ct2.open_commitments())
+
+
+def test_cortex_plugin_layered_config(tmp_path, monkeypatch):
+    """config-loader.ts semantics through the cortex plugin: configPath
+    pointer + inline enabled, legacy inline config wins outright."""
+    import json
+
+    from vainplex_openclaw_amd.core.api import HookBus, NullLogger, PluginApi
+    from vainplex_openclaw_amd.cortex.hooks import CortexPlugin
+
+    monkeypatch.setenv("OPENCLAW_HOME", str(tmp_path))
+    ext = tmp_path / "cortex.json"
+    ext.write_text(json.dumps({"workspace": str(tmp_path),
+                               "threadTracker": {"maxThreads": 7}}))
+    api = PluginApi(id="openclaw-cortex", plugin_config={"configPath": str(ext)},
+                    logger=NullLogger(), config={}, bus=HookBus())
+    p = CortexPlugin(str(tmp_path))
+    p.register(api)
+    assert p.hooks is not None
+    ws = p.hooks.ws(str(tmp_path))
+    assert ws.threads.config.max_threads == 7

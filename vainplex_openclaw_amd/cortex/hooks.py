@@ -14,7 +14,7 @@ import time
 from typing import Any, Dict, List, Optional
 
 from ..core.api import PluginApi, PluginLogger, NullLogger
-from ..core.config import load_plugin_config
+from ..core.config import load_raw_layered
 from .boot_context import BootContextConfig, BootContextGenerator
 from .commitment_tracker import CommitmentTracker, CommitmentTrackerConfig
 from .decision_tracker import DecisionTracker, DecisionTrackerConfig
@@ -181,7 +181,7 @@ class CortexPlugin:
         self._analyzer_timer = None
 
     def register(self, api: PluginApi) -> None:
-        config = load_plugin_config(self.id, fallback=api.plugin_config)
+        config = load_raw_layered(self.id, api.plugin_config)
         workspace = self.workspace or config.get("workspace") or "."
         h = CortexHooks(config, workspace, api.logger)
         self.hooks = h
