@@ -167,3 +167,104 @@ def test_tools_status_shape(tmp_path):
     assert st["threads"]["open"] >= 0
     assert "mood" in st["threads"] and "decisions" in st
     assert set(st["commitments"]) == {"open", "overdue"}
+
+
+# -- commitment patterns (commitment-patterns.test.ts, 11 its) ----------------
+
+@pytest.mark.parametrize("text,lang", [
+    ("I'll update the docs tomorrow", "en"),
+    ("I will deploy the fix tonight", "en"),
+    ("let me check the logs first", "en"),
+    ("I promise to review it today", "en"),
+    ("ich werde das morgen erledigen", "de"),
+    ("mach ich", "de"),
+    ("je vais corriger le bug", "fr"),
+    ("lo haré mañana", "es"),
+    ("pode deixar", "pt"),
+    ("ci penso io", "it"),
+    ("我会处理这个问题", "zh"),
+    ("対応します", "ja"),
+    ("제가 할게요", "ko"),
+    ("я займусь этим", "ru"),
+])
+def test_commitment_detection_per_language(text, lang):
+    from vainplex_openclaw_amd.cortex.commitment_tracker import detect_commitments
+
+    hits = detect_commitments(text)
+    assert hits and any(h["language"] == lang for h in hits)
+
+
+@pytest.mark.parametrize("text", ["sounds good", "agreed", "", "the weather is nice"])
+def test_commitment_no_false_positives(text):
+    from vainplex_openclaw_amd.cortex.commitment_tracker import detect_commitments
+
+    assert detect_commitments(text) == []
+
+
+def test_commitment_patterns_cover_ten_languages():
+    from vainplex_openclaw_amd.cortex.commitment_tracker import COMMITMENT_PATTERNS
+
+    langs = {lang for _, lang in COMMITMENT_PATTERNS}
+    assert langs >= {"en", "de", "fr", "es", "pt", "it", "zh", "ja", "ko", "ru"}
+
+
+# -- LLM enhancer (llm-enhance.test.ts, 9 its) --------------------------------
+
+def test_enhancer_disabled_returns_none():
+    from vainplex_openclaw_amd.cortex.llm_enhance import LlmEnhancer
+
+    e = LlmEnhancer({"enabled": False}, call_llm=lambda p: "{}")
+    assert e.enabled is False
+    assert e.add_message("anything") is None
+    assert e.flush() is None
+
+
+def test_enhancer_requires_call_llm():
+    from vainplex_openclaw_amd.cortex.llm_enhance import LlmEnhancer
+
+    e = LlmEnhancer({"enabled": True}, call_llm=None)
+    assert e.enabled is False
+
+
+def test_enhancer_buffers_until_batch_size():
+    import json
+
+    from vainplex_openclaw_amd.cortex.llm_enhance import LlmEnhancer
+
+    calls = []
+
+    def llm(prompt):
+        calls.append(prompt)
+        return json.dumps({"threads": [], "decisions": [], "closures": [],
+                           "mood": "focused"})
+
+    e = LlmEnhancer({"enabled": True, "batchSize": 3}, call_llm=llm)
+    assert e.add_message("one") is None
+    assert e.add_message("two") is None
+    out = e.add_message("three")
+    assert out is not None and out.get("mood") == "focused"
+    assert len(calls) == 1
+    assert "one" in calls[0] and "three" in calls[0]
+
+
+def test_enhancer_flush_empty_and_error():
+    from vainplex_openclaw_amd.cortex.llm_enhance import LlmEnhancer
+
+    e = LlmEnhancer({"enabled": True}, call_llm=lambda p: "{}")
+    assert e.flush() is None                            # nothing buffered
+
+    def boom(p):
+        raise ConnectionError("down")
+
+    e2 = LlmEnhancer({"enabled": True, "batchSize": 1}, call_llm=boom)
+    assert e2.add_message("msg") is None                # error swallowed
+
+
+def test_enhancer_config_defaults_merge():
+    from vainplex_openclaw_amd.cortex.llm_enhance import DEFAULT_CONFIG, LlmEnhancer
+
+    e = LlmEnhancer({"endpoint": "http://cloud/v1"})
+    assert e.config["endpoint"] == "http://cloud/v1"
+    for key, val in DEFAULT_CONFIG.items():
+        if key != "endpoint":
+            assert e.config[key] == val
