@@ -1,3 +1,4 @@
 from .cli import main
 
-raise SystemExit(main())
+if __name__ == "__main__":  # don't run on programmatic import
+    raise SystemExit(main())
