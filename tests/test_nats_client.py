@@ -9,6 +9,7 @@ and warn cadence, status block, drain-with-timeout force close.
 """
 
 import json
+import os
 import queue
 import threading
 
@@ -391,3 +392,30 @@ def test_eventstore_plugin_over_live_nats_backend():
     assert env["id"].startswith("evt-")
     st = api.gateway_methods["eventstore.status"]()
     assert st["connected"] is True and st["stream"] == "openclaw-events"
+
+
+@pytest.mark.skipif(not os.environ.get("NATS_URL"),
+                    reason="needs a live NATS server (set NATS_URL)")
+def test_live_jetstream_round_trip():
+    """Env-gated live-infrastructure test (the reference's only
+    infra-gated suite: nats-eventstore integration.test.ts skipIf
+    !NATS_URL): connect over real TCP, ensure the stream, publish an
+    envelope, read it back by sequence."""
+    import json as _json
+
+    from vainplex_openclaw_amd.eventstore.nats_client import JetStreamClient
+
+    cfg = dict(CFG)
+    cfg["natsUrl"] = os.environ["NATS_URL"]
+    cfg["streamName"] = "openclaw-events-test"
+    js = JetStreamClient(cfg)
+    js.connect()
+    try:
+        assert js.publish("openclaw.events.agent.live",
+                          _json.dumps({"id": "live-1", "ts": 1.0}))
+        info = js.stream_info()
+        assert info and info["state"]["messages"] >= 1
+        msg = js.get_message(int(info["state"]["last_seq"]))
+        assert msg and _json.loads(msg["data"].decode())["id"] == "live-1"
+    finally:
+        js.close()
