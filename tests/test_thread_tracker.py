@@ -175,3 +175,93 @@ def test_priority_from_impact_keywords(workspace, text, prio):
     tt = ThreadTracker(workspace)
     tt.process_message(text, "user")
     assert tt.get_threads()[0]["priority"] == prio
+
+
+# -- closure, pruning, persistence (thread-tracker.test.ts remainder) ---------
+
+def _tracker(tmp_path, clock=None, **cfg):
+    from vainplex_openclaw_amd.cortex.thread_tracker import (
+        ThreadTracker,
+        ThreadTrackerConfig,
+    )
+
+    return ThreadTracker(str(tmp_path), ThreadTrackerConfig(**cfg), "both",
+                         clock=clock or (lambda: 1_700_000_000.0))
+
+
+def test_closure_pattern_closes_matching_thread(tmp_path):
+    t = _tracker(tmp_path)
+    t.process_message("let's talk about the payment deployment", "user")
+    open_before = [x for x in t.get_threads() if x["status"] == "open"]
+    assert open_before
+    t.process_message("payment deployment is done", "user")
+    target = [x for x in t.get_threads()
+              if "payment" in x["title"].lower()]
+    assert target and target[0]["status"] == "closed"
+
+
+def test_prune_drops_old_closed_keeps_recent_and_open(tmp_path):
+    now = [1_700_000_000.0]
+    t = _tracker(tmp_path, clock=lambda: now[0], prune_days=14)
+    t.threads = [
+        {"id": "a", "title": "old closed", "status": "closed",
+         "last_activity": "2023-10-01T00:00:00Z"},
+        {"id": "b", "title": "fresh closed", "status": "closed",
+         "last_activity": "2023-11-14T00:00:00Z"},
+        {"id": "c", "title": "ancient open", "status": "open",
+         "last_activity": "2023-01-01T00:00:00Z"},
+    ]
+    t._prune_and_cap()
+    ids = {x["id"] for x in t.threads}
+    assert ids == {"b", "c"}                           # old closed pruned only
+
+
+def test_max_threads_cap_evicts_oldest_closed_first(tmp_path):
+    t = _tracker(tmp_path, max_threads=3)
+    t.threads = [
+        {"id": f"o{i}", "title": f"open {i}", "status": "open",
+         "last_activity": "2023-11-14T00:00:00Z"} for i in range(2)
+    ] + [
+        {"id": f"c{i}", "title": f"closed {i}", "status": "closed",
+         "last_activity": f"2023-11-{10 + i:02d}T00:00:00Z"} for i in range(3)
+    ]
+    t._prune_and_cap()
+    assert len(t.threads) == 3
+    ids = {x["id"] for x in t.threads}
+    assert {"o0", "o1"} <= ids                          # open always kept
+    assert "c2" in ids and "c0" not in ids              # newest closed survives
+
+
+def test_persistence_round_trip_and_bad_files(tmp_path):
+    import json
+    import os
+
+    t = _tracker(tmp_path)
+    t.process_message("working on the database migration", "user")
+    t.flush()
+    path = t.file_path
+    assert os.path.isfile(path)
+    # fresh tracker loads the same threads
+    t2 = _tracker(tmp_path)
+    assert [x["title"] for x in t2.get_threads()] == \
+           [x["title"] for x in t.get_threads()]
+    # corrupt file -> empty start, no raise
+    with open(path, "w") as fh:
+        fh.write("{corrupt")
+    t3 = _tracker(tmp_path)
+    assert t3.get_threads() == []
+    # missing file -> empty start
+    os.remove(path)
+    t4 = _tracker(tmp_path)
+    assert t4.get_threads() == []
+
+
+def test_matches_thread_case_and_short_words():
+    from vainplex_openclaw_amd.cortex.thread_tracker import matches_thread
+
+    th = {"title": "Payment API Deployment"}
+    assert matches_thread(th, "the PAYMENT api is failing") is True
+    assert matches_thread(th, "api is up") is False     # 'api' <3? no: 1 word only
+    assert matches_thread({"title": "do it on ax"}, "it on ax please") is False
+    assert matches_thread({"title": ""}, "anything at all") is False
+    assert matches_thread(th, "") is False
