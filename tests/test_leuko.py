@@ -437,3 +437,114 @@ def test_resolve_config_unknown_collector_added(tmp_path):
     c = resolve_config({"collectors": {"brandnew": {"enabled": True}}},
                        home=str(tmp_path))
     assert c["collectors"]["brandnew"]["enabled"] is True
+
+
+# -- goal quality checks + LLM recommendations (SURVEY §2.7 Leuko surface) ----
+
+def _goals_file(tmp_path, goals):
+    import json
+
+    p = tmp_path / "goals.json"
+    p.write_text(json.dumps({"goals": goals}))
+    return str(p)
+
+
+def test_goal_quality_flags_vague_unmeasurable_unowned(tmp_path):
+    from vainplex_openclaw_amd.leuko.quality import check_goal_quality
+
+    path = _goals_file(tmp_path, [
+        {"id": "g1", "title": "improve things", "status": "open"},
+        {"id": "g2", "title": "", "status": "open"},
+        {"id": "g3", "title": "reduce p95 latency below 200ms by friday",
+         "status": "open"},
+        {"id": "g4", "title": "migrate billing database to postgres 16",
+         "zone": "red", "status": "approved"},
+    ])
+    r = check_goal_quality({"goalsPath": path})
+    ids = [i["id"] for i in r["items"]]
+    assert "goalq-g1-vague" in ids
+    assert "goalq-g2-untitled" in ids
+    assert "goalq-g4-red-unowned" in ids
+    assert not any("g3" in i for i in ids)          # measurable goal is clean
+    assert r["status"] == "warn"
+
+
+def test_goal_quality_overload_and_clean(tmp_path):
+    from vainplex_openclaw_amd.leuko.quality import check_goal_quality
+
+    many = [{"id": f"g{i}", "title": f"ship feature {i} to 100% of users",
+             "status": "open", "success_criteria": "rollout complete"}
+            for i in range(9)]
+    r = check_goal_quality({"goalsPath": _goals_file(tmp_path, many),
+                            "maxOpenGoals": 7})
+    assert any(i["id"] == "goalq-overload" for i in r["items"])
+    few = many[:3]
+    r2 = check_goal_quality({"goalsPath": _goals_file(tmp_path / "b"
+                             if (tmp_path / "b").mkdir() is None else tmp_path,
+                             few)})
+    assert r2["items"] == [] and r2["status"] == "ok"
+
+
+def test_goal_quality_missing_file_and_unconfigured(tmp_path):
+    from vainplex_openclaw_amd.leuko.quality import check_goal_quality
+
+    assert check_goal_quality({})["status"] == "ok"
+    assert check_goal_quality({"goalsPath": str(tmp_path / "ghost.json")})["status"] == "ok"
+
+
+def test_llm_recommendations_parse_validate_and_fail_soft():
+    import json
+
+    from vainplex_openclaw_amd.leuko.quality import llm_recommendations
+
+    report = {"summary": "2 warnings", "health": 80,
+              "items": [{"title": "timer failed", "severity": "warn",
+                         "category": "auto_fixable"}]}
+    good = json.dumps({"recommendations": [
+        {"action": "restart the failed timer", "priority": "high", "reason": "stuck"},
+        {"action": "archive stale goals", "priority": "weird", "reason": ""},
+    ]})
+    recs = llm_recommendations(report, lambda p: good)
+    assert len(recs) == 2
+    assert recs[0]["priority"] == "high"
+    assert recs[1]["priority"] == "medium"            # invalid -> medium
+    assert llm_recommendations(report, None) == []
+    assert llm_recommendations(report, lambda p: "not json") == []
+
+    def boom(p):
+        raise TimeoutError("llm down")
+
+    assert llm_recommendations(report, boom) == []
+    capped = json.dumps({"recommendations": [
+        {"action": f"a{i}"} for i in range(20)]})
+    assert len(llm_recommendations(report, lambda p: capped, max_items=5)) == 5
+
+
+def test_plugin_attaches_recommendations(tmp_path):
+    import json
+
+    from vainplex_openclaw_amd.leuko.plugin import LeukoPlugin, resolve_config
+
+    p = LeukoPlugin(str(tmp_path), clock=lambda: 1_700_000_000.0,
+                    call_llm=lambda prompt: json.dumps(
+                        {"recommendations": [{"action": "do the thing"}]}))
+    p.config = resolve_config({
+        "outputPath": str(tmp_path / "sitrep.json"),
+        "previousPath": str(tmp_path / "prev.json"),
+        "collectors": {"systemd_timers": {"enabled": False},
+                       "gpu_health": {"enabled": False}},
+        "anomaly": {"enabled": False},
+    }, home=str(tmp_path))
+    report = p.run_once()
+    assert report["recommendations"] == [
+        {"action": "do the thing", "priority": "medium", "reason": ""}]
+    on_disk = json.load(open(str(tmp_path / "sitrep.json")))
+    assert on_disk["recommendations"][0]["action"] == "do the thing"
+
+
+def test_goal_quality_collector_registered(tmp_path):
+    from vainplex_openclaw_amd.leuko.collectors import BUILT_IN_COLLECTORS
+
+    assert "goal_quality" in BUILT_IN_COLLECTORS
+    r = BUILT_IN_COLLECTORS["goal_quality"]({"goalsPath": ""})
+    assert r["status"] == "ok"

@@ -329,6 +329,12 @@ def run_custom_collector(defn: Dict) -> Dict:
                   f"{len(items)} issue(s)" if items else "ok")
 
 
+def _goal_quality(config: Dict) -> Dict:
+    from .quality import check_goal_quality
+
+    return check_goal_quality(config)
+
+
 BUILT_IN_COLLECTORS: Dict[str, Callable[[Dict], Dict]] = {
     "systemd_timers": collect_systemd_timers,
     "nats": collect_journal,
@@ -337,4 +343,5 @@ BUILT_IN_COLLECTORS: Dict[str, Callable[[Dict], Dict]] = {
     "errors": collect_errors,
     "calendar": collect_calendar,
     "gpu_health": collect_gpu_health,
+    "goal_quality": _goal_quality,
 }

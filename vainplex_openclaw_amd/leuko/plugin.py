@@ -33,6 +33,7 @@ def default_config(home: Optional[str] = None) -> Dict[str, Any]:
             "systemd_timers": {"enabled": True},
             "nats": {"enabled": False},
             "goals": {"enabled": False, "goalsPath": ""},
+            "goal_quality": {"enabled": False, "goalsPath": "", "maxOpenGoals": 7},
             "threads": {"enabled": False, "threadsPath": ""},
             "errors": {"enabled": False, "patternsPath": ""},
             "calendar": {"enabled": False, "command": ""},
@@ -79,10 +80,12 @@ class LeukoPlugin:
     description = "Cognitive immune system: health checks, anomaly detection, sitrep"
     version = "0.1.0"
 
-    def __init__(self, workspace: Optional[str] = None, journal=None, clock=time.time):
+    def __init__(self, workspace: Optional[str] = None, journal=None, clock=time.time,
+                 call_llm=None):
         self.workspace = workspace or "."
         self.journal = journal
         self._clock = clock
+        self.call_llm = call_llm
         self.config: Dict[str, Any] = {}
         self.detector: Optional[AnomalyDetector] = None
         self._timer: Optional[threading.Timer] = None
@@ -96,6 +99,10 @@ class LeukoPlugin:
                 acfg, journal=self.journal, workspace=self.workspace
             )
         report = generate_sitrep(self.config, extra=extra, clock=self._clock)
+        if self.call_llm is not None and self.config.get("llmRecommendations", True):
+            from .quality import llm_recommendations
+
+            report["recommendations"] = llm_recommendations(report, self.call_llm)
         write_sitrep(report, self.config["outputPath"], self.config["previousPath"])
         self.last_report = report
         return report
