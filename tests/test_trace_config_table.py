@@ -206,3 +206,42 @@ def test_analyzer_from_config_wires_detectors_gap_and_report_path(tmp_path):
     assert ta.classifier.redactor.engine.scan_string("hunter2")["output"] != "hunter2"
     report = ta.run()
     assert report["stats"]["findings"] == 0
+
+
+# -- top-level cortex resolve_config depth (cortex config.test.ts) -----------
+
+def test_top_level_defaults_none_and_empty():
+    for raw in (None, {}):
+        c = resolve_config(raw)
+        assert c["threadTracker"] == {"enabled": True, "pruneDays": 14,
+                                      "maxThreads": 50}
+        assert c["language"] == "both"
+        assert c["llm"] == {"enabled": False}
+
+
+def test_top_level_partial_nested_merge():
+    c = resolve_config({"threadTracker": {"maxThreads": 9},
+                        "bootContext": {"enabled": False}})
+    assert c["threadTracker"]["maxThreads"] == 9
+    assert c["threadTracker"]["pruneDays"] == 14       # untouched default
+    assert c["bootContext"]["enabled"] is False
+    assert c["decisionTracker"]["enabled"] is True     # other section intact
+
+
+def test_top_level_unknown_keys_preserved_and_language():
+    c = resolve_config({"language": "de", "workspace": "/w",
+                        "customPatterns": {"en": {}}})
+    assert c["language"] == "de"
+    assert c["workspace"] == "/w"                      # extra keys pass through
+    assert c["customPatterns"] == {"en": {}}
+
+
+def test_top_level_all_features_disabled():
+    c = resolve_config({"threadTracker": {"enabled": False},
+                        "decisionTracker": {"enabled": False},
+                        "commitmentTracker": {"enabled": False},
+                        "bootContext": {"enabled": False},
+                        "traceAnalyzer": {"enabled": False}})
+    for section in ("threadTracker", "decisionTracker", "commitmentTracker",
+                    "bootContext", "traceAnalyzer"):
+        assert c[section]["enabled"] is False
