@@ -603,3 +603,46 @@ def test_detect_timezone_never_raises():
 
     tz = detect_timezone()
     assert isinstance(tz, str) and tz
+
+
+# -- writer merge semantics (writer.test.ts merge section) --------------------
+
+def test_update_config_does_not_mutate_original():
+    import copy
+
+    from vainplex_openclaw_amd.brainplex.writer import update_openclaw_config
+
+    original = {"plugins": {"entries": {"x": {"enabled": False}},
+                            "allow": ["x"]},
+                "other": {"nested": [1, 2]}}
+    snapshot = copy.deepcopy(original)
+    res = update_openclaw_config("/nonexistent.json", original,
+                                 ["openclaw-governance"], dry_run=True)
+    assert original == snapshot                       # input untouched
+    merged = res["config"]
+    assert merged["plugins"]["entries"]["x"] == {"enabled": False}
+    assert merged["plugins"]["entries"]["openclaw-governance"] == {"enabled": True}
+    assert merged["plugins"]["allow"] == ["x", "openclaw-governance"]
+    assert merged["other"] == {"nested": [1, 2]}      # unrelated preserved
+
+
+def test_update_config_handles_malformed_sections():
+    from vainplex_openclaw_amd.brainplex.writer import update_openclaw_config
+
+    res = update_openclaw_config("/nonexistent.json",
+                                 {"plugins": "not-a-dict"},
+                                 ["p1"], dry_run=True)
+    assert res["updated"] is True
+    assert res["config"]["plugins"]["entries"]["p1"] == {"enabled": True}
+    res2 = update_openclaw_config("/nonexistent.json",
+                                  {"plugins": {"entries": [1], "allow": "x"}},
+                                  ["p1"], dry_run=True)
+    assert res2["config"]["plugins"]["allow"] == ["p1"]
+
+
+def test_update_config_empty_plugin_list_noop():
+    from vainplex_openclaw_amd.brainplex.writer import update_openclaw_config
+
+    res = update_openclaw_config("/nonexistent.json", {}, [], dry_run=True)
+    assert res == {"updated": False, "backed_up": False,
+                   "added_entries": [], "added_allow": []}
