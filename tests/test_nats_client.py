@@ -419,3 +419,46 @@ def test_live_jetstream_round_trip():
         assert msg and _json.loads(msg["data"].decode())["id"] == "live-1"
     finally:
         js.close()
+
+
+def test_auto_reconnect_on_dead_connection():
+    """A dead connection triggers a bounded reconnect before publish;
+    the attempt counter resets on success (reference maxReconnectAttempts)."""
+    servers = []
+
+    def factory():
+        srv = FakeNatsServer(existing_streams={"openclaw-events":
+                                               {"name": "openclaw-events"}})
+        servers.append(srv)
+        return srv.transport()
+
+    js = JetStreamClient(CFG, transport_factory=factory)
+    js.connect()
+    assert js.publish("openclaw.events.a", "{}")
+    js.nc.close()                                    # kill the connection
+    assert js.publish("openclaw.events.b", "{}")     # reconnected + published
+    assert len(servers) == 2
+    assert servers[1].messages                       # landed on the NEW server
+    assert js.reconnect_attempts == 0                # reset on success
+
+
+def test_reconnect_attempts_bounded():
+    cfg = dict(CFG)
+    cfg["maxReconnectAttempts"] = 2
+
+    calls = []
+
+    def bad_factory():
+        calls.append(1)
+        raise ConnectionRefusedError("down")
+
+    srv = FakeNatsServer(existing_streams={"openclaw-events":
+                                           {"name": "openclaw-events"}})
+    js = JetStreamClient(cfg, transport=srv.transport())
+    js.connect()
+    js.transport_factory = bad_factory
+    js.nc.close()
+    for _ in range(5):
+        assert js.publish("openclaw.events.x", "{}") is False
+    assert len(calls) == 2                           # capped at 2 attempts
+    assert js.reconnect_attempts == 2

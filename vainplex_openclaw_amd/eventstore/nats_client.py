@@ -250,13 +250,19 @@ class JetStreamClient:
     """JetStream layer mirroring nats-client.ts createNatsClient."""
 
     def __init__(self, config: Dict, logger=None, transport=None,
-                 clock=time.time):
+                 transport_factory=None, clock=time.time):
         self.config = config
         self.logger = logger
         self.clock = clock
-        self.transport = transport if transport is not None else TcpTransport()
+        # transport_factory produces a FRESH transport per (re)connect;
+        # a plain injected transport is reused (loopback tests)
+        self.transport_factory = transport_factory
+        self.transport = transport if transport is not None else (
+            transport_factory() if transport_factory is not None else TcpTransport())
         self.nc: Optional[NatsConnection] = None
         self.publish_failures = 0
+        self.reconnect_attempts = 0
+        self.max_reconnect_attempts = int(config.get("maxReconnectAttempts", 10))
         self.stream = config.get("streamName", "openclaw-events")
         self.subject_prefix = config.get("subjectPrefix", "openclaw.events")
         self.publish_timeout_s = float(config.get("publishTimeoutMs", 5000)) / 1000.0
@@ -348,19 +354,48 @@ class JetStreamClient:
             raise NatsError(f"bad $JS.API.{op} reply: {exc}")
 
     # -- publish (never blocks agent operations) ---------------------------
+    def _try_reconnect(self) -> bool:
+        """Bounded auto-reconnect (the reference connects with
+        reconnect: true / maxReconnectAttempts: 10 — nats.js handles it
+        in-library; this client re-dials on a dead connection). The
+        attempt counter resets on a successful publish."""
+        if self.reconnect_attempts >= self.max_reconnect_attempts:
+            return False
+        self.reconnect_attempts += 1
+        try:
+            if self.transport_factory is not None:
+                self.transport = self.transport_factory()
+            elif isinstance(self.transport, TcpTransport):
+                self.transport = TcpTransport()
+            self.connect()
+            if self.logger:
+                self.logger.info(
+                    f"[nats-eventstore] Reconnected "
+                    f"(attempt {self.reconnect_attempts})")
+            return True
+        except Exception as exc:
+            if self.logger:
+                self.logger.warn(
+                    f"[nats-eventstore] Reconnect attempt "
+                    f"{self.reconnect_attempts} failed: {exc}")
+            return False
+
     def publish(self, subject: str, data: str) -> bool:
         """JetStream publish with ack; failures are swallowed, counted,
         and warn on the 1st + every 10th consecutive failure; the counter
-        resets on success (nats-client.ts:147-177)."""
+        resets on success (nats-client.ts:147-177). A dead connection
+        triggers one bounded reconnect attempt before the publish."""
         try:
             if self.nc is None or self.nc.is_closed:
-                raise NatsError("not connected")
+                if not self._try_reconnect():
+                    raise NatsError("not connected")
             ack_raw = self.nc.request(subject, data.encode(),
                                       timeout_s=self.publish_timeout_s)
             ack = json.loads(ack_raw.decode())
             if ack.get("error"):
                 raise NatsError(f"publish nack: {ack['error']}")
             self.publish_failures = 0
+            self.reconnect_attempts = 0
             return True
         except Exception as exc:
             self.publish_failures += 1
