@@ -381,3 +381,41 @@ def test_validation_trust_thresholds_end_to_end(workspace):
     assert eng.validate_output(text, "high")["verdict"] == "pass"
     # matching claims pass at any trust
     assert eng.validate_output("nginx is running.", "low")["verdict"] == "pass"
+
+
+def test_trace_findings_become_checkable_facts_end_to_end(tmp_path):
+    """Capstone loop (SURVEY §1 flow): journal events -> trace analyzer
+    detects a hallucination with an SPO payload -> trace-to-facts bridge
+    registers the fact -> the governance output validator then catches an
+    agent REPEATING the hallucinated claim, trust-proportionally."""
+    from vainplex_openclaw_amd.cortex.trace.analyzer import JournalTraceSource, TraceAnalyzer
+    from vainplex_openclaw_amd.eventstore import EventJournal
+    from vainplex_openclaw_amd.governance.facts import FactRegistry
+    from vainplex_openclaw_amd.governance.output_validator import OutputValidator
+    from vainplex_openclaw_amd.governance.trace_to_facts import apply_report_to_registry
+
+    j = EventJournal(durable=False)
+    base = {"actor": {"id": "forge"}, "scope": {"sessionKey": "agent:forge:s1"}}
+    j.publish("s.1", {**base, "id": "m1", "ts": 1000.0,
+                      "canonicalType": "message.out.sent",
+                      "data": {"content": "nginx-service is running"}})
+    j.publish("s.2", {**base, "id": "t1", "ts": 1001.0,
+                      "canonicalType": "tool.call.failed",
+                      "data": {"toolName": "exec",
+                               "error": "nginx-service: connection refused"}})
+
+    report = TraceAnalyzer(str(tmp_path), JournalTraceSource(j)).run()
+    halls = [f for f in report["findings"] if f["signalType"] == "hallucination"]
+    assert halls and halls[0]["evidence"]["subject"] == "nginx-service"
+
+    registry = FactRegistry()
+    assert apply_report_to_registry(report, registry) >= 1
+    fact = registry.lookup("nginx-service", "state")
+    assert fact and fact["value"] == "error"
+
+    validator = OutputValidator({"enabled": True})
+    validator.fact_registry = registry
+    low = validator.validate("the nginx-service is running fine", trust_score=20)
+    assert low["verdict"] == "block"                  # contradiction + low trust
+    high = validator.validate("the nginx-service is running fine", trust_score=90)
+    assert high["verdict"] == "pass"                  # trust-proportional
