@@ -52,6 +52,8 @@ def test_registers_analyze_and_status_commands(tmp_path):
     p, api, j = _plugin(tmp_path, ta={"enabled": True})
     assert "cortexanalyze" in api.commands
     assert "cortextracestatus" in api.commands
+    assert "trace-analyze" in api.commands           # reference names
+    assert "trace-status" in api.commands
     assert "cortex.analyze" in api.gateway_methods
     assert "cortex.trace.status" in api.gateway_methods
 

@@ -240,8 +240,11 @@ class CortexPlugin:
                 signal_registry=registry,
             )
         api.register_command("cortexanalyze", lambda *a, **kw: self.analyzer.run())
+        # reference command names (trace-analyzer/hooks.ts:176,182)
+        api.register_command("trace-analyze", lambda *a, **kw: self.analyzer.run())
         api.register_gateway_method("cortex.analyze", lambda *a, **kw: self.analyzer.run())
         api.register_command("cortextracestatus", lambda *a, **kw: self.trace_status())
+        api.register_command("trace-status", lambda *a, **kw: self.trace_status())
         api.register_gateway_method("cortex.trace.status", lambda *a, **kw: self.trace_status())
 
         schedule_cfg = ta_cfg.get("schedule") or {}
