@@ -142,3 +142,32 @@ def test_service_sync_check_latency_budget_gpu():
         assert lats[0] >= 2.0, lats
     finally:
         svc.close()
+
+
+def test_cancelled_future_does_not_poison_batch():
+    """A caller that cancels its future (e.g. a timed-out sync check)
+    must not divert the rest of the batch to the exception path."""
+    from vainplex_openclaw_amd.pipeline.service import FirewallService
+
+    import torch
+
+    class FakePipe:
+        cfg = type("C", (), {"batch": 64})()
+
+        def step(self, batch, staged=None):
+            n = len(batch.messages)
+            z = torch.zeros(n, dtype=torch.int64)
+            return {"verdict": torch.zeros(n, dtype=torch.int8),
+                    "risk": torch.zeros(n),
+                    "hits": {"injection": z},
+                    "recall_ids": torch.zeros(n, 4, dtype=torch.int32)}
+
+    svc = FirewallService(FakePipe(), max_batch=8, max_wait_ms=50.0)
+    try:
+        futs = [svc.submit(f"m{i}".encode()) for i in range(6)]
+        futs[2].cancel()                       # caller gave up
+        results = [f.result(timeout=10) for i, f in enumerate(futs) if i != 2]
+        assert all(r["verdict"] == "allow" for r in results)
+        assert futs[2].cancelled()
+    finally:
+        svc.close()

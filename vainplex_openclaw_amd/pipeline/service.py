@@ -101,11 +101,20 @@ class FirewallService:
             try:
                 out = self._process(batch)
                 for i, p in enumerate(batch):
-                    p.future.set_result(out[i])
+                    # a caller may have cancelled / timed out its future;
+                    # that must never poison the rest of the batch
+                    if not p.future.done():
+                        try:
+                            p.future.set_result(out[i])
+                        except Exception:
+                            pass
             except Exception as exc:
                 for p in batch:
                     if not p.future.done():
-                        p.future.set_exception(exc)
+                        try:
+                            p.future.set_exception(exc)
+                        except Exception:
+                            pass
 
     def _process(self, batch: List[_Pending]) -> List[Dict[str, Any]]:
         sb = SynthBatch(
